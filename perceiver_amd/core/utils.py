@@ -1,0 +1,65 @@
+"""Small building blocks shared by all Perceiver models.
+
+Functional parity with /root/reference/perceiver/model/core/utils.py (ModuleOutput,
+Residual, init_parameters, freeze) — reimplemented for the MI355X-native stack.
+State-dict key layout (``module.*`` inside Residual) is kept compatible with the
+reference checkpoint format.
+"""
+from __future__ import annotations
+
+from collections import OrderedDict
+
+import torch.nn as nn
+
+
+class ModuleOutput(OrderedDict):
+    """Ordered dict with attribute access; the uniform return type of all core modules."""
+
+    def __getattr__(self, name):
+        try:
+            return self[name]
+        except KeyError:
+            raise AttributeError(f"No such attribute: {name}") from None
+
+    def __setattr__(self, name, value):
+        self[name] = value
+
+    def __delattr__(self, name):
+        try:
+            del self[name]
+        except KeyError:
+            raise AttributeError(f"No such attribute: {name}") from None
+
+
+class Residual(nn.Module):
+    """Residual connection around ``module`` with dropout on the module output.
+
+    ``module`` must return a :class:`ModuleOutput`; the residual is added to its
+    ``last_hidden_state`` and the (possibly present) ``kv_cache`` is passed through.
+    """
+
+    def __init__(self, module: nn.Module, dropout: float = 0.0):
+        super().__init__()
+        self.module = module
+        self.dropout = nn.Dropout(dropout)
+
+    def forward(self, *args, **kwargs):
+        output = self.module(*args, **kwargs)
+        output.last_hidden_state = self.dropout(output.last_hidden_state) + args[0]
+        return output
+
+
+def init_parameters(module: nn.Module, init_scale: float) -> None:
+    """Normal(0, init_scale) init of every Linear/Embedding weight; zero biases."""
+    for m in module.modules():
+        if isinstance(m, nn.Linear):
+            m.weight.data.normal_(mean=0.0, std=init_scale)
+            if m.bias is not None:
+                m.bias.data.zero_()
+        elif isinstance(m, nn.Embedding):
+            m.weight.data.normal_(mean=0.0, std=init_scale)
+
+
+def freeze(module: nn.Module) -> None:
+    for p in module.parameters():
+        p.requires_grad = False
