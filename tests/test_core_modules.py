@@ -131,7 +131,7 @@ def test_positions_shift_and_clamp():
     shift = torch.tensor([[0], [3]])
     pos = positions(2, 5, shift=shift)
     assert pos[0].tolist() == [0, 1, 2, 3, 4]
-    assert pos[1].tolist() == [0, 0, 0, 1, 2]
+    assert pos[1].tolist() == [0, 0, 0, 0, 1]
 
 
 def test_fourier_position_encoding_channels():
