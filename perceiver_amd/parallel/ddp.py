@@ -1,0 +1,115 @@
+"""Native data-parallel gradient reduction over RCCL/xGMI.
+
+Replaces Lightning/torch-DDP's implicit reducer with an explicit
+bucketed all-reduce: parameters are grouped (in reverse registration order, which
+approximates backward completion order) into flat buckets; a bucket's all-reduce
+launches asynchronously as soon as its last gradient materializes, overlapping the
+remaining backward compute. On ROCm the "nccl" backend IS RCCL and the collectives
+ride xGMI; each MI355X has 7 point-to-point xGMI links (~153 GB/s each), so ring
+all-reduce is per-link bound — bucket sizes default to 50 MiB so several buckets
+pipeline across the links while backward proceeds.
+"""
+from __future__ import annotations
+
+from typing import List, Optional
+
+import torch
+import torch.distributed as dist
+
+
+class _Bucket:
+    def __init__(self, params: List[torch.nn.Parameter], device, dtype):
+        self.params = params
+        self.numel = sum(p.numel() for p in params)
+        self.flat = torch.zeros(self.numel, device=device, dtype=dtype)
+        self.offsets = []
+        off = 0
+        for p in params:
+            self.offsets.append(off)
+            off += p.numel()
+        self.ready = 0
+        self.work: Optional[dist.Work] = None
+
+    def reset(self):
+        self.ready = 0
+        self.work = None
+
+
+class BucketedGradReducer:
+    """Explicit bucketed gradient all-reduce with backward overlap.
+
+    Usage per step::
+
+        loss.backward()          # hooks fire, buckets all-reduce asynchronously
+        reducer.finalize()       # wait + write averaged grads back
+        optimizer.step()
+
+    Gradients are averaged over the world size. Parameters that do not require
+    grad are skipped. Works on any torch.distributed backend (RCCL on GPU,
+    gloo in CPU tests).
+    """
+
+    def __init__(self, module: torch.nn.Module, bucket_cap_mb: float = 50.0,
+                 process_group: Optional[dist.ProcessGroup] = None):
+        if not dist.is_initialized():
+            raise RuntimeError("torch.distributed must be initialized before BucketedGradReducer")
+        self.group = process_group
+        self.world_size = dist.get_world_size(process_group)
+
+        # unique params in reverse registration order (approximate backward order);
+        # shared/tied params appear once
+        seen = set()
+        params: List[torch.nn.Parameter] = []
+        for p in reversed(list(module.parameters())):
+            if p.requires_grad and id(p) not in seen:
+                seen.add(id(p))
+                params.append(p)
+
+        cap = int(bucket_cap_mb * 1024 * 1024)
+        self.buckets: List[_Bucket] = []
+        cur: List[torch.nn.Parameter] = []
+        cur_bytes = 0
+        for p in params:
+            sz = p.numel() * p.element_size()
+            if cur and cur_bytes + sz > cap:
+                self.buckets.append(_Bucket(cur, p.device, p.dtype))
+                cur, cur_bytes = [], 0
+            cur.append(p)
+            cur_bytes += sz
+        if cur:
+            self.buckets.append(_Bucket(cur, cur[0].device, cur[0].dtype))
+
+        self._param_bucket = {}
+        for b in self.buckets:
+            for p, off in zip(b.params, b.offsets):
+                self._param_bucket[id(p)] = (b, off)
+
+        self._hooks = [
+            p.register_post_accumulate_grad_hook(self._on_grad_ready) for p in params
+        ]
+
+    def _on_grad_ready(self, p: torch.nn.Parameter):
+        b, off = self._param_bucket[id(p)]
+        b.flat[off: off + p.numel()].copy_(p.grad.detach().reshape(-1))
+        b.ready += 1
+        if b.ready == len(b.params):
+            b.flat.div_(self.world_size)
+            b.work = dist.all_reduce(b.flat, group=self.group, async_op=True)
+
+    def finalize(self):
+        """Wait for all bucket reductions and write averaged gradients back."""
+        for b in self.buckets:
+            if b.ready != len(b.params):
+                raise RuntimeError(
+                    f"bucket incomplete at finalize: {b.ready}/{len(b.params)} grads ready "
+                    "(a parameter did not receive a gradient this step)"
+                )
+            b.work.wait()
+            for p, off in zip(b.params, b.offsets):
+                p.grad.detach().reshape(-1).copy_(b.flat[off: off + p.numel()])
+            b.reset()
+
+    def remove(self):
+        for h in self._hooks:
+            h.remove()
+        self._hooks = []
