@@ -137,7 +137,7 @@ class MLMBench:
 
     def config_json(self):
         return {"model": "perceiver-io-mlm-201M", "global_batch": None, "seq_len": self.seq,
-                "num_latents": 256, "parallelism": None}
+                "num_latents": 512, "parallelism": None}
 
 
 class CLMDecodeBench:

@@ -18,9 +18,11 @@ from perceiver_amd.models.vision.optical_flow import (
 )
 
 
-def mlm_flagship(vocab_size: int = 262, max_seq_len: int = 2048) -> MaskedLanguageModelConfig:
-    """201M-param Perceiver IO MLM: 256 latents x 1280, 26 self-attention layers,
-    qk 256 / v 1280, UTF-8 bytes seq 2048."""
+def mlm_flagship(vocab_size: int = 262, max_seq_len: int = 2048,
+                 num_latents: int = 512) -> MaskedLanguageModelConfig:
+    """Perceiver IO MLM flagship: 1280-channel latents, 26 self-attention layers,
+    qk 256 / v 1280, UTF-8 bytes seq 2048. BASELINE.json names 512 latents (the
+    reference CLI default is 256; same architecture otherwise)."""
     return MaskedLanguageModelConfig(
         encoder=TextEncoderConfig(
             vocab_size=vocab_size,
@@ -46,7 +48,7 @@ def mlm_flagship(vocab_size: int = 262, max_seq_len: int = 2048) -> MaskedLangua
             cross_attention_residual=False,
             dropout=0.1,
         ),
-        num_latents=256,
+        num_latents=num_latents,
         num_latent_channels=1280,
     )
 

@@ -1,0 +1,20 @@
+// Python bindings for the perceiver_amd CDNA4 (gfx950) kernel extension.
+#include <torch/extension.h>
+
+torch::Tensor gelu_bias_fwd(torch::Tensor x, c10::optional<torch::Tensor> bias);
+torch::Tensor gelu_bias_bwd(torch::Tensor x, c10::optional<torch::Tensor> bias, torch::Tensor dy);
+
+std::vector<torch::Tensor> flash_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                     c10::optional<torch::Tensor> pad_mask, bool causal);
+std::vector<torch::Tensor> flash_bwd(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
+                                     torch::Tensor v, torch::Tensor out, torch::Tensor lse,
+                                     c10::optional<torch::Tensor> pad_mask, bool causal);
+bool flash_supported_impl(long d_qk, long d_v, long needs_dropout);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("gelu_bias_fwd", &gelu_bias_fwd, "fused bias+GELU forward (bf16)");
+    m.def("gelu_bias_bwd", &gelu_bias_bwd, "fused bias+GELU backward (bf16)");
+    m.def("flash_fwd", &flash_fwd, "fused flash attention forward");
+    m.def("flash_bwd", &flash_bwd, "fused flash attention backward");
+    m.def("flash_supported", &flash_supported_impl, "shape gate for the flash kernel");
+}
