@@ -1,0 +1,39 @@
+"""Autograd binding for the fused CDNA4 flash attention kernels."""
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+from perceiver_amd.ops import hip
+
+
+def _to_bf16(t: torch.Tensor) -> torch.Tensor:
+    return t if t.dtype == torch.bfloat16 else t.to(torch.bfloat16)
+
+
+class FlashAttention(torch.autograd.Function):
+    """out = softmax(mask(q k^T)) v with online softmax on device.
+
+    q arrives pre-scaled. pad_mask: (B, Lk) bool, True = padding. causal uses the
+    right-aligned convention (j > Lk - Nq + i masked). Saves (q, k, v, out, lse)
+    for the recompute-based backward.
+    """
+
+    @staticmethod
+    def forward(ctx, q, k, v, pad_mask: Optional[torch.Tensor], causal: bool,
+                dropout_p: float, training: bool):
+        q, k, v = _to_bf16(q.contiguous()), _to_bf16(k.contiguous()), _to_bf16(v.contiguous())
+        out, lse = hip.ext().flash_fwd(q, k, v, pad_mask, causal)
+        ctx.save_for_backward(q, k, v, out, lse)
+        ctx.pad_mask = pad_mask
+        ctx.causal = causal
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        q, k, v, out, lse = ctx.saved_tensors
+        dq, dk, dv = hip.ext().flash_bwd(
+            _to_bf16(dout), q, k, v, out, lse, ctx.pad_mask, ctx.causal
+        )
+        return dq, dk, dv, None, None, None, None

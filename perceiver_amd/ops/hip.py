@@ -61,7 +61,12 @@ def can_use_flash(q, k, v, dropout_p: float = 0.0, training: bool = False) -> bo
     if ext is None:
         return False
     d_qk, d_v = q.shape[-1], v.shape[-1]
-    if q.dtype not in (torch.bfloat16, torch.float16, torch.float32):
+    # bf16 only (the kernels are bf16-I/O, fp32-accumulate); fp32 inputs are allowed
+    # under autocast where casting to bf16 is the contract anyway
+    bf16_ok = q.dtype == torch.bfloat16 or (
+        q.dtype == torch.float32 and torch.is_autocast_enabled()
+    )
+    if not bf16_ok:
         return False
     return bool(ext.flash_supported(d_qk, d_v, int(training and dropout_p > 0)))
 
