@@ -94,9 +94,16 @@ class MLMBench:
         self.model = MaskedLanguageModel(cfg).to(device)
         self.device = device
         self.cfg = cfg
+        self.pure_bf16 = device.type == "cuda"
+        if self.pure_bf16:
+            # pure-bf16 weights + fp32-master AdamW: no autocast cast traffic
+            from perceiver_amd.train.optim import MasterAdamW, convert_to_bf16_training
 
-        self.opt = torch.optim.AdamW(self.model.parameters(), lr=2e-4, weight_decay=0.01,
-                                     betas=(0.9, 0.999), foreach=True)
+            self.model = convert_to_bf16_training(self.model)
+            self.opt = MasterAdamW(self.model.parameters(), lr=2e-4, weight_decay=0.01)
+        else:
+            self.opt = torch.optim.AdamW(self.model.parameters(), lr=2e-4, weight_decay=0.01,
+                                         betas=(0.9, 0.999), foreach=True)
         self.model.train()
 
         # synthetic batch: random bytes, 15% positions carry labels (masked-LM style)
@@ -121,10 +128,8 @@ class MLMBench:
             self.reducer = BucketedGradReducer(self.model)
 
     def step(self):
-        amp_dtype = torch.bfloat16
-        with torch.autocast(self.device.type, dtype=amp_dtype, enabled=self.device.type == "cuda"):
-            logits = self.model(self.x, self.pad)
-            loss = F.cross_entropy(logits.flatten(0, 1).float(), self.labels.flatten())
+        logits = self.model(self.x, self.pad)
+        loss = F.cross_entropy(logits.flatten(0, 1).float(), self.labels.flatten())
         self.opt.zero_grad(set_to_none=True)
         loss.backward()
         if getattr(self, "reducer", None) is not None:

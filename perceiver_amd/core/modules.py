@@ -35,6 +35,7 @@ from perceiver_amd.core.adapter import (
 from perceiver_amd.core.config import CausalSequenceModelConfig
 from perceiver_amd.core.position import positions
 from perceiver_amd.ops.attention import scaled_dot_attention
+from perceiver_amd.ops.norm import LayerNorm
 
 KVCache = Tuple[torch.Tensor, torch.Tensor]
 
@@ -162,8 +163,8 @@ class CrossAttention(nn.Module):
         out_bias: bool = True,
     ):
         super().__init__()
-        self.q_norm = nn.LayerNorm(num_q_input_channels)
-        self.kv_norm = nn.LayerNorm(num_kv_input_channels)
+        self.q_norm = LayerNorm(num_q_input_channels)
+        self.kv_norm = LayerNorm(num_kv_input_channels)
         self.attention = MultiHeadAttention(
             num_heads=num_heads,
             num_q_input_channels=num_q_input_channels,
@@ -215,7 +216,7 @@ class SelfAttention(nn.Module):
         out_bias: bool = True,
     ):
         super().__init__()
-        self.norm = nn.LayerNorm(num_channels)
+        self.norm = LayerNorm(num_channels)
         self.attention = MultiHeadAttention(
             num_heads=num_heads,
             num_q_input_channels=num_channels,
@@ -442,7 +443,7 @@ class MLP(nn.Sequential):
 
     def __init__(self, num_channels: int, widening_factor: int, bias: bool = True):
         super().__init__(
-            nn.LayerNorm(num_channels),
+            LayerNorm(num_channels),
             nn.Linear(num_channels, widening_factor * num_channels, bias=bias),
             nn.GELU(),
             nn.Linear(widening_factor * num_channels, num_channels, bias=bias),
@@ -792,7 +793,7 @@ class CausalSequenceModel(PerceiverAR):
         self.config = config
 
         if config.output_norm:
-            self.out_norm = nn.LayerNorm(config.num_channels)
+            self.out_norm = LayerNorm(config.num_channels)
 
         self.output_adapter = TiedTokenOutputAdapter(vocab_size=config.vocab_size, emb_bias=config.output_bias)
 

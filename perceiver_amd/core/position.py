@@ -61,7 +61,9 @@ class RotaryPositionEmbedding:
             pos_enc = self.frq_pos_enc[..., :seq_len, :]
 
         t_rot, t_pass = t[..., : self.rotate_dim], t[..., self.rotate_dim :]
-        t_rot = t_rot * pos_enc.cos() + rotate_half_interleaved(t_rot) * pos_enc.sin()
+        # rotation in the encoding dtype (fp32 tables under bf16 training), result
+        # cast back to t's dtype so the bf16 fused-kernel path stays bf16
+        t_rot = (t_rot * pos_enc.cos() + rotate_half_interleaved(t_rot) * pos_enc.sin()).to(t.dtype)
         return torch.cat((t_rot, t_pass), dim=-1)
 
 

@@ -55,7 +55,7 @@ class ImageInputAdapter(InputAdapter):
         if tuple(d) != self.image_shape:
             raise ValueError(f"Input vision shape {tuple(d)} different from required shape {self.image_shape}")
         x = x.flatten(1, len(self.image_shape) - 1)
-        return torch.cat([x, self.position_encoding(b)], dim=-1)
+        return torch.cat([x, self.position_encoding(b).to(x.dtype)], dim=-1)
 
 
 class ImageClassifier(PerceiverIO):

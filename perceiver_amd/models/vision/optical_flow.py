@@ -74,7 +74,7 @@ class OpticalFlowInputAdapter(InputAdapter):
         x = x.permute(0, 3, 4, 1, 2).reshape(b, h, w, t * c)
         x = self.linear(x)
         x = x.flatten(1, 2)
-        return torch.cat([x, self.position_encoding(b)], dim=-1)
+        return torch.cat([x, self.position_encoding(b).to(x.dtype)], dim=-1)
 
 
 class OpticalFlowOutputAdapter(OutputAdapter):

@@ -30,6 +30,19 @@ DEVINL unsigned short f2bf(float f) {
     return (unsigned short)(u >> 16);
 }
 
+// counter-based RNG (splitmix64 finalizer) for attention dropout: the same
+// (seed, bh, i, j) always yields the same draw, so the backward regenerates the
+// forward's mask exactly without storing it.
+DEVINL unsigned int rng_hash(unsigned long long seed, int bh, int i, int j) {
+    unsigned long long z = seed + (unsigned long long)(unsigned)bh * 0x9E3779B97F4A7C15ull +
+                           (unsigned long long)(unsigned)i * 0xBF58476D1CE4E5B9ull +
+                           (unsigned long long)(unsigned)j * 0x94D049BB133111EBull;
+    z ^= z >> 30; z *= 0xBF58476D1CE4E5B9ull;
+    z ^= z >> 27; z *= 0x94D049BB133111EBull;
+    z ^= z >> 31;
+    return (unsigned int)z;
+}
+
 // exact-erf GELU, matches torch.nn.GELU default
 DEVINL float gelu_f(float x) { return 0.5f * x * (1.0f + erff(x * 0.70710678118654752440f)); }
 

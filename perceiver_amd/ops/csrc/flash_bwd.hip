@@ -110,7 +110,8 @@ __global__ void flash_dq_kernel(
     const float* __restrict__ lsep, const float* __restrict__ deltap,
     const bool* __restrict__ pad,
     unsigned short* __restrict__ dqp,
-    int B, int H, int Nq, int Lk, int D, int Dv, int causal) {
+    int B, int H, int Nq, int Lk, int D, int Dv, int causal,
+    float drop_p, unsigned long long drop_seed) {
     const int d_pad = (D + 31) & ~31;
     const int dv_pad = (Dv + 31) & ~31;
     const int d_blocks = d_pad / 32;
@@ -239,7 +240,13 @@ __global__ void flash_dq_kernel(
                 bool masked = j >= Lk || (padrow && j < Lk && padrow[j]) ||
                               (causal && j > Lk - Nq + qi);
                 float p = masked ? 0.f : expf(s_acc[kb][r] - lse_r[r]);
-                float ds = p * (dp_acc[kb][r] - delta_r[r]);
+                float dprobs = dp_acc[kb][r];
+                if (drop_p > 0.f) {
+                    unsigned int thresh = (unsigned int)(drop_p * 4294967296.0);
+                    bool kept = rng_hash(drop_seed, bh, qi, j) >= thresh;
+                    dprobs = kept ? dprobs / (1.0f - drop_p) : 0.f;
+                }
+                float ds = p * (dprobs - delta_r[r]);
                 // store dS to per-wave LDS for A-layout reload (bf16)
                 *reinterpret_cast<unsigned short*>(p_mine + (hi4 * 4 + r) * kt_stride + (kb * 16 + lo16) * 2) =
                     f2bf(ds);
@@ -285,7 +292,8 @@ __global__ void flash_dkv_kernel(
     const float* __restrict__ lsep, const float* __restrict__ deltap,
     const bool* __restrict__ pad,
     unsigned short* __restrict__ dkp, unsigned short* __restrict__ dvp,
-    int B, int H, int Nq, int Lk, int D, int Dv, int causal) {
+    int B, int H, int Nq, int Lk, int D, int Dv, int causal,
+    float drop_p, unsigned long long drop_seed) {
     constexpr int QTILE = 32;
     const int d_pad = (D + 31) & ~31;
     const int dv_pad = (Dv + 31) & ~31;
@@ -433,11 +441,19 @@ __global__ void flash_dkv_kernel(
                 float lse_i = (qi < Nq) ? lse_row[qi] : 0.f;
                 float delta_i = (qi < Nq) ? delta_row[qi] : 0.f;
                 float p = masked ? 0.f : expf(st_acc[qb][r] - lse_i);
-                float ds = p * (dpt_acc[qb][r] - delta_i);
+                float p_eff = p;         // probs actually used in the forward PV
+                float dprobs = dpt_acc[qb][r];
+                if (drop_p > 0.f) {
+                    unsigned int thresh = (unsigned int)(drop_p * 4294967296.0);
+                    bool kept = rng_hash(drop_seed, bh, qi, ki) >= thresh;
+                    float inv_keep = 1.0f / (1.0f - drop_p);
+                    p_eff = kept ? p * inv_keep : 0.f;
+                    dprobs = kept ? dprobs * inv_keep : 0.f;
+                }
+                float ds = p * (dprobs - delta_i);
                 char* slot = p_mine + (hi4 * 4 + r) * qt_stride + (qb * 16 + lo16) * 2;
-                // pack p (low half rows 0..15? no): we need TWO buffers; reuse one
-                // sequentially: first P^T pass for dV, then dS^T pass for dK.
-                *reinterpret_cast<unsigned short*>(slot) = f2bf(p);
+                // one buffer used twice: first P^T (for dV), then dS^T (for dK)
+                *reinterpret_cast<unsigned short*>(slot) = f2bf(p_eff);
                 // stash ds in registers for the second pass
                 st_acc[qb][r] = ds;  // reuse st_acc as ds storage
             }
@@ -508,6 +524,7 @@ template <int DMAX, int DVMAX>
 void launch_flash_bwd(const torch::Tensor& dout, const torch::Tensor& q, const torch::Tensor& k,
                       const torch::Tensor& v, const torch::Tensor& lse, const torch::Tensor& delta,
                       const c10::optional<torch::Tensor>& pad_mask, bool causal,
+                      float drop_p, unsigned long long drop_seed,
                       torch::Tensor& dq, torch::Tensor& dk, torch::Tensor& dv) {
     int B = q.size(0), H = q.size(1), Nq = q.size(2), D = q.size(3);
     int Lk = k.size(2), Dv = v.size(3);
@@ -529,7 +546,7 @@ void launch_flash_bwd(const torch::Tensor& dout, const torch::Tensor& q, const t
                            reinterpret_cast<const unsigned short*>(dout.data_ptr()),
                            lse.data_ptr<float>(), delta.data_ptr<float>(), padp,
                            reinterpret_cast<unsigned short*>(dq.data_ptr()),
-                           B, H, Nq, Lk, D, Dv, (int)causal);
+                           B, H, Nq, Lk, D, Dv, (int)causal, drop_p, drop_seed);
         HIP_CHECK_LAST();
     }
     {   // dK/dV
@@ -548,7 +565,7 @@ void launch_flash_bwd(const torch::Tensor& dout, const torch::Tensor& q, const t
                            lse.data_ptr<float>(), delta.data_ptr<float>(), padp,
                            reinterpret_cast<unsigned short*>(dk.data_ptr()),
                            reinterpret_cast<unsigned short*>(dv.data_ptr()),
-                           B, H, Nq, Lk, D, Dv, (int)causal);
+                           B, H, Nq, Lk, D, Dv, (int)causal, drop_p, drop_seed);
         HIP_CHECK_LAST();
     }
 }
@@ -557,7 +574,8 @@ void launch_flash_bwd(const torch::Tensor& dout, const torch::Tensor& q, const t
 
 std::vector<torch::Tensor> flash_bwd(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
                                      torch::Tensor v, torch::Tensor out, torch::Tensor lse,
-                                     c10::optional<torch::Tensor> pad_mask, bool causal) {
+                                     c10::optional<torch::Tensor> pad_mask, bool causal,
+                                     double dropout_p, int64_t seed) {
     TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
     dout = dout.contiguous(); q = q.contiguous(); k = k.contiguous(); v = v.contiguous();
     out = out.contiguous();
@@ -584,11 +602,13 @@ std::vector<torch::Tensor> flash_bwd(torch::Tensor dout, torch::Tensor q, torch:
     c10::optional<torch::Tensor> pm;
     if (pad_mask.has_value() && pad_mask->defined()) pm = pad_mask->contiguous();
 
-    if (D <= 32 && Dv <= 160)       launch_flash_bwd<32, 160>(dout, q, k, v, lse, delta, pm, causal, dq, dk, dv);
-    else if (D <= 64 && Dv <= 64)   launch_flash_bwd<64, 64>(dout, q, k, v, lse, delta, pm, causal, dq, dk, dv);
-    else if (D <= 128 && Dv <= 128) launch_flash_bwd<128, 128>(dout, q, k, v, lse, delta, pm, causal, dq, dk, dv);
-    else if (D <= 160 && Dv <= 160) launch_flash_bwd<160, 160>(dout, q, k, v, lse, delta, pm, causal, dq, dk, dv);
-    else                            launch_flash_bwd<352, 352>(dout, q, k, v, lse, delta, pm, causal, dq, dk, dv);
+    float dp = (float)dropout_p;
+    unsigned long long sd = (unsigned long long)seed;
+    if (D <= 32 && Dv <= 160)       launch_flash_bwd<32, 160>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
+    else if (D <= 64 && Dv <= 64)   launch_flash_bwd<64, 64>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
+    else if (D <= 128 && Dv <= 128) launch_flash_bwd<128, 128>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
+    else if (D <= 160 && Dv <= 160) launch_flash_bwd<160, 160>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
+    else                            launch_flash_bwd<352, 352>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
 
     return {dq, dk, dv};
 }

@@ -111,7 +111,8 @@ __global__ void flash_fwd_kernel(
     const bool* __restrict__ pad,           // (B,Lk) or null
     unsigned short* __restrict__ op,        // (B,H,Nq,Dv)
     float* __restrict__ lsep,               // (B,H,Nq)
-    int B, int H, int Nq, int Lk, int D, int Dv, int causal) {
+    int B, int H, int Nq, int Lk, int D, int Dv, int causal,
+    float drop_p, unsigned long long drop_seed) {
     const int d_pad = (D + 31) & ~31;
     const int dv_pad = (Dv + 15) & ~15;
     const int d_blocks = d_pad / 32;       // QK^T k-steps
@@ -249,6 +250,21 @@ __global__ void flash_fwd_kernel(
             for (int r = 0; r < 4; ++r) o_acc[cb][r] *= alpha[r];
         }
 
+        // ---- attention dropout on the (unnormalized) probabilities; the softmax
+        // denominator l uses the undropped sum, matching softmax->dropout order ----
+        if (drop_p > 0.f) {
+            unsigned int thresh = (unsigned int)(drop_p * 4294967296.0);
+#pragma unroll
+            for (int kb = 0; kb < 2; ++kb) {
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    int qi = q0 + hi4 * 4 + r;
+                    int j = kv0 + kb * 16 + lo16;
+                    if (rng_hash(drop_seed, bh, qi, j) < thresh) pvals[kb][r] = 0.f;
+                }
+            }
+        }
+
         // ---- redistribute P (C layout) -> A layout via per-wave LDS ----
         // write: lane holds P[row hi4*4+r][key lo16+16*kb]
 #pragma unroll
@@ -282,7 +298,8 @@ __global__ void flash_fwd_kernel(
     for (int r = 0; r < 4; ++r) {
         int qi = q0 + hi4 * 4 + r;
         if (qi >= Nq) continue;
-        float inv_l = (l_run[r] > 0.f) ? 1.0f / l_run[r] : 0.f;
+        float l_eff = l_run[r] * (drop_p > 0.f ? (1.0f - drop_p) : 1.0f);
+        float inv_l = (l_eff > 0.f) ? 1.0f / l_eff : 0.f;
         unsigned short* orow = op + ((long)bh * Nq + qi) * Dv;
 #pragma unroll
         for (int cb = 0; cb < DVMAX / 16; ++cb) {
@@ -298,6 +315,7 @@ __global__ void flash_fwd_kernel(
 template <int DMAX, int DVMAX>
 void launch_flash_fwd(const torch::Tensor& q, const torch::Tensor& k, const torch::Tensor& v,
                       const c10::optional<torch::Tensor>& pad_mask, bool causal,
+                      float drop_p, unsigned long long drop_seed,
                       torch::Tensor& out, torch::Tensor& lse) {
     int B = q.size(0), H = q.size(1), Nq = q.size(2), D = q.size(3);
     int Lk = k.size(2), Dv = v.size(3);
@@ -319,19 +337,21 @@ void launch_flash_fwd(const torch::Tensor& q, const torch::Tensor& k, const torc
                        reinterpret_cast<const unsigned short*>(v.data_ptr()),
                        padp,
                        reinterpret_cast<unsigned short*>(out.data_ptr()),
-                       lse.data_ptr<float>(), B, H, Nq, Lk, D, Dv, (int)causal);
+                       lse.data_ptr<float>(), B, H, Nq, Lk, D, Dv, (int)causal,
+                       drop_p, drop_seed);
     HIP_CHECK_LAST();
 }
 
 }  // namespace
 
 bool flash_supported_impl(long d_qk, long d_v, long needs_dropout) {
-    if (needs_dropout) return false;  // dropout handled by the eager path for now
+    (void)needs_dropout;  // in-kernel counter-hash dropout
     return d_qk <= 352 && d_v <= 352;
 }
 
 std::vector<torch::Tensor> flash_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
-                                     c10::optional<torch::Tensor> pad_mask, bool causal) {
+                                     c10::optional<torch::Tensor> pad_mask, bool causal,
+                                     double dropout_p, int64_t seed) {
     TORCH_CHECK(q.is_cuda() && k.is_cuda() && v.is_cuda());
     TORCH_CHECK(q.scalar_type() == torch::kBFloat16, "flash_fwd: bf16 only");
     TORCH_CHECK(q.dim() == 4 && k.dim() == 4 && v.dim() == 4);
@@ -347,11 +367,13 @@ std::vector<torch::Tensor> flash_fwd(torch::Tensor q, torch::Tensor k, torch::Te
         pm = pad_mask->contiguous();
     }
 
-    if (D <= 32 && Dv <= 160)       launch_flash_fwd<32, 160>(q, k, v, pm, causal, out, lse);
-    else if (D <= 64 && Dv <= 64)   launch_flash_fwd<64, 64>(q, k, v, pm, causal, out, lse);
-    else if (D <= 128 && Dv <= 128) launch_flash_fwd<128, 128>(q, k, v, pm, causal, out, lse);
-    else if (D <= 160 && Dv <= 160) launch_flash_fwd<160, 160>(q, k, v, pm, causal, out, lse);
-    else                            launch_flash_fwd<352, 352>(q, k, v, pm, causal, out, lse);
+    float dp = (float)dropout_p;
+    unsigned long long sd = (unsigned long long)seed;
+    if (D <= 32 && Dv <= 160)       launch_flash_fwd<32, 160>(q, k, v, pm, causal, dp, sd, out, lse);
+    else if (D <= 64 && Dv <= 64)   launch_flash_fwd<64, 64>(q, k, v, pm, causal, dp, sd, out, lse);
+    else if (D <= 128 && Dv <= 128) launch_flash_fwd<128, 128>(q, k, v, pm, causal, dp, sd, out, lse);
+    else if (D <= 160 && Dv <= 160) launch_flash_fwd<160, 160>(q, k, v, pm, causal, dp, sd, out, lse);
+    else                            launch_flash_fwd<352, 352>(q, k, v, pm, causal, dp, sd, out, lse);
 
     return {out, lse};
 }

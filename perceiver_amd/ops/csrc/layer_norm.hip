@@ -1,0 +1,264 @@
+// Fused bf16 LayerNorm forward/backward for CDNA4 (the LN half of SURVEY.md §2.3 K6).
+// Memory-bound: one wave64 per row, ushort8 vector loads, fp32 accumulation,
+// cross-lane shfl_xor reductions; saves per-row mean/rstd for the backward.
+// Register caching uses compile-time CHUNKS (static indexing — guide §5.4 rule 20);
+// dW/db come from a per-block partial-reduction kernel + a final torch sum.
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+#include "common.h"
+
+namespace {
+
+DEVINL float wave_sum(float x) {
+#pragma unroll
+    for (int m = 1; m < 64; m <<= 1) x += __shfl_xor(x, m, 64);
+    return x;
+}
+
+// CHUNKS = ceil(C / 512); lane handles 8 contiguous elems per chunk.
+template <int CHUNKS>
+__global__ void ln_fwd_kernel(const unsigned short* __restrict__ x,
+                              const unsigned short* __restrict__ w,
+                              const unsigned short* __restrict__ b,
+                              unsigned short* __restrict__ y,
+                              float* __restrict__ mean_out, float* __restrict__ rstd_out,
+                              long rows, int C, float eps) {
+    long row = (long)blockIdx.x * (blockDim.x / 64) + threadIdx.x / 64;
+    int lane = threadIdx.x % 64;
+    if (row >= rows) return;
+    const unsigned short* xrow = x + row * C;
+    unsigned short* yrow = y + row * C;
+
+    float vals[CHUNKS][8];
+    float sum = 0.f, sumsq = 0.f;
+#pragma unroll
+    for (int i = 0; i < CHUNKS; ++i) {
+        int c0 = lane * 8 + i * 512;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) vals[i][e] = 0.f;
+        if (c0 + 8 <= C) {
+            short8v v = *reinterpret_cast<const short8v*>(xrow + c0);
+#pragma unroll
+            for (int e = 0; e < 8; ++e) {
+                float f = bf2f((unsigned short)v[e]);
+                vals[i][e] = f;
+                sum += f;
+                sumsq += f * f;
+            }
+        } else if (c0 < C) {
+            for (int e = 0; e < C - c0; ++e) {
+                float f = bf2f(xrow[c0 + e]);
+                vals[i][e] = f;
+                sum += f;
+                sumsq += f * f;
+            }
+        }
+    }
+    sum = wave_sum(sum);
+    sumsq = wave_sum(sumsq);
+    float mean = sum / C;
+    float var = sumsq / C - mean * mean;
+    float rstd = rsqrtf(fmaxf(var, 0.f) + eps);
+    if (lane == 0) {
+        mean_out[row] = mean;
+        rstd_out[row] = rstd;
+    }
+
+#pragma unroll
+    for (int i = 0; i < CHUNKS; ++i) {
+        int c0 = lane * 8 + i * 512;
+        if (c0 + 8 <= C) {
+            short8v wv = *reinterpret_cast<const short8v*>(w + c0);
+            short8v o;
+#pragma unroll
+            for (int e = 0; e < 8; ++e) {
+                float bb = b ? bf2f(b[c0 + e]) : 0.f;
+                o[e] = (short)f2bf((vals[i][e] - mean) * rstd * bf2f((unsigned short)wv[e]) + bb);
+            }
+            *reinterpret_cast<short8v*>(yrow + c0) = o;
+        } else if (c0 < C) {
+            for (int e = 0; e < C - c0; ++e) {
+                float bb = b ? bf2f(b[c0 + e]) : 0.f;
+                yrow[c0 + e] = f2bf((vals[i][e] - mean) * rstd * bf2f(w[c0 + e]) + bb);
+            }
+        }
+    }
+}
+
+template <int CHUNKS>
+__global__ void ln_bwd_dx_kernel(const unsigned short* __restrict__ dy,
+                                 const unsigned short* __restrict__ x,
+                                 const unsigned short* __restrict__ w,
+                                 const float* __restrict__ mean_in,
+                                 const float* __restrict__ rstd_in,
+                                 unsigned short* __restrict__ dx,
+                                 long rows, int C) {
+    long row = (long)blockIdx.x * (blockDim.x / 64) + threadIdx.x / 64;
+    int lane = threadIdx.x % 64;
+    if (row >= rows) return;
+    const unsigned short* dyrow = dy + row * C;
+    const unsigned short* xrow = x + row * C;
+    unsigned short* dxrow = dx + row * C;
+    float mean = mean_in[row], rstd = rstd_in[row];
+
+    float g[CHUNKS][8], xh[CHUNKS][8];
+    float s1 = 0.f, s2 = 0.f;
+#pragma unroll
+    for (int i = 0; i < CHUNKS; ++i) {
+        int c0 = lane * 8 + i * 512;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) { g[i][e] = 0.f; xh[i][e] = 0.f; }
+        if (c0 + 8 <= C) {
+            short8v dyv = *reinterpret_cast<const short8v*>(dyrow + c0);
+            short8v xv = *reinterpret_cast<const short8v*>(xrow + c0);
+            short8v wv = *reinterpret_cast<const short8v*>(w + c0);
+#pragma unroll
+            for (int e = 0; e < 8; ++e) {
+                float gg = bf2f((unsigned short)dyv[e]) * bf2f((unsigned short)wv[e]);
+                float xhat = (bf2f((unsigned short)xv[e]) - mean) * rstd;
+                g[i][e] = gg;
+                xh[i][e] = xhat;
+                s1 += gg;
+                s2 += gg * xhat;
+            }
+        } else if (c0 < C) {
+            for (int e = 0; e < C - c0; ++e) {
+                float gg = bf2f(dyrow[c0 + e]) * bf2f(w[c0 + e]);
+                float xhat = (bf2f(xrow[c0 + e]) - mean) * rstd;
+                g[i][e] = gg;
+                xh[i][e] = xhat;
+                s1 += gg;
+                s2 += gg * xhat;
+            }
+        }
+    }
+    s1 = wave_sum(s1) / C;
+    s2 = wave_sum(s2) / C;
+
+#pragma unroll
+    for (int i = 0; i < CHUNKS; ++i) {
+        int c0 = lane * 8 + i * 512;
+        if (c0 + 8 <= C) {
+            short8v o;
+#pragma unroll
+            for (int e = 0; e < 8; ++e)
+                o[e] = (short)f2bf(rstd * (g[i][e] - s1 - xh[i][e] * s2));
+            *reinterpret_cast<short8v*>(dxrow + c0) = o;
+        } else if (c0 < C) {
+            for (int e = 0; e < C - c0; ++e)
+                dxrow[c0 + e] = f2bf(rstd * (g[i][e] - s1 - xh[i][e] * s2));
+        }
+    }
+}
+
+constexpr int RB = 64;  // rows per partial-reduce block
+
+__global__ void ln_bwd_dwdb_kernel(const unsigned short* __restrict__ dy,
+                                   const unsigned short* __restrict__ x,
+                                   const float* __restrict__ mean_in,
+                                   const float* __restrict__ rstd_in,
+                                   float* __restrict__ dw_part, float* __restrict__ db_part,
+                                   long rows, int C) {
+    long r0 = (long)blockIdx.x * RB;
+    long r1 = min(rows, r0 + RB);
+    for (int c = threadIdx.x; c < C; c += blockDim.x) {
+        float dw = 0.f, db = 0.f;
+        for (long r = r0; r < r1; ++r) {
+            float dyv = bf2f(dy[r * C + c]);
+            float xhat = (bf2f(x[r * C + c]) - mean_in[r]) * rstd_in[r];
+            dw += dyv * xhat;
+            db += dyv;
+        }
+        dw_part[(long)blockIdx.x * C + c] = dw;
+        db_part[(long)blockIdx.x * C + c] = db;
+    }
+}
+
+template <int CHUNKS>
+void launch_ln_fwd(const torch::Tensor& x, const torch::Tensor& w, const unsigned short* bp,
+                   torch::Tensor& y, torch::Tensor& mean, torch::Tensor& rstd,
+                   long rows, int C, float eps) {
+    int wpb = 4;
+    long blocks = (rows + wpb - 1) / wpb;
+    hipLaunchKernelGGL((ln_fwd_kernel<CHUNKS>), dim3(blocks), dim3(64 * wpb), 0,
+                       at::cuda::getCurrentCUDAStream(),
+                       reinterpret_cast<const unsigned short*>(x.data_ptr()),
+                       reinterpret_cast<const unsigned short*>(w.data_ptr()), bp,
+                       reinterpret_cast<unsigned short*>(y.data_ptr()),
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(), rows, C, eps);
+    HIP_CHECK_LAST();
+}
+
+template <int CHUNKS>
+void launch_ln_bwd_dx(const torch::Tensor& dy, const torch::Tensor& x, const torch::Tensor& w,
+                      const torch::Tensor& mean, const torch::Tensor& rstd, torch::Tensor& dx,
+                      long rows, int C) {
+    int wpb = 4;
+    long blocks = (rows + wpb - 1) / wpb;
+    hipLaunchKernelGGL((ln_bwd_dx_kernel<CHUNKS>), dim3(blocks), dim3(64 * wpb), 0,
+                       at::cuda::getCurrentCUDAStream(),
+                       reinterpret_cast<const unsigned short*>(dy.data_ptr()),
+                       reinterpret_cast<const unsigned short*>(x.data_ptr()),
+                       reinterpret_cast<const unsigned short*>(w.data_ptr()),
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       reinterpret_cast<unsigned short*>(dx.data_ptr()), rows, C);
+    HIP_CHECK_LAST();
+}
+
+}  // namespace
+
+std::vector<torch::Tensor> ln_fwd(torch::Tensor x, torch::Tensor w, c10::optional<torch::Tensor> b,
+                                  double eps) {
+    TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16);
+    x = x.contiguous();
+    auto wc = w.contiguous();
+    int C = x.size(-1);
+    TORCH_CHECK(C <= 2048, "ln_fwd: C must be <= 2048");
+    long rows = x.numel() / C;
+    auto y = torch::empty_like(x);
+    auto mean = torch::empty({rows}, x.options().dtype(torch::kFloat32));
+    auto rstd = torch::empty_like(mean);
+    const unsigned short* bp = nullptr;
+    torch::Tensor bc;
+    if (b.has_value() && b->defined()) {
+        bc = b->contiguous();
+        bp = reinterpret_cast<const unsigned short*>(bc.data_ptr());
+    }
+    int chunks = (C + 511) / 512;
+    if (chunks == 1)      launch_ln_fwd<1>(x, wc, bp, y, mean, rstd, rows, C, (float)eps);
+    else if (chunks == 2) launch_ln_fwd<2>(x, wc, bp, y, mean, rstd, rows, C, (float)eps);
+    else if (chunks == 3) launch_ln_fwd<3>(x, wc, bp, y, mean, rstd, rows, C, (float)eps);
+    else                  launch_ln_fwd<4>(x, wc, bp, y, mean, rstd, rows, C, (float)eps);
+    return {y, mean, rstd};
+}
+
+std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor w,
+                                  torch::Tensor mean, torch::Tensor rstd, bool needs_dwdb) {
+    dy = dy.contiguous(); x = x.contiguous();
+    auto wc = w.contiguous();
+    int C = x.size(-1);
+    long rows = x.numel() / C;
+    auto dx = torch::empty_like(x);
+    int chunks = (C + 511) / 512;
+    if (chunks == 1)      launch_ln_bwd_dx<1>(dy, x, wc, mean, rstd, dx, rows, C);
+    else if (chunks == 2) launch_ln_bwd_dx<2>(dy, x, wc, mean, rstd, dx, rows, C);
+    else if (chunks == 3) launch_ln_bwd_dx<3>(dy, x, wc, mean, rstd, dx, rows, C);
+    else                  launch_ln_bwd_dx<4>(dy, x, wc, mean, rstd, dx, rows, C);
+
+    torch::Tensor dw, db;
+    if (needs_dwdb) {
+        long nblocks = (rows + RB - 1) / RB;
+        auto dw_part = torch::empty({nblocks, (long)C}, x.options().dtype(torch::kFloat32));
+        auto db_part = torch::empty_like(dw_part);
+        hipLaunchKernelGGL(ln_bwd_dwdb_kernel, dim3(nblocks), dim3(256), 0,
+                           at::cuda::getCurrentCUDAStream(),
+                           reinterpret_cast<const unsigned short*>(dy.data_ptr()),
+                           reinterpret_cast<const unsigned short*>(x.data_ptr()),
+                           mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                           dw_part.data_ptr<float>(), db_part.data_ptr<float>(), rows, C);
+        HIP_CHECK_LAST();
+        dw = dw_part.sum(0).to(x.scalar_type());
+        db = db_part.sum(0).to(x.scalar_type());
+    }
+    return {dx, dw, db};
+}

@@ -5,10 +5,16 @@ torch::Tensor gelu_bias_fwd(torch::Tensor x, c10::optional<torch::Tensor> bias);
 torch::Tensor gelu_bias_bwd(torch::Tensor x, c10::optional<torch::Tensor> bias, torch::Tensor dy);
 
 std::vector<torch::Tensor> flash_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
-                                     c10::optional<torch::Tensor> pad_mask, bool causal);
+                                     c10::optional<torch::Tensor> pad_mask, bool causal,
+                                     double dropout_p, int64_t seed);
 std::vector<torch::Tensor> flash_bwd(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
                                      torch::Tensor v, torch::Tensor out, torch::Tensor lse,
-                                     c10::optional<torch::Tensor> pad_mask, bool causal);
+                                     c10::optional<torch::Tensor> pad_mask, bool causal,
+                                     double dropout_p, int64_t seed);
+std::vector<torch::Tensor> ln_fwd(torch::Tensor x, torch::Tensor w, c10::optional<torch::Tensor> b,
+                                  double eps);
+std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor w,
+                                  torch::Tensor mean, torch::Tensor rstd, bool needs_dwdb);
 bool flash_supported_impl(long d_qk, long d_v, long needs_dropout);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -17,4 +23,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("flash_fwd", &flash_fwd, "fused flash attention forward");
     m.def("flash_bwd", &flash_bwd, "fused flash attention backward");
     m.def("flash_supported", &flash_supported_impl, "shape gate for the flash kernel");
+    m.def("ln_fwd", &ln_fwd, "fused bf16 LayerNorm forward");
+    m.def("ln_bwd", &ln_bwd, "fused bf16 LayerNorm backward");
 }

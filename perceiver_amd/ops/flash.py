@@ -24,16 +24,20 @@ class FlashAttention(torch.autograd.Function):
     def forward(ctx, q, k, v, pad_mask: Optional[torch.Tensor], causal: bool,
                 dropout_p: float, training: bool):
         q, k, v = _to_bf16(q.contiguous()), _to_bf16(k.contiguous()), _to_bf16(v.contiguous())
-        out, lse = hip.ext().flash_fwd(q, k, v, pad_mask, causal)
+        p = float(dropout_p) if training else 0.0
+        seed = int(torch.randint(0, 2**62, (1,)).item()) if p > 0 else 0
+        out, lse = hip.ext().flash_fwd(q, k, v, pad_mask, causal, p, seed)
         ctx.save_for_backward(q, k, v, out, lse)
         ctx.pad_mask = pad_mask
         ctx.causal = causal
+        ctx.dropout = (p, seed)
         return out
 
     @staticmethod
     def backward(ctx, dout):
         q, k, v, out, lse = ctx.saved_tensors
+        p, seed = ctx.dropout
         dq, dk, dv = hip.ext().flash_bwd(
-            _to_bf16(dout), q, k, v, out, lse, ctx.pad_mask, ctx.causal
+            _to_bf16(dout), q, k, v, out, lse, ctx.pad_mask, ctx.causal, p, seed
         )
         return dq, dk, dv, None, None, None, None
