@@ -72,7 +72,18 @@ def scaled_dot_attention(
     training: bool = False,
     max_heads_parallel: Optional[int] = None,
 ) -> torch.Tensor:
-    """Dispatching attention core. Shapes as in :func:`eager_attention`."""
+    """Dispatching attention core. Shapes as in :func:`eager_attention`.
+
+    Batch-1 queries (learned latent/output-query arrays) broadcast against
+    batch-N keys, matching the reference's einsum broadcasting."""
+    if q.shape[0] != k.shape[0]:
+        if q.shape[0] == 1:
+            q = q.expand(k.shape[0], *q.shape[1:])
+        elif k.shape[0] == 1:
+            k = k.expand(q.shape[0], *k.shape[1:])
+            v = v.expand(q.shape[0], *v.shape[1:])
+        else:
+            raise ValueError(f"incompatible batch sizes {q.shape[0]} vs {k.shape[0]}")
     if q.is_cuda:
         from perceiver_amd.ops import hip
 

@@ -159,6 +159,7 @@ def test_model_forward_uses_flash_and_matches_cpu():
         gm = model.cuda()
         with torch.autocast("cuda", dtype=torch.bfloat16):
             got = gm(x.cuda(), torch.zeros(2, 256, dtype=torch.bool, device="cuda"))
+    assert got.shape == ref.shape, f"shape mismatch {got.shape} vs {ref.shape}"
     err = (got.float().cpu() - ref).abs().max().item()
     assert err < 0.25, f"GPU/CPU logits diverge: {err}"
 
