@@ -361,6 +361,12 @@ std::vector<torch::Tensor> flash_fwd(torch::Tensor q, torch::Tensor k, torch::Te
 
     auto out = torch::empty({q.size(0), q.size(1), q.size(2), (long)Dv}, q.options());
     auto lse = torch::empty({q.size(0), q.size(1), q.size(2)}, q.options().dtype(torch::kFloat32));
+    if (q.numel() == 0 || k.numel() == 0) {
+        // degenerate shapes fall back to the eager path upstream; just return zeros
+        out.zero_();
+        lse.zero_();
+        return {out, lse};
+    }
 
     c10::optional<torch::Tensor> pm;
     if (pad_mask.has_value() && pad_mask->defined()) {

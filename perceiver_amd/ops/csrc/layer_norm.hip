@@ -218,6 +218,7 @@ std::vector<torch::Tensor> ln_fwd(torch::Tensor x, torch::Tensor w, c10::optiona
     auto y = torch::empty_like(x);
     auto mean = torch::empty({rows}, x.options().dtype(torch::kFloat32));
     auto rstd = torch::empty_like(mean);
+    if (rows == 0) return {y, mean, rstd};
     const unsigned short* bp = nullptr;
     torch::Tensor bc;
     if (b.has_value() && b->defined()) {
@@ -239,6 +240,7 @@ std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor x, torch::Tens
     int C = x.size(-1);
     long rows = x.numel() / C;
     auto dx = torch::empty_like(x);
+    if (rows == 0) return {dx, torch::zeros_like(w), torch::zeros_like(w)};
     int chunks = (C + 511) / 512;
     if (chunks == 1)      launch_ln_bwd_dx<1>(dy, x, wc, mean, rstd, dx, rows, C);
     else if (chunks == 2) launch_ln_bwd_dx<2>(dy, x, wc, mean, rstd, dx, rows, C);
