@@ -2,24 +2,27 @@
 // (atomic-free): dQ parallel over Q blocks, dK/dV parallel over KV blocks.
 //
 //   delta_i = sum_c dO[i][c] * O[i][c]
-//   P       = exp(QK^T - lse_i)            (masks reapplied)
-//   dS      = P * (dO V^T - delta_i)
-//   dQ      = dS K          dK = dS^T Q          dV = P^T dO
+//   P       = exp(QK^T - lse_i)            (masks + dropout reapplied)
+//   dS      = P * (dO V^T * mask/(1-p) - delta_i)
+//   dQ      = dS K          dK = dS^T Q          dV = (P*mask/(1-p))^T dO
 //
-// Layouts mirror flash_fwd.hip: 4 waves x 16 rows, KVBLK=32 tiles, LDS staging with
-// +16 B row padding, mfma_f32_16x16x32_bf16, C-layout -> A-layout redistribution of
-// P/dS through per-wave LDS. q arrives pre-scaled so no extra scale appears here.
+// v2 layouts mirror flash_fwd.hip: 4 waves x 16 rows, 64-key/row tiles, vectorized
+// transposed staging (4-source-row packed b64 writes), mfma_f32_16x16x32_bf16,
+// C-layout -> A-layout redistribution of P/dS through per-wave LDS.
+// q arrives pre-scaled so no extra scale appears here.
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
 #include <cfloat>
 #include "common.h"
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) short short4x;
 
 namespace {
 
-constexpr int KVBLK = 32;
-constexpr int QROWS = 16;
+constexpr int TILE = 64;       // keys per tile (dq) / q rows per tile (dkv)
+constexpr int TBLKS = TILE / 16;
+constexpr int QROWS = 16;      // rows per wave (q rows in dq, key rows in dkv)
 constexpr int NWAVES = 4;
 constexpr int QBLK = QROWS * NWAVES;
 
@@ -27,8 +30,7 @@ constexpr int QBLK = QROWS * NWAVES;
 __global__ void delta_kernel(const unsigned short* __restrict__ dout,
                              const unsigned short* __restrict__ out,
                              float* __restrict__ delta, long rows, int dv) {
-    // one 64-lane wave per row chunk: thread covers one row with a strided loop
-    long row = blockIdx.x * (blockDim.x / 64) + threadIdx.x / 64;
+    long row = (long)blockIdx.x * (blockDim.x / 64) + threadIdx.x / 64;
     int lane = threadIdx.x % 64;
     if (row >= rows) return;
     const unsigned short* a = dout + row * dv;
@@ -43,19 +45,13 @@ __global__ void delta_kernel(const unsigned short* __restrict__ dout,
     if (lane == 0) delta[row] = acc;
 }
 
-DEVINL float warp16_sum(float x) {
-#pragma unroll
-    for (int m = 1; m < 16; m <<= 1) x += __shfl_xor(x, m, 64);
-    return x;
-}
-
-// stage (rows x d) tile row-major into LDS (row stride ldst_bytes), zero-pad
+// stage (rows_tile x d) tile row-major into LDS (row stride ldst_bytes), zero-pad
 DEVINL void stage_rm(const unsigned short* __restrict__ src, long src_stride,
                      int rows_valid, int rows_tile, int d, int d_pad,
-                     char* lds, int ldst_bytes, int tid, int nthreads) {
+                     char* lds, int ldst_bytes, int tid) {
     const int gpr = d_pad / 8;
     const int total = rows_tile * gpr;
-    for (int g = tid; g < total; g += nthreads) {
+    for (int g = tid; g < total; g += 256) {
         int row = g / gpr;
         int c0 = (g % gpr) * 8;
         short8v val = {};
@@ -72,35 +68,42 @@ DEVINL void stage_rm(const unsigned short* __restrict__ src, long src_stride,
     }
 }
 
-// stage (rows x d) tile TRANSPOSED into LDS: lds row = channel (d_pad rows),
-// col = source row (rows_tile cols, stride ldst_bytes = rows_tile*2+16)
+// stage (rows_tile x d) tile TRANSPOSED: lds row = channel (d_pad rows), col =
+// source row; 4-source-row packed b64 writes.
 DEVINL void stage_tr(const unsigned short* __restrict__ src, long src_stride,
                      int rows_valid, int rows_tile, int d, int d_pad,
-                     char* ldsT, int ldst_bytes, int tid, int nthreads) {
+                     char* ldsT, int ldst_bytes, int tid) {
     const int gpr = d_pad / 8;
-    const int total = rows_tile * gpr;
-    for (int g = tid; g < total; g += nthreads) {
-        int row = g / gpr;
+    const int total = (rows_tile / 4) * gpr;
+    for (int g = tid; g < total; g += 256) {
+        int row0 = (g / gpr) * 4;
         int c0 = (g % gpr) * 8;
-        short8v val = {};
-        if (row < rows_valid && c0 < d) {
-            if (c0 + 8 <= d) {
-                val = *reinterpret_cast<const short8v*>(src + (long)row * src_stride + c0);
-            } else {
+        short8v rows[4];
 #pragma unroll
-                for (int e = 0; e < 8; ++e)
-                    val[e] = (c0 + e < d) ? (short)src[(long)row * src_stride + c0 + e] : (short)0;
+        for (int r = 0; r < 4; ++r) {
+            short8v val = {};
+            int row = row0 + r;
+            if (row < rows_valid && c0 < d) {
+                if (c0 + 8 <= d) {
+                    val = *reinterpret_cast<const short8v*>(src + (long)row * src_stride + c0);
+                } else {
+#pragma unroll
+                    for (int e = 0; e < 8; ++e)
+                        val[e] = (c0 + e < d) ? (short)src[(long)row * src_stride + c0 + e] : (short)0;
+                }
             }
+            rows[r] = val;
         }
 #pragma unroll
-        for (int e = 0; e < 8; ++e)
-            *reinterpret_cast<unsigned short*>(ldsT + (c0 + e) * ldst_bytes + row * 2) =
-                (unsigned short)val[e];
+        for (int e = 0; e < 8; ++e) {
+            short4x pack = {rows[0][e], rows[1][e], rows[2][e], rows[3][e]};
+            *reinterpret_cast<short4x*>(ldsT + (c0 + e) * ldst_bytes + row0 * 2) = pack;
+        }
     }
 }
 
 // ---------------------------------------------------------------- dQ kernel
-// grid.x over Q blocks, grid.y = B*H. Stages per KV tile: K row-major (for S),
+// grid.x over Q blocks, grid.y = B*H. Stages per 64-key tile: K row-major (for S),
 // K^T (for dQ = dS K), V row-major (for dP = dO V^T).
 template <int DMAX, int DVMAX>
 __launch_bounds__(256)
@@ -132,15 +135,14 @@ __global__ void flash_dq_kernel(
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
     const int k_stride = d_pad * 2 + 16;
-    const int kt_stride = KVBLK * 2 + 16;   // K^T: d_pad rows x 32 keys
+    const int kt_stride = TILE * 2 + 16;   // K^T: d_pad rows x 64 keys
     const int v_stride = dv_pad * 2 + 16;
-    char* k_lds = smem;                                  // KVBLK * k_stride
-    char* kt_lds = k_lds + KVBLK * k_stride;             // DMAX * kt_stride
-    char* v_lds = kt_lds + DMAX * kt_stride;             // KVBLK * v_stride
-    char* p_lds = v_lds + KVBLK * v_stride;              // NWAVES * QROWS * kt_stride
+    char* k_lds = smem;                                  // TILE * k_stride
+    char* kt_lds = k_lds + TILE * k_stride;              // DMAX * kt_stride
+    char* v_lds = kt_lds + DMAX * kt_stride;             // TILE * v_stride
+    char* p_lds = v_lds + TILE * v_stride;               // NWAVES * QROWS * kt_stride
     char* p_mine = p_lds + wave * QROWS * kt_stride;
 
-    // Q and dO fragments (A layout: lane = row lo16, k = hi4*8+e)
     short8v q_frag[DMAX / 32];
     short8v do_frag[DVMAX / 32];
     float lse_r[4], delta_r[4];
@@ -176,7 +178,6 @@ __global__ void flash_dq_kernel(
             }
             do_frag[kb] = val;
         }
-        // per C-layout rows
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
             int qi2 = q0 + hi4 * 4 + r;
@@ -191,78 +192,88 @@ __global__ void flash_dq_kernel(
 
     int kv_end = Lk;
     if (causal) kv_end = min(Lk, Lk - Nq + blockIdx.x * QBLK + QBLK);
+    const unsigned int drop_thresh = (unsigned int)(drop_p * 4294967296.0);
 
-    for (int kv0 = 0; kv0 < kv_end; kv0 += KVBLK) {
-        int rows_valid = min(KVBLK, Lk - kv0);
+    for (int kv0 = 0; kv0 < kv_end; kv0 += TILE) {
+        int rows_valid = min(TILE, Lk - kv0);
         __syncthreads();
-        stage_rm(kbase + (long)kv0 * D, D, rows_valid, KVBLK, D, d_pad, k_lds, k_stride, tid, 256);
-        stage_tr(kbase + (long)kv0 * D, D, rows_valid, KVBLK, D, d_pad, kt_lds, kt_stride, tid, 256);
-        stage_rm(vbase + (long)kv0 * Dv, Dv, rows_valid, KVBLK, Dv, dv_pad, v_lds, v_stride, tid, 256);
+        stage_rm(kbase + (long)kv0 * D, D, rows_valid, TILE, D, d_pad, k_lds, k_stride, tid);
+        stage_tr(kbase + (long)kv0 * D, D, rows_valid, TILE, D, d_pad, kt_lds, kt_stride, tid);
+        stage_rm(vbase + (long)kv0 * Dv, Dv, rows_valid, TILE, Dv, dv_pad, v_lds, v_stride, tid);
         __syncthreads();
 
-        // S = Q K^T (16 x 32)
-        float4v s_acc[2] = {float4v{0.f, 0.f, 0.f, 0.f}, float4v{0.f, 0.f, 0.f, 0.f}};
+        // S = Q K^T (16 x 64)
+        float4v s_acc[TBLKS];
+#pragma unroll
+        for (int t = 0; t < TBLKS; ++t) s_acc[t] = float4v{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
         for (int kb = 0; kb < DMAX / 32; ++kb) {
             if (kb < d_blocks) {
 #pragma unroll
-                for (int keyblk = 0; keyblk < 2; ++keyblk) {
-                    const char* src = k_lds + (keyblk * 16 + lo16) * k_stride + (kb * 32 + hi4 * 8) * 2;
+                for (int t = 0; t < TBLKS; ++t) {
+                    const char* src = k_lds + (t * 16 + lo16) * k_stride + (kb * 32 + hi4 * 8) * 2;
                     bf16x8 bfrag = (bf16x8)(*reinterpret_cast<const short8v*>(src));
-                    s_acc[keyblk] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        (bf16x8)q_frag[kb], bfrag, s_acc[keyblk], 0, 0, 0);
+                    s_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        (bf16x8)q_frag[kb], bfrag, s_acc[t], 0, 0, 0);
                 }
             }
         }
 
-        // dP = dO V^T (16 x 32)
-        float4v dp_acc[2] = {float4v{0.f, 0.f, 0.f, 0.f}, float4v{0.f, 0.f, 0.f, 0.f}};
+        // dP = dO V^T (16 x 64)
+        float4v dp_acc[TBLKS];
+#pragma unroll
+        for (int t = 0; t < TBLKS; ++t) dp_acc[t] = float4v{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
         for (int kb = 0; kb < DVMAX / 32; ++kb) {
             if (kb < dv_blocks32) {
 #pragma unroll
-                for (int keyblk = 0; keyblk < 2; ++keyblk) {
-                    const char* src = v_lds + (keyblk * 16 + lo16) * v_stride + (kb * 32 + hi4 * 8) * 2;
+                for (int t = 0; t < TBLKS; ++t) {
+                    const char* src = v_lds + (t * 16 + lo16) * v_stride + (kb * 32 + hi4 * 8) * 2;
                     bf16x8 bfrag = (bf16x8)(*reinterpret_cast<const short8v*>(src));
-                    dp_acc[keyblk] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        (bf16x8)do_frag[kb], bfrag, dp_acc[keyblk], 0, 0, 0);
+                    dp_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        (bf16x8)do_frag[kb], bfrag, dp_acc[t], 0, 0, 0);
                 }
             }
         }
 
-        // dS = P * (dP - delta), P = exp(S - lse)  [C layout]
+        // dS = P * (dprobs - delta), P = exp(S - lse)  [C layout] -> per-wave LDS
 #pragma unroll
-        for (int kb = 0; kb < 2; ++kb) {
+        for (int t = 0; t < TBLKS; ++t) {
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 int qi = q0 + hi4 * 4 + r;
-                int j = kv0 + kb * 16 + lo16;
+                int j = kv0 + t * 16 + lo16;
                 bool masked = j >= Lk || (padrow && j < Lk && padrow[j]) ||
                               (causal && j > Lk - Nq + qi);
-                float p = masked ? 0.f : expf(s_acc[kb][r] - lse_r[r]);
-                float dprobs = dp_acc[kb][r];
+                float p = masked ? 0.f : expf(s_acc[t][r] - lse_r[r]);
+                float dprobs = dp_acc[t][r];
                 if (drop_p > 0.f) {
-                    unsigned int thresh = (unsigned int)(drop_p * 4294967296.0);
-                    bool kept = rng_hash(drop_seed, bh, qi, j) >= thresh;
+                    bool kept = rng_hash(drop_seed, bh, qi, j) >= drop_thresh;
                     dprobs = kept ? dprobs / (1.0f - drop_p) : 0.f;
                 }
                 float ds = p * (dprobs - delta_r[r]);
-                // store dS to per-wave LDS for A-layout reload (bf16)
-                *reinterpret_cast<unsigned short*>(p_mine + (hi4 * 4 + r) * kt_stride + (kb * 16 + lo16) * 2) =
-                    f2bf(ds);
+                *reinterpret_cast<unsigned short*>(
+                    p_mine + (hi4 * 4 + r) * kt_stride + (t * 16 + lo16) * 2) = f2bf(ds);
             }
         }
         __builtin_amdgcn_s_waitcnt(0);
-        bf16x8 ds_frag = (bf16x8)(*reinterpret_cast<const short8v*>(
-            p_mine + lo16 * kt_stride + hi4 * 8 * 2));
+        bf16x8 ds_frag[TBLKS / 2];
+#pragma unroll
+        for (int t32 = 0; t32 < TBLKS / 2; ++t32) {
+            ds_frag[t32] = (bf16x8)(*reinterpret_cast<const short8v*>(
+                p_mine + lo16 * kt_stride + (t32 * 32 + hi4 * 8) * 2));
+        }
 
         // dQ += dS K : B[key k][col d] = K^T_lds[d][k]
 #pragma unroll
         for (int cb = 0; cb < DMAX / 16; ++cb) {
             if (cb * 16 < d_pad) {
-                const char* src = kt_lds + (cb * 16 + lo16) * kt_stride + hi4 * 8 * 2;
-                bf16x8 bfrag = (bf16x8)(*reinterpret_cast<const short8v*>(src));
-                dq_acc[cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ds_frag, bfrag, dq_acc[cb], 0, 0, 0);
+#pragma unroll
+                for (int t32 = 0; t32 < TBLKS / 2; ++t32) {
+                    const char* src = kt_lds + (cb * 16 + lo16) * kt_stride + (t32 * 32 + hi4 * 8) * 2;
+                    bf16x8 bfrag = (bf16x8)(*reinterpret_cast<const short8v*>(src));
+                    dq_acc[cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ds_frag[t32], bfrag, dq_acc[cb], 0, 0, 0);
+                }
             }
         }
     }
@@ -283,7 +294,7 @@ __global__ void flash_dq_kernel(
 
 // ---------------------------------------------------------------- dK/dV kernel
 // grid.x over KV blocks (64 keys per workgroup, 16 per wave), grid.y = B*H.
-// Loops over Q tiles of 32 rows; stages Q row-major + Q^T, dO row-major + dO^T.
+// Loops over 64-row Q tiles; stages Q row-major + Q^T, dO row-major + dO^T.
 template <int DMAX, int DVMAX>
 __launch_bounds__(256)
 __global__ void flash_dkv_kernel(
@@ -294,7 +305,6 @@ __global__ void flash_dkv_kernel(
     unsigned short* __restrict__ dkp, unsigned short* __restrict__ dvp,
     int B, int H, int Nq, int Lk, int D, int Dv, int causal,
     float drop_p, unsigned long long drop_seed) {
-    constexpr int QTILE = 32;
     const int d_pad = (D + 31) & ~31;
     const int dv_pad = (Dv + 31) & ~31;
     const int d_blocks = d_pad / 32;
@@ -314,18 +324,16 @@ __global__ void flash_dkv_kernel(
     const bool* padrow = pad ? pad + (long)b * Lk : nullptr;
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
-    const int q_stride = d_pad * 2 + 16;      // Q row-major: QTILE rows
-    const int qt_stride = QTILE * 2 + 16;     // Q^T: d_pad rows
-    const int do_stride = dv_pad * 2 + 16;    // dO row-major: QTILE rows
-    const int dot_stride = QTILE * 2 + 16;    // dO^T: dv_pad rows
+    const int q_stride = d_pad * 2 + 16;      // Q row-major: TILE rows
+    const int qt_stride = TILE * 2 + 16;      // Q^T: d_pad rows
+    const int do_stride = dv_pad * 2 + 16;    // dO row-major: TILE rows
     char* q_lds = smem;
-    char* qt_lds = q_lds + QTILE * q_stride;
+    char* qt_lds = q_lds + TILE * q_stride;
     char* do_lds = qt_lds + DMAX * qt_stride;
-    char* dot_lds = do_lds + QTILE * do_stride;
-    char* p_lds = dot_lds + DVMAX * dot_stride;   // NWAVES * QROWS * qt_stride
+    char* dot_lds = do_lds + TILE * do_stride;            // DVMAX * qt_stride
+    char* p_lds = dot_lds + DVMAX * qt_stride;            // NWAVES * QROWS * qt_stride
     char* p_mine = p_lds + wave * QROWS * qt_stride;
 
-    // K and V fragments (A layout: lane = key row lo16, k = ch hi4*8+e)
     short8v k_frag[DMAX / 32];
     short8v v_frag[DVMAX / 32];
     bool key_pad[4];
@@ -375,127 +383,131 @@ __global__ void flash_dkv_kernel(
 #pragma unroll
     for (int cb = 0; cb < DVMAX / 16; ++cb) dv_acc[cb] = float4v{0.f, 0.f, 0.f, 0.f};
 
-    // causal: key j interacts with q rows i >= j - (Lk - Nq); start q tile there
     int q_start = 0;
     if (causal) {
-        int j_lo = blockIdx.x * QBLK;                 // workgroup-min key
+        int j_lo = blockIdx.x * QBLK;
         q_start = max(0, j_lo - (Lk - Nq));
-        q_start = (q_start / QTILE) * QTILE;
+        q_start = (q_start / TILE) * TILE;
     }
+    const unsigned int drop_thresh = (unsigned int)(drop_p * 4294967296.0);
 
-    for (int qt0 = q_start; qt0 < Nq; qt0 += QTILE) {
-        int rows_valid = min(QTILE, Nq - qt0);
+    for (int qt0 = q_start; qt0 < Nq; qt0 += TILE) {
+        int rows_valid = min(TILE, Nq - qt0);
         __syncthreads();
-        stage_rm(qbase + (long)qt0 * D, D, rows_valid, QTILE, D, d_pad, q_lds, q_stride, tid, 256);
-        stage_tr(qbase + (long)qt0 * D, D, rows_valid, QTILE, D, d_pad, qt_lds, qt_stride, tid, 256);
-        stage_rm(dobase + (long)qt0 * Dv, Dv, rows_valid, QTILE, Dv, dv_pad, do_lds, do_stride, tid, 256);
-        stage_tr(dobase + (long)qt0 * Dv, Dv, rows_valid, QTILE, Dv, dv_pad, dot_lds, dot_stride, tid, 256);
+        stage_rm(qbase + (long)qt0 * D, D, rows_valid, TILE, D, d_pad, q_lds, q_stride, tid);
+        stage_tr(qbase + (long)qt0 * D, D, rows_valid, TILE, D, d_pad, qt_lds, qt_stride, tid);
+        stage_rm(dobase + (long)qt0 * Dv, Dv, rows_valid, TILE, Dv, dv_pad, do_lds, do_stride, tid);
+        stage_tr(dobase + (long)qt0 * Dv, Dv, rows_valid, TILE, Dv, dv_pad, dot_lds, qt_stride, tid);
         __syncthreads();
 
-        // S^T = K Q^T (16 keys x 32 qrows): B[ch k][col i] = Q^T_lds[ch][i]
-        float4v st_acc[2] = {float4v{0.f, 0.f, 0.f, 0.f}, float4v{0.f, 0.f, 0.f, 0.f}};
+        // S^T = K Q^T (16 keys x 64 qrows); B[ch k][qrow j]: q_lds row j contiguous ch
+        float4v st_acc[TBLKS];
+        float4v dpt_acc[TBLKS];
+#pragma unroll
+        for (int t = 0; t < TBLKS; ++t) {
+            st_acc[t] = float4v{0.f, 0.f, 0.f, 0.f};
+            dpt_acc[t] = float4v{0.f, 0.f, 0.f, 0.f};
+        }
 #pragma unroll
         for (int kb = 0; kb < DMAX / 32; ++kb) {
             if (kb < d_blocks) {
 #pragma unroll
-                for (int qb = 0; qb < 2; ++qb) {
-                    // B fragment: B[k][i] where k = ch, i = q row; read Q^T row (ch) ... but
-                    // fragment wants lane col = i (q row), k = hi4*8+e (ch): element = Q[i][ch]
-                    // = qt_lds[ch][i] -> lane reads COLUMN of qt_lds. Instead read from q_lds:
-                    // q_lds[i][ch] with i = qb*16+lo16, ch = kb*32+hi4*8.. contiguous. B[k][j]
-                    // wants contiguous k per lane -> that's qt_lds[...]. We need B[k=ch][j=qrow]:
-                    // lane lo16 = qrow j, elements e over ch: qt wrong; q_lds row j gives Q[j][ch]
-                    // contiguous in ch: exactly B[k][j] elements for fixed j. Use q_lds.
-                    const char* src = q_lds + (qb * 16 + lo16) * q_stride + (kb * 32 + hi4 * 8) * 2;
+                for (int t = 0; t < TBLKS; ++t) {
+                    const char* src = q_lds + (t * 16 + lo16) * q_stride + (kb * 32 + hi4 * 8) * 2;
                     bf16x8 bfrag = (bf16x8)(*reinterpret_cast<const short8v*>(src));
-                    st_acc[qb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        (bf16x8)k_frag[kb], bfrag, st_acc[qb], 0, 0, 0);
+                    st_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        (bf16x8)k_frag[kb], bfrag, st_acc[t], 0, 0, 0);
                 }
             }
         }
-
-        // dP^T = V dO^T (16 keys x 32 qrows): B[ch][qrow] from do_lds rows
-        float4v dpt_acc[2] = {float4v{0.f, 0.f, 0.f, 0.f}, float4v{0.f, 0.f, 0.f, 0.f}};
+        // dP^T = V dO^T (16 keys x 64 qrows)
 #pragma unroll
         for (int kb = 0; kb < DVMAX / 32; ++kb) {
             if (kb < dv_blocks32) {
 #pragma unroll
-                for (int qb = 0; qb < 2; ++qb) {
-                    const char* src = do_lds + (qb * 16 + lo16) * do_stride + (kb * 32 + hi4 * 8) * 2;
+                for (int t = 0; t < TBLKS; ++t) {
+                    const char* src = do_lds + (t * 16 + lo16) * do_stride + (kb * 32 + hi4 * 8) * 2;
                     bf16x8 bfrag = (bf16x8)(*reinterpret_cast<const short8v*>(src));
-                    dpt_acc[qb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        (bf16x8)v_frag[kb], bfrag, dpt_acc[qb], 0, 0, 0);
+                    dpt_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        (bf16x8)v_frag[kb], bfrag, dpt_acc[t], 0, 0, 0);
                 }
             }
         }
 
-        // P^T and dS^T in C layout: row = key hi4*4+r (global k0+...), col = qrow qb*16+lo16
-        // write both to per-wave LDS (P^T for dV, dS^T for dK), A-layout reload
+        // P^T (dropped, rescaled -> for dV) to LDS; dS^T stashed in st_acc
 #pragma unroll
-        for (int qb = 0; qb < 2; ++qb) {
+        for (int t = 0; t < TBLKS; ++t) {
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 int ki = k0 + hi4 * 4 + r;
-                int qi = qt0 + qb * 16 + lo16;
+                int qi = qt0 + t * 16 + lo16;
                 bool masked = key_pad[r] || qi >= Nq || (causal && ki > Lk - Nq + qi);
                 float lse_i = (qi < Nq) ? lse_row[qi] : 0.f;
                 float delta_i = (qi < Nq) ? delta_row[qi] : 0.f;
-                float p = masked ? 0.f : expf(st_acc[qb][r] - lse_i);
-                float p_eff = p;         // probs actually used in the forward PV
-                float dprobs = dpt_acc[qb][r];
+                float p = masked ? 0.f : expf(st_acc[t][r] - lse_i);
+                float p_eff = p;          // probs actually used in the forward PV
+                float dprobs = dpt_acc[t][r];
                 if (drop_p > 0.f) {
-                    unsigned int thresh = (unsigned int)(drop_p * 4294967296.0);
-                    bool kept = rng_hash(drop_seed, bh, qi, ki) >= thresh;
+                    bool kept = rng_hash(drop_seed, bh, qi, ki) >= drop_thresh;
                     float inv_keep = 1.0f / (1.0f - drop_p);
                     p_eff = kept ? p * inv_keep : 0.f;
                     dprobs = kept ? dprobs * inv_keep : 0.f;
                 }
                 float ds = p * (dprobs - delta_i);
-                char* slot = p_mine + (hi4 * 4 + r) * qt_stride + (qb * 16 + lo16) * 2;
-                // one buffer used twice: first P^T (for dV), then dS^T (for dK)
-                *reinterpret_cast<unsigned short*>(slot) = f2bf(p_eff);
-                // stash ds in registers for the second pass
-                st_acc[qb][r] = ds;  // reuse st_acc as ds storage
+                *reinterpret_cast<unsigned short*>(
+                    p_mine + (hi4 * 4 + r) * qt_stride + (t * 16 + lo16) * 2) = f2bf(p_eff);
+                st_acc[t][r] = ds;  // reuse as dS^T storage
             }
         }
         __builtin_amdgcn_s_waitcnt(0);
-        bf16x8 pt_frag = (bf16x8)(*reinterpret_cast<const short8v*>(
-            p_mine + lo16 * qt_stride + hi4 * 8 * 2));
+        bf16x8 pt_frag[TBLKS / 2];
+#pragma unroll
+        for (int t32 = 0; t32 < TBLKS / 2; ++t32) {
+            pt_frag[t32] = (bf16x8)(*reinterpret_cast<const short8v*>(
+                p_mine + lo16 * qt_stride + (t32 * 32 + hi4 * 8) * 2));
+        }
 
-        // dV += P^T dO : B[qrow i][ch c] = do_lds... B[k=i][j=c]: lane j=c col, k=i:
-        // element = dO[i][c] = dot_lds[c][i] contiguous in i. Use dO^T.
+        // dV += P^T dO : B[qrow i][ch c] = dO^T_lds[c][i] (contiguous in i)
 #pragma unroll
         for (int cb = 0; cb < DVMAX / 16; ++cb) {
             if (cb * 16 < dv_pad) {
-                const char* src = dot_lds + (cb * 16 + lo16) * dot_stride + hi4 * 8 * 2;
-                bf16x8 bfrag = (bf16x8)(*reinterpret_cast<const short8v*>(src));
-                dv_acc[cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pt_frag, bfrag, dv_acc[cb], 0, 0, 0);
+#pragma unroll
+                for (int t32 = 0; t32 < TBLKS / 2; ++t32) {
+                    const char* src = dot_lds + (cb * 16 + lo16) * qt_stride + (t32 * 32 + hi4 * 8) * 2;
+                    bf16x8 bfrag = (bf16x8)(*reinterpret_cast<const short8v*>(src));
+                    dv_acc[cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pt_frag[t32], bfrag, dv_acc[cb], 0, 0, 0);
+                }
             }
         }
 
-        // second pass: dS^T through LDS
-        __builtin_amdgcn_s_waitcnt(0);
-        __syncthreads();  // ensure all waves finished reading P before overwrite (same buffer, wave-local actually)
+        // second pass: dS^T through the same per-wave buffer
+        __syncthreads();  // uniform: all waves done reading their pt_frag (wave-local, but keep tiles in sync)
 #pragma unroll
-        for (int qb = 0; qb < 2; ++qb) {
+        for (int t = 0; t < TBLKS; ++t) {
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 *reinterpret_cast<unsigned short*>(
-                    p_mine + (hi4 * 4 + r) * qt_stride + (qb * 16 + lo16) * 2) = f2bf(st_acc[qb][r]);
+                    p_mine + (hi4 * 4 + r) * qt_stride + (t * 16 + lo16) * 2) = f2bf(st_acc[t][r]);
             }
         }
         __builtin_amdgcn_s_waitcnt(0);
-        bf16x8 dst_frag = (bf16x8)(*reinterpret_cast<const short8v*>(
-            p_mine + lo16 * qt_stride + hi4 * 8 * 2));
+        bf16x8 dst_frag[TBLKS / 2];
+#pragma unroll
+        for (int t32 = 0; t32 < TBLKS / 2; ++t32) {
+            dst_frag[t32] = (bf16x8)(*reinterpret_cast<const short8v*>(
+                p_mine + lo16 * qt_stride + (t32 * 32 + hi4 * 8) * 2));
+        }
 
-        // dK += dS^T Q : B[qrow i][ch d] = Q[i][d] = qt_lds[d][i]... need contiguous k=i:
-        // qt_lds row d holds Q[.][d] over i contiguous -> use qt_lds.
+        // dK += dS^T Q : B[qrow i][ch d] = Q^T_lds[d][i] (contiguous in i)
 #pragma unroll
         for (int cb = 0; cb < DMAX / 16; ++cb) {
             if (cb * 16 < d_pad) {
-                const char* src = qt_lds + (cb * 16 + lo16) * qt_stride + hi4 * 8 * 2;
-                bf16x8 bfrag = (bf16x8)(*reinterpret_cast<const short8v*>(src));
-                dk_acc[cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dst_frag, bfrag, dk_acc[cb], 0, 0, 0);
+#pragma unroll
+                for (int t32 = 0; t32 < TBLKS / 2; ++t32) {
+                    const char* src = qt_lds + (cb * 16 + lo16) * qt_stride + (t32 * 32 + hi4 * 8) * 2;
+                    bf16x8 bfrag = (bf16x8)(*reinterpret_cast<const short8v*>(src));
+                    dk_acc[cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dst_frag[t32], bfrag, dk_acc[cb], 0, 0, 0);
+                }
             }
         }
     }
@@ -535,9 +547,9 @@ void launch_flash_bwd(const torch::Tensor& dout, const torch::Tensor& q, const t
     if (pad_mask.has_value() && pad_mask->defined()) padp = pad_mask->data_ptr<bool>();
 
     {   // dQ
-        int k_stride = d_pad * 2 + 16, kt_stride = KVBLK * 2 + 16, v_stride = dv_pad * 2 + 16;
-        size_t smem = (size_t)KVBLK * k_stride + (size_t)DMAX * kt_stride +
-                      (size_t)KVBLK * v_stride + (size_t)NWAVES * QROWS * kt_stride;
+        int k_stride = d_pad * 2 + 16, kt_stride = TILE * 2 + 16, v_stride = dv_pad * 2 + 16;
+        size_t smem = (size_t)TILE * k_stride + (size_t)DMAX * kt_stride +
+                      (size_t)TILE * v_stride + (size_t)NWAVES * QROWS * kt_stride;
         dim3 grid((Nq + QBLK - 1) / QBLK, B * H);
         hipLaunchKernelGGL((flash_dq_kernel<DMAX, DVMAX>), grid, dim3(256), smem, stream,
                            reinterpret_cast<const unsigned short*>(q.data_ptr()),
@@ -550,11 +562,10 @@ void launch_flash_bwd(const torch::Tensor& dout, const torch::Tensor& q, const t
         HIP_CHECK_LAST();
     }
     {   // dK/dV
-        constexpr int QTILE = 32;
-        int q_stride = d_pad * 2 + 16, qt_stride = QTILE * 2 + 16;
-        int do_stride = dv_pad * 2 + 16, dot_stride = QTILE * 2 + 16;
-        size_t smem = (size_t)QTILE * q_stride + (size_t)DMAX * qt_stride +
-                      (size_t)QTILE * do_stride + (size_t)DVMAX * dot_stride +
+        int q_stride = d_pad * 2 + 16, qt_stride = TILE * 2 + 16;
+        int do_stride = dv_pad * 2 + 16;
+        size_t smem = (size_t)TILE * q_stride + (size_t)DMAX * qt_stride +
+                      (size_t)TILE * do_stride + (size_t)DVMAX * qt_stride +
                       (size_t)NWAVES * QROWS * qt_stride;
         dim3 grid((Lk + QBLK - 1) / QBLK, B * H);
         hipLaunchKernelGGL((flash_dkv_kernel<DMAX, DVMAX>), grid, dim3(256), smem, stream,

@@ -9,13 +9,14 @@
 // Semantics match perceiver_amd.ops.attention.eager_attention: q arrives pre-scaled,
 // pad_mask (B, Lk) bool True=pad masked with -FLT_MAX (so fully-masked rows degrade
 // to uniform attention exactly like the reference's -finfo.max fill), causal mask
-// j > Lk - Nq + i (right-aligned).
+// j > Lk - Nq + i (right-aligned), optional in-kernel dropout on the probabilities
+// (counter-hash RNG regenerated in the backward).
 //
-// Structure (v1, plain HIP, compiler-scheduled):
+// Structure (v2):
 //   workgroup = 4 waves x 64 lanes; each wave owns 16 q rows (64 rows/workgroup);
-//   K/V tiles (KVBLK=32 keys) cooperatively staged in LDS, K row-major and V
-//   transposed (channels x keys) so both QK^T and PV read contiguous 16-B LDS
-//   fragments; +16 B row padding makes the 16-lane b128 groups bank-conflict-free;
+//   KVBLK=64-key K/V tiles cooperatively staged in LDS — K row-major (contiguous
+//   b128 fragment reads), V transposed channels x keys with 4-key-packed b64
+//   writes; +16 B row padding keeps 16-lane b128 groups bank-conflict-free;
 //   QK^T and PV on mfma_f32_16x16x32_bf16 with fp32 accumulation; online softmax
 //   with cross-lane shfl_xor row reductions; P redistributed C-layout -> A-layout
 //   through a per-wave LDS buffer. Outputs O and logsumexp (for the backward).
@@ -28,7 +29,8 @@ typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 
 namespace {
 
-constexpr int KVBLK = 32;       // keys per tile
+constexpr int KVBLK = 64;       // keys per tile
+constexpr int KEYBLKS = KVBLK / 16;
 constexpr int QROWS = 16;       // q rows per wave
 constexpr int NWAVES = 4;       // waves per workgroup
 constexpr int QBLK = QROWS * NWAVES;
@@ -45,9 +47,8 @@ DEVINL float warp16_sum(float x) {
     return x;
 }
 
-// Stage a (KVBLK x D) bf16 tile from global (row-major, row stride src_stride elems)
-// into LDS with row stride ldst_bytes, zero-padding cols >= d and rows >= rows_valid.
-// All 256 threads participate; 8 bf16 (16 B) per thread per iteration.
+// Stage a (KVBLK x D) bf16 tile from global (row stride src_stride elems) into LDS
+// row-major (row stride ldst_bytes), zero-padding col >= d / row >= rows_valid.
 DEVINL void stage_tile_rowmajor(const unsigned short* __restrict__ src, long src_stride,
                                 int rows_valid, int d, int d_pad,
                                 char* lds, int ldst_bytes, int tid) {
@@ -71,33 +72,40 @@ DEVINL void stage_tile_rowmajor(const unsigned short* __restrict__ src, long src
     }
 }
 
-// Stage a (KVBLK x Dv) bf16 tile TRANSPOSED into LDS: ldsT row = channel, col = key.
-// ldsT row stride = KVBLK*2 + 16 bytes.
+typedef __attribute__((ext_vector_type(4))) short short4x;
+
+// Stage a (KVBLK x Dv) tile TRANSPOSED: ldsT row = channel, col = key. Each thread
+// iteration covers 4 keys x 8 channels: 4 coalesced 16-B global reads, 8 packed
+// 8-B LDS writes (keys k..k+3 of one channel).
 DEVINL void stage_tile_transposed(const unsigned short* __restrict__ src, long src_stride,
                                   int rows_valid, int dv, int dv_pad,
                                   char* ldsT, int ldst_bytes, int tid) {
-    // thread t handles elements in (key, ch) order for coalesced global reads:
-    // 256 threads -> 8 consecutive channels per thread per key row
-    const int granules_per_row = dv_pad / 8;
-    const int total = KVBLK * granules_per_row;
+    const int gpr = dv_pad / 8;           // 8-channel granules
+    const int total = (KVBLK / 4) * gpr;
     for (int g = tid; g < total; g += 256) {
-        int key = g / granules_per_row;
-        int c0 = (g % granules_per_row) * 8;
-        short8v val = {};
-        if (key < rows_valid && c0 < dv) {
-            if (c0 + 8 <= dv) {
-                val = *reinterpret_cast<const short8v*>(src + (long)key * src_stride + c0);
-            } else {
+        int key0 = (g / gpr) * 4;
+        int c0 = (g % gpr) * 8;
+        short8v rows[4];
 #pragma unroll
-                for (int e = 0; e < 8; ++e) {
-                    val[e] = (c0 + e < dv) ? (short)src[(long)key * src_stride + c0 + e] : (short)0;
+        for (int r = 0; r < 4; ++r) {
+            short8v val = {};
+            int key = key0 + r;
+            if (key < rows_valid && c0 < dv) {
+                if (c0 + 8 <= dv) {
+                    val = *reinterpret_cast<const short8v*>(src + (long)key * src_stride + c0);
+                } else {
+#pragma unroll
+                    for (int e = 0; e < 8; ++e) {
+                        val[e] = (c0 + e < dv) ? (short)src[(long)key * src_stride + c0 + e] : (short)0;
+                    }
                 }
             }
+            rows[r] = val;
         }
 #pragma unroll
         for (int e = 0; e < 8; ++e) {
-            *reinterpret_cast<unsigned short*>(ldsT + (c0 + e) * ldst_bytes + key * 2) =
-                (unsigned short)val[e];
+            short4x pack = {rows[0][e], rows[1][e], rows[2][e], rows[3][e]};
+            *reinterpret_cast<short4x*>(ldsT + (c0 + e) * ldst_bytes + key0 * 2) = pack;
         }
     }
 }
@@ -121,12 +129,12 @@ __global__ void flash_fwd_kernel(
     const int tid = threadIdx.x;
     const int wave = tid / 64;
     const int lane = tid % 64;
-    const int lo16 = lane & 15;            // col index within fragments
-    const int hi4 = lane >> 4;             // k-subblock / row-subgroup index
+    const int lo16 = lane & 15;
+    const int hi4 = lane >> 4;
 
     const int bh = blockIdx.y;
     const int b = bh / H;
-    const int q0 = blockIdx.x * QBLK + wave * QROWS;   // this wave's first q row
+    const int q0 = blockIdx.x * QBLK + wave * QROWS;
 
     const unsigned short* qbase = qp + ((long)bh * Nq) * D;
     const unsigned short* kbase = kp + ((long)bh * Lk) * D;
@@ -139,10 +147,10 @@ __global__ void flash_fwd_kernel(
     const int vt_stride = KVBLK * 2 + 16;
     char* k_lds = smem;                                   // KVBLK * k_stride
     char* vt_lds = k_lds + KVBLK * k_stride;              // dv_pad * vt_stride
-    char* p_lds = vt_lds + DVMAX * vt_stride;             // per wave: QROWS * (KVBLK*2+16)
+    char* p_lds = vt_lds + DVMAX * vt_stride;             // per wave: QROWS * vt_stride
     char* p_mine = p_lds + wave * QROWS * vt_stride;
 
-    // ---- load Q fragments: lane holds A[i=lo16][k = hi4*8 + e] per 32-wide k-block
+    // ---- Q fragments: lane holds A[i=lo16][k = hi4*8 + e] per 32-wide k-block ----
     short8v q_frag[DMAX / 32];
     {
         int qi = q0 + lo16;
@@ -180,6 +188,8 @@ __global__ void flash_fwd_kernel(
         kv_end = min(Lk, Lk - Nq + q_hi + 1);
     }
 
+    const unsigned int drop_thresh = (unsigned int)(drop_p * 4294967296.0);
+
     for (int kv0 = 0; kv0 < kv_end; kv0 += KVBLK) {
         int rows_valid = min(KVBLK, Lk - kv0);
         __syncthreads();
@@ -187,18 +197,17 @@ __global__ void flash_fwd_kernel(
         stage_tile_transposed(vbase + (long)kv0 * Dv, Dv, rows_valid, Dv, dv_pad, vt_lds, vt_stride, tid);
         __syncthreads();
 
-        // ---- S = Q K^T for this wave's 16 rows x 32 keys ----
-        float4v s_acc[2];
-        s_acc[0] = float4v{0.f, 0.f, 0.f, 0.f};
-        s_acc[1] = float4v{0.f, 0.f, 0.f, 0.f};
+        // ---- S = Q K^T (16 rows x KVBLK keys) ----
+        float4v s_acc[KEYBLKS];
+#pragma unroll
+        for (int kb = 0; kb < KEYBLKS; ++kb) s_acc[kb] = float4v{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
         for (int kb = 0; kb < DMAX / 32; ++kb) {
             if (kb < d_blocks) {
                 // B fragment: B[k][j] = K[key j][ch k]; lane reads K row (lo16 + 16*keyblk)
 #pragma unroll
-                for (int keyblk = 0; keyblk < 2; ++keyblk) {
-                    int krow = keyblk * 16 + lo16;
-                    const char* src = k_lds + krow * k_stride + (kb * 32 + hi4 * 8) * 2;
+                for (int keyblk = 0; keyblk < KEYBLKS; ++keyblk) {
+                    const char* src = k_lds + (keyblk * 16 + lo16) * k_stride + (kb * 32 + hi4 * 8) * 2;
                     bf16x8 bfrag = (bf16x8)(*reinterpret_cast<const short8v*>(src));
                     s_acc[keyblk] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         (bf16x8)q_frag[kb], bfrag, s_acc[keyblk], 0, 0, 0);
@@ -206,15 +215,15 @@ __global__ void flash_fwd_kernel(
             }
         }
 
-        // ---- mask + online softmax (C layout: row hi4*4+r, col lo16+16*kb) ----
-        float pvals[2][4];
+        // ---- mask + online softmax (C layout: row hi4*4+r, col kv0 + kb*16 + lo16) ----
+        float pvals[KEYBLKS][4];
         float rowmax[4];
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
             int qi = q0 + hi4 * 4 + r;
             float mx = -FLT_MAX;
 #pragma unroll
-            for (int kb = 0; kb < 2; ++kb) {
+            for (int kb = 0; kb < KEYBLKS; ++kb) {
                 int j = kv0 + kb * 16 + lo16;
                 float s = s_acc[kb][r];
                 bool masked = j >= Lk;
@@ -235,7 +244,7 @@ __global__ void flash_fwd_kernel(
             m_run[r] = m_new;
             float psum = 0.f;
 #pragma unroll
-            for (int kb = 0; kb < 2; ++kb) {
+            for (int kb = 0; kb < KEYBLKS; ++kb) {
                 float p = expf(pvals[kb][r] - m_new);
                 pvals[kb][r] = p;
                 psum += p;
@@ -250,25 +259,22 @@ __global__ void flash_fwd_kernel(
             for (int r = 0; r < 4; ++r) o_acc[cb][r] *= alpha[r];
         }
 
-        // ---- attention dropout on the (unnormalized) probabilities; the softmax
-        // denominator l uses the undropped sum, matching softmax->dropout order ----
+        // ---- dropout on (unnormalized) probabilities; l uses the undropped sum ----
         if (drop_p > 0.f) {
-            unsigned int thresh = (unsigned int)(drop_p * 4294967296.0);
 #pragma unroll
-            for (int kb = 0; kb < 2; ++kb) {
+            for (int kb = 0; kb < KEYBLKS; ++kb) {
 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
                     int qi = q0 + hi4 * 4 + r;
                     int j = kv0 + kb * 16 + lo16;
-                    if (rng_hash(drop_seed, bh, qi, j) < thresh) pvals[kb][r] = 0.f;
+                    if (rng_hash(drop_seed, bh, qi, j) < drop_thresh) pvals[kb][r] = 0.f;
                 }
             }
         }
 
         // ---- redistribute P (C layout) -> A layout via per-wave LDS ----
-        // write: lane holds P[row hi4*4+r][key lo16+16*kb]
 #pragma unroll
-        for (int kb = 0; kb < 2; ++kb) {
+        for (int kb = 0; kb < KEYBLKS; ++kb) {
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 int prow = hi4 * 4 + r;
@@ -278,22 +284,29 @@ __global__ void flash_fwd_kernel(
             }
         }
         __builtin_amdgcn_s_waitcnt(0);  // lgkmcnt(0): wave-local LDS ordering
-        // read: lane needs A[i=lo16][k=hi4*8+e], 8 contiguous bf16
-        bf16x8 p_frag = (bf16x8)(*reinterpret_cast<const short8v*>(
-            p_mine + lo16 * vt_stride + hi4 * 8 * 2));
+        // A-layout fragments per 32-key block: lane reads row lo16, k = kb32*32 + hi4*8
+        bf16x8 p_frag[KEYBLKS / 2];
+#pragma unroll
+        for (int kb32 = 0; kb32 < KEYBLKS / 2; ++kb32) {
+            p_frag[kb32] = (bf16x8)(*reinterpret_cast<const short8v*>(
+                p_mine + lo16 * vt_stride + (kb32 * 32 + hi4 * 8) * 2));
+        }
 
         // ---- O += P V : B[k][j] = V[key k][ch j] = vt_lds[ch j][key k] ----
 #pragma unroll
         for (int cb = 0; cb < DVMAX / 16; ++cb) {
             if (cb < dv_blocks) {
-                const char* src = vt_lds + (cb * 16 + lo16) * vt_stride + hi4 * 8 * 2;
-                bf16x8 bfrag = (bf16x8)(*reinterpret_cast<const short8v*>(src));
-                o_acc[cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(p_frag, bfrag, o_acc[cb], 0, 0, 0);
+#pragma unroll
+                for (int kb32 = 0; kb32 < KEYBLKS / 2; ++kb32) {
+                    const char* src = vt_lds + (cb * 16 + lo16) * vt_stride + (kb32 * 32 + hi4 * 8) * 2;
+                    bf16x8 bfrag = (bf16x8)(*reinterpret_cast<const short8v*>(src));
+                    o_acc[cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(p_frag[kb32], bfrag, o_acc[cb], 0, 0, 0);
+                }
             }
         }
     }
 
-    // ---- epilogue: O /= l, store O (bf16) and lse ----
+    // ---- epilogue: O /= l (x dropout keep-rate), store O (bf16) and lse ----
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
         int qi = q0 + hi4 * 4 + r;
@@ -320,7 +333,6 @@ void launch_flash_fwd(const torch::Tensor& q, const torch::Tensor& k, const torc
     int B = q.size(0), H = q.size(1), Nq = q.size(2), D = q.size(3);
     int Lk = k.size(2), Dv = v.size(3);
     const int d_pad = (D + 31) & ~31;
-    const int dv_pad = (Dv + 15) & ~15;
     int k_stride = d_pad * 2 + 16;
     int vt_stride = KVBLK * 2 + 16;
     size_t smem = (size_t)KVBLK * k_stride + (size_t)DVMAX * vt_stride +

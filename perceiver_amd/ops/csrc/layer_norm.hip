@@ -151,7 +151,7 @@ __global__ void ln_bwd_dx_kernel(const unsigned short* __restrict__ dy,
     }
 }
 
-constexpr int RB = 64;  // rows per partial-reduce block
+constexpr int RB = 16;  // rows per partial-reduce block (small: parallelism over 256 CUs)
 
 __global__ void ln_bwd_dwdb_kernel(const unsigned short* __restrict__ dy,
                                    const unsigned short* __restrict__ x,

@@ -28,7 +28,7 @@ class MasterAdamW(torch.optim.Optimizer):
                 loss = closure()
 
         for group in self.param_groups:
-            params, grads, masters, exp_avgs, exp_avg_sqs, steps = [], [], [], [], [], []
+            params, raw_grads, grads, masters, exp_avgs, exp_avg_sqs, steps = [], [], [], [], [], [], []
             for p in group["params"]:
                 if p.grad is None:
                     continue
@@ -38,9 +38,11 @@ class MasterAdamW(torch.optim.Optimizer):
                     state["master"] = p.detach().to(torch.float32, copy=True)
                     state["exp_avg"] = torch.zeros_like(state["master"])
                     state["exp_avg_sq"] = torch.zeros_like(state["master"])
+                    state["grad32"] = torch.empty_like(state["master"])
                 state["step"] += 1
                 params.append(p)
-                grads.append(p.grad.to(torch.float32))
+                raw_grads.append(p.grad)
+                grads.append(state["grad32"])
                 masters.append(state["master"])
                 exp_avgs.append(state["exp_avg"])
                 exp_avg_sqs.append(state["exp_avg_sq"])
@@ -48,6 +50,9 @@ class MasterAdamW(torch.optim.Optimizer):
 
             if not params:
                 continue
+
+            # single fused multi-tensor cast instead of one cast kernel per tensor
+            torch._foreach_copy_(grads, raw_grads)
 
             beta1, beta2 = group["betas"]
             lr, eps, wd = group["lr"], group["eps"], group["weight_decay"]
@@ -72,9 +77,8 @@ class MasterAdamW(torch.optim.Optimizer):
             torch._foreach_mul_(updates, step_sizes)
             torch._foreach_sub_(masters, updates)
 
-            # write back to the live (possibly bf16) parameters
-            for p, m in zip(params, masters):
-                p.copy_(m)
+            # write back to the live (possibly bf16) parameters (fused multi-tensor)
+            torch._foreach_copy_(params, masters)
 
         return loss
 
