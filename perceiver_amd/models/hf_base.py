@@ -1,0 +1,284 @@
+"""🤗 integration base: PreTrainedModel wrapper for causal Perceiver-AR models with
+the sliding latent/prefix generation schedule, plus weight-copy utilities mapping
+``transformers`` Perceiver checkpoints onto this library's modules.
+
+Parity: reference perceiver/model/core/huggingface.py:21-230. The generation loop is
+implemented natively (transformers 5.x removed the legacy list-of-tuples cache path
+its GenerationMixin integration relied on): ``generate`` supports greedy, temperature/
+top-k/top-p sampling and the grow-latents -> grow-prefix -> slide-window schedule with
+self/cross cache truncation, with cached and uncached paths producing identical
+tokens (the reference's equality contract).
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import Any, List, Optional, Sequence
+
+import torch
+import torch.nn as nn
+from transformers import PreTrainedModel
+from transformers.modeling_outputs import CausalLMOutputWithPast
+
+from perceiver_amd.core import (
+    CrossAttentionLayer,
+    MLP,
+    MultiHeadAttention,
+    PerceiverDecoder,
+    PerceiverEncoder,
+    SelfAttentionLayer,
+)
+from perceiver_amd.core.modules import KVCache
+
+
+# ----------------------------------------------------------------- param copying
+def copy_param(src: nn.Parameter, tgt: nn.Parameter):
+    with torch.no_grad():
+        tgt.copy_(src)
+
+
+def copy_params(src: nn.Module, tgt: nn.Module):
+    tgt.load_state_dict(src.state_dict())
+
+
+def copy_attention_params(src, tgt: MultiHeadAttention):
+    copy_params(src.attention.self.query, tgt.q_proj)
+    copy_params(src.attention.self.key, tgt.k_proj)
+    copy_params(src.attention.self.value, tgt.v_proj)
+    copy_params(src.attention.output.dense, tgt.o_proj)
+
+
+def copy_mlp_params(src, tgt: MLP):
+    copy_params(src.layernorm, tgt[0])
+    copy_params(src.mlp.dense1, tgt[1])
+    copy_params(src.mlp.dense2, tgt[3])
+
+
+def copy_cross_attention_layer_params(src, tgt: CrossAttentionLayer, query_residual: bool):
+    att_tgt = tgt[0].module if query_residual else tgt[0]
+    mlp_tgt = tgt[1].module
+    copy_params(src.attention.self.layernorm1, att_tgt.q_norm)
+    copy_params(src.attention.self.layernorm2, att_tgt.kv_norm)
+    copy_attention_params(src, att_tgt.attention)
+    copy_mlp_params(src, mlp_tgt)
+
+
+def copy_self_attention_layer_params(src, tgt: SelfAttentionLayer):
+    att_tgt = tgt[0].module
+    mlp_tgt = tgt[1].module
+    copy_params(src.attention.self.layernorm1, att_tgt.norm)
+    copy_attention_params(src, att_tgt.attention)
+    copy_mlp_params(src, mlp_tgt)
+
+
+def copy_self_attention_block_params(src: Sequence, tgt: Sequence):
+    assert len(src) == len(tgt)
+    for src_layer, tgt_layer in zip(src, tgt):
+        copy_self_attention_layer_params(src_layer, tgt_layer)
+
+
+def copy_latent_provider_params(src, tgt: PerceiverEncoder):
+    copy_param(src.embeddings.latents, tgt.latent_provider._query)
+
+
+def copy_classification_decoder_params(src, tgt: PerceiverDecoder, query_residual=True):
+    copy_cross_attention_layer_params(
+        src.decoder.decoder.decoding_cross_attention, tgt.cross_attn, query_residual=query_residual
+    )
+    copy_params(src.decoder.decoder.final_layer, tgt.output_adapter.linear)
+    copy_param(src.decoder.decoder.output_position_encodings.position_embeddings,
+               tgt.output_query_provider._query)
+
+
+# ----------------------------------------------------------------- causal base
+@dataclass
+class PerceiverCausalSequenceModelOutput(CausalLMOutputWithPast):
+    prefix_len: Optional[int] = None
+
+
+class PerceiverCausalSequenceModel(PreTrainedModel):
+    """Wraps a CausalSequenceModel backend for 🤗 inference. Subclasses set
+    ``self.backend_model`` in __init__."""
+
+    def forward(
+        self,
+        input_ids: torch.LongTensor,
+        prefix_len: int,
+        attention_mask: Optional[torch.FloatTensor] = None,
+        past_key_values: Optional[List[KVCache]] = None,
+        use_cache: Optional[bool] = None,
+        labels: Optional[torch.LongTensor] = None,
+        **kwargs: Any,
+    ):
+        if labels is not None:
+            raise ValueError("Loss computation from labels not supported yet")
+        pad_mask = None if attention_mask is None else ~attention_mask.type(torch.bool)
+        if use_cache and past_key_values is None:
+            past_key_values = []
+        output = self.backend_model(input_ids, prefix_len=prefix_len, pad_mask=pad_mask,
+                                    kv_cache=past_key_values)
+        return PerceiverCausalSequenceModelOutput(
+            logits=output.logits,
+            hidden_states=(output.last_hidden_state,),
+            past_key_values=output.kv_cache,
+            prefix_len=prefix_len,
+        )
+
+    # -------------------------------------------------- cache maintenance
+    def _reorder_cache(self, past_key_values, beam_idx):
+        return [
+            tuple(t.index_select(0, beam_idx.to(t.device)) for t in layer_past)
+            for layer_past in past_key_values
+        ]
+
+    def _truncate_cross_attention_past_key_values(self, past_key_values):
+        max_ca_cache_len = self.backend_model.max_seq_len - 1
+        (k_cache, v_cache), *sa_cache = past_key_values
+        ca_cache = (k_cache[:, -max_ca_cache_len:], v_cache[:, -max_ca_cache_len:])
+        return [ca_cache] + sa_cache
+
+    def _truncate_self_attention_past_key_values(self, past_key_values):
+        max_sa_cache_len = self.backend_model.max_latents - 1
+        ca_cache, *sa_cache = past_key_values
+        sa_cache = [(k[:, -max_sa_cache_len:], v[:, -max_sa_cache_len:]) for k, v in sa_cache]
+        return [ca_cache] + sa_cache
+
+    def prepare_inputs_for_generation(self, input_ids, past_key_values=None, **kwargs):
+        """The sliding schedule: grow latents to max_latents, then grow the prefix to
+        max_prefix_len, then slide the window (discarding the left-most prefix token).
+        Kept API-compatible with the reference (core/huggingface.py:89-138)."""
+        attention_mask = kwargs.get("attention_mask", None)
+        use_cache = kwargs.get("use_cache", None)
+        prefix_len = kwargs.get("prefix_len", None)
+
+        if past_key_values is None:
+            input_len = input_ids.shape[1]
+        else:
+            # contrastive-search workaround: derive input length from the cache
+            input_len = past_key_values[0][0].shape[1] + 1
+
+        max_seq_len = self.backend_model.max_seq_len
+        num_latents = input_len - prefix_len
+
+        max_seq_len_exceeded = input_len > max_seq_len
+        max_latents_exceeded = num_latents > self.backend_model.max_latents
+
+        if max_latents_exceeded and prefix_len < self.backend_model.max_prefix_len:
+            prefix_len += 1
+
+        if past_key_values:
+            input_ids = input_ids[:, -1:]
+        else:
+            input_ids = input_ids[:, -max_seq_len:]
+
+        if attention_mask is not None and attention_mask.shape[1] > max_seq_len:
+            attention_mask = attention_mask[:, -max_seq_len:]
+
+        if past_key_values:
+            if max_latents_exceeded:
+                past_key_values = self._truncate_self_attention_past_key_values(past_key_values)
+            if max_seq_len_exceeded:
+                past_key_values = self._truncate_cross_attention_past_key_values(past_key_values)
+
+        return {
+            "input_ids": input_ids,
+            "attention_mask": attention_mask,
+            "past_key_values": past_key_values,
+            "use_cache": use_cache,
+            "prefix_len": prefix_len,
+        }
+
+    # -------------------------------------------------- generation
+    @torch.no_grad()
+    def generate(
+        self,
+        inputs: Optional[torch.Tensor] = None,
+        input_ids: Optional[torch.Tensor] = None,
+        num_latents: int = 1,
+        max_new_tokens: int = 64,
+        do_sample: bool = False,
+        temperature: float = 1.0,
+        top_k: Optional[int] = None,
+        top_p: Optional[float] = None,
+        use_cache: bool = True,
+        attention_mask: Optional[torch.Tensor] = None,
+        pad_token_id: Optional[int] = None,
+        eos_token_id: Optional[int] = None,
+        generator: Optional[torch.Generator] = None,
+        **kwargs,
+    ) -> torch.Tensor:
+        """Native generation loop with the latent/prefix sliding schedule.
+
+        ``num_latents``: initial number of latent positions assigned to the end of
+        the prompt. Cached and uncached paths produce identical tokens.
+        """
+        if input_ids is None:
+            input_ids = inputs
+        if input_ids is None:
+            raise ValueError("Either inputs or input_ids must be defined")
+
+        seq_len = input_ids.shape[1]
+        if not 0 < seq_len <= self.backend_model.max_seq_len:
+            raise ValueError(
+                f"Input sequence length out of valid range [1..{self.backend_model.max_seq_len}]"
+            )
+        if not 0 < num_latents <= self.backend_model.max_latents:
+            raise ValueError(
+                f"num_latents={num_latents} out of valid range [1..{self.backend_model.max_latents}]"
+            )
+        num_latents = min(seq_len, num_latents)
+        prefix_len = seq_len - num_latents
+        if prefix_len > self.backend_model.max_prefix_len:
+            num_latents_min = num_latents + prefix_len - self.backend_model.max_prefix_len
+            raise ValueError(
+                f"For given sequence of length={seq_len}, num_latents must "
+                f"be in range [{num_latents_min}..{self.backend_model.max_latents}]"
+            )
+
+        if attention_mask is None:
+            attention_mask = torch.ones_like(input_ids)
+        done = torch.zeros(input_ids.shape[0], dtype=torch.bool, device=input_ids.device)
+        past = None
+
+        for _ in range(max_new_tokens):
+            model_inputs = self.prepare_inputs_for_generation(
+                input_ids, past_key_values=past, attention_mask=attention_mask,
+                use_cache=use_cache, prefix_len=prefix_len,
+            )
+            out = self(**model_inputs)
+            prefix_len = out.prefix_len
+            past = out.past_key_values if use_cache else None
+
+            logits = out.logits[:, -1, :].float()
+            next_token = self._select_next(logits, do_sample, temperature, top_k, top_p, generator)
+            if eos_token_id is not None:
+                fill = pad_token_id if pad_token_id is not None else eos_token_id
+                next_token = torch.where(done, torch.full_like(next_token, fill), next_token)
+                done = done | (next_token == eos_token_id)
+
+            input_ids = torch.cat([input_ids, next_token[:, None]], dim=1)
+            attention_mask = torch.cat(
+                [attention_mask, torch.ones_like(next_token[:, None])], dim=1
+            )
+            if eos_token_id is not None and bool(done.all()):
+                break
+
+        return input_ids
+
+    @staticmethod
+    def _select_next(logits, do_sample, temperature, top_k, top_p, generator):
+        if not do_sample:
+            return logits.argmax(dim=-1)
+        if temperature != 1.0:
+            logits = logits / temperature
+        if top_k is not None and top_k > 0:
+            kth = torch.topk(logits, min(top_k, logits.shape[-1]))[0][..., -1, None]
+            logits = logits.masked_fill(logits < kth, float("-inf"))
+        if top_p is not None and 0 < top_p < 1:
+            sorted_logits, sorted_idx = torch.sort(logits, descending=True)
+            probs = sorted_logits.softmax(-1).cumsum(-1)
+            remove = probs - probs.gather(-1, torch.zeros_like(sorted_idx[..., :1])) > top_p
+            remove[..., 0] = False
+            scatter_mask = remove.scatter(-1, sorted_idx, remove)
+            logits = logits.masked_fill(scatter_mask, float("-inf"))
+        probs = logits.softmax(dim=-1)
+        return torch.multinomial(probs, 1, generator=generator).squeeze(-1)

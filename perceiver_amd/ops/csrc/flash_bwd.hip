@@ -20,8 +20,6 @@ typedef __attribute__((ext_vector_type(4))) short short4x;
 
 namespace {
 
-constexpr int TILE = 64;       // keys per tile (dq) / q rows per tile (dkv)
-constexpr int TBLKS = TILE / 16;
 constexpr int QROWS = 16;      // rows per wave (q rows in dq, key rows in dkv)
 constexpr int NWAVES = 4;
 constexpr int QBLK = QROWS * NWAVES;
@@ -105,7 +103,7 @@ DEVINL void stage_tr(const unsigned short* __restrict__ src, long src_stride,
 // ---------------------------------------------------------------- dQ kernel
 // grid.x over Q blocks, grid.y = B*H. Stages per 64-key tile: K row-major (for S),
 // K^T (for dQ = dS K), V row-major (for dP = dO V^T).
-template <int DMAX, int DVMAX>
+template <int DMAX, int DVMAX, int TILE>
 __launch_bounds__(256)
 __global__ void flash_dq_kernel(
     const unsigned short* __restrict__ qp, const unsigned short* __restrict__ kp,
@@ -115,6 +113,7 @@ __global__ void flash_dq_kernel(
     unsigned short* __restrict__ dqp,
     int B, int H, int Nq, int Lk, int D, int Dv, int causal,
     float drop_p, unsigned long long drop_seed) {
+    constexpr int TBLKS = TILE / 16;
     const int d_pad = (D + 31) & ~31;
     const int dv_pad = (Dv + 31) & ~31;
     const int d_blocks = d_pad / 32;
@@ -295,7 +294,7 @@ __global__ void flash_dq_kernel(
 // ---------------------------------------------------------------- dK/dV kernel
 // grid.x over KV blocks (64 keys per workgroup, 16 per wave), grid.y = B*H.
 // Loops over 64-row Q tiles; stages Q row-major + Q^T, dO row-major + dO^T.
-template <int DMAX, int DVMAX>
+template <int DMAX, int DVMAX, int TILE>
 __launch_bounds__(256)
 __global__ void flash_dkv_kernel(
     const unsigned short* __restrict__ qp, const unsigned short* __restrict__ kp,
@@ -305,6 +304,7 @@ __global__ void flash_dkv_kernel(
     unsigned short* __restrict__ dkp, unsigned short* __restrict__ dvp,
     int B, int H, int Nq, int Lk, int D, int Dv, int causal,
     float drop_p, unsigned long long drop_seed) {
+    constexpr int TBLKS = TILE / 16;
     const int d_pad = (D + 31) & ~31;
     const int dv_pad = (Dv + 31) & ~31;
     const int d_blocks = d_pad / 32;
@@ -532,7 +532,7 @@ __global__ void flash_dkv_kernel(
     }
 }
 
-template <int DMAX, int DVMAX>
+template <int DMAX, int DVMAX, int TILE>
 void launch_flash_bwd(const torch::Tensor& dout, const torch::Tensor& q, const torch::Tensor& k,
                       const torch::Tensor& v, const torch::Tensor& lse, const torch::Tensor& delta,
                       const c10::optional<torch::Tensor>& pad_mask, bool causal,
@@ -547,11 +547,11 @@ void launch_flash_bwd(const torch::Tensor& dout, const torch::Tensor& q, const t
     if (pad_mask.has_value() && pad_mask->defined()) padp = pad_mask->data_ptr<bool>();
 
     {   // dQ
-        int k_stride = d_pad * 2 + 16, kt_stride = TILE * 2 + 16, v_stride = dv_pad * 2 + 16;
+        const int k_stride = d_pad * 2 + 16, kt_stride = TILE * 2 + 16, v_stride = dv_pad * 2 + 16;
         size_t smem = (size_t)TILE * k_stride + (size_t)DMAX * kt_stride +
                       (size_t)TILE * v_stride + (size_t)NWAVES * QROWS * kt_stride;
         dim3 grid((Nq + QBLK - 1) / QBLK, B * H);
-        hipLaunchKernelGGL((flash_dq_kernel<DMAX, DVMAX>), grid, dim3(256), smem, stream,
+        hipLaunchKernelGGL((flash_dq_kernel<DMAX, DVMAX, TILE>), grid, dim3(256), smem, stream,
                            reinterpret_cast<const unsigned short*>(q.data_ptr()),
                            reinterpret_cast<const unsigned short*>(k.data_ptr()),
                            reinterpret_cast<const unsigned short*>(v.data_ptr()),
@@ -568,7 +568,7 @@ void launch_flash_bwd(const torch::Tensor& dout, const torch::Tensor& q, const t
                       (size_t)TILE * do_stride + (size_t)DVMAX * qt_stride +
                       (size_t)NWAVES * QROWS * qt_stride;
         dim3 grid((Lk + QBLK - 1) / QBLK, B * H);
-        hipLaunchKernelGGL((flash_dkv_kernel<DMAX, DVMAX>), grid, dim3(256), smem, stream,
+        hipLaunchKernelGGL((flash_dkv_kernel<DMAX, DVMAX, TILE>), grid, dim3(256), smem, stream,
                            reinterpret_cast<const unsigned short*>(q.data_ptr()),
                            reinterpret_cast<const unsigned short*>(k.data_ptr()),
                            reinterpret_cast<const unsigned short*>(v.data_ptr()),
@@ -615,11 +615,11 @@ std::vector<torch::Tensor> flash_bwd(torch::Tensor dout, torch::Tensor q, torch:
 
     float dp = (float)dropout_p;
     unsigned long long sd = (unsigned long long)seed;
-    if (D <= 32 && Dv <= 160)       launch_flash_bwd<32, 160>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
-    else if (D <= 64 && Dv <= 64)   launch_flash_bwd<64, 64>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
-    else if (D <= 128 && Dv <= 128) launch_flash_bwd<128, 128>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
-    else if (D <= 160 && Dv <= 160) launch_flash_bwd<160, 160>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
-    else                            launch_flash_bwd<352, 352>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
+    if (D <= 32 && Dv <= 160)       launch_flash_bwd<32, 160, 64>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
+    else if (D <= 64 && Dv <= 64)   launch_flash_bwd<64, 64, 64>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
+    else if (D <= 128 && Dv <= 128) launch_flash_bwd<128, 128, 64>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
+    else if (D <= 160 && Dv <= 160) launch_flash_bwd<160, 160, 64>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
+    else                            launch_flash_bwd<352, 352, 32>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
 
     return {dq, dk, dv};
 }
