@@ -6,10 +6,9 @@
 //   dS      = P * (dO V^T * mask/(1-p) - delta_i)
 //   dQ      = dS K          dK = dS^T Q          dV = (P*mask/(1-p))^T dO
 //
-// v2 layouts mirror flash_fwd.hip: 4 waves x 16 rows, 64-key/row tiles, vectorized
-// transposed staging (4-source-row packed b64 writes), mfma_f32_16x16x32_bf16,
-// C-layout -> A-layout redistribution of P/dS through per-wave LDS.
-// q arrives pre-scaled so no extra scale appears here.
+// v3: QH 16-row A-fragment groups per wave (one B-fragment LDS read feeds QH MFMAs),
+// vectorized transposed staging (4-source-row packed b64 writes), per-template TILE
+// sizes balancing LDS occupancy against the register budget.
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
 #include <cfloat>
@@ -20,9 +19,7 @@ typedef __attribute__((ext_vector_type(4))) short short4x;
 
 namespace {
 
-constexpr int QROWS = 16;      // rows per wave (q rows in dq, key rows in dkv)
 constexpr int NWAVES = 4;
-constexpr int QBLK = QROWS * NWAVES;
 
 // ---------------------------------------------------------------- delta kernel
 __global__ void delta_kernel(const unsigned short* __restrict__ dout,
@@ -44,11 +41,12 @@ __global__ void delta_kernel(const unsigned short* __restrict__ dout,
 }
 
 // stage (rows_tile x d) tile row-major into LDS (row stride ldst_bytes), zero-pad
+template <int ROWS_TILE>
 DEVINL void stage_rm(const unsigned short* __restrict__ src, long src_stride,
-                     int rows_valid, int rows_tile, int d, int d_pad,
+                     int rows_valid, int d, int d_pad,
                      char* lds, int ldst_bytes, int tid) {
     const int gpr = d_pad / 8;
-    const int total = rows_tile * gpr;
+    const int total = ROWS_TILE * gpr;
     for (int g = tid; g < total; g += 256) {
         int row = g / gpr;
         int c0 = (g % gpr) * 8;
@@ -68,11 +66,12 @@ DEVINL void stage_rm(const unsigned short* __restrict__ src, long src_stride,
 
 // stage (rows_tile x d) tile TRANSPOSED: lds row = channel (d_pad rows), col =
 // source row; 4-source-row packed b64 writes.
+template <int ROWS_TILE>
 DEVINL void stage_tr(const unsigned short* __restrict__ src, long src_stride,
-                     int rows_valid, int rows_tile, int d, int d_pad,
+                     int rows_valid, int d, int d_pad,
                      char* ldsT, int ldst_bytes, int tid) {
     const int gpr = d_pad / 8;
-    const int total = (rows_tile / 4) * gpr;
+    const int total = (ROWS_TILE / 4) * gpr;
     for (int g = tid; g < total; g += 256) {
         int row0 = (g / gpr) * 4;
         int c0 = (g % gpr) * 8;
@@ -101,9 +100,9 @@ DEVINL void stage_tr(const unsigned short* __restrict__ src, long src_stride,
 }
 
 // ---------------------------------------------------------------- dQ kernel
-// grid.x over Q blocks, grid.y = B*H. Stages per 64-key tile: K row-major (for S),
-// K^T (for dQ = dS K), V row-major (for dP = dO V^T).
-template <int DMAX, int DVMAX, int TILE>
+// grid.x over Q blocks (QH*16 rows per wave), grid.y = B*H. Stages per TILE-key
+// tile: K row-major (for S), K^T (for dQ = dS K), V row-major (for dP = dO V^T).
+template <int DMAX, int DVMAX, int TILE, int QH>
 __launch_bounds__(256)
 __global__ void flash_dq_kernel(
     const unsigned short* __restrict__ qp, const unsigned short* __restrict__ kp,
@@ -114,6 +113,8 @@ __global__ void flash_dq_kernel(
     int B, int H, int Nq, int Lk, int D, int Dv, int causal,
     float drop_p, unsigned long long drop_seed) {
     constexpr int TBLKS = TILE / 16;
+    constexpr int QROWS = 16 * QH;
+    constexpr int QBLK = QROWS * NWAVES;
     const int d_pad = (D + 31) & ~31;
     const int dv_pad = (Dv + 31) & ~31;
     const int d_blocks = d_pad / 32;
@@ -134,7 +135,7 @@ __global__ void flash_dq_kernel(
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
     const int k_stride = d_pad * 2 + 16;
-    const int kt_stride = TILE * 2 + 16;   // K^T: d_pad rows x 64 keys
+    const int kt_stride = TILE * 2 + 16;   // K^T: d_pad rows x TILE keys
     const int v_stride = dv_pad * 2 + 16;
     char* k_lds = smem;                                  // TILE * k_stride
     char* kt_lds = k_lds + TILE * k_stride;              // DMAX * kt_stride
@@ -142,11 +143,12 @@ __global__ void flash_dq_kernel(
     char* p_lds = v_lds + TILE * v_stride;               // NWAVES * QROWS * kt_stride
     char* p_mine = p_lds + wave * QROWS * kt_stride;
 
-    short8v q_frag[DMAX / 32];
-    short8v do_frag[DVMAX / 32];
-    float lse_r[4], delta_r[4];
-    {
-        int qi = q0 + lo16;
+    short8v q_frag[QH][DMAX / 32];
+    short8v do_frag[QH][DVMAX / 32];
+    float lse_r[QH][4], delta_r[QH][4];
+#pragma unroll
+    for (int h = 0; h < QH; ++h) {
+        int qi = q0 + h * 16 + lo16;
         bool valid = qi < Nq;
         int qc = valid ? qi : Nq - 1;
         const unsigned short* qrow = qbase + (long)qc * D;
@@ -162,7 +164,7 @@ __global__ void flash_dq_kernel(
                     for (int e = 0; e < 8; ++e) val[e] = (c0 + e < D) ? (short)qrow[c0 + e] : (short)0;
                 }
             }
-            q_frag[kb] = val;
+            q_frag[h][kb] = val;
         }
 #pragma unroll
         for (int kb = 0; kb < DVMAX / 32; ++kb) {
@@ -175,19 +177,21 @@ __global__ void flash_dq_kernel(
                     for (int e = 0; e < 8; ++e) val[e] = (c0 + e < Dv) ? (short)dorow[c0 + e] : (short)0;
                 }
             }
-            do_frag[kb] = val;
+            do_frag[h][kb] = val;
         }
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
-            int qi2 = q0 + hi4 * 4 + r;
-            lse_r[r] = (qi2 < Nq) ? lse_row[qi2] : 0.f;
-            delta_r[r] = (qi2 < Nq) ? delta_row[qi2] : 0.f;
+            int qi2 = q0 + h * 16 + hi4 * 4 + r;
+            lse_r[h][r] = (qi2 < Nq) ? lse_row[qi2] : 0.f;
+            delta_r[h][r] = (qi2 < Nq) ? delta_row[qi2] : 0.f;
         }
     }
 
-    float4v dq_acc[DMAX / 16];
+    float4v dq_acc[QH][DMAX / 16];
 #pragma unroll
-    for (int cb = 0; cb < DMAX / 16; ++cb) dq_acc[cb] = float4v{0.f, 0.f, 0.f, 0.f};
+    for (int h = 0; h < QH; ++h)
+#pragma unroll
+        for (int cb = 0; cb < DMAX / 16; ++cb) dq_acc[h][cb] = float4v{0.f, 0.f, 0.f, 0.f};
 
     int kv_end = Lk;
     if (causal) kv_end = min(Lk, Lk - Nq + blockIdx.x * QBLK + QBLK);
@@ -196,15 +200,21 @@ __global__ void flash_dq_kernel(
     for (int kv0 = 0; kv0 < kv_end; kv0 += TILE) {
         int rows_valid = min(TILE, Lk - kv0);
         __syncthreads();
-        stage_rm(kbase + (long)kv0 * D, D, rows_valid, TILE, D, d_pad, k_lds, k_stride, tid);
-        stage_tr(kbase + (long)kv0 * D, D, rows_valid, TILE, D, d_pad, kt_lds, kt_stride, tid);
-        stage_rm(vbase + (long)kv0 * Dv, Dv, rows_valid, TILE, Dv, dv_pad, v_lds, v_stride, tid);
+        stage_rm<TILE>(kbase + (long)kv0 * D, D, rows_valid, D, d_pad, k_lds, k_stride, tid);
+        stage_tr<TILE>(kbase + (long)kv0 * D, D, rows_valid, D, d_pad, kt_lds, kt_stride, tid);
+        stage_rm<TILE>(vbase + (long)kv0 * Dv, Dv, rows_valid, Dv, dv_pad, v_lds, v_stride, tid);
         __syncthreads();
 
-        // S = Q K^T (16 x 64)
-        float4v s_acc[TBLKS];
+        // S = Q K^T (QHx16 x TILE)
+        float4v s_acc[QH][TBLKS];
+        float4v dp_acc[QH][TBLKS];
 #pragma unroll
-        for (int t = 0; t < TBLKS; ++t) s_acc[t] = float4v{0.f, 0.f, 0.f, 0.f};
+        for (int h = 0; h < QH; ++h)
+#pragma unroll
+            for (int t = 0; t < TBLKS; ++t) {
+                s_acc[h][t] = float4v{0.f, 0.f, 0.f, 0.f};
+                dp_acc[h][t] = float4v{0.f, 0.f, 0.f, 0.f};
+            }
 #pragma unroll
         for (int kb = 0; kb < DMAX / 32; ++kb) {
             if (kb < d_blocks) {
@@ -212,16 +222,14 @@ __global__ void flash_dq_kernel(
                 for (int t = 0; t < TBLKS; ++t) {
                     const char* src = k_lds + (t * 16 + lo16) * k_stride + (kb * 32 + hi4 * 8) * 2;
                     bf16x8 bfrag = (bf16x8)(*reinterpret_cast<const short8v*>(src));
-                    s_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        (bf16x8)q_frag[kb], bfrag, s_acc[t], 0, 0, 0);
+#pragma unroll
+                    for (int h = 0; h < QH; ++h)
+                        s_acc[h][t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            (bf16x8)q_frag[h][kb], bfrag, s_acc[h][t], 0, 0, 0);
                 }
             }
         }
-
-        // dP = dO V^T (16 x 64)
-        float4v dp_acc[TBLKS];
-#pragma unroll
-        for (int t = 0; t < TBLKS; ++t) dp_acc[t] = float4v{0.f, 0.f, 0.f, 0.f};
+        // dP = dO V^T
 #pragma unroll
         for (int kb = 0; kb < DVMAX / 32; ++kb) {
             if (kb < dv_blocks32) {
@@ -229,39 +237,43 @@ __global__ void flash_dq_kernel(
                 for (int t = 0; t < TBLKS; ++t) {
                     const char* src = v_lds + (t * 16 + lo16) * v_stride + (kb * 32 + hi4 * 8) * 2;
                     bf16x8 bfrag = (bf16x8)(*reinterpret_cast<const short8v*>(src));
-                    dp_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        (bf16x8)do_frag[kb], bfrag, dp_acc[t], 0, 0, 0);
+#pragma unroll
+                    for (int h = 0; h < QH; ++h)
+                        dp_acc[h][t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            (bf16x8)do_frag[h][kb], bfrag, dp_acc[h][t], 0, 0, 0);
                 }
             }
         }
 
         // dS = P * (dprobs - delta), P = exp(S - lse)  [C layout] -> per-wave LDS
 #pragma unroll
-        for (int t = 0; t < TBLKS; ++t) {
+        for (int h = 0; h < QH; ++h)
 #pragma unroll
-            for (int r = 0; r < 4; ++r) {
-                int qi = q0 + hi4 * 4 + r;
-                int j = kv0 + t * 16 + lo16;
-                bool masked = j >= Lk || (padrow && j < Lk && padrow[j]) ||
-                              (causal && j > Lk - Nq + qi);
-                float p = masked ? 0.f : expf(s_acc[t][r] - lse_r[r]);
-                float dprobs = dp_acc[t][r];
-                if (drop_p > 0.f) {
-                    bool kept = rng_hash(drop_seed, bh, qi, j) >= drop_thresh;
-                    dprobs = kept ? dprobs / (1.0f - drop_p) : 0.f;
+            for (int t = 0; t < TBLKS; ++t)
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    int qi = q0 + h * 16 + hi4 * 4 + r;
+                    int j = kv0 + t * 16 + lo16;
+                    bool masked = j >= Lk || (padrow && j < Lk && padrow[j]) ||
+                                  (causal && j > Lk - Nq + qi);
+                    float p = masked ? 0.f : expf(s_acc[h][t][r] - lse_r[h][r]);
+                    float dprobs = dp_acc[h][t][r];
+                    if (drop_p > 0.f) {
+                        bool kept = rng_hash(drop_seed, bh, qi, j) >= drop_thresh;
+                        dprobs = kept ? dprobs / (1.0f - drop_p) : 0.f;
+                    }
+                    float ds = p * (dprobs - delta_r[h][r]);
+                    *reinterpret_cast<unsigned short*>(
+                        p_mine + (h * 16 + hi4 * 4 + r) * kt_stride + (t * 16 + lo16) * 2) = f2bf(ds);
                 }
-                float ds = p * (dprobs - delta_r[r]);
-                *reinterpret_cast<unsigned short*>(
-                    p_mine + (hi4 * 4 + r) * kt_stride + (t * 16 + lo16) * 2) = f2bf(ds);
-            }
-        }
         __builtin_amdgcn_s_waitcnt(0);
-        bf16x8 ds_frag[TBLKS / 2];
+        bf16x8 ds_frag[QH][TBLKS / 2];
 #pragma unroll
-        for (int t32 = 0; t32 < TBLKS / 2; ++t32) {
-            ds_frag[t32] = (bf16x8)(*reinterpret_cast<const short8v*>(
-                p_mine + lo16 * kt_stride + (t32 * 32 + hi4 * 8) * 2));
-        }
+        for (int h = 0; h < QH; ++h)
+#pragma unroll
+            for (int t32 = 0; t32 < TBLKS / 2; ++t32)
+                ds_frag[h][t32] = (bf16x8)(*reinterpret_cast<const short8v*>(
+                    p_mine + (h * 16 + lo16) * kt_stride + (t32 * 32 + hi4 * 8) * 2));
 
         // dQ += dS K : B[key k][col d] = K^T_lds[d][k]
 #pragma unroll
@@ -271,30 +283,35 @@ __global__ void flash_dq_kernel(
                 for (int t32 = 0; t32 < TBLKS / 2; ++t32) {
                     const char* src = kt_lds + (cb * 16 + lo16) * kt_stride + (t32 * 32 + hi4 * 8) * 2;
                     bf16x8 bfrag = (bf16x8)(*reinterpret_cast<const short8v*>(src));
-                    dq_acc[cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ds_frag[t32], bfrag, dq_acc[cb], 0, 0, 0);
+#pragma unroll
+                    for (int h = 0; h < QH; ++h)
+                        dq_acc[h][cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            ds_frag[h][t32], bfrag, dq_acc[h][cb], 0, 0, 0);
                 }
             }
         }
     }
 
-    // store dQ (C layout rows hi4*4+r, col lo16+16cb)
+    // store dQ (C layout rows h*16 + hi4*4+r, col lo16+16cb)
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-        int qi = q0 + hi4 * 4 + r;
-        if (qi >= Nq) continue;
-        unsigned short* dqrow = dqp + ((long)bh * Nq + qi) * D;
+    for (int h = 0; h < QH; ++h)
 #pragma unroll
-        for (int cb = 0; cb < DMAX / 16; ++cb) {
-            int c = cb * 16 + lo16;
-            if (c < D) dqrow[c] = f2bf(dq_acc[cb][r]);
+        for (int r = 0; r < 4; ++r) {
+            int qi = q0 + h * 16 + hi4 * 4 + r;
+            if (qi >= Nq) continue;
+            unsigned short* dqrow = dqp + ((long)bh * Nq + qi) * D;
+#pragma unroll
+            for (int cb = 0; cb < DMAX / 16; ++cb) {
+                int c = cb * 16 + lo16;
+                if (c < D) dqrow[c] = f2bf(dq_acc[h][cb][r]);
+            }
         }
-    }
 }
 
 // ---------------------------------------------------------------- dK/dV kernel
-// grid.x over KV blocks (64 keys per workgroup, 16 per wave), grid.y = B*H.
-// Loops over 64-row Q tiles; stages Q row-major + Q^T, dO row-major + dO^T.
-template <int DMAX, int DVMAX, int TILE>
+// grid.x over KV blocks (QH*16 keys per wave), grid.y = B*H.
+// Loops over TILE-row Q tiles; stages Q row-major + Q^T, dO row-major + dO^T.
+template <int DMAX, int DVMAX, int TILE, int QH>
 __launch_bounds__(256)
 __global__ void flash_dkv_kernel(
     const unsigned short* __restrict__ qp, const unsigned short* __restrict__ kp,
@@ -305,6 +322,8 @@ __global__ void flash_dkv_kernel(
     int B, int H, int Nq, int Lk, int D, int Dv, int causal,
     float drop_p, unsigned long long drop_seed) {
     constexpr int TBLKS = TILE / 16;
+    constexpr int KROWS = 16 * QH;
+    constexpr int KBLK = KROWS * NWAVES;
     const int d_pad = (D + 31) & ~31;
     const int dv_pad = (Dv + 31) & ~31;
     const int d_blocks = d_pad / 32;
@@ -313,7 +332,7 @@ __global__ void flash_dkv_kernel(
     const int tid = threadIdx.x, wave = tid / 64, lane = tid % 64;
     const int lo16 = lane & 15, hi4 = lane >> 4;
     const int bh = blockIdx.y, b = bh / H;
-    const int k0 = blockIdx.x * QBLK + wave * QROWS;  // this wave's first key row
+    const int k0 = blockIdx.x * KBLK + wave * KROWS;  // this wave's first key row
 
     const unsigned short* qbase = qp + (long)bh * Nq * D;
     const unsigned short* kbase = kp + (long)bh * Lk * D;
@@ -331,14 +350,15 @@ __global__ void flash_dkv_kernel(
     char* qt_lds = q_lds + TILE * q_stride;
     char* do_lds = qt_lds + DMAX * qt_stride;
     char* dot_lds = do_lds + TILE * do_stride;            // DVMAX * qt_stride
-    char* p_lds = dot_lds + DVMAX * qt_stride;            // NWAVES * QROWS * qt_stride
-    char* p_mine = p_lds + wave * QROWS * qt_stride;
+    char* p_lds = dot_lds + DVMAX * qt_stride;            // NWAVES * KROWS * qt_stride
+    char* p_mine = p_lds + wave * KROWS * qt_stride;
 
-    short8v k_frag[DMAX / 32];
-    short8v v_frag[DVMAX / 32];
-    bool key_pad[4];
-    {
-        int ki = k0 + lo16;
+    short8v k_frag[QH][DMAX / 32];
+    short8v v_frag[QH][DVMAX / 32];
+    bool key_pad[QH][4];
+#pragma unroll
+    for (int h = 0; h < QH; ++h) {
+        int ki = k0 + h * 16 + lo16;
         bool valid = ki < Lk;
         int kc = valid ? ki : Lk - 1;
         const unsigned short* krow = kbase + (long)kc * D;
@@ -354,7 +374,7 @@ __global__ void flash_dkv_kernel(
                     for (int e = 0; e < 8; ++e) val[e] = (c0 + e < D) ? (short)krow[c0 + e] : (short)0;
                 }
             }
-            k_frag[kb] = val;
+            k_frag[h][kb] = val;
         }
 #pragma unroll
         for (int kb = 0; kb < DVMAX / 32; ++kb) {
@@ -367,25 +387,28 @@ __global__ void flash_dkv_kernel(
                     for (int e = 0; e < 8; ++e) val[e] = (c0 + e < Dv) ? (short)vrow[c0 + e] : (short)0;
                 }
             }
-            v_frag[kb] = val;
+            v_frag[h][kb] = val;
         }
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
-            int ki2 = k0 + hi4 * 4 + r;
-            key_pad[r] = (ki2 >= Lk) || (padrow && padrow[min(ki2, Lk - 1)]);
+            int ki2 = k0 + h * 16 + hi4 * 4 + r;
+            key_pad[h][r] = (ki2 >= Lk) || (padrow && padrow[min(ki2, Lk - 1)]);
         }
     }
 
-    float4v dk_acc[DMAX / 16];
-    float4v dv_acc[DVMAX / 16];
+    float4v dk_acc[QH][DMAX / 16];
+    float4v dv_acc[QH][DVMAX / 16];
 #pragma unroll
-    for (int cb = 0; cb < DMAX / 16; ++cb) dk_acc[cb] = float4v{0.f, 0.f, 0.f, 0.f};
+    for (int h = 0; h < QH; ++h) {
 #pragma unroll
-    for (int cb = 0; cb < DVMAX / 16; ++cb) dv_acc[cb] = float4v{0.f, 0.f, 0.f, 0.f};
+        for (int cb = 0; cb < DMAX / 16; ++cb) dk_acc[h][cb] = float4v{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int cb = 0; cb < DVMAX / 16; ++cb) dv_acc[h][cb] = float4v{0.f, 0.f, 0.f, 0.f};
+    }
 
     int q_start = 0;
     if (causal) {
-        int j_lo = blockIdx.x * QBLK;
+        int j_lo = blockIdx.x * KBLK;
         q_start = max(0, j_lo - (Lk - Nq));
         q_start = (q_start / TILE) * TILE;
     }
@@ -394,20 +417,22 @@ __global__ void flash_dkv_kernel(
     for (int qt0 = q_start; qt0 < Nq; qt0 += TILE) {
         int rows_valid = min(TILE, Nq - qt0);
         __syncthreads();
-        stage_rm(qbase + (long)qt0 * D, D, rows_valid, TILE, D, d_pad, q_lds, q_stride, tid);
-        stage_tr(qbase + (long)qt0 * D, D, rows_valid, TILE, D, d_pad, qt_lds, qt_stride, tid);
-        stage_rm(dobase + (long)qt0 * Dv, Dv, rows_valid, TILE, Dv, dv_pad, do_lds, do_stride, tid);
-        stage_tr(dobase + (long)qt0 * Dv, Dv, rows_valid, TILE, Dv, dv_pad, dot_lds, qt_stride, tid);
+        stage_rm<TILE>(qbase + (long)qt0 * D, D, rows_valid, D, d_pad, q_lds, q_stride, tid);
+        stage_tr<TILE>(qbase + (long)qt0 * D, D, rows_valid, D, d_pad, qt_lds, qt_stride, tid);
+        stage_rm<TILE>(dobase + (long)qt0 * Dv, Dv, rows_valid, Dv, dv_pad, do_lds, do_stride, tid);
+        stage_tr<TILE>(dobase + (long)qt0 * Dv, Dv, rows_valid, Dv, dv_pad, dot_lds, qt_stride, tid);
         __syncthreads();
 
-        // S^T = K Q^T (16 keys x 64 qrows); B[ch k][qrow j]: q_lds row j contiguous ch
-        float4v st_acc[TBLKS];
-        float4v dpt_acc[TBLKS];
+        // S^T = K Q^T (KROWS x TILE); dP^T = V dO^T
+        float4v st_acc[QH][TBLKS];
+        float4v dpt_acc[QH][TBLKS];
 #pragma unroll
-        for (int t = 0; t < TBLKS; ++t) {
-            st_acc[t] = float4v{0.f, 0.f, 0.f, 0.f};
-            dpt_acc[t] = float4v{0.f, 0.f, 0.f, 0.f};
-        }
+        for (int h = 0; h < QH; ++h)
+#pragma unroll
+            for (int t = 0; t < TBLKS; ++t) {
+                st_acc[h][t] = float4v{0.f, 0.f, 0.f, 0.f};
+                dpt_acc[h][t] = float4v{0.f, 0.f, 0.f, 0.f};
+            }
 #pragma unroll
         for (int kb = 0; kb < DMAX / 32; ++kb) {
             if (kb < d_blocks) {
@@ -415,12 +440,13 @@ __global__ void flash_dkv_kernel(
                 for (int t = 0; t < TBLKS; ++t) {
                     const char* src = q_lds + (t * 16 + lo16) * q_stride + (kb * 32 + hi4 * 8) * 2;
                     bf16x8 bfrag = (bf16x8)(*reinterpret_cast<const short8v*>(src));
-                    st_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        (bf16x8)k_frag[kb], bfrag, st_acc[t], 0, 0, 0);
+#pragma unroll
+                    for (int h = 0; h < QH; ++h)
+                        st_acc[h][t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            (bf16x8)k_frag[h][kb], bfrag, st_acc[h][t], 0, 0, 0);
                 }
             }
         }
-        // dP^T = V dO^T (16 keys x 64 qrows)
 #pragma unroll
         for (int kb = 0; kb < DVMAX / 32; ++kb) {
             if (kb < dv_blocks32) {
@@ -428,44 +454,48 @@ __global__ void flash_dkv_kernel(
                 for (int t = 0; t < TBLKS; ++t) {
                     const char* src = do_lds + (t * 16 + lo16) * do_stride + (kb * 32 + hi4 * 8) * 2;
                     bf16x8 bfrag = (bf16x8)(*reinterpret_cast<const short8v*>(src));
-                    dpt_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        (bf16x8)v_frag[kb], bfrag, dpt_acc[t], 0, 0, 0);
+#pragma unroll
+                    for (int h = 0; h < QH; ++h)
+                        dpt_acc[h][t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            (bf16x8)v_frag[h][kb], bfrag, dpt_acc[h][t], 0, 0, 0);
                 }
             }
         }
 
         // P^T (dropped, rescaled -> for dV) to LDS; dS^T stashed in st_acc
 #pragma unroll
-        for (int t = 0; t < TBLKS; ++t) {
+        for (int h = 0; h < QH; ++h)
 #pragma unroll
-            for (int r = 0; r < 4; ++r) {
-                int ki = k0 + hi4 * 4 + r;
-                int qi = qt0 + t * 16 + lo16;
-                bool masked = key_pad[r] || qi >= Nq || (causal && ki > Lk - Nq + qi);
-                float lse_i = (qi < Nq) ? lse_row[qi] : 0.f;
-                float delta_i = (qi < Nq) ? delta_row[qi] : 0.f;
-                float p = masked ? 0.f : expf(st_acc[t][r] - lse_i);
-                float p_eff = p;          // probs actually used in the forward PV
-                float dprobs = dpt_acc[t][r];
-                if (drop_p > 0.f) {
-                    bool kept = rng_hash(drop_seed, bh, qi, ki) >= drop_thresh;
-                    float inv_keep = 1.0f / (1.0f - drop_p);
-                    p_eff = kept ? p * inv_keep : 0.f;
-                    dprobs = kept ? dprobs * inv_keep : 0.f;
+            for (int t = 0; t < TBLKS; ++t)
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    int ki = k0 + h * 16 + hi4 * 4 + r;
+                    int qi = qt0 + t * 16 + lo16;
+                    bool masked = key_pad[h][r] || qi >= Nq || (causal && ki > Lk - Nq + qi);
+                    float lse_i = (qi < Nq) ? lse_row[qi] : 0.f;
+                    float delta_i = (qi < Nq) ? delta_row[qi] : 0.f;
+                    float p = masked ? 0.f : expf(st_acc[h][t][r] - lse_i);
+                    float p_eff = p;
+                    float dprobs = dpt_acc[h][t][r];
+                    if (drop_p > 0.f) {
+                        bool kept = rng_hash(drop_seed, bh, qi, ki) >= drop_thresh;
+                        float inv_keep = 1.0f / (1.0f - drop_p);
+                        p_eff = kept ? p * inv_keep : 0.f;
+                        dprobs = kept ? dprobs * inv_keep : 0.f;
+                    }
+                    float ds = p * (dprobs - delta_i);
+                    *reinterpret_cast<unsigned short*>(
+                        p_mine + (h * 16 + hi4 * 4 + r) * qt_stride + (t * 16 + lo16) * 2) = f2bf(p_eff);
+                    st_acc[h][t][r] = ds;
                 }
-                float ds = p * (dprobs - delta_i);
-                *reinterpret_cast<unsigned short*>(
-                    p_mine + (hi4 * 4 + r) * qt_stride + (t * 16 + lo16) * 2) = f2bf(p_eff);
-                st_acc[t][r] = ds;  // reuse as dS^T storage
-            }
-        }
         __builtin_amdgcn_s_waitcnt(0);
-        bf16x8 pt_frag[TBLKS / 2];
+        bf16x8 pt_frag[QH][TBLKS / 2];
 #pragma unroll
-        for (int t32 = 0; t32 < TBLKS / 2; ++t32) {
-            pt_frag[t32] = (bf16x8)(*reinterpret_cast<const short8v*>(
-                p_mine + lo16 * qt_stride + (t32 * 32 + hi4 * 8) * 2));
-        }
+        for (int h = 0; h < QH; ++h)
+#pragma unroll
+            for (int t32 = 0; t32 < TBLKS / 2; ++t32)
+                pt_frag[h][t32] = (bf16x8)(*reinterpret_cast<const short8v*>(
+                    p_mine + (h * 16 + lo16) * qt_stride + (t32 * 32 + hi4 * 8) * 2));
 
         // dV += P^T dO : B[qrow i][ch c] = dO^T_lds[c][i] (contiguous in i)
 #pragma unroll
@@ -475,28 +505,32 @@ __global__ void flash_dkv_kernel(
                 for (int t32 = 0; t32 < TBLKS / 2; ++t32) {
                     const char* src = dot_lds + (cb * 16 + lo16) * qt_stride + (t32 * 32 + hi4 * 8) * 2;
                     bf16x8 bfrag = (bf16x8)(*reinterpret_cast<const short8v*>(src));
-                    dv_acc[cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pt_frag[t32], bfrag, dv_acc[cb], 0, 0, 0);
+#pragma unroll
+                    for (int h = 0; h < QH; ++h)
+                        dv_acc[h][cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            pt_frag[h][t32], bfrag, dv_acc[h][cb], 0, 0, 0);
                 }
             }
         }
 
-        // second pass: dS^T through the same per-wave buffer
-        __syncthreads();  // uniform: all waves done reading their pt_frag (wave-local, but keep tiles in sync)
+        // second pass: dS^T through the same per-wave buffer (wave-local)
 #pragma unroll
-        for (int t = 0; t < TBLKS; ++t) {
+        for (int h = 0; h < QH; ++h)
 #pragma unroll
-            for (int r = 0; r < 4; ++r) {
-                *reinterpret_cast<unsigned short*>(
-                    p_mine + (hi4 * 4 + r) * qt_stride + (t * 16 + lo16) * 2) = f2bf(st_acc[t][r]);
-            }
-        }
+            for (int t = 0; t < TBLKS; ++t)
+#pragma unroll
+                for (int r = 0; r < 4; ++r)
+                    *reinterpret_cast<unsigned short*>(
+                        p_mine + (h * 16 + hi4 * 4 + r) * qt_stride + (t * 16 + lo16) * 2) =
+                        f2bf(st_acc[h][t][r]);
         __builtin_amdgcn_s_waitcnt(0);
-        bf16x8 dst_frag[TBLKS / 2];
+        bf16x8 dst_frag[QH][TBLKS / 2];
 #pragma unroll
-        for (int t32 = 0; t32 < TBLKS / 2; ++t32) {
-            dst_frag[t32] = (bf16x8)(*reinterpret_cast<const short8v*>(
-                p_mine + lo16 * qt_stride + (t32 * 32 + hi4 * 8) * 2));
-        }
+        for (int h = 0; h < QH; ++h)
+#pragma unroll
+            for (int t32 = 0; t32 < TBLKS / 2; ++t32)
+                dst_frag[h][t32] = (bf16x8)(*reinterpret_cast<const short8v*>(
+                    p_mine + (h * 16 + lo16) * qt_stride + (t32 * 32 + hi4 * 8) * 2));
 
         // dK += dS^T Q : B[qrow i][ch d] = Q^T_lds[d][i] (contiguous in i)
 #pragma unroll
@@ -506,33 +540,38 @@ __global__ void flash_dkv_kernel(
                 for (int t32 = 0; t32 < TBLKS / 2; ++t32) {
                     const char* src = qt_lds + (cb * 16 + lo16) * qt_stride + (t32 * 32 + hi4 * 8) * 2;
                     bf16x8 bfrag = (bf16x8)(*reinterpret_cast<const short8v*>(src));
-                    dk_acc[cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dst_frag[t32], bfrag, dk_acc[cb], 0, 0, 0);
+#pragma unroll
+                    for (int h = 0; h < QH; ++h)
+                        dk_acc[h][cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            dst_frag[h][t32], bfrag, dk_acc[h][cb], 0, 0, 0);
                 }
             }
         }
     }
 
-    // store dK/dV (C layout: key row hi4*4+r, col lo16+16cb)
+    // store dK/dV (C layout: key row h*16 + hi4*4+r, col lo16+16cb)
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-        int ki = k0 + hi4 * 4 + r;
-        if (ki >= Lk) continue;
-        unsigned short* dkrow = dkp + ((long)bh * Lk + ki) * D;
-        unsigned short* dvrow = dvp + ((long)bh * Lk + ki) * Dv;
+    for (int h = 0; h < QH; ++h)
 #pragma unroll
-        for (int cb = 0; cb < DMAX / 16; ++cb) {
-            int c = cb * 16 + lo16;
-            if (c < D) dkrow[c] = f2bf(dk_acc[cb][r]);
+        for (int r = 0; r < 4; ++r) {
+            int ki = k0 + h * 16 + hi4 * 4 + r;
+            if (ki >= Lk) continue;
+            unsigned short* dkrow = dkp + ((long)bh * Lk + ki) * D;
+            unsigned short* dvrow = dvp + ((long)bh * Lk + ki) * Dv;
+#pragma unroll
+            for (int cb = 0; cb < DMAX / 16; ++cb) {
+                int c = cb * 16 + lo16;
+                if (c < D) dkrow[c] = f2bf(dk_acc[h][cb][r]);
+            }
+#pragma unroll
+            for (int cb = 0; cb < DVMAX / 16; ++cb) {
+                int c = cb * 16 + lo16;
+                if (c < Dv) dvrow[c] = f2bf(dv_acc[h][cb][r]);
+            }
         }
-#pragma unroll
-        for (int cb = 0; cb < DVMAX / 16; ++cb) {
-            int c = cb * 16 + lo16;
-            if (c < Dv) dvrow[c] = f2bf(dv_acc[cb][r]);
-        }
-    }
 }
 
-template <int DMAX, int DVMAX, int TILE>
+template <int DMAX, int DVMAX, int DQ_TILE, int DQ_QH, int DKV_TILE, int DKV_QH>
 void launch_flash_bwd(const torch::Tensor& dout, const torch::Tensor& q, const torch::Tensor& k,
                       const torch::Tensor& v, const torch::Tensor& lse, const torch::Tensor& delta,
                       const c10::optional<torch::Tensor>& pad_mask, bool causal,
@@ -547,11 +586,12 @@ void launch_flash_bwd(const torch::Tensor& dout, const torch::Tensor& q, const t
     if (pad_mask.has_value() && pad_mask->defined()) padp = pad_mask->data_ptr<bool>();
 
     {   // dQ
-        const int k_stride = d_pad * 2 + 16, kt_stride = TILE * 2 + 16, v_stride = dv_pad * 2 + 16;
-        size_t smem = (size_t)TILE * k_stride + (size_t)DMAX * kt_stride +
-                      (size_t)TILE * v_stride + (size_t)NWAVES * QROWS * kt_stride;
-        dim3 grid((Nq + QBLK - 1) / QBLK, B * H);
-        hipLaunchKernelGGL((flash_dq_kernel<DMAX, DVMAX, TILE>), grid, dim3(256), smem, stream,
+        const int k_stride = d_pad * 2 + 16, kt_stride = DQ_TILE * 2 + 16, v_stride = dv_pad * 2 + 16;
+        const int qblk = 16 * DQ_QH * NWAVES;
+        size_t smem = (size_t)DQ_TILE * k_stride + (size_t)DMAX * kt_stride +
+                      (size_t)DQ_TILE * v_stride + (size_t)NWAVES * 16 * DQ_QH * kt_stride;
+        dim3 grid((Nq + qblk - 1) / qblk, B * H);
+        hipLaunchKernelGGL((flash_dq_kernel<DMAX, DVMAX, DQ_TILE, DQ_QH>), grid, dim3(256), smem, stream,
                            reinterpret_cast<const unsigned short*>(q.data_ptr()),
                            reinterpret_cast<const unsigned short*>(k.data_ptr()),
                            reinterpret_cast<const unsigned short*>(v.data_ptr()),
@@ -562,13 +602,14 @@ void launch_flash_bwd(const torch::Tensor& dout, const torch::Tensor& q, const t
         HIP_CHECK_LAST();
     }
     {   // dK/dV
-        int q_stride = d_pad * 2 + 16, qt_stride = TILE * 2 + 16;
-        int do_stride = dv_pad * 2 + 16;
-        size_t smem = (size_t)TILE * q_stride + (size_t)DMAX * qt_stride +
-                      (size_t)TILE * do_stride + (size_t)DVMAX * qt_stride +
-                      (size_t)NWAVES * QROWS * qt_stride;
-        dim3 grid((Lk + QBLK - 1) / QBLK, B * H);
-        hipLaunchKernelGGL((flash_dkv_kernel<DMAX, DVMAX, TILE>), grid, dim3(256), smem, stream,
+        const int q_stride = d_pad * 2 + 16, qt_stride = DKV_TILE * 2 + 16;
+        const int do_stride = dv_pad * 2 + 16;
+        const int kblk = 16 * DKV_QH * NWAVES;
+        size_t smem = (size_t)DKV_TILE * q_stride + (size_t)DMAX * qt_stride +
+                      (size_t)DKV_TILE * do_stride + (size_t)DVMAX * qt_stride +
+                      (size_t)NWAVES * 16 * DKV_QH * qt_stride;
+        dim3 grid((Lk + kblk - 1) / kblk, B * H);
+        hipLaunchKernelGGL((flash_dkv_kernel<DMAX, DVMAX, DKV_TILE, DKV_QH>), grid, dim3(256), smem, stream,
                            reinterpret_cast<const unsigned short*>(q.data_ptr()),
                            reinterpret_cast<const unsigned short*>(k.data_ptr()),
                            reinterpret_cast<const unsigned short*>(v.data_ptr()),
@@ -615,11 +656,18 @@ std::vector<torch::Tensor> flash_bwd(torch::Tensor dout, torch::Tensor q, torch:
 
     float dp = (float)dropout_p;
     unsigned long long sd = (unsigned long long)seed;
-    if (D <= 32 && Dv <= 160)       launch_flash_bwd<32, 160, 64>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
-    else if (D <= 64 && Dv <= 64)   launch_flash_bwd<64, 64, 64>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
-    else if (D <= 128 && Dv <= 128) launch_flash_bwd<128, 128, 64>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
-    else if (D <= 160 && Dv <= 160) launch_flash_bwd<160, 160, 64>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
-    else                            launch_flash_bwd<352, 352, 32>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
+    // <DMAX, DVMAX, DQ_TILE, DQ_QH, DKV_TILE, DKV_QH> — tiles/QH balance LDS
+    // occupancy against the 256-VGPR budget per template
+    if (D <= 32 && Dv <= 160)
+        launch_flash_bwd<32, 160, 64, 2, 32, 2>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
+    else if (D <= 64 && Dv <= 64)
+        launch_flash_bwd<64, 64, 64, 2, 64, 2>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
+    else if (D <= 128 && Dv <= 128)
+        launch_flash_bwd<128, 128, 64, 2, 32, 1>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
+    else if (D <= 160 && Dv <= 160)
+        launch_flash_bwd<160, 160, 64, 1, 32, 1>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
+    else
+        launch_flash_bwd<352, 352, 32, 1, 32, 1>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
 
     return {dq, dk, dv};
 }

@@ -31,9 +31,7 @@ namespace {
 
 constexpr int KVBLK = 64;       // keys per tile
 constexpr int KEYBLKS = KVBLK / 16;
-constexpr int QROWS = 16;       // q rows per wave
 constexpr int NWAVES = 4;       // waves per workgroup
-constexpr int QBLK = QROWS * NWAVES;
 
 DEVINL float warp16_max(float x) {
 #pragma unroll
@@ -110,7 +108,9 @@ DEVINL void stage_tile_transposed(const unsigned short* __restrict__ src, long s
     }
 }
 
-template <int DMAX, int DVMAX>
+// QH: 16-row A-fragments per wave (2 doubles MFMA work per B-fragment LDS read;
+// 1 for the large-D templates where the register budget is spent on O accumulators)
+template <int DMAX, int DVMAX, int QH>
 __launch_bounds__(256)
 __global__ void flash_fwd_kernel(
     const unsigned short* __restrict__ qp,  // (B,H,Nq,D) bf16, pre-scaled
@@ -126,6 +126,8 @@ __global__ void flash_fwd_kernel(
     const int d_blocks = d_pad / 32;       // QK^T k-steps
     const int dv_blocks = dv_pad / 16;     // O column blocks
 
+    constexpr int QROWS = 16 * QH;
+    constexpr int QBLK = QROWS * NWAVES;
     const int tid = threadIdx.x;
     const int wave = tid / 64;
     const int lane = tid % 64;
@@ -151,9 +153,10 @@ __global__ void flash_fwd_kernel(
     char* p_mine = p_lds + wave * QROWS * vt_stride;
 
     // ---- Q fragments: lane holds A[i=lo16][k = hi4*8 + e] per 32-wide k-block ----
-    short8v q_frag[DMAX / 32];
-    {
-        int qi = q0 + lo16;
+    short8v q_frag[QH][DMAX / 32];
+#pragma unroll
+    for (int h = 0; h < QH; ++h) {
+        int qi = q0 + h * 16 + lo16;
         bool valid = qi < Nq;
         int qclamp = valid ? qi : Nq - 1;
         const unsigned short* qrow = qbase + (long)qclamp * D;
@@ -169,17 +172,20 @@ __global__ void flash_fwd_kernel(
                     for (int e = 0; e < 8; ++e) val[e] = (c0 + e < D) ? (short)qrow[c0 + e] : (short)0;
                 }
             }
-            q_frag[kb] = val;
+            q_frag[h][kb] = val;
         }
     }
 
-    // ---- accumulators (C layout: lane holds rows hi4*4+r, col lo16 + 16*cb) ----
-    float4v o_acc[DVMAX / 16];
+    // ---- accumulators (C layout: lane holds rows h*16 + hi4*4+r, col lo16 + 16*cb) ----
+    float4v o_acc[QH][DVMAX / 16];
+    float m_run[QH][4], l_run[QH][4];
 #pragma unroll
-    for (int cb = 0; cb < DVMAX / 16; ++cb) o_acc[cb] = float4v{0.f, 0.f, 0.f, 0.f};
-    float m_run[4], l_run[4];
+    for (int h = 0; h < QH; ++h) {
 #pragma unroll
-    for (int r = 0; r < 4; ++r) { m_run[r] = -INFINITY; l_run[r] = 0.f; }
+        for (int cb = 0; cb < DVMAX / 16; ++cb) o_acc[h][cb] = float4v{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int r = 0; r < 4; ++r) { m_run[h][r] = -INFINITY; l_run[h][r] = 0.f; }
+    }
 
     // causal: q row qi may attend keys j <= Lk - Nq + qi
     int kv_end = Lk;
@@ -197,10 +203,12 @@ __global__ void flash_fwd_kernel(
         stage_tile_transposed(vbase + (long)kv0 * Dv, Dv, rows_valid, Dv, dv_pad, vt_lds, vt_stride, tid);
         __syncthreads();
 
-        // ---- S = Q K^T (16 rows x KVBLK keys) ----
-        float4v s_acc[KEYBLKS];
+        // ---- S = Q K^T (QH x 16 rows x KVBLK keys); one B read feeds QH MFMAs ----
+        float4v s_acc[QH][KEYBLKS];
 #pragma unroll
-        for (int kb = 0; kb < KEYBLKS; ++kb) s_acc[kb] = float4v{0.f, 0.f, 0.f, 0.f};
+        for (int h = 0; h < QH; ++h)
+#pragma unroll
+            for (int kb = 0; kb < KEYBLKS; ++kb) s_acc[h][kb] = float4v{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
         for (int kb = 0; kb < DMAX / 32; ++kb) {
             if (kb < d_blocks) {
@@ -209,88 +217,90 @@ __global__ void flash_fwd_kernel(
                 for (int keyblk = 0; keyblk < KEYBLKS; ++keyblk) {
                     const char* src = k_lds + (keyblk * 16 + lo16) * k_stride + (kb * 32 + hi4 * 8) * 2;
                     bf16x8 bfrag = (bf16x8)(*reinterpret_cast<const short8v*>(src));
-                    s_acc[keyblk] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        (bf16x8)q_frag[kb], bfrag, s_acc[keyblk], 0, 0, 0);
+#pragma unroll
+                    for (int h = 0; h < QH; ++h) {
+                        s_acc[h][keyblk] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            (bf16x8)q_frag[h][kb], bfrag, s_acc[h][keyblk], 0, 0, 0);
+                    }
                 }
             }
         }
 
-        // ---- mask + online softmax (C layout: row hi4*4+r, col kv0 + kb*16 + lo16) ----
-        float pvals[KEYBLKS][4];
-        float rowmax[4];
+        // ---- mask + online softmax per 16-row group ----
+        float pvals[QH][KEYBLKS][4];
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-            int qi = q0 + hi4 * 4 + r;
-            float mx = -FLT_MAX;
+        for (int h = 0; h < QH; ++h) {
+            float rowmax[4];
 #pragma unroll
-            for (int kb = 0; kb < KEYBLKS; ++kb) {
-                int j = kv0 + kb * 16 + lo16;
-                float s = s_acc[kb][r];
-                bool masked = j >= Lk;
-                if (padrow && j < Lk) masked |= padrow[j];
-                if (causal && j > Lk - Nq + qi) masked = true;
-                s = masked ? -FLT_MAX : s;
-                pvals[kb][r] = s;
-                mx = fmaxf(mx, s);
+            for (int r = 0; r < 4; ++r) {
+                int qi = q0 + h * 16 + hi4 * 4 + r;
+                float mx = -FLT_MAX;
+#pragma unroll
+                for (int kb = 0; kb < KEYBLKS; ++kb) {
+                    int j = kv0 + kb * 16 + lo16;
+                    float sv = s_acc[h][kb][r];
+                    bool masked = j >= Lk;
+                    if (padrow && j < Lk) masked |= padrow[j];
+                    if (causal && j > Lk - Nq + qi) masked = true;
+                    sv = masked ? -FLT_MAX : sv;
+                    pvals[h][kb][r] = sv;
+                    mx = fmaxf(mx, sv);
+                }
+                rowmax[r] = warp16_max(mx);
             }
-            rowmax[r] = warp16_max(mx);
-        }
-
-        float alpha[4];
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-            float m_new = fmaxf(m_run[r], rowmax[r]);
-            alpha[r] = expf(m_run[r] - m_new);
-            m_run[r] = m_new;
-            float psum = 0.f;
+            for (int r = 0; r < 4; ++r) {
+                float m_new = fmaxf(m_run[h][r], rowmax[r]);
+                float alpha = expf(m_run[h][r] - m_new);
+                m_run[h][r] = m_new;
+                float psum = 0.f;
 #pragma unroll
-            for (int kb = 0; kb < KEYBLKS; ++kb) {
-                float p = expf(pvals[kb][r] - m_new);
-                pvals[kb][r] = p;
-                psum += p;
+                for (int kb = 0; kb < KEYBLKS; ++kb) {
+                    float pv = expf(pvals[h][kb][r] - m_new);
+                    pvals[h][kb][r] = pv;
+                    psum += pv;
+                }
+                l_run[h][r] = l_run[h][r] * alpha + warp16_sum(psum);
+#pragma unroll
+                for (int cb = 0; cb < DVMAX / 16; ++cb) o_acc[h][cb][r] *= alpha;
             }
-            l_run[r] = l_run[r] * alpha[r] + warp16_sum(psum);
-        }
-
-        // rescale O
-#pragma unroll
-        for (int cb = 0; cb < DVMAX / 16; ++cb) {
-#pragma unroll
-            for (int r = 0; r < 4; ++r) o_acc[cb][r] *= alpha[r];
         }
 
         // ---- dropout on (unnormalized) probabilities; l uses the undropped sum ----
         if (drop_p > 0.f) {
 #pragma unroll
-            for (int kb = 0; kb < KEYBLKS; ++kb) {
+            for (int h = 0; h < QH; ++h)
 #pragma unroll
-                for (int r = 0; r < 4; ++r) {
-                    int qi = q0 + hi4 * 4 + r;
-                    int j = kv0 + kb * 16 + lo16;
-                    if (rng_hash(drop_seed, bh, qi, j) < drop_thresh) pvals[kb][r] = 0.f;
-                }
-            }
+                for (int kb = 0; kb < KEYBLKS; ++kb)
+#pragma unroll
+                    for (int r = 0; r < 4; ++r) {
+                        int qi = q0 + h * 16 + hi4 * 4 + r;
+                        int j = kv0 + kb * 16 + lo16;
+                        if (rng_hash(drop_seed, bh, qi, j) < drop_thresh) pvals[h][kb][r] = 0.f;
+                    }
         }
 
         // ---- redistribute P (C layout) -> A layout via per-wave LDS ----
 #pragma unroll
-        for (int kb = 0; kb < KEYBLKS; ++kb) {
+        for (int h = 0; h < QH; ++h)
 #pragma unroll
-            for (int r = 0; r < 4; ++r) {
-                int prow = hi4 * 4 + r;
-                int pcol = kb * 16 + lo16;
-                *reinterpret_cast<unsigned short*>(p_mine + prow * vt_stride + pcol * 2) =
-                    f2bf(pvals[kb][r]);
-            }
-        }
+            for (int kb = 0; kb < KEYBLKS; ++kb)
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    int prow = h * 16 + hi4 * 4 + r;
+                    int pcol = kb * 16 + lo16;
+                    *reinterpret_cast<unsigned short*>(p_mine + prow * vt_stride + pcol * 2) =
+                        f2bf(pvals[h][kb][r]);
+                }
         __builtin_amdgcn_s_waitcnt(0);  // lgkmcnt(0): wave-local LDS ordering
-        // A-layout fragments per 32-key block: lane reads row lo16, k = kb32*32 + hi4*8
-        bf16x8 p_frag[KEYBLKS / 2];
+        bf16x8 p_frag[QH][KEYBLKS / 2];
 #pragma unroll
-        for (int kb32 = 0; kb32 < KEYBLKS / 2; ++kb32) {
-            p_frag[kb32] = (bf16x8)(*reinterpret_cast<const short8v*>(
-                p_mine + lo16 * vt_stride + (kb32 * 32 + hi4 * 8) * 2));
-        }
+        for (int h = 0; h < QH; ++h)
+#pragma unroll
+            for (int kb32 = 0; kb32 < KEYBLKS / 2; ++kb32) {
+                p_frag[h][kb32] = (bf16x8)(*reinterpret_cast<const short8v*>(
+                    p_mine + (h * 16 + lo16) * vt_stride + (kb32 * 32 + hi4 * 8) * 2));
+            }
 
         // ---- O += P V : B[k][j] = V[key k][ch j] = vt_lds[ch j][key k] ----
 #pragma unroll
@@ -300,7 +310,11 @@ __global__ void flash_fwd_kernel(
                 for (int kb32 = 0; kb32 < KEYBLKS / 2; ++kb32) {
                     const char* src = vt_lds + (cb * 16 + lo16) * vt_stride + (kb32 * 32 + hi4 * 8) * 2;
                     bf16x8 bfrag = (bf16x8)(*reinterpret_cast<const short8v*>(src));
-                    o_acc[cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(p_frag[kb32], bfrag, o_acc[cb], 0, 0, 0);
+#pragma unroll
+                    for (int h = 0; h < QH; ++h) {
+                        o_acc[h][cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            p_frag[h][kb32], bfrag, o_acc[h][cb], 0, 0, 0);
+                    }
                 }
             }
         }
@@ -308,28 +322,32 @@ __global__ void flash_fwd_kernel(
 
     // ---- epilogue: O /= l (x dropout keep-rate), store O (bf16) and lse ----
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-        int qi = q0 + hi4 * 4 + r;
-        if (qi >= Nq) continue;
-        float l_eff = l_run[r] * (drop_p > 0.f ? (1.0f - drop_p) : 1.0f);
-        float inv_l = (l_eff > 0.f) ? 1.0f / l_eff : 0.f;
-        unsigned short* orow = op + ((long)bh * Nq + qi) * Dv;
+    for (int h = 0; h < QH; ++h)
 #pragma unroll
-        for (int cb = 0; cb < DVMAX / 16; ++cb) {
-            int c = cb * 16 + lo16;
-            if (cb < dv_blocks && c < Dv) orow[c] = f2bf(o_acc[cb][r] * inv_l);
+        for (int r = 0; r < 4; ++r) {
+            int qi = q0 + h * 16 + hi4 * 4 + r;
+            if (qi >= Nq) continue;
+            float l_eff = l_run[h][r] * (drop_p > 0.f ? (1.0f - drop_p) : 1.0f);
+            float inv_l = (l_eff > 0.f) ? 1.0f / l_eff : 0.f;
+            unsigned short* orow = op + ((long)bh * Nq + qi) * Dv;
+#pragma unroll
+            for (int cb = 0; cb < DVMAX / 16; ++cb) {
+                int c = cb * 16 + lo16;
+                if (cb < dv_blocks && c < Dv) orow[c] = f2bf(o_acc[h][cb][r] * inv_l);
+            }
+            if (lo16 == 0) {
+                lsep[(long)bh * Nq + qi] = m_run[h][r] + logf(fmaxf(l_run[h][r], 1e-37f));
+            }
         }
-        if (lo16 == 0) {
-            lsep[(long)bh * Nq + qi] = m_run[r] + logf(fmaxf(l_run[r], 1e-37f));
-        }
-    }
 }
 
-template <int DMAX, int DVMAX>
+template <int DMAX, int DVMAX, int QH>
 void launch_flash_fwd(const torch::Tensor& q, const torch::Tensor& k, const torch::Tensor& v,
                       const c10::optional<torch::Tensor>& pad_mask, bool causal,
                       float drop_p, unsigned long long drop_seed,
                       torch::Tensor& out, torch::Tensor& lse) {
+    constexpr int QROWS = 16 * QH;
+    constexpr int QBLK = QROWS * NWAVES;
     int B = q.size(0), H = q.size(1), Nq = q.size(2), D = q.size(3);
     int Lk = k.size(2), Dv = v.size(3);
     const int d_pad = (D + 31) & ~31;
@@ -342,7 +360,7 @@ void launch_flash_fwd(const torch::Tensor& q, const torch::Tensor& k, const torc
     if (pad_mask.has_value() && pad_mask->defined()) {
         padp = pad_mask->data_ptr<bool>();
     }
-    hipLaunchKernelGGL((flash_fwd_kernel<DMAX, DVMAX>), grid, dim3(256), smem,
+    hipLaunchKernelGGL((flash_fwd_kernel<DMAX, DVMAX, QH>), grid, dim3(256), smem,
                        at::cuda::getCurrentCUDAStream(),
                        reinterpret_cast<const unsigned short*>(q.data_ptr()),
                        reinterpret_cast<const unsigned short*>(k.data_ptr()),
@@ -387,11 +405,11 @@ std::vector<torch::Tensor> flash_fwd(torch::Tensor q, torch::Tensor k, torch::Te
 
     float dp = (float)dropout_p;
     unsigned long long sd = (unsigned long long)seed;
-    if (D <= 32 && Dv <= 160)       launch_flash_fwd<32, 160>(q, k, v, pm, causal, dp, sd, out, lse);
-    else if (D <= 64 && Dv <= 64)   launch_flash_fwd<64, 64>(q, k, v, pm, causal, dp, sd, out, lse);
-    else if (D <= 128 && Dv <= 128) launch_flash_fwd<128, 128>(q, k, v, pm, causal, dp, sd, out, lse);
-    else if (D <= 160 && Dv <= 160) launch_flash_fwd<160, 160>(q, k, v, pm, causal, dp, sd, out, lse);
-    else                            launch_flash_fwd<352, 352>(q, k, v, pm, causal, dp, sd, out, lse);
+    if (D <= 32 && Dv <= 160)       launch_flash_fwd<32, 160, 2>(q, k, v, pm, causal, dp, sd, out, lse);
+    else if (D <= 64 && Dv <= 64)   launch_flash_fwd<64, 64, 2>(q, k, v, pm, causal, dp, sd, out, lse);
+    else if (D <= 128 && Dv <= 128) launch_flash_fwd<128, 128, 2>(q, k, v, pm, causal, dp, sd, out, lse);
+    else if (D <= 160 && Dv <= 160) launch_flash_fwd<160, 160, 1>(q, k, v, pm, causal, dp, sd, out, lse);
+    else                            launch_flash_fwd<352, 352, 1>(q, k, v, pm, causal, dp, sd, out, lse);
 
     return {out, lse};
 }
