@@ -29,6 +29,16 @@ os.environ.setdefault("TOKENIZERS_PARALLELISM", "false")
 PERCEIVER_TOKENIZERS = ["krasserm/perceiver-io-mlm", "deepmind/language-perceiver"]
 
 
+def resolve_tokenizer(tokenizer):
+    """Accepts a 🤗 hub name OR a tokenizer instance (offline-friendly: the
+    byte-level PerceiverTokenizer constructs without network access)."""
+    if isinstance(tokenizer, str):
+        from transformers import AutoTokenizer
+
+        return AutoTokenizer.from_pretrained(tokenizer, verbose=False)
+    return tokenizer
+
+
 class Hparams(dict):
     """Attribute-access hyperparameter store (stand-in for Lightning's hparams)."""
 
@@ -48,10 +58,8 @@ class Task(Enum):
 class TextPreprocessor:
     """Inference-side text -> (input_ids, pad_mask) preprocessing."""
 
-    def __init__(self, tokenizer: str, max_seq_len: int, add_special_tokens: bool):
-        from transformers import AutoTokenizer
-
-        self.tokenizer = AutoTokenizer.from_pretrained(tokenizer, verbose=False)
+    def __init__(self, tokenizer, max_seq_len: int, add_special_tokens: bool):
+        self.tokenizer = resolve_tokenizer(tokenizer)
         self.max_seq_len = max_seq_len
         self.add_special_tokens = add_special_tokens
 
@@ -99,8 +107,6 @@ class TextDataModule:
         pin_memory: bool = True,
         **extra_hparams,
     ):
-        from transformers import AutoTokenizer
-
         self.hparams = Hparams(
             dataset_dir=dataset_dir, tokenizer=tokenizer, max_seq_len=max_seq_len, task=task,
             mask_prob=mask_prob, mask_words=mask_words, static_masking=static_masking,
@@ -116,13 +122,17 @@ class TextDataModule:
         if static_masking and not mask_words:
             raise ValueError("static_masking=true is only supported for mask_words=true")
 
-        self.tokenizer = AutoTokenizer.from_pretrained(tokenizer, verbose=False)
+        self.tokenizer = resolve_tokenizer(tokenizer)
         if padding_side is not None:
             self.tokenizer.padding_side = padding_side
 
         # the PerceiverTokenizer is not a fast tokenizer: word ids come from
         # whitespace boundaries instead of encoding.word_ids()
-        self.perceiver_tokenizer_configured = tokenizer in PERCEIVER_TOKENIZERS
+        from transformers import PerceiverTokenizer
+
+        self.perceiver_tokenizer_configured = (
+            tokenizer in PERCEIVER_TOKENIZERS or isinstance(self.tokenizer, PerceiverTokenizer)
+        )
         if self.perceiver_tokenizer_configured:
             self.perceiver_tokenizer_util = PerceiverTokenizerUtil(self.tokenizer)
 

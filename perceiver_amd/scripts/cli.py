@@ -1,0 +1,169 @@
+"""Training CLI — the LightningCLI-equivalent of the reference's scripts/cli.py,
+built on the native Trainer.
+
+Configuration model:
+  - per-task scripts supply flagship defaults (the reference's ``set_defaults``)
+    and link functions (the reference's ``link_arguments``, e.g. data.vocab_size ->
+    model.encoder.vocab_size);
+  - flags use dotted paths: ``--model.num_latents 256 --data.batch_size 32
+    --trainer.max_steps 1000 --optimizer.lr 1e-3``;
+  - ``--config file.yaml`` merges a YAML tree under the same keys; later flags win;
+  - the resolved config is saved to ``<out_dir>/config.yaml``.
+
+Subcommands: fit (default), validate.
+"""
+from __future__ import annotations
+
+import copy
+import os
+import sys
+from dataclasses import fields, is_dataclass
+from typing import Any, Callable, Dict, Optional
+
+import yaml
+
+from perceiver_amd.train.trainer import TrainConfig, Trainer
+
+
+def _set_path(tree: Dict, path: str, value: Any) -> None:
+    keys = path.split(".")
+    node = tree
+    for k in keys[:-1]:
+        node = node.setdefault(k, {})
+    node[keys[-1]] = value
+
+
+def _merge(base: Dict, override: Dict) -> Dict:
+    out = copy.deepcopy(base)
+    for k, v in override.items():
+        if isinstance(v, dict) and isinstance(out.get(k), dict):
+            out[k] = _merge(out[k], v)
+        else:
+            out[k] = v
+    return out
+
+
+def _parse_value(raw: str) -> Any:
+    try:
+        return yaml.safe_load(raw)
+    except yaml.YAMLError:
+        return raw
+
+
+def parse_cli_config(argv, defaults: Dict) -> (str, Dict):
+    """Returns (subcommand, merged config tree)."""
+    subcommand = "fit"
+    if argv and not argv[0].startswith("-"):
+        subcommand = argv[0]
+        argv = argv[1:]
+
+    tree: Dict = {}
+    i = 0
+    while i < len(argv):
+        arg = argv[i]
+        if not arg.startswith("--"):
+            raise SystemExit(f"unexpected argument: {arg}")
+        key = arg[2:]
+        if "=" in key:
+            key, raw = key.split("=", 1)
+            i += 1
+        else:
+            if i + 1 >= len(argv):
+                raise SystemExit(f"missing value for --{key}")
+            raw = argv[i + 1]
+            i += 2
+        if key == "config":
+            with open(raw) as f:
+                tree = _merge(tree, yaml.safe_load(f) or {})
+        else:
+            _set_path(tree, key, _parse_value(raw))
+
+    return subcommand, _merge(defaults, tree)
+
+
+def build_dataclass(cls, cfg: Dict):
+    """Instantiate a (possibly nested) dataclass from a config dict, ignoring
+    unknown keys (mirrors jsonargparse's dataclass binding)."""
+    if not is_dataclass(cls):
+        return cfg
+    kwargs = {}
+    for f in fields(cls):
+        if f.name in cfg:
+            kwargs[f.name] = cfg[f.name]
+    return cls(**kwargs)
+
+
+class CLI:
+    """Per-task training CLI driver.
+
+    :param lit_cls: task wrapper class (perceiver_amd.train.lit.*)
+    :param datamodule_cls: data module class
+    :param defaults: {"model": {...}, "data": {...}, "trainer": {...},
+                      "optimizer": {...}} flagship defaults
+    :param build_model: callable(model_cfg: dict, datamodule) -> lit task instance
+    :param link: callable(cfg, datamodule) mutating cfg["model"] from data properties
+    """
+
+    def __init__(self, lit_cls, datamodule_cls, defaults: Dict,
+                 build_model: Callable, link: Optional[Callable] = None,
+                 argv=None, run: bool = True):
+        self.lit_cls = lit_cls
+        self.datamodule_cls = datamodule_cls
+        argv = sys.argv[1:] if argv is None else argv
+        self.subcommand, self.config = parse_cli_config(argv, defaults)
+        self.build_model = build_model
+        self.link = link
+        if run:
+            self.run()
+
+    def run(self):
+        cfg = self.config
+        datamodule = self.datamodule_cls(**cfg.get("data", {}))
+        if self.link is not None:
+            self.link(cfg, datamodule)
+
+        trainer_cfg = dict(cfg.get("trainer", {}))
+        opt_cfg = dict(cfg.get("optimizer", {}))
+        for k in ("lr", "weight_decay", "warmup_steps", "lr_schedule", "min_lr_fraction"):
+            if k in opt_cfg:
+                trainer_cfg[k] = opt_cfg[k]
+        ckpt_path = trainer_cfg.pop("ckpt_path", None)
+        tc = TrainConfig(**trainer_cfg)
+        trainer = Trainer(tc)
+
+        task = self.build_model(cfg.get("model", {}), datamodule)
+
+        from perceiver_amd.parallel import is_main_process
+
+        if is_main_process():
+            os.makedirs(tc.out_dir, exist_ok=True)
+            with open(os.path.join(tc.out_dir, "config.yaml"), "w") as f:
+                yaml.safe_dump(_sanitize(cfg), f)
+
+        if self.subcommand == "fit":
+            trainer.fit(task, datamodule=datamodule, ckpt_path=ckpt_path)
+        elif self.subcommand == "validate":
+            datamodule.prepare_data()
+            datamodule.setup("validate")
+            if ckpt_path:
+                trainer._restore(task, None, None, ckpt_path)
+            task = task.to(trainer.device)
+            metrics = trainer._validate(task, datamodule.val_dataloader())
+            print(metrics)
+        else:
+            raise SystemExit(f"unknown subcommand {self.subcommand}")
+        return trainer
+
+
+def _sanitize(obj):
+    from dataclasses import asdict
+
+    if is_dataclass(obj) and not isinstance(obj, type):
+        return asdict(obj)
+    if isinstance(obj, dict):
+        return {k: _sanitize(v) for k, v in obj.items()}
+    if isinstance(obj, (list, tuple)):
+        return [_sanitize(v) for v in obj]
+    if isinstance(obj, (str, int, float, bool)) or obj is None:
+        return obj
+    return str(obj)
