@@ -205,59 +205,48 @@ __global__ void flash_dq_kernel(
         stage_rm<TILE>(vbase + (long)kv0 * Dv, Dv, rows_valid, Dv, dv_pad, v_lds, v_stride, tid);
         __syncthreads();
 
-        // S = Q K^T (QHx16 x TILE)
-        float4v s_acc[QH][TBLKS];
-        float4v dp_acc[QH][TBLKS];
+        // t-outer: per 16-key block compute S and dP with short-lived accumulators,
+        // convert to dS and spill to the per-wave LDS buffer immediately
 #pragma unroll
-        for (int h = 0; h < QH; ++h)
+        for (int t = 0; t < TBLKS; ++t) {
+            float4v s_acc[QH], dp_acc[QH];
 #pragma unroll
-            for (int t = 0; t < TBLKS; ++t) {
-                s_acc[h][t] = float4v{0.f, 0.f, 0.f, 0.f};
-                dp_acc[h][t] = float4v{0.f, 0.f, 0.f, 0.f};
+            for (int h = 0; h < QH; ++h) {
+                s_acc[h] = float4v{0.f, 0.f, 0.f, 0.f};
+                dp_acc[h] = float4v{0.f, 0.f, 0.f, 0.f};
             }
 #pragma unroll
-        for (int kb = 0; kb < DMAX / 32; ++kb) {
-            if (kb < d_blocks) {
-#pragma unroll
-                for (int t = 0; t < TBLKS; ++t) {
+            for (int kb = 0; kb < DMAX / 32; ++kb) {
+                if (kb < d_blocks) {
                     const char* src = k_lds + (t * 16 + lo16) * k_stride + (kb * 32 + hi4 * 8) * 2;
                     bf16x8 bfrag = (bf16x8)(*reinterpret_cast<const short8v*>(src));
 #pragma unroll
                     for (int h = 0; h < QH; ++h)
-                        s_acc[h][t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                            (bf16x8)q_frag[h][kb], bfrag, s_acc[h][t], 0, 0, 0);
+                        s_acc[h] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            (bf16x8)q_frag[h][kb], bfrag, s_acc[h], 0, 0, 0);
                 }
             }
-        }
-        // dP = dO V^T
 #pragma unroll
-        for (int kb = 0; kb < DVMAX / 32; ++kb) {
-            if (kb < dv_blocks32) {
-#pragma unroll
-                for (int t = 0; t < TBLKS; ++t) {
+            for (int kb = 0; kb < DVMAX / 32; ++kb) {
+                if (kb < dv_blocks32) {
                     const char* src = v_lds + (t * 16 + lo16) * v_stride + (kb * 32 + hi4 * 8) * 2;
                     bf16x8 bfrag = (bf16x8)(*reinterpret_cast<const short8v*>(src));
 #pragma unroll
                     for (int h = 0; h < QH; ++h)
-                        dp_acc[h][t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                            (bf16x8)do_frag[h][kb], bfrag, dp_acc[h][t], 0, 0, 0);
+                        dp_acc[h] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            (bf16x8)do_frag[h][kb], bfrag, dp_acc[h], 0, 0, 0);
                 }
             }
-        }
-
-        // dS = P * (dprobs - delta), P = exp(S - lse)  [C layout] -> per-wave LDS
 #pragma unroll
-        for (int h = 0; h < QH; ++h)
-#pragma unroll
-            for (int t = 0; t < TBLKS; ++t)
+            for (int h = 0; h < QH; ++h)
 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
                     int qi = q0 + h * 16 + hi4 * 4 + r;
                     int j = kv0 + t * 16 + lo16;
                     bool masked = j >= Lk || (padrow && j < Lk && padrow[j]) ||
                                   (causal && j > Lk - Nq + qi);
-                    float p = masked ? 0.f : expf(s_acc[h][t][r] - lse_r[h][r]);
-                    float dprobs = dp_acc[h][t][r];
+                    float p = masked ? 0.f : expf(s_acc[h][r] - lse_r[h][r]);
+                    float dprobs = dp_acc[h][r];
                     if (drop_p > 0.f) {
                         bool kept = rng_hash(drop_seed, bh, qi, j) >= drop_thresh;
                         dprobs = kept ? dprobs / (1.0f - drop_p) : 0.f;
@@ -266,6 +255,7 @@ __global__ void flash_dq_kernel(
                     *reinterpret_cast<unsigned short*>(
                         p_mine + (h * 16 + hi4 * 4 + r) * kt_stride + (t * 16 + lo16) * 2) = f2bf(ds);
                 }
+        }
         __builtin_amdgcn_s_waitcnt(0);
         bf16x8 ds_frag[QH][TBLKS / 2];
 #pragma unroll
@@ -423,50 +413,41 @@ __global__ void flash_dkv_kernel(
         stage_tr<TILE>(dobase + (long)qt0 * Dv, Dv, rows_valid, Dv, dv_pad, dot_lds, qt_stride, tid);
         __syncthreads();
 
-        // S^T = K Q^T (KROWS x TILE); dP^T = V dO^T
-        float4v st_acc[QH][TBLKS];
-        float4v dpt_acc[QH][TBLKS];
+        // t-outer: per 16-q-row block compute S^T and dP^T with short-lived
+        // accumulators; P^T goes to LDS now, dS^T is kept in a small register array
+        float ds_keep[QH][TBLKS][4];
 #pragma unroll
-        for (int h = 0; h < QH; ++h)
+        for (int t = 0; t < TBLKS; ++t) {
+            float4v st_acc[QH], dpt_acc[QH];
 #pragma unroll
-            for (int t = 0; t < TBLKS; ++t) {
-                st_acc[h][t] = float4v{0.f, 0.f, 0.f, 0.f};
-                dpt_acc[h][t] = float4v{0.f, 0.f, 0.f, 0.f};
+            for (int h = 0; h < QH; ++h) {
+                st_acc[h] = float4v{0.f, 0.f, 0.f, 0.f};
+                dpt_acc[h] = float4v{0.f, 0.f, 0.f, 0.f};
             }
 #pragma unroll
-        for (int kb = 0; kb < DMAX / 32; ++kb) {
-            if (kb < d_blocks) {
-#pragma unroll
-                for (int t = 0; t < TBLKS; ++t) {
+            for (int kb = 0; kb < DMAX / 32; ++kb) {
+                if (kb < d_blocks) {
                     const char* src = q_lds + (t * 16 + lo16) * q_stride + (kb * 32 + hi4 * 8) * 2;
                     bf16x8 bfrag = (bf16x8)(*reinterpret_cast<const short8v*>(src));
 #pragma unroll
                     for (int h = 0; h < QH; ++h)
-                        st_acc[h][t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                            (bf16x8)k_frag[h][kb], bfrag, st_acc[h][t], 0, 0, 0);
+                        st_acc[h] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            (bf16x8)k_frag[h][kb], bfrag, st_acc[h], 0, 0, 0);
                 }
             }
-        }
 #pragma unroll
-        for (int kb = 0; kb < DVMAX / 32; ++kb) {
-            if (kb < dv_blocks32) {
-#pragma unroll
-                for (int t = 0; t < TBLKS; ++t) {
+            for (int kb = 0; kb < DVMAX / 32; ++kb) {
+                if (kb < dv_blocks32) {
                     const char* src = do_lds + (t * 16 + lo16) * do_stride + (kb * 32 + hi4 * 8) * 2;
                     bf16x8 bfrag = (bf16x8)(*reinterpret_cast<const short8v*>(src));
 #pragma unroll
                     for (int h = 0; h < QH; ++h)
-                        dpt_acc[h][t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                            (bf16x8)v_frag[h][kb], bfrag, dpt_acc[h][t], 0, 0, 0);
+                        dpt_acc[h] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            (bf16x8)v_frag[h][kb], bfrag, dpt_acc[h], 0, 0, 0);
                 }
             }
-        }
-
-        // P^T (dropped, rescaled -> for dV) to LDS; dS^T stashed in st_acc
 #pragma unroll
-        for (int h = 0; h < QH; ++h)
-#pragma unroll
-            for (int t = 0; t < TBLKS; ++t)
+            for (int h = 0; h < QH; ++h)
 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
                     int ki = k0 + h * 16 + hi4 * 4 + r;
@@ -474,20 +455,20 @@ __global__ void flash_dkv_kernel(
                     bool masked = key_pad[h][r] || qi >= Nq || (causal && ki > Lk - Nq + qi);
                     float lse_i = (qi < Nq) ? lse_row[qi] : 0.f;
                     float delta_i = (qi < Nq) ? delta_row[qi] : 0.f;
-                    float p = masked ? 0.f : expf(st_acc[h][t][r] - lse_i);
+                    float p = masked ? 0.f : expf(st_acc[h][r] - lse_i);
                     float p_eff = p;
-                    float dprobs = dpt_acc[h][t][r];
+                    float dprobs = dpt_acc[h][r];
                     if (drop_p > 0.f) {
                         bool kept = rng_hash(drop_seed, bh, qi, ki) >= drop_thresh;
                         float inv_keep = 1.0f / (1.0f - drop_p);
                         p_eff = kept ? p * inv_keep : 0.f;
                         dprobs = kept ? dprobs * inv_keep : 0.f;
                     }
-                    float ds = p * (dprobs - delta_i);
+                    ds_keep[h][t][r] = p * (dprobs - delta_i);
                     *reinterpret_cast<unsigned short*>(
                         p_mine + (h * 16 + hi4 * 4 + r) * qt_stride + (t * 16 + lo16) * 2) = f2bf(p_eff);
-                    st_acc[h][t][r] = ds;
                 }
+        }
         __builtin_amdgcn_s_waitcnt(0);
         bf16x8 pt_frag[QH][TBLKS / 2];
 #pragma unroll
@@ -522,7 +503,7 @@ __global__ void flash_dkv_kernel(
                 for (int r = 0; r < 4; ++r)
                     *reinterpret_cast<unsigned short*>(
                         p_mine + (h * 16 + hi4 * 4 + r) * qt_stride + (t * 16 + lo16) * 2) =
-                        f2bf(st_acc[h][t][r]);
+                        f2bf(ds_keep[h][t][r]);
         __builtin_amdgcn_s_waitcnt(0);
         bf16x8 dst_frag[QH][TBLKS / 2];
 #pragma unroll
@@ -659,13 +640,13 @@ std::vector<torch::Tensor> flash_bwd(torch::Tensor dout, torch::Tensor q, torch:
     // <DMAX, DVMAX, DQ_TILE, DQ_QH, DKV_TILE, DKV_QH> — tiles/QH balance LDS
     // occupancy against the 256-VGPR budget per template
     if (D <= 32 && Dv <= 160)
-        launch_flash_bwd<32, 160, 64, 2, 32, 2>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
+        launch_flash_bwd<32, 160, 64, 2, 64, 2>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
     else if (D <= 64 && Dv <= 64)
         launch_flash_bwd<64, 64, 64, 2, 64, 2>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
     else if (D <= 128 && Dv <= 128)
-        launch_flash_bwd<128, 128, 64, 2, 32, 1>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
+        launch_flash_bwd<128, 128, 64, 2, 64, 1>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
     else if (D <= 160 && Dv <= 160)
-        launch_flash_bwd<160, 160, 64, 1, 32, 1>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
+        launch_flash_bwd<160, 160, 64, 1, 64, 1>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
     else
         launch_flash_bwd<352, 352, 32, 1, 32, 1>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
 
