@@ -644,9 +644,9 @@ std::vector<torch::Tensor> flash_bwd(torch::Tensor dout, torch::Tensor q, torch:
     else if (D <= 64 && Dv <= 64)
         launch_flash_bwd<64, 64, 64, 2, 64, 2>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
     else if (D <= 128 && Dv <= 128)
-        launch_flash_bwd<128, 128, 64, 2, 64, 1>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
+        launch_flash_bwd<128, 128, 64, 2, 32, 1>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
     else if (D <= 160 && Dv <= 160)
-        launch_flash_bwd<160, 160, 64, 1, 64, 1>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
+        launch_flash_bwd<160, 160, 64, 1, 32, 1>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
     else
         launch_flash_bwd<352, 352, 32, 1, 32, 1>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
 
