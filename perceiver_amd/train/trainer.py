@@ -99,8 +99,9 @@ class Trainer:
             self._log_file.flush()
 
     def _reduce_mean(self, value: torch.Tensor) -> torch.Tensor:
+        value = value.detach()
         if dist.is_initialized():
-            value = value.detach().clone()
+            value = value.clone()
             dist.all_reduce(value)
             value /= get_world_size()
         return value

@@ -1,0 +1,155 @@
+"""Native trainer tests: end-to-end CLI-driven training on a synthetic MNIST-style
+dataset (the CPU-plumbing config of BASELINE.json), checkpoint save/restore/resume,
+and loss decrease over a few steps."""
+import json
+import os
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+from perceiver_amd.core import ClassificationDecoderConfig
+from perceiver_amd.models.vision.image_classifier import ImageEncoderConfig
+from perceiver_amd.train.lit import LitImageClassifier
+from perceiver_amd.train.trainer import TrainConfig, Trainer
+
+
+class SyntheticMNISTDataModule:
+    """MNIST-shaped synthetic data module (offline stand-in for MNISTDataModule)."""
+
+    def __init__(self, n=64, batch_size=16):
+        g = torch.Generator().manual_seed(0)
+        self.images = torch.randn(n, 28, 28, 1, generator=g)
+        self.labels = torch.randint(0, 10, (n,), generator=g)
+        self.batch_size = batch_size
+
+    @property
+    def num_classes(self):
+        return 10
+
+    @property
+    def image_shape(self):
+        return (28, 28, 1)
+
+    def prepare_data(self):
+        pass
+
+    def setup(self, stage=None):
+        pass
+
+    def _loader(self):
+        ds = torch.utils.data.TensorDataset(self.images, self.labels)
+
+        def collate(batch):
+            xs, ys = zip(*batch)
+            return {"image": torch.stack(xs), "label": torch.stack(ys)}
+
+        return torch.utils.data.DataLoader(ds, batch_size=self.batch_size, collate_fn=collate)
+
+    def train_dataloader(self):
+        return self._loader()
+
+    def val_dataloader(self):
+        return self._loader()
+
+
+def tiny_task():
+    torch.manual_seed(1)
+    return LitImageClassifier(
+        ImageEncoderConfig(image_shape=(28, 28, 1), num_frequency_bands=8,
+                           num_cross_attention_heads=1, num_self_attention_heads=2,
+                           num_self_attention_layers_per_block=2, num_self_attention_blocks=2,
+                           first_cross_attention_layer_shared=False,
+                           num_cross_attention_layers=2),
+        ClassificationDecoderConfig(num_classes=10, num_output_query_channels=32),
+        num_latents=16, num_latent_channels=32,
+    )
+
+
+def test_fit_and_checkpoint(tmp_path):
+    dm = SyntheticMNISTDataModule()
+    cfg = TrainConfig(max_steps=6, log_every=2, out_dir=str(tmp_path / "run"), lr=1e-3)
+    trainer = Trainer(cfg)
+    task = tiny_task()
+    trainer.fit(task, datamodule=dm)
+
+    ckpt_dir = tmp_path / "run" / "checkpoints"
+    assert (ckpt_dir / "last.ckpt").exists()
+    ckpt = torch.load(ckpt_dir / "last.ckpt", map_location="cpu", weights_only=False)
+    assert ckpt["global_step"] == 6
+    assert "state_dict" in ckpt and "hyper_parameters" in ckpt
+    # metric log written
+    lines = [json.loads(l) for l in open(tmp_path / "run" / "metrics.jsonl")]
+    assert any("train_loss" in l for l in lines)
+    assert any("val_loss" in l and "val_acc" in l for l in lines)
+
+    # Lightning-compatible reconstruction from hparams embedded in the checkpoint
+    restored = LitImageClassifier.load_from_checkpoint(str(ckpt_dir / "last.ckpt"))
+    x = dm.images[:2]
+    task.eval()
+    restored.eval()
+    with torch.no_grad():
+        assert torch.allclose(task(x), restored(x), atol=1e-6)
+
+
+def test_resume_from_checkpoint(tmp_path):
+    dm = SyntheticMNISTDataModule()
+    cfg = TrainConfig(max_steps=3, log_every=10, out_dir=str(tmp_path / "a"))
+    t1 = Trainer(cfg)
+    task = tiny_task()
+    t1.fit(task, datamodule=dm)
+
+    cfg2 = TrainConfig(max_steps=6, log_every=10, out_dir=str(tmp_path / "b"))
+    t2 = Trainer(cfg2)
+    task2 = tiny_task()
+    t2.fit(task2, datamodule=dm, ckpt_path=str(tmp_path / "a" / "checkpoints" / "last.ckpt"))
+    assert t2.global_step == 6
+
+
+def test_loss_decreases_on_learnable_problem(tmp_path):
+    torch.manual_seed(0)
+    from perceiver_amd.models.text.clm import CausalLanguageModel, CausalLanguageModelConfig
+
+    model = CausalLanguageModel(CausalLanguageModelConfig(
+        vocab_size=16, max_seq_len=32, max_latents=16, num_channels=32, num_heads=4,
+        num_self_attention_layers=2, cross_attention_dropout=0.0,
+    ))
+    opt = torch.optim.AdamW(model.parameters(), lr=3e-3)
+    x = torch.arange(32).remainder(16).unsqueeze(0).repeat(4, 1)  # periodic -> learnable
+    losses = []
+    for _ in range(30):
+        out = model(x, prefix_len=16)
+        loss = F.cross_entropy(out.logits.flatten(0, 1), x[:, 16:].flatten())
+        opt.zero_grad()
+        loss.backward()
+        opt.step()
+        losses.append(float(loss))
+    assert losses[-1] < losses[0] * 0.5, losses[::10]
+
+
+def test_cli_fit_runs(tmp_path, monkeypatch):
+    """CLI plumbing end-to-end: image-classifier CLI on the synthetic data module."""
+    from perceiver_amd.scripts.cli import CLI
+    from perceiver_amd.scripts.vision.image_classifier import DEFAULTS, build_model, link
+
+    defaults = json.loads(json.dumps({k: v for k, v in DEFAULTS.items() if k != "data"}, default=str))
+    defaults = DEFAULTS  # use as-is; data ignored by synthetic module
+    argv = [
+        "fit",
+        "--model.num_latents", "16",
+        "--model.num_latent_channels", "32",
+        "--model.encoder.num_frequency_bands", "4",
+        "--model.encoder.num_self_attention_layers_per_block", "1",
+        "--model.encoder.num_self_attention_blocks", "1",
+        "--model.encoder.num_cross_attention_heads", "1",
+        "--model.decoder.num_output_query_channels", "16",
+        "--trainer.max_steps", "2",
+        "--trainer.out_dir", str(tmp_path / "cli_run"),
+    ]
+    cli = CLI(LitImageClassifier, SyntheticMNISTDataModule, defaults, build_model, link,
+              argv=argv, run=False)
+    cli.config["data"] = {}
+    trainer = cli.run()
+    assert trainer.global_step == 2
+    assert (tmp_path / "cli_run" / "config.yaml").exists()
+    assert (tmp_path / "cli_run" / "checkpoints" / "last.ckpt").exists()
