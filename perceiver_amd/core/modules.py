@@ -438,8 +438,9 @@ class SelfAttentionBlock(nn.Sequential):
 
 
 class MLP(nn.Sequential):
-    """LayerNorm -> Linear(x widening) -> GELU -> Linear. On GPU this maps to the
-    fused LN+GEMM+GELU+GEMM kernel."""
+    """LayerNorm -> Linear(x widening) -> GELU -> Linear.
+    On GPU in bf16 the LayerNorm is the fused HIP LN and the widening Linear's bias
+    add fuses into the GELU kernel (the GEMM itself stays on hipBLASLt)."""
 
     def __init__(self, num_channels: int, widening_factor: int, bias: bool = True):
         super().__init__(
@@ -450,6 +451,14 @@ class MLP(nn.Sequential):
         )
 
     def forward(self, x: torch.Tensor) -> ModuleOutput:
+        from perceiver_amd.ops.gelu import GeluBias, can_fuse_gelu_bias
+
+        if can_fuse_gelu_bias(x):
+            h = self[0](x)
+            h = torch.nn.functional.linear(h, self[1].weight)  # bias folded into GELU
+            h = GeluBias.apply(h.contiguous(), self[1].bias)
+            h = self[3](h)
+            return ModuleOutput(last_hidden_state=h)
         return ModuleOutput(last_hidden_state=super().forward(x))
 
 

@@ -16,6 +16,9 @@ std::vector<torch::Tensor> ln_fwd(torch::Tensor x, torch::Tensor w, c10::optiona
 std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor w,
                                   torch::Tensor mean, torch::Tensor rstd, bool needs_dwdb);
 bool flash_supported_impl(long d_qk, long d_v, long needs_dropout);
+void adamw_step(torch::Tensor master, torch::Tensor m, torch::Tensor v, torch::Tensor g,
+                double lr, double beta1, double beta2, double eps, double weight_decay,
+                int64_t step);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("gelu_bias_fwd", &gelu_bias_fwd, "fused bias+GELU forward (bf16)");
@@ -25,4 +28,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("flash_supported", &flash_supported_impl, "shape gate for the flash kernel");
     m.def("ln_fwd", &ln_fwd, "fused bf16 LayerNorm forward");
     m.def("ln_bwd", &ln_bwd, "fused bf16 LayerNorm backward");
+    m.def("adamw_step", &adamw_step, "single-pass fused AdamW on flat fp32 state");
 }

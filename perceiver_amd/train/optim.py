@@ -15,10 +15,66 @@ import torch
 
 
 class MasterAdamW(torch.optim.Optimizer):
+    """On CUDA with the HIP extension, the whole update runs as: one fused
+    multi-tensor grad cast into a flat fp32 buffer, ONE single-pass adamw_step
+    kernel over the flat master/m/v/grad buffers, one fused write-back to the
+    bf16 parameters. Falls back to torch._foreach_* elsewhere."""
+
     def __init__(self, params: Iterable[torch.nn.Parameter], lr: float = 1e-3,
                  betas=(0.9, 0.999), eps: float = 1e-8, weight_decay: float = 0.01):
         defaults = dict(lr=lr, betas=betas, eps=eps, weight_decay=weight_decay)
         super().__init__(params, defaults)
+        self._flat = None  # lazy per-group flat state
+
+    def _build_flat(self, group):
+        params = [p for p in group["params"] if p.requires_grad]
+        n = sum(p.numel() for p in params)
+        dev = params[0].device
+        flat = {
+            "params": params,
+            "master": torch.empty(n, dtype=torch.float32, device=dev),
+            "m": torch.zeros(n, dtype=torch.float32, device=dev),
+            "v": torch.zeros(n, dtype=torch.float32, device=dev),
+            "g": torch.empty(n, dtype=torch.float32, device=dev),
+            "step": 0,
+        }
+        views_master, views_g = [], []
+        off = 0
+        for p in params:
+            k = p.numel()
+            mv = flat["master"][off: off + k].view_as(p)
+            mv.copy_(p.detach().to(torch.float32))
+            views_master.append(mv)
+            views_g.append(flat["g"][off: off + k].view_as(p))
+            off += k
+        flat["views_master"] = views_master
+        flat["views_g"] = views_g
+        return flat
+
+    def _fused_available(self, group):
+        from perceiver_amd.ops import hip
+
+        p0 = group["params"][0]
+        return p0.is_cuda and hip.is_available()
+
+    def _step_fused(self, group):
+        from perceiver_amd.ops import hip
+
+        if self._flat is None:
+            self._flat = {}
+        gid = id(group)
+        if gid not in self._flat:
+            self._flat[gid] = self._build_flat(group)
+        fl = self._flat[gid]
+        fl["step"] += 1
+        grads = [p.grad for p in fl["params"]]
+        assert all(g is not None for g in grads), "missing gradient in fused AdamW step"
+        torch._foreach_copy_(fl["views_g"], grads)
+        beta1, beta2 = group["betas"]
+        hip.ext().adamw_step(fl["master"], fl["m"], fl["v"], fl["g"],
+                             group["lr"], beta1, beta2, group["eps"],
+                             group["weight_decay"], fl["step"])
+        torch._foreach_copy_(fl["params"], fl["views_master"])
 
     @torch.no_grad()
     def step(self, closure=None):
@@ -28,6 +84,9 @@ class MasterAdamW(torch.optim.Optimizer):
                 loss = closure()
 
         for group in self.param_groups:
+            if self._fused_available(group):
+                self._step_fused(group)
+                continue
             params, raw_grads, grads, masters, exp_avgs, exp_avg_sqs, steps = [], [], [], [], [], [], []
             for p in group["params"]:
                 if p.grad is None:
