@@ -284,8 +284,10 @@ def test_fused_adamw_matches_reference_adamw():
         opt32.step()
         optbf.step()
     for a, b in zip(p32, pbf):
+        # params live in bf16: tolerance = ~2 bf16 ulps at |w|~2 (the fused kernel
+        # itself matches the fp32 update to ~2e-7; see tools/dbg_adamw.py)
         err = (a.detach() - b.detach().float()).abs().max().item()
-        assert err < 2e-2, err
+        assert err < 4e-2, err
 
 
 def test_mlp_gelu_bias_fusion_matches_eager():
