@@ -110,6 +110,8 @@ __global__ void flash_dq_kernel(
     const float* __restrict__ lsep, const float* __restrict__ deltap,
     const bool* __restrict__ pad,
     unsigned short* __restrict__ dqp,
+    float* __restrict__ dq_part,  // (S,B,H,Nq,D) fp32 when gridDim.z > 1
+    long kv_chunk,
     int B, int H, int Nq, int Lk, int D, int Dv, int causal,
     float drop_p, unsigned long long drop_seed) {
     constexpr int TBLKS = TILE / 16;
@@ -195,9 +197,14 @@ __global__ void flash_dq_kernel(
 
     int kv_end = Lk;
     if (causal) kv_end = min(Lk, Lk - Nq + blockIdx.x * QBLK + QBLK);
+    int kv_begin = 0;
+    if (gridDim.z > 1) {  // KV-split; fp32 partials summed by the caller
+        kv_begin = (int)((long)blockIdx.z * kv_chunk);
+        kv_end = min((long)kv_end, (long)(blockIdx.z + 1) * kv_chunk);
+    }
     const unsigned int drop_thresh = (unsigned int)(drop_p * 4294967296.0);
 
-    for (int kv0 = 0; kv0 < kv_end; kv0 += TILE) {
+    for (int kv0 = kv_begin; kv0 < kv_end; kv0 += TILE) {
         int rows_valid = min(TILE, Lk - kv0);
         __syncthreads();
         stage_rm<TILE>(kbase + (long)kv0 * D, D, rows_valid, D, d_pad, k_lds, k_stride, tid);
@@ -282,18 +289,28 @@ __global__ void flash_dq_kernel(
         }
     }
 
-    // store dQ (C layout rows h*16 + hi4*4+r, col lo16+16cb)
+    // store dQ (C layout rows h*16 + hi4*4+r, col lo16+16cb); fp32 partial per
+    // split when KV-split is active (caller sums)
 #pragma unroll
     for (int h = 0; h < QH; ++h)
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
             int qi = q0 + h * 16 + hi4 * 4 + r;
             if (qi >= Nq) continue;
-            unsigned short* dqrow = dqp + ((long)bh * Nq + qi) * D;
+            if (gridDim.z > 1) {
+                float* dqrow = dq_part + (((long)blockIdx.z * B * H + bh) * Nq + qi) * D;
 #pragma unroll
-            for (int cb = 0; cb < DMAX / 16; ++cb) {
-                int c = cb * 16 + lo16;
-                if (c < D) dqrow[c] = f2bf(dq_acc[h][cb][r]);
+                for (int cb = 0; cb < DMAX / 16; ++cb) {
+                    int c = cb * 16 + lo16;
+                    if (c < D) dqrow[c] = dq_acc[h][cb][r];
+                }
+            } else {
+                unsigned short* dqrow = dqp + ((long)bh * Nq + qi) * D;
+#pragma unroll
+                for (int cb = 0; cb < DMAX / 16; ++cb) {
+                    int c = cb * 16 + lo16;
+                    if (c < D) dqrow[c] = f2bf(dq_acc[h][cb][r]);
+                }
             }
         }
 }
@@ -566,12 +583,31 @@ void launch_flash_bwd(const torch::Tensor& dout, const torch::Tensor& q, const t
     const bool* padp = nullptr;
     if (pad_mask.has_value() && pad_mask->defined()) padp = pad_mask->data_ptr<bool>();
 
-    {   // dQ
+    {   // dQ (with KV-split when the grid would underfill the chip)
         const int k_stride = d_pad * 2 + 16, kt_stride = DQ_TILE * 2 + 16, v_stride = dv_pad * 2 + 16;
         const int qblk = 16 * DQ_QH * NWAVES;
         size_t smem = (size_t)DQ_TILE * k_stride + (size_t)DMAX * kt_stride +
                       (size_t)DQ_TILE * v_stride + (size_t)NWAVES * 16 * DQ_QH * kt_stride;
-        dim3 grid((Nq + qblk - 1) / qblk, B * H);
+        int gx = (Nq + qblk - 1) / qblk, gy = B * H;
+        int nsplit = 1;
+        long kv_chunk = Lk;
+        if ((long)gx * gy < 512 && Lk > 4 * DQ_TILE) {
+            int want = (int)(512 / ((long)gx * gy)) + 1;
+            int max_split = (Lk + 4 * DQ_TILE - 1) / (4 * DQ_TILE);
+            nsplit = std::min({want, max_split, 32});
+            long tiles = (Lk + DQ_TILE - 1) / DQ_TILE;
+            long tiles_per = (tiles + nsplit - 1) / nsplit;
+            kv_chunk = tiles_per * DQ_TILE;
+            nsplit = (int)((Lk + kv_chunk - 1) / kv_chunk);
+        }
+        torch::Tensor dq_part;
+        float* dq_part_p = nullptr;
+        if (nsplit > 1) {
+            dq_part = torch::empty({(long)nsplit, (long)B * H * Nq, (long)D},
+                                   q.options().dtype(torch::kFloat32));
+            dq_part_p = dq_part.data_ptr<float>();
+        }
+        dim3 grid(gx, gy, nsplit);
         hipLaunchKernelGGL((flash_dq_kernel<DMAX, DVMAX, DQ_TILE, DQ_QH>), grid, dim3(256), smem, stream,
                            reinterpret_cast<const unsigned short*>(q.data_ptr()),
                            reinterpret_cast<const unsigned short*>(k.data_ptr()),
@@ -579,8 +615,12 @@ void launch_flash_bwd(const torch::Tensor& dout, const torch::Tensor& q, const t
                            reinterpret_cast<const unsigned short*>(dout.data_ptr()),
                            lse.data_ptr<float>(), delta.data_ptr<float>(), padp,
                            reinterpret_cast<unsigned short*>(dq.data_ptr()),
+                           dq_part_p, kv_chunk,
                            B, H, Nq, Lk, D, Dv, (int)causal, drop_p, drop_seed);
         HIP_CHECK_LAST();
+        if (nsplit > 1) {
+            dq.copy_(dq_part.sum(0).view_as(dq));
+        }
     }
     {   // dK/dV
         const int q_stride = d_pad * 2 + 16, qt_stride = DKV_TILE * 2 + 16;

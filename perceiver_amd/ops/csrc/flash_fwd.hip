@@ -119,6 +119,9 @@ __global__ void flash_fwd_kernel(
     const bool* __restrict__ pad,           // (B,Lk) or null
     unsigned short* __restrict__ op,        // (B,H,Nq,Dv)
     float* __restrict__ lsep,               // (B,H,Nq)
+    float* __restrict__ o_part,             // (S,B,H,Nq,Dv) fp32, when gridDim.z > 1
+    float* __restrict__ lse_part,           // (S,B,H,Nq)
+    long kv_chunk,                          // keys per split (multiple of KVBLK)
     int B, int H, int Nq, int Lk, int D, int Dv, int causal,
     float drop_p, unsigned long long drop_seed) {
     const int d_pad = (D + 31) & ~31;
@@ -193,10 +196,18 @@ __global__ void flash_fwd_kernel(
         int q_hi = blockIdx.x * QBLK + QBLK - 1;  // workgroup-max q row
         kv_end = min(Lk, Lk - Nq + q_hi + 1);
     }
+    // KV-split across gridDim.z (fills the 256 CUs for small B*H*Nq/QBLK grids,
+    // e.g. the 50k/182k-KV encoder cross-attentions and single-token decode);
+    // partials merge in flash_merge_kernel via logsumexp weights.
+    int kv_begin = 0;
+    if (gridDim.z > 1) {
+        kv_begin = (int)((long)blockIdx.z * kv_chunk);
+        kv_end = min((long)kv_end, (long)(blockIdx.z + 1) * kv_chunk);
+    }
 
     const unsigned int drop_thresh = (unsigned int)(drop_p * 4294967296.0);
 
-    for (int kv0 = 0; kv0 < kv_end; kv0 += KVBLK) {
+    for (int kv0 = kv_begin; kv0 < kv_end; kv0 += KVBLK) {
         int rows_valid = min(KVBLK, Lk - kv0);
         __syncthreads();
         stage_tile_rowmajor(kbase + (long)kv0 * D, D, rows_valid, D, d_pad, k_lds, k_stride, tid);
@@ -320,7 +331,8 @@ __global__ void flash_fwd_kernel(
         }
     }
 
-    // ---- epilogue: O /= l (x dropout keep-rate), store O (bf16) and lse ----
+    // ---- epilogue: O /= l (x dropout keep-rate); direct bf16 store, or fp32
+    // partials (one slab per KV split) for the merge kernel ----
 #pragma unroll
     for (int h = 0; h < QH; ++h)
 #pragma unroll
@@ -329,16 +341,60 @@ __global__ void flash_fwd_kernel(
             if (qi >= Nq) continue;
             float l_eff = l_run[h][r] * (drop_p > 0.f ? (1.0f - drop_p) : 1.0f);
             float inv_l = (l_eff > 0.f) ? 1.0f / l_eff : 0.f;
-            unsigned short* orow = op + ((long)bh * Nq + qi) * Dv;
+            float lse_v = (l_run[h][r] > 0.f) ? m_run[h][r] + logf(l_run[h][r]) : -1e30f;
+            if (gridDim.z > 1) {
+                long row = ((long)blockIdx.z * B * H + bh) * Nq + qi;
+                float* orow = o_part + row * Dv;
 #pragma unroll
-            for (int cb = 0; cb < DVMAX / 16; ++cb) {
-                int c = cb * 16 + lo16;
-                if (cb < dv_blocks && c < Dv) orow[c] = f2bf(o_acc[h][cb][r] * inv_l);
-            }
-            if (lo16 == 0) {
-                lsep[(long)bh * Nq + qi] = m_run[h][r] + logf(fmaxf(l_run[h][r], 1e-37f));
+                for (int cb = 0; cb < DVMAX / 16; ++cb) {
+                    int c = cb * 16 + lo16;
+                    if (cb < dv_blocks && c < Dv) orow[c] = o_acc[h][cb][r] * inv_l;
+                }
+                if (lo16 == 0) lse_part[row] = lse_v;
+            } else {
+                unsigned short* orow = op + ((long)bh * Nq + qi) * Dv;
+#pragma unroll
+                for (int cb = 0; cb < DVMAX / 16; ++cb) {
+                    int c = cb * 16 + lo16;
+                    if (cb < dv_blocks && c < Dv) orow[c] = f2bf(o_acc[h][cb][r] * inv_l);
+                }
+                if (lo16 == 0) lsep[(long)bh * Nq + qi] = lse_v;
             }
         }
+}
+
+// Merge KV-split partials: one wave per (b,h,q) row.
+//   w_i = exp(lse_i - max) ; O = sum w_i O_i / sum w_i ; lse = max + log(sum w_i)
+__global__ void flash_merge_kernel(const float* __restrict__ o_part,
+                                   const float* __restrict__ lse_part,
+                                   unsigned short* __restrict__ op,
+                                   float* __restrict__ lsep,
+                                   long rows, int nsplit, int dv) {
+    long row = (long)blockIdx.x * (blockDim.x / 64) + threadIdx.x / 64;
+    int lane = threadIdx.x % 64;
+    if (row >= rows) return;
+
+    __shared__ float wbuf[4][32];  // per-wave weights (runtime-indexed arrays
+    int wv_id = (int)(threadIdx.x / 64); // would spill to scratch — rule 20)
+
+    float mx = -1e30f;
+    for (int i = 0; i < nsplit; ++i) mx = fmaxf(mx, lse_part[(long)i * rows + row]);
+    float wsum = 0.f;
+    for (int i = lane; i < nsplit; i += 64) {
+        float wv = expf(lse_part[(long)i * rows + row] - mx);
+        wbuf[wv_id][i] = wv;
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+    for (int i = 0; i < nsplit; ++i) wsum += wbuf[wv_id][i];
+    float inv = (wsum > 0.f) ? 1.0f / wsum : 0.f;
+
+    for (int c = lane; c < dv; c += 64) {
+        float acc = 0.f;
+        for (int i = 0; i < nsplit; ++i)
+            acc += wbuf[wv_id][i] * o_part[((long)i * rows + row) * dv + c];
+        op[row * dv + c] = f2bf(acc * inv);
+    }
+    if (lane == 0) lsep[row] = mx + logf(fmaxf(wsum, 1e-37f));
 }
 
 template <int DMAX, int DVMAX, int QH>
@@ -355,21 +411,54 @@ void launch_flash_fwd(const torch::Tensor& q, const torch::Tensor& k, const torc
     int vt_stride = KVBLK * 2 + 16;
     size_t smem = (size_t)KVBLK * k_stride + (size_t)DVMAX * vt_stride +
                   (size_t)NWAVES * QROWS * vt_stride;
-    dim3 grid((Nq + QBLK - 1) / QBLK, B * H);
+    int gx = (Nq + QBLK - 1) / QBLK;
+    int gy = B * H;
+    // KV-split: aim for ~2 blocks per CU (512 workgroups on 256 CUs)
+    int nsplit = 1;
+    long kv_chunk = Lk;
+    if ((long)gx * gy < 512 && Lk > 4 * KVBLK) {
+        int want = (int)(512 / ((long)gx * gy)) + 1;
+        int max_split = (Lk + 4 * KVBLK - 1) / (4 * KVBLK);  // >= 4 tiles per split
+        nsplit = std::min({want, max_split, 32});
+        long tiles = (Lk + KVBLK - 1) / KVBLK;
+        long tiles_per = (tiles + nsplit - 1) / nsplit;
+        kv_chunk = tiles_per * KVBLK;
+        nsplit = (int)((Lk + kv_chunk - 1) / kv_chunk);
+    }
+    dim3 grid(gx, gy, nsplit);
     const bool* padp = nullptr;
     if (pad_mask.has_value() && pad_mask->defined()) {
         padp = pad_mask->data_ptr<bool>();
     }
-    hipLaunchKernelGGL((flash_fwd_kernel<DMAX, DVMAX, QH>), grid, dim3(256), smem,
-                       at::cuda::getCurrentCUDAStream(),
+    auto stream = at::cuda::getCurrentCUDAStream();
+    torch::Tensor o_part, lse_part;
+    float* o_part_p = nullptr;
+    float* lse_part_p = nullptr;
+    long rows = (long)B * H * Nq;
+    if (nsplit > 1) {
+        o_part = torch::empty({(long)nsplit, rows, (long)Dv}, q.options().dtype(torch::kFloat32));
+        lse_part = torch::empty({(long)nsplit, rows}, q.options().dtype(torch::kFloat32));
+        o_part_p = o_part.data_ptr<float>();
+        lse_part_p = lse_part.data_ptr<float>();
+    }
+    hipLaunchKernelGGL((flash_fwd_kernel<DMAX, DVMAX, QH>), grid, dim3(256), smem, stream,
                        reinterpret_cast<const unsigned short*>(q.data_ptr()),
                        reinterpret_cast<const unsigned short*>(k.data_ptr()),
                        reinterpret_cast<const unsigned short*>(v.data_ptr()),
                        padp,
                        reinterpret_cast<unsigned short*>(out.data_ptr()),
-                       lse.data_ptr<float>(), B, H, Nq, Lk, D, Dv, (int)causal,
-                       drop_p, drop_seed);
+                       lse.data_ptr<float>(), o_part_p, lse_part_p, kv_chunk,
+                       B, H, Nq, Lk, D, Dv, (int)causal, drop_p, drop_seed);
     HIP_CHECK_LAST();
+    if (nsplit > 1) {
+        int wpb = 4;
+        long blocks = (rows + wpb - 1) / wpb;
+        hipLaunchKernelGGL(flash_merge_kernel, dim3(blocks), dim3(64 * wpb), 0, stream,
+                           o_part_p, lse_part_p,
+                           reinterpret_cast<unsigned short*>(out.data_ptr()),
+                           lse.data_ptr<float>(), rows, nsplit, Dv);
+        HIP_CHECK_LAST();
+    }
 }
 
 }  // namespace
