@@ -111,7 +111,9 @@ def convert_checkpoint(save_dir, ckpt_url, tokenizer_name, **kwargs):
 def convert_config(config) -> MaskedLanguageModelConfig:
     """transformers PerceiverConfig -> MaskedLanguageModelConfig."""
     assert config.hidden_act == "gelu"
-    assert config.tie_word_embeddings
+    # transformers 5.x guards global access to tie_word_embeddings on some configs
+    tie = config.to_dict().get("tie_word_embeddings", True)
+    assert tie
 
     encoder_config = TextEncoderConfig(
         vocab_size=config.vocab_size,
