@@ -112,6 +112,8 @@ __global__ void flash_dq_kernel(
     unsigned short* __restrict__ dqp,
     float* __restrict__ dq_part,  // (S,B,H,Nq,D) fp32 when gridDim.z > 1
     long kv_chunk,
+    long qsb, long qsh, long qsn, long ksb, long ksh, long ksn,
+    long vsb, long vsh, long vsn,
     int B, int H, int Nq, int Lk, int D, int Dv, int causal,
     float drop_p, unsigned long long drop_seed) {
     constexpr int TBLKS = TILE / 16;
@@ -124,12 +126,12 @@ __global__ void flash_dq_kernel(
 
     const int tid = threadIdx.x, wave = tid / 64, lane = tid % 64;
     const int lo16 = lane & 15, hi4 = lane >> 4;
-    const int bh = blockIdx.y, b = bh / H;
+    const int bh = blockIdx.y, b = bh / H, hh = bh % H;
     const int q0 = blockIdx.x * QBLK + wave * QROWS;
 
-    const unsigned short* qbase = qp + (long)bh * Nq * D;
-    const unsigned short* kbase = kp + (long)bh * Lk * D;
-    const unsigned short* vbase = vp + (long)bh * Lk * Dv;
+    const unsigned short* qbase = qp + (long)b * qsb + (long)hh * qsh;
+    const unsigned short* kbase = kp + (long)b * ksb + (long)hh * ksh;
+    const unsigned short* vbase = vp + (long)b * vsb + (long)hh * vsh;
     const unsigned short* dobase = dop + (long)bh * Nq * Dv;
     const float* lse_row = lsep + (long)bh * Nq;
     const float* delta_row = deltap + (long)bh * Nq;
@@ -153,7 +155,7 @@ __global__ void flash_dq_kernel(
         int qi = q0 + h * 16 + lo16;
         bool valid = qi < Nq;
         int qc = valid ? qi : Nq - 1;
-        const unsigned short* qrow = qbase + (long)qc * D;
+        const unsigned short* qrow = qbase + (long)qc * qsn;
         const unsigned short* dorow = dobase + (long)qc * Dv;
 #pragma unroll
         for (int kb = 0; kb < DMAX / 32; ++kb) {
@@ -207,9 +209,9 @@ __global__ void flash_dq_kernel(
     for (int kv0 = kv_begin; kv0 < kv_end; kv0 += TILE) {
         int rows_valid = min(TILE, Lk - kv0);
         __syncthreads();
-        stage_rm<TILE>(kbase + (long)kv0 * D, D, rows_valid, D, d_pad, k_lds, k_stride, tid);
-        stage_tr<TILE>(kbase + (long)kv0 * D, D, rows_valid, D, d_pad, kt_lds, kt_stride, tid);
-        stage_rm<TILE>(vbase + (long)kv0 * Dv, Dv, rows_valid, Dv, dv_pad, v_lds, v_stride, tid);
+        stage_rm<TILE>(kbase + (long)kv0 * ksn, ksn, rows_valid, D, d_pad, k_lds, k_stride, tid);
+        stage_tr<TILE>(kbase + (long)kv0 * ksn, ksn, rows_valid, D, d_pad, kt_lds, kt_stride, tid);
+        stage_rm<TILE>(vbase + (long)kv0 * vsn, vsn, rows_valid, Dv, dv_pad, v_lds, v_stride, tid);
         __syncthreads();
 
         // t-outer: per 16-key block compute S and dP with short-lived accumulators,
@@ -326,6 +328,8 @@ __global__ void flash_dkv_kernel(
     const float* __restrict__ lsep, const float* __restrict__ deltap,
     const bool* __restrict__ pad,
     unsigned short* __restrict__ dkp, unsigned short* __restrict__ dvp,
+    long qsb, long qsh, long qsn, long ksb, long ksh, long ksn,
+    long vsb, long vsh, long vsn,
     int B, int H, int Nq, int Lk, int D, int Dv, int causal,
     float drop_p, unsigned long long drop_seed) {
     constexpr int TBLKS = TILE / 16;
@@ -338,12 +342,12 @@ __global__ void flash_dkv_kernel(
 
     const int tid = threadIdx.x, wave = tid / 64, lane = tid % 64;
     const int lo16 = lane & 15, hi4 = lane >> 4;
-    const int bh = blockIdx.y, b = bh / H;
+    const int bh = blockIdx.y, b = bh / H, hh = bh % H;
     const int k0 = blockIdx.x * KBLK + wave * KROWS;  // this wave's first key row
 
-    const unsigned short* qbase = qp + (long)bh * Nq * D;
-    const unsigned short* kbase = kp + (long)bh * Lk * D;
-    const unsigned short* vbase = vp + (long)bh * Lk * Dv;
+    const unsigned short* qbase = qp + (long)b * qsb + (long)hh * qsh;
+    const unsigned short* kbase = kp + (long)b * ksb + (long)hh * ksh;
+    const unsigned short* vbase = vp + (long)b * vsb + (long)hh * vsh;
     const unsigned short* dobase = dop + (long)bh * Nq * Dv;
     const float* lse_row = lsep + (long)bh * Nq;
     const float* delta_row = deltap + (long)bh * Nq;
@@ -368,8 +372,8 @@ __global__ void flash_dkv_kernel(
         int ki = k0 + h * 16 + lo16;
         bool valid = ki < Lk;
         int kc = valid ? ki : Lk - 1;
-        const unsigned short* krow = kbase + (long)kc * D;
-        const unsigned short* vrow = vbase + (long)kc * Dv;
+        const unsigned short* krow = kbase + (long)kc * ksn;
+        const unsigned short* vrow = vbase + (long)kc * vsn;
 #pragma unroll
         for (int kb = 0; kb < DMAX / 32; ++kb) {
             short8v val = {};
@@ -424,8 +428,8 @@ __global__ void flash_dkv_kernel(
     for (int qt0 = q_start; qt0 < Nq; qt0 += TILE) {
         int rows_valid = min(TILE, Nq - qt0);
         __syncthreads();
-        stage_rm<TILE>(qbase + (long)qt0 * D, D, rows_valid, D, d_pad, q_lds, q_stride, tid);
-        stage_tr<TILE>(qbase + (long)qt0 * D, D, rows_valid, D, d_pad, qt_lds, qt_stride, tid);
+        stage_rm<TILE>(qbase + (long)qt0 * qsn, qsn, rows_valid, D, d_pad, q_lds, q_stride, tid);
+        stage_tr<TILE>(qbase + (long)qt0 * qsn, qsn, rows_valid, D, d_pad, qt_lds, qt_stride, tid);
         stage_rm<TILE>(dobase + (long)qt0 * Dv, Dv, rows_valid, Dv, dv_pad, do_lds, do_stride, tid);
         stage_tr<TILE>(dobase + (long)qt0 * Dv, Dv, rows_valid, Dv, dv_pad, dot_lds, qt_stride, tid);
         __syncthreads();
@@ -616,6 +620,9 @@ void launch_flash_bwd(const torch::Tensor& dout, const torch::Tensor& q, const t
                            lse.data_ptr<float>(), delta.data_ptr<float>(), padp,
                            reinterpret_cast<unsigned short*>(dq.data_ptr()),
                            dq_part_p, kv_chunk,
+                           q.stride(0), q.stride(1), q.stride(2),
+                           k.stride(0), k.stride(1), k.stride(2),
+                           v.stride(0), v.stride(1), v.stride(2),
                            B, H, Nq, Lk, D, Dv, (int)causal, drop_p, drop_seed);
         HIP_CHECK_LAST();
         if (nsplit > 1) {
@@ -638,6 +645,9 @@ void launch_flash_bwd(const torch::Tensor& dout, const torch::Tensor& q, const t
                            lse.data_ptr<float>(), delta.data_ptr<float>(), padp,
                            reinterpret_cast<unsigned short*>(dk.data_ptr()),
                            reinterpret_cast<unsigned short*>(dv.data_ptr()),
+                           q.stride(0), q.stride(1), q.stride(2),
+                           k.stride(0), k.stride(1), k.stride(2),
+                           v.stride(0), v.stride(1), v.stride(2),
                            B, H, Nq, Lk, D, Dv, (int)causal, drop_p, drop_seed);
         HIP_CHECK_LAST();
     }
@@ -650,8 +660,11 @@ std::vector<torch::Tensor> flash_bwd(torch::Tensor dout, torch::Tensor q, torch:
                                      c10::optional<torch::Tensor> pad_mask, bool causal,
                                      double dropout_p, int64_t seed) {
     TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
-    dout = dout.contiguous(); q = q.contiguous(); k = k.contiguous(); v = v.contiguous();
+    dout = dout.contiguous();
     out = out.contiguous();
+    if (q.stride(3) != 1) q = q.contiguous();
+    if (k.stride(3) != 1) k = k.contiguous();
+    if (v.stride(3) != 1) v = v.contiguous();
     int B = q.size(0), H = q.size(1), Nq = q.size(2), D = q.size(3);
     int Lk = k.size(2), Dv = v.size(3);
 
@@ -668,9 +681,9 @@ std::vector<torch::Tensor> flash_bwd(torch::Tensor dout, torch::Tensor q, torch:
         HIP_CHECK_LAST();
     }
 
-    auto dq = torch::empty_like(q);
-    auto dk = torch::empty_like(k);
-    auto dv = torch::empty_like(v);
+    auto dq = torch::empty(q.sizes(), q.options());
+    auto dk = torch::empty(k.sizes(), k.options());
+    auto dv = torch::empty(v.sizes(), v.options());
 
     c10::optional<torch::Tensor> pm;
     if (pad_mask.has_value() && pad_mask->defined()) pm = pad_mask->contiguous();

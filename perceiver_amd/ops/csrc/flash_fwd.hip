@@ -122,6 +122,9 @@ __global__ void flash_fwd_kernel(
     float* __restrict__ o_part,             // (S,B,H,Nq,Dv) fp32, when gridDim.z > 1
     float* __restrict__ lse_part,           // (S,B,H,Nq)
     long kv_chunk,                          // keys per split (multiple of KVBLK)
+    long qsb, long qsh, long qsn,           // q strides (elements; last dim contiguous)
+    long ksb, long ksh, long ksn,           // k strides
+    long vsb, long vsh, long vsn,           // v strides
     int B, int H, int Nq, int Lk, int D, int Dv, int causal,
     float drop_p, unsigned long long drop_seed) {
     const int d_pad = (D + 31) & ~31;
@@ -139,11 +142,12 @@ __global__ void flash_fwd_kernel(
 
     const int bh = blockIdx.y;
     const int b = bh / H;
+    const int hh = bh % H;
     const int q0 = blockIdx.x * QBLK + wave * QROWS;
 
-    const unsigned short* qbase = qp + ((long)bh * Nq) * D;
-    const unsigned short* kbase = kp + ((long)bh * Lk) * D;
-    const unsigned short* vbase = vp + ((long)bh * Lk) * Dv;
+    const unsigned short* qbase = qp + (long)b * qsb + (long)hh * qsh;
+    const unsigned short* kbase = kp + (long)b * ksb + (long)hh * ksh;
+    const unsigned short* vbase = vp + (long)b * vsb + (long)hh * vsh;
     const bool* padrow = pad ? pad + (long)b * Lk : nullptr;
 
     // ---- dynamic LDS carve (guideline 17: 16-B aligned offsets) ----
@@ -162,7 +166,7 @@ __global__ void flash_fwd_kernel(
         int qi = q0 + h * 16 + lo16;
         bool valid = qi < Nq;
         int qclamp = valid ? qi : Nq - 1;
-        const unsigned short* qrow = qbase + (long)qclamp * D;
+        const unsigned short* qrow = qbase + (long)qclamp * qsn;
 #pragma unroll
         for (int kb = 0; kb < DMAX / 32; ++kb) {
             short8v val = {};
@@ -210,8 +214,8 @@ __global__ void flash_fwd_kernel(
     for (int kv0 = kv_begin; kv0 < kv_end; kv0 += KVBLK) {
         int rows_valid = min(KVBLK, Lk - kv0);
         __syncthreads();
-        stage_tile_rowmajor(kbase + (long)kv0 * D, D, rows_valid, D, d_pad, k_lds, k_stride, tid);
-        stage_tile_transposed(vbase + (long)kv0 * Dv, Dv, rows_valid, Dv, dv_pad, vt_lds, vt_stride, tid);
+        stage_tile_rowmajor(kbase + (long)kv0 * ksn, ksn, rows_valid, D, d_pad, k_lds, k_stride, tid);
+        stage_tile_transposed(vbase + (long)kv0 * vsn, vsn, rows_valid, Dv, dv_pad, vt_lds, vt_stride, tid);
         __syncthreads();
 
         // ---- S = Q K^T (QH x 16 rows x KVBLK keys); one B read feeds QH MFMAs ----
@@ -448,6 +452,9 @@ void launch_flash_fwd(const torch::Tensor& q, const torch::Tensor& k, const torc
                        padp,
                        reinterpret_cast<unsigned short*>(out.data_ptr()),
                        lse.data_ptr<float>(), o_part_p, lse_part_p, kv_chunk,
+                       q.stride(0), q.stride(1), q.stride(2),
+                       k.stride(0), k.stride(1), k.stride(2),
+                       v.stride(0), v.stride(1), v.stride(2),
                        B, H, Nq, Lk, D, Dv, (int)causal, drop_p, drop_seed);
     HIP_CHECK_LAST();
     if (nsplit > 1) {
@@ -474,7 +481,11 @@ std::vector<torch::Tensor> flash_fwd(torch::Tensor q, torch::Tensor k, torch::Te
     TORCH_CHECK(q.is_cuda() && k.is_cuda() && v.is_cuda());
     TORCH_CHECK(q.scalar_type() == torch::kBFloat16, "flash_fwd: bf16 only");
     TORCH_CHECK(q.dim() == 4 && k.dim() == 4 && v.dim() == 4);
-    q = q.contiguous(); k = k.contiguous(); v = v.contiguous();
+    // last dim must be contiguous; batch/head/seq strides are free (head-transposed
+    // views and preallocated KV-cache buffers pass through without copies)
+    if (q.stride(3) != 1) q = q.contiguous();
+    if (k.stride(3) != 1) k = k.contiguous();
+    if (v.stride(3) != 1) v = v.contiguous();
     int D = q.size(3), Dv = v.size(3);
     TORCH_CHECK(flash_supported_impl(D, Dv, 0), "flash_fwd: unsupported head dims ", D, " ", Dv);
 

@@ -23,7 +23,9 @@ class FlashAttention(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, pad_mask: Optional[torch.Tensor], causal: bool,
                 dropout_p: float, training: bool):
-        q, k, v = _to_bf16(q.contiguous()), _to_bf16(k.contiguous()), _to_bf16(v.contiguous())
+        # last-dim contiguity is enough (the kernels take batch/head/seq strides);
+        # head-transposed views and preallocated cache buffers pass zero-copy
+        q, k, v = _to_bf16(q), _to_bf16(k), _to_bf16(v)
         p = float(dropout_p) if training else 0.0
         seed = int(torch.randint(0, 2**62, (1,)).item()) if p > 0 else 0
         out, lse = hip.ext().flash_fwd(q, k, v, pad_mask, causal, p, seed)
