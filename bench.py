@@ -184,13 +184,19 @@ class CLMDecodeBench:
 
     @torch.no_grad()
     def step(self):
-        # prefill once per step then decode_steps cached single-token steps
-        out = self.model(self.prompt, prefix_len=self.prompt_len - 1, kv_cache=[])
-        kv = out.kv_cache
+        # prefill once per step then decode_steps cached single-token steps,
+        # using the preallocated in-place KV cache (no per-step concat)
+        from perceiver_amd.core.cache import allocate_kv_cache
+
+        if getattr(self, "_kv", None) is None:
+            p = next(self.model.parameters())
+            self._kv = allocate_kv_cache(self.model, self.batch, device=p.device, dtype=p.dtype)
+        for c in self._kv:
+            c.reset()
+        out = self.model(self.prompt, prefix_len=self.prompt_len - 1, kv_cache=self._kv)
         tok = out.logits[:, -1:].argmax(-1)
         for _ in range(self.decode_steps - 1):
-            out = self.model(tok, prefix_len=0, kv_cache=kv)
-            kv = out.kv_cache
+            out = self.model(tok, prefix_len=0, kv_cache=self._kv)
             tok = out.logits[:, -1:].argmax(-1)
         return None
 

@@ -1,0 +1,82 @@
+"""Preallocated KV cache for autoregressive decoding.
+
+The legacy protocol concatenates (k, v) tuples each step — an O(cache) reallocating
+copy per layer per token. StaticKVCache preallocates (B, capacity, C) buffers once
+and appends in place; the flash kernels read the live prefix as a strided view
+(batch stride = capacity * C), so no copies happen on the decode path. Sliding-
+window truncation shifts in place (only active once the window is full).
+"""
+from __future__ import annotations
+
+from typing import List, Tuple, Union
+
+import torch
+
+
+class StaticKVCache:
+    def __init__(self, batch: int, capacity: int, k_channels: int, v_channels: int,
+                 device=None, dtype=torch.float32):
+        self.k_buf = torch.empty(batch, capacity, k_channels, device=device, dtype=dtype)
+        self.v_buf = torch.empty(batch, capacity, v_channels, device=device, dtype=dtype)
+        self.length = 0
+
+    @property
+    def capacity(self) -> int:
+        return self.k_buf.shape[1]
+
+    def append(self, k_new: torch.Tensor, v_new: torch.Tensor):
+        """Write new keys/values in place; returns (k, v) views over the full live
+        prefix (length includes the new entries)."""
+        n = k_new.shape[1]
+        if self.length + n > self.capacity:
+            raise RuntimeError(
+                f"KV cache overflow: {self.length}+{n} > capacity {self.capacity}"
+            )
+        self.k_buf[:, self.length: self.length + n].copy_(k_new)
+        self.v_buf[:, self.length: self.length + n].copy_(v_new)
+        self.length += n
+        return self.k_buf[:, : self.length], self.v_buf[:, : self.length]
+
+    def reset(self):
+        self.length = 0
+
+    def truncate_front_to(self, max_len: int):
+        """Keep only the last ``max_len`` entries (window slide)."""
+        if self.length > max_len:
+            drop = self.length - max_len
+            self.k_buf[:, :max_len].copy_(self.k_buf[:, drop: self.length].clone())
+            self.v_buf[:, :max_len].copy_(self.v_buf[:, drop: self.length].clone())
+            self.length = max_len
+
+    def index_select_batch(self, idx: torch.Tensor) -> "StaticKVCache":
+        """Beam-search reorder."""
+        out = StaticKVCache.__new__(StaticKVCache)
+        out.k_buf = self.k_buf.index_select(0, idx)
+        out.v_buf = self.v_buf.index_select(0, idx)
+        out.length = self.length
+        return out
+
+
+KVCacheEntry = Union[StaticKVCache, Tuple[torch.Tensor, torch.Tensor]]
+
+
+def cache_len(entry: KVCacheEntry) -> int:
+    if isinstance(entry, StaticKVCache):
+        return entry.length
+    return entry[0].shape[1]
+
+
+def allocate_kv_cache(model, batch: int, device=None, dtype=torch.float32) -> List[StaticKVCache]:
+    """Preallocate the [cross-attention, *self-attention] cache list for a
+    PerceiverAR-family ``model`` (capacities: max_seq_len / max_latents)."""
+    def chans(layer):
+        if not hasattr(layer, "num_qk_channels"):  # activation-checkpoint wrapper
+            layer = layer.module
+        return layer.num_qk_channels, layer.num_v_channels
+
+    qk, vc = chans(model.cross_attention)
+    caches = [StaticKVCache(batch, model.max_seq_len, qk, vc, device=device, dtype=dtype)]
+    for layer in model.self_attention:
+        qk, vc = chans(layer)
+        caches.append(StaticKVCache(batch, model.max_latents, qk, vc, device=device, dtype=dtype))
+    return caches

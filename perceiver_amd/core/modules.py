@@ -34,6 +34,7 @@ from perceiver_amd.core.adapter import (
 )
 from perceiver_amd.core.config import CausalSequenceModelConfig
 from perceiver_amd.core.position import positions
+from perceiver_amd.core.cache import StaticKVCache, cache_len
 from perceiver_amd.ops.attention import scaled_dot_attention
 from perceiver_amd.ops.norm import LayerNorm
 
@@ -116,10 +117,14 @@ class MultiHeadAttention(nn.Module):
         v = self.v_proj(x_kv)
 
         if kv_cache is not None:
-            k_cache, v_cache = kv_cache
-            k = torch.cat([k_cache, k], dim=1)
-            v = torch.cat([v_cache, v], dim=1)
-            kv_cache = (k, v)
+            if isinstance(kv_cache, StaticKVCache):
+                # in-place append; k/v become strided views over the live prefix
+                k, v = kv_cache.append(k, v)
+            else:
+                k_cache, v_cache = kv_cache
+                k = torch.cat([k_cache, k], dim=1)
+                v = torch.cat([v_cache, v], dim=1)
+                kv_cache = (k, v)
 
         q = self._split_heads(q) * self.dp_scale
         k = self._split_heads(k)
@@ -706,18 +711,21 @@ class PerceiverAR(nn.Module):
         # left-pad correction for absolute positions (caller guarantees left padding)
         shift = None if pad_mask is None else pad_mask.sum(dim=1, keepdim=True)
 
-        if kv_cache is None or len(kv_cache) == 0:
+        cache_fresh = kv_cache is not None and (
+            len(kv_cache) == 0 or cache_len(kv_cache[0]) == 0
+        )
+        if kv_cache is None or cache_fresh:
             b, n = x.shape
         else:
             b = x.shape[0]
-            n = kv_cache[0][0].shape[1] + x.shape[1]
+            n = cache_len(kv_cache[0]) + x.shape[1]
 
         if not 0 <= prefix_len < n:
             raise ValueError(f"prefix_len ({prefix_len}) out of valid range [0..{n})")
 
         x, frq_pos_enc = self.input_adapter(x, abs_pos=positions(b, n, shift=shift, device=x.device))
 
-        if kv_cache is None or len(kv_cache) == 0:
+        if kv_cache is None or cache_fresh:
             x_latent, x_prefix = x[:, prefix_len:], x[:, :prefix_len]
         else:
             # cached decode: every new token is a latent
@@ -756,6 +764,7 @@ class PerceiverAR(nn.Module):
         elif len(kv_cache) == 0:
             ca_kv_cache, sa_kv_cache, kv_cache_updated = self.cross_attention.empty_kv_cache(x_latent), [], []
         else:
+            # populated tuples, or preallocated StaticKVCache objects (possibly fresh)
             ca_kv_cache, *sa_kv_cache = kv_cache
             kv_cache_updated = []
 

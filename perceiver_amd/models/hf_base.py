@@ -27,6 +27,7 @@ from perceiver_amd.core import (
     PerceiverEncoder,
     SelfAttentionLayer,
 )
+from perceiver_amd.core.cache import StaticKVCache, allocate_kv_cache, cache_len
 from perceiver_amd.core.modules import KVCache
 
 
@@ -125,22 +126,31 @@ class PerceiverCausalSequenceModel(PreTrainedModel):
 
     # -------------------------------------------------- cache maintenance
     def _reorder_cache(self, past_key_values, beam_idx):
-        return [
-            tuple(t.index_select(0, beam_idx.to(t.device)) for t in layer_past)
-            for layer_past in past_key_values
-        ]
+        out = []
+        for layer_past in past_key_values:
+            if isinstance(layer_past, StaticKVCache):
+                out.append(layer_past.index_select_batch(beam_idx.to(layer_past.k_buf.device)))
+            else:
+                out.append(tuple(t.index_select(0, beam_idx.to(t.device)) for t in layer_past))
+        return out
+
+    @staticmethod
+    def _truncate_entry(entry, max_len):
+        if isinstance(entry, StaticKVCache):
+            entry.truncate_front_to(max_len)
+            return entry
+        k, v = entry
+        return (k[:, -max_len:], v[:, -max_len:])
 
     def _truncate_cross_attention_past_key_values(self, past_key_values):
         max_ca_cache_len = self.backend_model.max_seq_len - 1
-        (k_cache, v_cache), *sa_cache = past_key_values
-        ca_cache = (k_cache[:, -max_ca_cache_len:], v_cache[:, -max_ca_cache_len:])
-        return [ca_cache] + sa_cache
+        ca_cache, *sa_cache = past_key_values
+        return [self._truncate_entry(ca_cache, max_ca_cache_len)] + sa_cache
 
     def _truncate_self_attention_past_key_values(self, past_key_values):
         max_sa_cache_len = self.backend_model.max_latents - 1
         ca_cache, *sa_cache = past_key_values
-        sa_cache = [(k[:, -max_sa_cache_len:], v[:, -max_sa_cache_len:]) for k, v in sa_cache]
-        return [ca_cache] + sa_cache
+        return [ca_cache] + [self._truncate_entry(e, max_sa_cache_len) for e in sa_cache]
 
     def prepare_inputs_for_generation(self, input_ids, past_key_values=None, **kwargs):
         """The sliding schedule: grow latents to max_latents, then grow the prefix to
@@ -150,11 +160,13 @@ class PerceiverCausalSequenceModel(PreTrainedModel):
         use_cache = kwargs.get("use_cache", None)
         prefix_len = kwargs.get("prefix_len", None)
 
-        if past_key_values is None:
+        has_cached = (past_key_values is not None and len(past_key_values) > 0
+                      and cache_len(past_key_values[0]) > 0)
+        if not has_cached:
             input_len = input_ids.shape[1]
         else:
             # contrastive-search workaround: derive input length from the cache
-            input_len = past_key_values[0][0].shape[1] + 1
+            input_len = cache_len(past_key_values[0]) + 1
 
         max_seq_len = self.backend_model.max_seq_len
         num_latents = input_len - prefix_len
@@ -165,7 +177,7 @@ class PerceiverCausalSequenceModel(PreTrainedModel):
         if max_latents_exceeded and prefix_len < self.backend_model.max_prefix_len:
             prefix_len += 1
 
-        if past_key_values:
+        if has_cached:
             input_ids = input_ids[:, -1:]
         else:
             input_ids = input_ids[:, -max_seq_len:]
@@ -173,7 +185,7 @@ class PerceiverCausalSequenceModel(PreTrainedModel):
         if attention_mask is not None and attention_mask.shape[1] > max_seq_len:
             attention_mask = attention_mask[:, -max_seq_len:]
 
-        if past_key_values:
+        if has_cached:
             if max_latents_exceeded:
                 past_key_values = self._truncate_self_attention_past_key_values(past_key_values)
             if max_seq_len_exceeded:
@@ -200,6 +212,7 @@ class PerceiverCausalSequenceModel(PreTrainedModel):
         top_k: Optional[int] = None,
         top_p: Optional[float] = None,
         use_cache: bool = True,
+        static_cache: bool = True,
         attention_mask: Optional[torch.Tensor] = None,
         pad_token_id: Optional[int] = None,
         eos_token_id: Optional[int] = None,
@@ -238,6 +251,11 @@ class PerceiverCausalSequenceModel(PreTrainedModel):
             attention_mask = torch.ones_like(input_ids)
         done = torch.zeros(input_ids.shape[0], dtype=torch.bool, device=input_ids.device)
         past = None
+        if use_cache and static_cache:
+            # preallocated in-place cache: no per-step concat/realloc (K9)
+            p = next(self.backend_model.parameters())
+            past = allocate_kv_cache(self.backend_model, input_ids.shape[0],
+                                     device=p.device, dtype=p.dtype)
 
         for _ in range(max_new_tokens):
             model_inputs = self.prepare_inputs_for_generation(

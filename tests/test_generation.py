@@ -118,3 +118,35 @@ def test_lit_checkpoint_to_hf_conversion(tmp_path):
     a = lit.model(x, prefix_len=6).logits
     b = hf(x, prefix_len=6).logits
     assert torch.allclose(a, b, atol=1e-6)
+
+
+def test_static_cache_matches_legacy_cache(model):
+    a = model.generate(input_ids=_prompt(2, 10, seed=4), num_latents=4,
+                       max_new_tokens=15, use_cache=True, static_cache=True)
+    b = model.generate(input_ids=_prompt(2, 10, seed=4), num_latents=4,
+                       max_new_tokens=15, use_cache=True, static_cache=False)
+    assert torch.equal(a, b)
+
+
+def test_static_cache_direct_model_loop_matches_full_forward():
+    from perceiver_amd.core.cache import allocate_kv_cache
+    from perceiver_amd.models.text.clm import CausalLanguageModel, CausalLanguageModelConfig
+
+    torch.manual_seed(9)
+    cfg = CausalLanguageModelConfig(vocab_size=50, max_seq_len=24, max_latents=8,
+                                    num_channels=24, num_heads=4, num_self_attention_layers=2,
+                                    cross_attention_dropout=0.0)
+    m = CausalLanguageModel(cfg).eval()
+    x = torch.randint(0, 50, (2, 16))
+    full = m(x, prefix_len=8).logits
+
+    kv = allocate_kv_cache(m, batch=2)
+    outs = []
+    with torch.no_grad():
+        out = m(x[:, :9], prefix_len=8, kv_cache=kv)
+        outs.append(out.logits)
+        for i in range(9, 16):
+            out = m(x[:, i:i + 1], prefix_len=0, kv_cache=kv)
+            outs.append(out.logits)
+    inc = torch.cat(outs, dim=1)
+    assert torch.allclose(full, inc, atol=1e-4)
