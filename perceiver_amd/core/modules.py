@@ -112,9 +112,19 @@ class MultiHeadAttention(nn.Module):
         rot_pos_emb_k: Optional[RotaryPositionEmbedding] = None,
         kv_cache: Optional[KVCache] = None,
     ) -> ModuleOutput:
-        q = self.q_proj(x_q)
-        k = self.k_proj(x_kv)
-        v = self.v_proj(x_kv)
+        if x_q is x_kv:
+            # self-attention: one merged QKV GEMM instead of three skinny ones
+            # (weights concatenated at call time; state-dict layout unchanged)
+            w = torch.cat([self.q_proj.weight, self.k_proj.weight, self.v_proj.weight], dim=0)
+            b = None
+            if self.q_proj.bias is not None:
+                b = torch.cat([self.q_proj.bias, self.k_proj.bias, self.v_proj.bias], dim=0)
+            qkv = torch.nn.functional.linear(x_q, w, b)
+            q, k, v = qkv.split([self.num_qk_channels, self.num_qk_channels, self.num_v_channels], dim=-1)
+        else:
+            q = self.q_proj(x_q)
+            k = self.k_proj(x_kv)
+            v = self.v_proj(x_kv)
 
         if kv_cache is not None:
             if isinstance(kv_cache, StaticKVCache):
