@@ -194,6 +194,31 @@ class LitCausalSequenceModel(TrainableModule):
     def validation_step(self, batch):
         return {"val_loss": float(self.step(batch))}
 
+    @torch.no_grad()
+    def on_validation_end(self, trainer, metrics):
+        """Qualitative sampling at validation end (parity: reference
+        text/clm/lightning.py:54-100): top-k continuation of a fixed prompt or a
+        validation record, logged to the metrics stream (rank zero)."""
+        from perceiver_amd.parallel import is_main_process
+
+        prompt_text = self.hparams.get("validation_sample_prompt")
+        if not is_main_process() or prompt_text is None:
+            return
+        dm = getattr(trainer, "datamodule", None)
+        if dm is None or not hasattr(dm, "text_preprocessor"):
+            return
+        try:
+            preproc = dm.text_preprocessor()
+            ids, _ = preproc.preprocess(prompt_text)
+            device = next(self.parameters()).device
+            hgf = self.to_hgf_model()
+            out = hgf.generate(input_ids=ids[None].to(device), num_latents=1,
+                               max_new_tokens=64, do_sample=True, top_k=10)
+            text = preproc.tokenizer.decode(out[0].tolist())
+            trainer.log_metrics({"sample_generation": text})
+        except Exception as e:  # sampling must never kill training
+            trainer.log_metrics({"sample_generation_error": str(e)})
+
 
 # ---------------------------------------------------------------------- text
 class LitMaskedLanguageModel(LitPerceiverIO):
@@ -227,6 +252,30 @@ class LitMaskedLanguageModel(LitPerceiverIO):
     @torch.no_grad()
     def validation_step(self, batch):
         return {"val_loss": float(self.step(batch))}
+
+    @torch.no_grad()
+    def on_validation_end(self, trainer, metrics):
+        """Mask-filling table at validation end (parity: reference
+        text/mlm/lightning.py:77-94), rank zero, logged to the metrics stream."""
+        from perceiver_amd.parallel import is_main_process
+
+        samples = self.hparams.get("masked_samples")
+        if not is_main_process() or not samples:
+            return
+        dm = getattr(trainer, "datamodule", None)
+        if dm is None or not hasattr(dm, "text_preprocessor"):
+            return
+        try:
+            from perceiver_amd.models.text.mlm_utils import MaskFiller
+
+            filler = MaskFiller(dm.text_preprocessor())
+            device = next(self.parameters()).device
+            masked, preds = filler.fill(self.model, list(samples),
+                                        int(self.hparams.get("num_predictions", 3)),
+                                        device=device)
+            trainer.log_metrics({"masked_samples": masked, "predictions": preds})
+        except Exception as e:
+            trainer.log_metrics({"mask_filling_error": str(e)})
 
 
 class LitTextClassifier(LitClassifier):

@@ -159,6 +159,7 @@ class Trainer:
             datamodule.setup("fit")
             train_loader = datamodule.train_dataloader()
             val_loader = datamodule.val_dataloader() if hasattr(datamodule, "val_dataloader") else None
+        self.datamodule = datamodule
 
         reducer = None
         if get_world_size() > 1:
