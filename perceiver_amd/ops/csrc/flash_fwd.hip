@@ -111,7 +111,7 @@ DEVINL void stage_tile_transposed(const unsigned short* __restrict__ src, long s
 // QH: 16-row A-fragments per wave (2 doubles MFMA work per B-fragment LDS read;
 // 1 for the large-D templates where the register budget is spent on O accumulators)
 template <int DMAX, int DVMAX, int QH>
-__launch_bounds__(256)
+__launch_bounds__(256, 2)   // force an allocation that admits 2 waves/SIMD
 __global__ void flash_fwd_kernel(
     const unsigned short* __restrict__ qp,  // (B,H,Nq,D) bf16, pre-scaled
     const unsigned short* __restrict__ kp,  // (B,H,Lk,D)
@@ -241,8 +241,8 @@ __global__ void flash_fwd_kernel(
             }
         }
 
-        // ---- mask + online softmax per 16-row group ----
-        float pvals[QH][KEYBLKS][4];
+        // ---- mask + online softmax per 16-row group (P computed in place in
+        // s_acc, dropped, and written straight to the per-wave LDS buffer) ----
 #pragma unroll
         for (int h = 0; h < QH; ++h) {
             float rowmax[4];
@@ -258,7 +258,7 @@ __global__ void flash_fwd_kernel(
                     if (padrow && j < Lk) masked |= padrow[j];
                     if (causal && j > Lk - Nq + qi) masked = true;
                     sv = masked ? -FLT_MAX : sv;
-                    pvals[h][kb][r] = sv;
+                    s_acc[h][kb][r] = sv;
                     mx = fmaxf(mx, sv);
                 }
                 rowmax[r] = warp16_max(mx);
@@ -271,42 +271,22 @@ __global__ void flash_fwd_kernel(
                 float psum = 0.f;
 #pragma unroll
                 for (int kb = 0; kb < KEYBLKS; ++kb) {
-                    float pv = expf(pvals[h][kb][r] - m_new);
-                    pvals[h][kb][r] = pv;
+                    float pv = expf(s_acc[h][kb][r] - m_new);
                     psum += pv;
+                    if (drop_p > 0.f) {
+                        int qi = q0 + h * 16 + hi4 * 4 + r;
+                        int j = kv0 + kb * 16 + lo16;
+                        if (rng_hash(drop_seed, bh, qi, j) < drop_thresh) pv = 0.f;
+                    }
+                    *reinterpret_cast<unsigned short*>(
+                        p_mine + (h * 16 + hi4 * 4 + r) * vt_stride + (kb * 16 + lo16) * 2) =
+                        f2bf(pv);
                 }
                 l_run[h][r] = l_run[h][r] * alpha + warp16_sum(psum);
 #pragma unroll
                 for (int cb = 0; cb < DVMAX / 16; ++cb) o_acc[h][cb][r] *= alpha;
             }
         }
-
-        // ---- dropout on (unnormalized) probabilities; l uses the undropped sum ----
-        if (drop_p > 0.f) {
-#pragma unroll
-            for (int h = 0; h < QH; ++h)
-#pragma unroll
-                for (int kb = 0; kb < KEYBLKS; ++kb)
-#pragma unroll
-                    for (int r = 0; r < 4; ++r) {
-                        int qi = q0 + h * 16 + hi4 * 4 + r;
-                        int j = kv0 + kb * 16 + lo16;
-                        if (rng_hash(drop_seed, bh, qi, j) < drop_thresh) pvals[h][kb][r] = 0.f;
-                    }
-        }
-
-        // ---- redistribute P (C layout) -> A layout via per-wave LDS ----
-#pragma unroll
-        for (int h = 0; h < QH; ++h)
-#pragma unroll
-            for (int kb = 0; kb < KEYBLKS; ++kb)
-#pragma unroll
-                for (int r = 0; r < 4; ++r) {
-                    int prow = h * 16 + hi4 * 4 + r;
-                    int pcol = kb * 16 + lo16;
-                    *reinterpret_cast<unsigned short*>(p_mine + prow * vt_stride + pcol * 2) =
-                        f2bf(pvals[h][kb][r]);
-                }
         __builtin_amdgcn_s_waitcnt(0);  // lgkmcnt(0): wave-local LDS ordering
         bf16x8 p_frag[QH][KEYBLKS / 2];
 #pragma unroll
