@@ -216,6 +216,8 @@ class PerceiverCausalSequenceModel(PreTrainedModel):
         attention_mask: Optional[torch.Tensor] = None,
         pad_token_id: Optional[int] = None,
         eos_token_id: Optional[int] = None,
+        num_beams: int = 1,
+        length_penalty: float = 1.0,
         generator: Optional[torch.Generator] = None,
         **kwargs,
     ) -> torch.Tensor:
@@ -249,6 +251,10 @@ class PerceiverCausalSequenceModel(PreTrainedModel):
 
         if attention_mask is None:
             attention_mask = torch.ones_like(input_ids)
+        if num_beams > 1:
+            return self._beam_search(input_ids, attention_mask, prefix_len, num_beams,
+                                     max_new_tokens, eos_token_id, pad_token_id,
+                                     length_penalty, use_cache, static_cache)
         done = torch.zeros(input_ids.shape[0], dtype=torch.bool, device=input_ids.device)
         past = None
         if use_cache and static_cache:
@@ -281,6 +287,78 @@ class PerceiverCausalSequenceModel(PreTrainedModel):
                 break
 
         return input_ids
+
+    @torch.no_grad()
+    def _beam_search(self, input_ids, attention_mask, prefix_len, num_beams,
+                     max_new_tokens, eos_token_id, pad_token_id, length_penalty,
+                     use_cache, static_cache):
+        """Compact beam search over the sliding latent/prefix schedule: beams stay
+        live until EOS (frozen beams propose only EOS at unchanged score); the
+        highest length-penalized score per batch wins."""
+        bsz, seq_len = input_ids.shape
+        device = input_ids.device
+        # expand to (bsz * num_beams)
+        input_ids = input_ids.repeat_interleave(num_beams, dim=0)
+        attention_mask = attention_mask.repeat_interleave(num_beams, dim=0)
+        beam_scores = torch.full((bsz, num_beams), -1e9, device=device)
+        beam_scores[:, 0] = 0.0
+        done = torch.zeros(bsz * num_beams, dtype=torch.bool, device=device)
+        gen_len = torch.zeros(bsz * num_beams, device=device)
+
+        past = None
+        if use_cache and static_cache:
+            p = next(self.backend_model.parameters())
+            past = allocate_kv_cache(self.backend_model, bsz * num_beams,
+                                     device=p.device, dtype=p.dtype)
+
+        fill = pad_token_id if pad_token_id is not None else (eos_token_id or 0)
+        for _ in range(max_new_tokens):
+            model_inputs = self.prepare_inputs_for_generation(
+                input_ids, past_key_values=past, attention_mask=attention_mask,
+                use_cache=use_cache, prefix_len=prefix_len,
+            )
+            out = self(**model_inputs)
+            prefix_len = out.prefix_len
+            past = out.past_key_values if use_cache else None
+
+            logp = out.logits[:, -1, :].float().log_softmax(-1)   # (bsz*beams, V)
+            vocab = logp.shape[-1]
+            if eos_token_id is not None:
+                # frozen beams propose only the fill token at unchanged score
+                frozen = torch.zeros_like(logp)
+                frozen[:] = -1e9
+                frozen[:, fill] = 0.0
+                logp = torch.where(done[:, None], frozen, logp)
+
+            cand = beam_scores.view(-1, 1) + logp                 # (bsz*beams, V)
+            cand = cand.view(bsz, num_beams * vocab)
+            top_scores, top_idx = cand.topk(num_beams, dim=-1)    # (bsz, beams)
+            src_beam = top_idx // vocab                            # (bsz, beams)
+            next_tok = top_idx % vocab
+
+            beam_idx = (src_beam + torch.arange(bsz, device=device)[:, None] * num_beams).view(-1)
+            input_ids = input_ids.index_select(0, beam_idx)
+            attention_mask = attention_mask.index_select(0, beam_idx)
+            done = done.index_select(0, beam_idx)
+            gen_len = gen_len.index_select(0, beam_idx)
+            if past is not None:
+                past = self._reorder_cache(past, beam_idx)
+
+            next_tok = next_tok.view(-1)
+            beam_scores = top_scores
+            gen_len = gen_len + (~done).float()
+            if eos_token_id is not None:
+                done = done | (next_tok == eos_token_id)
+
+            input_ids = torch.cat([input_ids, next_tok[:, None]], dim=1)
+            attention_mask = torch.cat([attention_mask, torch.ones_like(next_tok[:, None])], dim=1)
+            if eos_token_id is not None and bool(done.all()):
+                break
+
+        norm = gen_len.view(bsz, num_beams).clamp(min=1.0) ** length_penalty
+        best = (beam_scores / norm).argmax(dim=-1)
+        idx = best + torch.arange(bsz, device=device) * num_beams
+        return input_ids.index_select(0, idx)
 
     @staticmethod
     def _select_next(logits, do_sample, temperature, top_k, top_p, generator):

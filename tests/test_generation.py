@@ -150,3 +150,38 @@ def test_static_cache_direct_model_loop_matches_full_forward():
             outs.append(out.logits)
     inc = torch.cat(outs, dim=1)
     assert torch.allclose(full, inc, atol=1e-4)
+
+
+def test_beam_search_shapes_and_determinism(model):
+    out = model.generate(input_ids=_prompt(2, 10, seed=5), num_latents=4,
+                         max_new_tokens=6, num_beams=3)
+    assert out.shape == (2, 16)
+    out2 = model.generate(input_ids=_prompt(2, 10, seed=5), num_latents=4,
+                          max_new_tokens=6, num_beams=3)
+    assert torch.equal(out, out2)
+
+
+def test_beam_search_beats_greedy_logprob(model):
+    """Beam search's sequence log-probability must be >= greedy's."""
+    import torch.nn.functional as F
+
+    prompt = _prompt(1, 10, seed=6)
+
+    def seq_logprob(tokens):
+        lp = 0.0
+        ids = prompt.clone()
+        prefix = 6
+        past = None
+        for t in tokens:
+            mi = model.prepare_inputs_for_generation(ids, past_key_values=past,
+                                                     attention_mask=torch.ones_like(ids),
+                                                     use_cache=False, prefix_len=prefix)
+            out = model(**mi)
+            prefix = out.prefix_len
+            lp += float(F.log_softmax(out.logits[:, -1].float(), -1)[0, t])
+            ids = torch.cat([ids, torch.tensor([[t]])], dim=1)
+        return lp
+
+    greedy = model.generate(input_ids=prompt, num_latents=4, max_new_tokens=5)[0, 10:]
+    beam = model.generate(input_ids=prompt, num_latents=4, max_new_tokens=5, num_beams=4)[0, 10:]
+    assert seq_logprob(beam.tolist()) >= seq_logprob(greedy.tolist()) - 1e-4
