@@ -111,8 +111,8 @@ DEVINL void stage_tile_transposed(const unsigned short* __restrict__ src, long s
 // QH: 16-row A-fragments per wave (2 doubles MFMA work per B-fragment LDS read;
 // 1 for the large-D templates where the register budget is spent on O accumulators)
 template <int DMAX, int DVMAX, int QH>
-__launch_bounds__(256, 2)   // force an allocation that admits 2 waves/SIMD
-__global__ void flash_fwd_kernel(
+__launch_bounds__(256, DMAX <= 160 ? 2 : 1)  // 2 waves/SIMD where the register
+__global__ void flash_fwd_kernel(            // budget allows (not the 352 template)
     const unsigned short* __restrict__ qp,  // (B,H,Nq,D) bf16, pre-scaled
     const unsigned short* __restrict__ kp,  // (B,H,Lk,D)
     const unsigned short* __restrict__ vp,  // (B,H,Lk,Dv)
