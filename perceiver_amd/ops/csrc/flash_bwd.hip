@@ -321,7 +321,7 @@ __global__ void flash_dq_kernel(
 // grid.x over KV blocks (QH*16 keys per wave), grid.y = B*H.
 // Loops over TILE-row Q tiles; stages Q row-major + Q^T, dO row-major + dO^T.
 template <int DMAX, int DVMAX, int TILE, int QH>
-__launch_bounds__(256, DMAX <= 160 ? 2 : 1)
+__launch_bounds__(256)
 __global__ void flash_dkv_kernel(
     const unsigned short* __restrict__ qp, const unsigned short* __restrict__ kp,
     const unsigned short* __restrict__ vp, const unsigned short* __restrict__ dop,
@@ -697,9 +697,9 @@ std::vector<torch::Tensor> flash_bwd(torch::Tensor dout, torch::Tensor q, torch:
     else if (D <= 64 && Dv <= 64)
         launch_flash_bwd<64, 64, 64, 2, 64, 1>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
     else if (D <= 128 && Dv <= 128)
-        launch_flash_bwd<128, 128, 64, 2, 64, 1>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
+        launch_flash_bwd<128, 128, 64, 2, 32, 1>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
     else if (D <= 160 && Dv <= 160)
-        launch_flash_bwd<160, 160, 64, 1, 64, 1>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
+        launch_flash_bwd<160, 160, 64, 1, 32, 1>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
     else
         launch_flash_bwd<352, 352, 32, 1, 32, 1>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
 
