@@ -33,6 +33,8 @@ def parse_args():
     p.add_argument("--batch", type=int, default=0, help="per-GPU batch size (0 = model default)")
     p.add_argument("--device", default=None, help="override device (e.g. cpu for local testing)")
     p.add_argument("--tiny", action="store_true", help="tiny config for CPU plumbing tests")
+    p.add_argument("--eager-decode", action="store_true",
+                   help="clm-decode: host-driven per-token steps instead of hipGraph replay")
     p.add_argument("--ddp-impl", default="native", choices=["native", "torch"],
                    help="gradient reducer: perceiver_amd.parallel bucketed RCCL reducer or torch DDP")
     p.add_argument("--seed", type=int, default=17)
@@ -178,6 +180,7 @@ class CLMDecodeBench:
         self.prompt = torch.randint(0, cfg.vocab_size, (self.batch, self.prompt_len), generator=g).to(device)
         # decode_steps tokens generated per bench "step"
         self.decode_steps = 16 if not args.tiny else 4
+        self.use_graph = device.type == "cuda" and not args.eager_decode
 
     def wrap_ddp(self, impl):
         pass  # inference benchmark: no gradient reduction
@@ -185,12 +188,21 @@ class CLMDecodeBench:
     @torch.no_grad()
     def step(self):
         # prefill once per step then decode_steps cached single-token steps,
-        # using the preallocated in-place KV cache (no per-step concat)
+        # using the preallocated in-place KV cache (no per-step concat);
+        # on GPU the per-token step is a single hipGraph replay
         from perceiver_amd.core.cache import allocate_kv_cache
 
         if getattr(self, "_kv", None) is None:
             p = next(self.model.parameters())
             self._kv = allocate_kv_cache(self.model, self.batch, device=p.device, dtype=p.dtype)
+            if self.use_graph:
+                from perceiver_amd.core.graph_decode import GraphedDecoder
+
+                self._gd = GraphedDecoder(self.model, self._kv)
+        if self.use_graph:
+            self._gd.prefill(self.prompt, prefix_len=self.prompt_len - 1)
+            self._gd.decode(self.decode_steps - 1)
+            return None
         for c in self._kv:
             c.reset()
         out = self.model(self.prompt, prefix_len=self.prompt_len - 1, kv_cache=self._kv)

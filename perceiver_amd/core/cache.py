@@ -24,9 +24,26 @@ class StaticKVCache:
     def capacity(self) -> int:
         return self.k_buf.shape[1]
 
+    def enable_graph_append(self, len_dev: torch.Tensor):
+        """Switch to device-indexed appends for hipGraph capture: the write index
+        lives in ``len_dev`` (1-elem int64 device tensor) instead of the host
+        ``length`` counter, and ``append`` returns FULL-capacity views (callers
+        mask the tail ``j > len`` themselves). No host sync anywhere on the path,
+        so the step can be stream-captured and replayed."""
+        self._len_dev = len_dev
+
+    def disable_graph_append(self):
+        self._len_dev = None
+
     def append(self, k_new: torch.Tensor, v_new: torch.Tensor):
         """Write new keys/values in place; returns (k, v) views over the full live
         prefix (length includes the new entries)."""
+        if getattr(self, "_len_dev", None) is not None:
+            if k_new.shape[1] != 1:
+                raise RuntimeError("graph-mode append is single-token only")
+            self.k_buf.index_copy_(1, self._len_dev, k_new)
+            self.v_buf.index_copy_(1, self._len_dev, v_new)
+            return self.k_buf, self.v_buf
         n = k_new.shape[1]
         if self.length + n > self.capacity:
             raise RuntimeError(

@@ -1,0 +1,102 @@
+"""GraphedDecoder: the device-driven single-token step (full-capacity masked
+attention, device length counters, static rotary tables) must reproduce the
+host-driven cached decode of CausalSequenceModel exactly. CPU tests run the
+un-captured step; the GPU test captures and replays a real hipGraph."""
+import pytest
+import torch
+
+from perceiver_amd.core.cache import allocate_kv_cache
+from perceiver_amd.core.graph_decode import GraphedDecoder
+from perceiver_amd.models.text.clm import CausalLanguageModel, CausalLanguageModelConfig
+
+
+def _cfg(**kw):
+    base = dict(vocab_size=61, max_seq_len=48, max_latents=16, num_channels=32,
+                num_heads=4, num_self_attention_layers=2, cross_attention_dropout=0.0)
+    base.update(kw)
+    return CausalLanguageModelConfig(**base)
+
+
+def _host_decode(model, prompt, prefix_len, steps, device="cpu"):
+    kv = allocate_kv_cache(model, prompt.shape[0], device=device,
+                           dtype=next(model.parameters()).dtype)
+    out = model(prompt, prefix_len=prefix_len, kv_cache=kv)
+    tok = out.logits[:, -1:].argmax(-1)
+    toks, logits = [], []
+    for _ in range(steps):
+        out = model(tok, prefix_len=0, kv_cache=kv)
+        logits.append(out.logits[:, -1])
+        tok = out.logits[:, -1:].argmax(-1)
+        toks.append(tok)
+    return torch.cat(toks, dim=1), logits
+
+
+@pytest.mark.parametrize("abs_pos_emb", [False, True])
+def test_graph_step_matches_host_decode_cpu(abs_pos_emb):
+    torch.manual_seed(0)
+    model = CausalLanguageModel(_cfg(abs_pos_emb=abs_pos_emb)).eval()
+    prompt = torch.randint(0, 61, (3, 30))
+    prefix_len = 29
+
+    with torch.no_grad():
+        host_toks, host_logits = _host_decode(model, prompt, prefix_len, steps=6)
+
+        gd = GraphedDecoder(model, allocate_kv_cache(model, 3), use_graph=False)
+        t0 = gd.prefill(prompt, prefix_len=prefix_len)
+        graph_toks = gd.decode(6)
+
+    torch.testing.assert_close(gd.last_logits[:, -1], host_logits[-1],
+                               rtol=1e-4, atol=1e-5)
+    assert torch.equal(graph_toks, host_toks)
+    # prefill token must equal the host prefill argmax by construction
+    assert t0.shape == (3, 1)
+
+
+def test_graph_decoder_refuses_cache_overflow():
+    model = CausalLanguageModel(_cfg()).eval()
+    gd = GraphedDecoder(model, allocate_kv_cache(model, 2), use_graph=False)
+    with torch.no_grad():
+        gd.prefill(torch.randint(0, 61, (2, 30)), prefix_len=29)
+        with pytest.raises(RuntimeError, match="overflow"):
+            gd.decode(16)  # sa cache: 1 + 16 > 16
+
+
+def test_repeated_prefill_resets_state():
+    torch.manual_seed(1)
+    model = CausalLanguageModel(_cfg(abs_pos_emb=False)).eval()
+    prompt = torch.randint(0, 61, (2, 30))
+    gd = GraphedDecoder(model, allocate_kv_cache(model, 2), use_graph=False)
+    with torch.no_grad():
+        gd.prefill(prompt, prefix_len=29)
+        a = gd.decode(5)
+        gd.prefill(prompt, prefix_len=29)
+        b = gd.decode(5)
+    assert torch.equal(a, b)
+
+
+@pytest.mark.gpu
+def test_graph_decode_gpu_matches_host():
+    torch.manual_seed(0)
+    dev = torch.device("cuda:0")
+    model = CausalLanguageModel(_cfg()).to(dev, torch.bfloat16).eval()
+    prompt = torch.randint(0, 61, (4, 30), device=dev)
+    prefix_len = 29
+
+    with torch.no_grad():
+        host_toks, _ = _host_decode(model, prompt, prefix_len, steps=8, device=dev)
+
+        gd = GraphedDecoder(model, allocate_kv_cache(model, 4, device=dev,
+                                                     dtype=torch.bfloat16))
+        gd.prefill(prompt, prefix_len=prefix_len)
+        graph_toks = gd.decode(8)
+
+    assert gd._graph is not None, "hipGraph was not captured"
+    # bf16 + differing KV-split merge order can flip an argmax near a tie;
+    # require the sequences to agree on the overwhelming majority of tokens
+    match = (graph_toks == host_toks).float().mean().item()
+    assert match >= 0.75, f"graphed decode diverged from host decode: {match:.2f}"
+
+    # replays after a fresh prefill reuse the captured graph and stay coherent
+    gd.prefill(prompt, prefix_len=prefix_len)
+    again = gd.decode(8)
+    assert torch.equal(again, graph_toks)
