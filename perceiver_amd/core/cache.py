@@ -19,6 +19,12 @@ class StaticKVCache:
         self.k_buf = torch.empty(batch, capacity, k_channels, device=device, dtype=dtype)
         self.v_buf = torch.empty(batch, capacity, v_channels, device=device, dtype=dtype)
         self.length = 0
+        # pre_rotated: keys are stored ALREADY rotary-rotated at their absolute
+        # positions, so attention skips the O(cache) re-rotation every step.
+        # Scores are identical (RoPE depends on position differences only), but
+        # baked-in rotations are incompatible with window sliding — only the
+        # graph decoder (which forbids truncation) sets this.
+        self.pre_rotated = False
 
     @property
     def capacity(self) -> int:
@@ -71,6 +77,8 @@ class StaticKVCache:
         out.k_buf = self.k_buf.index_select(0, idx)
         out.v_buf = self.v_buf.index_select(0, idx)
         out.length = self.length
+        out.pre_rotated = self.pre_rotated
+        out._len_dev = None
         return out
 
 

@@ -79,14 +79,6 @@ class GraphedDecoder:
         self.ar_sa = torch.arange(self.sa_cap, device=dev).unsqueeze(0)
         self.out_buf = torch.zeros(self.batch, self.sa_cap or self.ca_cap,
                                    dtype=torch.long, device=dev)
-        # full-capacity rotary tables (fp32, like the host path); SA table is
-        # filled at prefill time once prefix0 is known, and its LAST row is
-        # refreshed in-graph with the current query position each step
-        frq = model.input_adapter.frq_pos_encoding
-        self.frq_ca = frq(self.ar_ca)                       # (1, ca_cap, rot)
-        self.frq_sa = torch.zeros(1, self.sa_cap, self.frq_ca.shape[-1],
-                                  dtype=self.frq_ca.dtype, device=dev)
-        self._prefix0: Optional[int] = None
         self._graph = None
 
     # ------------------------------------------------------------------ prefill
@@ -98,19 +90,14 @@ class GraphedDecoder:
         for c in self.caches:
             c.disable_graph_append()
             c.reset()
+            # store keys pre-rotated at their absolute positions: the decode
+            # step then never touches cached rows again (no O(cache) re-rotate)
+            c.pre_rotated = True
         out = self.model(prompt, prefix_len=prefix_len, kv_cache=self.caches)
         tok = out.logits[:, -1:].argmax(-1)
 
         n0 = prompt.shape[1]
         prefix0 = prefix_len
-        if self._prefix0 is not None and self._prefix0 != prefix0:
-            # SA rotary table depends on prefix0; a captured graph bakes its
-            # buffer POINTERS only, so refilling the table is enough
-            self._graph = None
-        self._prefix0 = prefix0
-        frq = self.model.input_adapter.frq_pos_encoding
-        self.frq_sa.copy_(frq(self.ar_sa + prefix0))
-
         self.tok.copy_(tok)
         self.ca_len.fill_(n0)
         self.sa_len.fill_(n0 - prefix0)
@@ -134,26 +121,26 @@ class GraphedDecoder:
         x, frq_q = m.input_adapter(self.tok, abs_pos=pos_q)
 
         # cross-attention: new token is the single latent; cached K/V at full
-        # capacity, tail j > ca_len masked (the new token lands AT index ca_len)
+        # capacity, tail j > ca_len masked (the new token lands AT index ca_len).
+        # Caches are pre-rotated, so BOTH rotary tables are just the current
+        # position's row: q uses it directly, k uses it to rotate the new row
+        # before it is appended.
+        rot_now = RotaryPositionEmbedding(frq_q, right_align=True)
         ca_mask = (self.ar_ca > self.ca_len).expand(self.batch, -1)
         ca_out = m.cross_attention(
             x,
             x_kv_prefix=x[:, :0],
             pad_mask=ca_mask,
-            rot_pos_emb_q=RotaryPositionEmbedding(frq_q, right_align=True),
-            rot_pos_emb_k=RotaryPositionEmbedding(self.frq_ca, right_align=True),
+            rot_pos_emb_q=rot_now,
+            rot_pos_emb_k=rot_now,
             kv_cache=self.caches[0],
         )
 
-        # self-attention stack: shared q/k rotary object — k rows read the
-        # static table (garbage tail rows are masked), the 1-row query right-
-        # aligns onto the last row, refreshed here to the current position
-        self.frq_sa[:, -1:].copy_(frq_q[:1])
         sa_mask = (self.ar_sa > self.sa_len).expand(self.batch, -1)
         sa_out = m.self_attention(
             ca_out.last_hidden_state,
             pad_mask=sa_mask,
-            rot_pos_emb=RotaryPositionEmbedding(self.frq_sa, right_align=True),
+            rot_pos_emb=rot_now,
             kv_cache=list(self.caches[1:]),
         )
 

@@ -126,6 +126,17 @@ class MultiHeadAttention(nn.Module):
             k = self.k_proj(x_kv)
             v = self.v_proj(x_kv)
 
+        pre_rotate = (
+            isinstance(kv_cache, StaticKVCache)
+            and kv_cache.pre_rotated
+            and rot_pos_emb_k is not None
+        )
+        if pre_rotate:
+            # rotate only the NEW rows (right-aligned table rows are exactly
+            # their positions) and store them rotated — the cached prefix keeps
+            # its baked-in rotation instead of being re-rotated every step
+            k = self._merge_heads(rot_pos_emb_k.rotate(self._split_heads(k)))
+
         if kv_cache is not None:
             if isinstance(kv_cache, StaticKVCache):
                 # in-place append; k/v become strided views over the live prefix
@@ -142,7 +153,7 @@ class MultiHeadAttention(nn.Module):
 
         if rot_pos_emb_q is not None:
             q = rot_pos_emb_q.rotate(q)
-        if rot_pos_emb_k is not None:
+        if rot_pos_emb_k is not None and not pre_rotate:
             k = rot_pos_emb_k.rotate(k)
 
         o = scaled_dot_attention(
