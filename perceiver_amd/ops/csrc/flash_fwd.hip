@@ -263,11 +263,26 @@ __global__ void flash_fwd_kernel(            // budget allows (not the 352 templ
                 }
                 rowmax[r] = warp16_max(mx);
             }
+            // rescale-skip: once the running max is established, most tiles do
+            // not raise it — skip the O-wide rescale + m update wholesale (the
+            // branch must be wave-uniform, hence __any over the wave's rows)
+            bool need = false;
+#pragma unroll
+            for (int r = 0; r < 4; ++r) need |= rowmax[r] > m_run[h][r];
+            if (__any(need)) {
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    float m_new = fmaxf(m_run[h][r], rowmax[r]);
+                    float alpha = expf(m_run[h][r] - m_new);
+                    m_run[h][r] = m_new;
+                    l_run[h][r] *= alpha;
+#pragma unroll
+                    for (int cb = 0; cb < DVMAX / 16; ++cb) o_acc[h][cb][r] *= alpha;
+                }
+            }
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
-                float m_new = fmaxf(m_run[h][r], rowmax[r]);
-                float alpha = expf(m_run[h][r] - m_new);
-                m_run[h][r] = m_new;
+                float m_new = m_run[h][r];
                 float psum = 0.f;
 #pragma unroll
                 for (int kb = 0; kb < KEYBLKS; ++kb) {
@@ -282,12 +297,11 @@ __global__ void flash_fwd_kernel(            // budget allows (not the 352 templ
                         p_mine + (h * 16 + hi4 * 4 + r) * vt_stride + (kb * 16 + lo16) * 2) =
                         f2bf(pv);
                 }
-                l_run[h][r] = l_run[h][r] * alpha + warp16_sum(psum);
-#pragma unroll
-                for (int cb = 0; cb < DVMAX / 16; ++cb) o_acc[h][cb][r] *= alpha;
+                l_run[h][r] += warp16_sum(psum);
             }
         }
-        __builtin_amdgcn_s_waitcnt(0);  // lgkmcnt(0): wave-local LDS ordering
+        // wave-local LDS ordering for the P roundtrip (lgkmcnt only)
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
         bf16x8 p_frag[QH][KEYBLKS / 2];
 #pragma unroll
         for (int h = 0; h < QH; ++h)
