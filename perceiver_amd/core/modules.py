@@ -7,8 +7,10 @@ Functional parity with /root/reference/perceiver/model/core/modules.py (see SURV
     which dispatches to the fused CDNA4 flash kernels on GPU (K1/K3/K4/K7/K9 of
     SURVEY.md §2.3) and to a plain PyTorch composition on CPU;
   - activation checkpointing uses ``torch.utils.checkpoint`` (no fairscale);
-  - state-dict key layout is kept compatible with the reference checkpoint format
-    (q_proj/k_proj/v_proj/o_proj, cross_attn_1/self_attn_1/..., latent_provider._query).
+  - state-dict key layout follows the reference checkpoint format
+    (cross_attn_1/self_attn_1/..., latent_provider._query); self-attention layers
+    store one merged ``qkv_proj`` parameter (rows [q|k|v]) and load legacy split
+    q_proj/k_proj/v_proj keys through a pre-hook.
 
 KV caches are (k, v) tuples of shape (B, L, num_qk_channels)/(B, L, num_v_channels),
 concatenated along the sequence dim *before* the head split
@@ -64,6 +66,7 @@ class MultiHeadAttention(nn.Module):
         dropout: float = 0.0,
         qkv_bias: bool = True,
         out_bias: bool = True,
+        merged_qkv: bool = False,
     ):
         super().__init__()
 
@@ -87,13 +90,38 @@ class MultiHeadAttention(nn.Module):
         self.attention_dropout = dropout
         self.max_heads_parallel = max_heads_parallel
 
-        self.q_proj = nn.Linear(num_q_input_channels, num_qk_channels, bias=qkv_bias)
-        self.k_proj = nn.Linear(num_kv_input_channels, num_qk_channels, bias=qkv_bias)
-        self.v_proj = nn.Linear(num_kv_input_channels, num_v_channels, bias=qkv_bias)
+        if merged_qkv and num_q_input_channels == num_kv_input_channels:
+            # self-attention: ONE merged projection parameter — one wide GEMM per
+            # forward instead of three skinny ones plus a per-call weight concat.
+            # qkv_proj.weight rows are [q | k | v] in the reference's per-matrix
+            # order (modules.py:50-55); the load hook accepts legacy checkpoints
+            # with split q_proj/k_proj/v_proj keys.
+            self.qkv_proj = nn.Linear(
+                num_kv_input_channels, 2 * num_qk_channels + num_v_channels, bias=qkv_bias
+            )
+            self._register_load_state_dict_pre_hook(
+                MultiHeadAttention._merge_sd_hook, with_module=True
+            )
+        else:
+            self.q_proj = nn.Linear(num_q_input_channels, num_qk_channels, bias=qkv_bias)
+            self.k_proj = nn.Linear(num_kv_input_channels, num_qk_channels, bias=qkv_bias)
+            self.v_proj = nn.Linear(num_kv_input_channels, num_v_channels, bias=qkv_bias)
         self.o_proj = nn.Linear(num_v_channels, num_output_channels, bias=out_bias)
         # kept as a module for state-dict/layout parity; the dispatch path applies
         # dropout functionally inside the attention core
         self.dropout = nn.Dropout(dropout)
+
+    @staticmethod
+    def _merge_sd_hook(module, state_dict, prefix, *args):
+        """Accept split q/k/v keys and fold them into the merged projection."""
+        if prefix + "q_proj.weight" in state_dict:
+            state_dict[prefix + "qkv_proj.weight"] = torch.cat(
+                [state_dict.pop(prefix + n + "_proj.weight") for n in ("q", "k", "v")], dim=0
+            )
+        if prefix + "q_proj.bias" in state_dict:
+            state_dict[prefix + "qkv_proj.bias"] = torch.cat(
+                [state_dict.pop(prefix + n + "_proj.bias") for n in ("q", "k", "v")], dim=0
+            )
 
     def _split_heads(self, x: torch.Tensor) -> torch.Tensor:
         b, n, _ = x.shape
@@ -112,15 +140,18 @@ class MultiHeadAttention(nn.Module):
         rot_pos_emb_k: Optional[RotaryPositionEmbedding] = None,
         kv_cache: Optional[KVCache] = None,
     ) -> ModuleOutput:
-        if x_q is x_kv:
-            # self-attention: one merged QKV GEMM instead of three skinny ones
-            # (weights concatenated at call time; state-dict layout unchanged)
-            w = torch.cat([self.q_proj.weight, self.k_proj.weight, self.v_proj.weight], dim=0)
-            b = None
-            if self.q_proj.bias is not None:
-                b = torch.cat([self.q_proj.bias, self.k_proj.bias, self.v_proj.bias], dim=0)
-            qkv = torch.nn.functional.linear(x_q, w, b)
-            q, k, v = qkv.split([self.num_qk_channels, self.num_qk_channels, self.num_v_channels], dim=-1)
+        qk, vc = self.num_qk_channels, self.num_v_channels
+        if hasattr(self, "qkv_proj"):
+            if x_q is x_kv:
+                qkv = self.qkv_proj(x_q)
+                q, k, v = qkv.split([qk, qk, vc], dim=-1)
+            else:
+                # same tensors, different inputs (e.g. cached decode through a
+                # SelfAttention layer): sliced views of the merged parameter
+                w, bias = self.qkv_proj.weight, self.qkv_proj.bias
+                q = torch.nn.functional.linear(x_q, w[:qk], None if bias is None else bias[:qk])
+                k = torch.nn.functional.linear(x_kv, w[qk: 2 * qk], None if bias is None else bias[qk: 2 * qk])
+                v = torch.nn.functional.linear(x_kv, w[2 * qk:], None if bias is None else bias[2 * qk:])
         else:
             q = self.q_proj(x_q)
             k = self.k_proj(x_kv)
@@ -254,6 +285,7 @@ class SelfAttention(nn.Module):
             dropout=dropout,
             qkv_bias=qkv_bias,
             out_bias=out_bias,
+            merged_qkv=True,
         )
 
     def forward(

@@ -42,9 +42,21 @@ def copy_params(src: nn.Module, tgt: nn.Module):
 
 
 def copy_attention_params(src, tgt: MultiHeadAttention):
-    copy_params(src.attention.self.query, tgt.q_proj)
-    copy_params(src.attention.self.key, tgt.k_proj)
-    copy_params(src.attention.self.value, tgt.v_proj)
+    if hasattr(tgt, "qkv_proj"):
+        # merged projection (self-attention layers): write the q/k/v slices
+        qk = tgt.num_qk_channels
+        with torch.no_grad():
+            tgt.qkv_proj.weight[:qk].copy_(src.attention.self.query.weight)
+            tgt.qkv_proj.weight[qk: 2 * qk].copy_(src.attention.self.key.weight)
+            tgt.qkv_proj.weight[2 * qk:].copy_(src.attention.self.value.weight)
+            if tgt.qkv_proj.bias is not None:
+                tgt.qkv_proj.bias[:qk].copy_(src.attention.self.query.bias)
+                tgt.qkv_proj.bias[qk: 2 * qk].copy_(src.attention.self.key.bias)
+                tgt.qkv_proj.bias[2 * qk:].copy_(src.attention.self.value.bias)
+    else:
+        copy_params(src.attention.self.query, tgt.q_proj)
+        copy_params(src.attention.self.key, tgt.k_proj)
+        copy_params(src.attention.self.value, tgt.v_proj)
     copy_params(src.attention.output.dense, tgt.o_proj)
 
 
