@@ -9,6 +9,7 @@ from __future__ import annotations
 
 from collections import OrderedDict
 
+import torch
 import torch.nn as nn
 
 
@@ -29,6 +30,17 @@ class ModuleOutput(OrderedDict):
             del self[name]
         except KeyError:
             raise AttributeError(f"No such attribute: {name}") from None
+
+
+# Register with torch's pytree so containers of tensors returned by module
+# forwards are traversed (FSDP2 attaches its pre-backward re-gather hooks to the
+# tensors it finds in the output tree; an unregistered OrderedDict SUBCLASS is a
+# leaf and the hooks would silently not attach).
+torch.utils._pytree.register_pytree_node(
+    ModuleOutput,
+    lambda mo: (list(mo.values()), list(mo.keys())),
+    lambda values, keys: ModuleOutput(zip(keys, values)),
+)
 
 
 class Residual(nn.Module):

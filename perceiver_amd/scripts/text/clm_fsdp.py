@@ -34,20 +34,34 @@ DEFAULTS = {
 }
 
 
+def shard_perceiver_ar(model, device_type: str = None):
+    """Shard a PerceiverAR-family model per attention layer with ``fully_shard``
+    (FSDP2) — the same layer granularity as the reference's
+    transformer_auto_wrap_policy over its attention layer classes
+    (reference scripts/text/clm_fsdp.py:15-23). Parameters become DTensors over a
+    1-D mesh of the full world; per-layer units overlap reshard/all-gather with
+    compute. Works over RCCL on GPU and gloo on CPU (the 2-rank CPU test)."""
+    import torch.distributed as dist
+    from torch.distributed.device_mesh import init_device_mesh
+    from torch.distributed.fsdp import fully_shard
+
+    if device_type is None:
+        device_type = "cuda" if torch.cuda.is_available() else "cpu"
+    mesh = init_device_mesh(device_type, (dist.get_world_size(),))
+    for m in model.modules():
+        if isinstance(m, (CrossAttentionLayer, SelfAttentionLayer)):
+            fully_shard(m, mesh=mesh)
+    fully_shard(model, mesh=mesh)
+    return model
+
+
 class LitCausalLanguageModelFSDP(LitCausalLanguageModel):
-    """Wraps the backend in torch FSDP with a transformer auto-wrap policy over the
-    attention layer classes (the reference's fsdp_perceiver_ar strategy)."""
+    """Wraps the backend with per-attention-layer fully_shard (FSDP2)."""
 
     def wrap_fsdp(self):
-        from torch.distributed.fsdp import FullyShardedDataParallel as FSDP
-        from torch.distributed.fsdp.wrap import transformer_auto_wrap_policy
-
-        policy = functools.partial(
-            transformer_auto_wrap_policy,
-            transformer_layer_cls={CrossAttentionLayer, SelfAttentionLayer},
-        )
-        self.model = FSDP(self.model, auto_wrap_policy=policy,
-                          device_id=torch.cuda.current_device() if torch.cuda.is_available() else None)
+        if torch.cuda.is_available():
+            self.model = self.model.to(torch.device("cuda", torch.cuda.current_device()))
+        shard_perceiver_ar(self.model)
         return self
 
 

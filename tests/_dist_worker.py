@@ -62,7 +62,36 @@ def trainer_worker(rank, world, out_dir):
     torch.save(model.state_dict(), os.path.join(out_dir, f"rank{rank}_weights.pt"))
 
 
-MODES = {"reducer": reducer_worker, "trainer": trainer_worker}
+def fsdp_worker(rank, world, out_dir):
+    """FSDP path (the scripts/text/clm_fsdp.py sharding layout) on gloo/CPU: shard
+    a tiny CLM per attention layer with fully_shard, run 2 optimizer steps on
+    identical data, dump the reassembled full state dict (must match across
+    ranks)."""
+    from torch.distributed.checkpoint.state_dict import StateDictOptions, get_model_state_dict
+
+    from perceiver_amd.scripts.text.clm_fsdp import shard_perceiver_ar
+
+    model = make_clm_model()
+    shard_perceiver_ar(model, device_type="cpu")
+    opt = torch.optim.AdamW(model.parameters(), lr=1e-3)
+
+    torch.manual_seed(7)  # same data on all ranks -> parameter trajectories identical
+    losses = []
+    for _ in range(2):
+        x = torch.randint(0, 50, (2, 16))
+        out = model(x, prefix_len=8)
+        loss = F.cross_entropy(out.logits.flatten(0, 1), x[:, 8:].flatten())
+        opt.zero_grad(set_to_none=True)
+        loss.backward()
+        opt.step()
+        losses.append(loss.item())
+
+    full = get_model_state_dict(model, options=StateDictOptions(full_state_dict=True))
+    torch.save({"state_dict": full, "losses": losses},
+               os.path.join(out_dir, f"rank{rank}_fsdp.pt"))
+
+
+MODES = {"reducer": reducer_worker, "trainer": trainer_worker, "fsdp": fsdp_worker}
 
 if __name__ == "__main__":
     name, rank, world, port, out_dir = sys.argv[1:6]
