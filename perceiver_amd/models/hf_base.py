@@ -230,6 +230,7 @@ class PerceiverCausalSequenceModel(PreTrainedModel):
         eos_token_id: Optional[int] = None,
         num_beams: int = 1,
         length_penalty: float = 1.0,
+        penalty_alpha: Optional[float] = None,
         generator: Optional[torch.Generator] = None,
         **kwargs,
     ) -> torch.Tensor:
@@ -267,6 +268,12 @@ class PerceiverCausalSequenceModel(PreTrainedModel):
             return self._beam_search(input_ids, attention_mask, prefix_len, num_beams,
                                      max_new_tokens, eos_token_id, pad_token_id,
                                      length_penalty, use_cache, static_cache)
+        if penalty_alpha is not None and penalty_alpha > 0 and top_k and top_k > 1 \
+                and not do_sample:
+            return self._contrastive_search(input_ids, attention_mask, prefix_len,
+                                            top_k, penalty_alpha, max_new_tokens,
+                                            eos_token_id, pad_token_id, use_cache,
+                                            static_cache)
         done = torch.zeros(input_ids.shape[0], dtype=torch.bool, device=input_ids.device)
         past = None
         if use_cache and static_cache:
@@ -371,6 +378,87 @@ class PerceiverCausalSequenceModel(PreTrainedModel):
         best = (beam_scores / norm).argmax(dim=-1)
         idx = best + torch.arange(bsz, device=device) * num_beams
         return input_ids.index_select(0, idx)
+
+    @torch.no_grad()
+    def _contrastive_search(self, input_ids, attention_mask, prefix_len, top_k,
+                            penalty_alpha, max_new_tokens, eos_token_id,
+                            pad_token_id, use_cache, static_cache):
+        """Contrastive search (Su et al. 2022, the 🤗 penalty_alpha decoding the
+        reference inherits from GenerationMixin): each step scores the top-k
+        candidates by (1-a)*p(v) - a*max_cos(h_v, previous hidden states) and
+        commits the best. The candidate forward runs the batch expanded k-fold
+        through the normal cached step, so the selected candidate's KV append is
+        reused as the real step (one extra forward per token, not k).
+
+        Perceiver-AR adaptation: only latent positions have output states, so the
+        degeneration penalty compares against the latent tail of the prompt plus
+        the generated tokens (prefix-only positions have no hidden state)."""
+        if not use_cache:
+            raise ValueError("contrastive search requires use_cache=True")
+        bsz = input_ids.shape[0]
+        device = input_ids.device
+        k = top_k
+
+        past = None
+        if static_cache:
+            p = next(self.backend_model.parameters())
+            past = allocate_kv_cache(self.backend_model, bsz, device=p.device, dtype=p.dtype)
+
+        # prompt pass: logits for step 0 + the latent hidden states as history
+        model_inputs = self.prepare_inputs_for_generation(
+            input_ids, past_key_values=past if past is not None else [],
+            attention_mask=attention_mask, use_cache=True, prefix_len=prefix_len)
+        out = self(**model_inputs)
+        prefix_len = out.prefix_len
+        past = out.past_key_values
+        logits = out.logits[:, -1, :].float()
+        hist = torch.nn.functional.normalize(out.hidden_states[-1].float(), dim=-1)
+
+        done = torch.zeros(bsz, dtype=torch.bool, device=device)
+        fill = pad_token_id if pad_token_id is not None else (eos_token_id or 0)
+        rep = torch.arange(bsz, device=device).repeat_interleave(k)
+
+        for _ in range(max_new_tokens):
+            probs = logits.softmax(-1)
+            top_p_vals, cand = probs.topk(k, dim=-1)             # (bsz, k)
+
+            # expanded candidate step: k copies of every sequence, one per candidate
+            past_k = self._reorder_cache(past, rep)
+            ids_k = torch.cat([input_ids.index_select(0, rep), cand.view(-1, 1)], dim=1)
+            mask_k = torch.cat([attention_mask.index_select(0, rep),
+                                torch.ones(bsz * k, 1, dtype=attention_mask.dtype,
+                                           device=device)], dim=1)
+            model_inputs = self.prepare_inputs_for_generation(
+                ids_k, past_key_values=past_k, attention_mask=mask_k,
+                use_cache=True, prefix_len=prefix_len)
+            out = self(**model_inputs)
+
+            h_cand = torch.nn.functional.normalize(
+                out.hidden_states[-1][:, -1, :].float(), dim=-1)  # (bsz*k, C)
+            # max cosine similarity against each candidate's own history
+            sim = torch.einsum("btc,bkc->bkt", hist,
+                               h_cand.view(bsz, k, -1)).amax(-1)  # (bsz, k)
+            score = (1.0 - penalty_alpha) * top_p_vals - penalty_alpha * sim
+            sel = score.argmax(-1)                                # (bsz,)
+            idx = torch.arange(bsz, device=device) * k + sel
+
+            past = self._reorder_cache(out.past_key_values, idx)
+            prefix_len = out.prefix_len
+            logits = out.logits.index_select(0, idx)[:, -1, :].float()
+            hist = torch.cat([hist, h_cand.index_select(0, idx)[:, None]], dim=1)
+
+            next_token = cand.gather(-1, sel[:, None]).squeeze(-1)
+            if eos_token_id is not None:
+                next_token = torch.where(done, torch.full_like(next_token, fill), next_token)
+                done = done | (next_token == eos_token_id)
+
+            input_ids = torch.cat([input_ids, next_token[:, None]], dim=1)
+            attention_mask = torch.cat([attention_mask,
+                                        torch.ones_like(next_token[:, None])], dim=1)
+            if eos_token_id is not None and bool(done.all()):
+                break
+
+        return input_ids
 
     @staticmethod
     def _select_next(logits, do_sample, temperature, top_k, top_p, generator):
