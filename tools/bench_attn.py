@@ -26,14 +26,16 @@ SHAPES = [
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--iters", type=int, default=30)
+    ap.add_argument("--only", default=None, help="comma-separated shape names")
     args = ap.parse_args()
+    shapes = SHAPES if args.only is None else [s for s in SHAPES if s[0] in args.only.split(",")]
 
     from perceiver_amd.ops import hip
 
     ext = hip.ext()
     dev = "cuda"
     results = {}
-    for name, b, h, nq, lk, d, dv, causal in SHAPES:
+    for name, b, h, nq, lk, d, dv, causal in shapes:
         q = (torch.randn(b, h, nq, d, device=dev) * (d ** -0.5)).bfloat16()
         k = torch.randn(b, h, lk, d, device=dev).bfloat16()
         v = torch.randn(b, h, lk, dv, device=dev).bfloat16()
