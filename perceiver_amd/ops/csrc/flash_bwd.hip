@@ -91,13 +91,8 @@ DEVINL void stage_tr(const unsigned short* __restrict__ src, long src_stride,
             }
             rows[r] = val;
         }
-        // XOR phase per thread — see flash_fwd.hip stage_tile_transposed: rows 8
-        // apart alias one LDS bank at 16-B-aligned strides; the phase keeps the
-        // wave's lanes on different rows per write instruction.
-        const int ph = g & 7;
 #pragma unroll
-        for (int e0 = 0; e0 < 8; ++e0) {
-            int e = e0 ^ ph;
+        for (int e = 0; e < 8; ++e) {
             short4x pack = {rows[0][e], rows[1][e], rows[2][e], rows[3][e]};
             *reinterpret_cast<short4x*>(ldsT + (c0 + e) * ldst_bytes + row0 * 2) = pack;
         }

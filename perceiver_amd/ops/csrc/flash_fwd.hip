@@ -100,14 +100,8 @@ DEVINL void stage_tile_transposed(const unsigned short* __restrict__ src, long s
             }
             rows[r] = val;
         }
-        // XOR phase per thread: with 16-B-aligned row strides, rows 8 apart
-        // always alias the same LDS bank, and a lockstep e-loop makes every
-        // lane's write land on one bank per iteration (measured ~11% of wave
-        // cycles). The phase keeps lanes on different rows per instruction.
-        const int ph = g & 7;
 #pragma unroll
-        for (int e0 = 0; e0 < 8; ++e0) {
-            int e = e0 ^ ph;
+        for (int e = 0; e < 8; ++e) {
             short4x pack = {rows[0][e], rows[1][e], rows[2][e], rows[3][e]};
             *reinterpret_cast<short4x*>(ldsT + (c0 + e) * ldst_bytes + key0 * 2) = pack;
         }
