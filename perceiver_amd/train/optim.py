@@ -21,9 +21,15 @@ class MasterAdamW(torch.optim.Optimizer):
     bf16 parameters. Falls back to torch._foreach_* elsewhere."""
 
     def __init__(self, params: Iterable[torch.nn.Parameter], lr: float = 1e-3,
-                 betas=(0.9, 0.999), eps: float = 1e-8, weight_decay: float = 0.01):
+                 betas=(0.9, 0.999), eps: float = 1e-8, weight_decay: float = 0.01,
+                 max_grad_norm: float = 0.0):
+        """``max_grad_norm``: 0 disables; otherwise global-norm gradient clipping
+        is applied INSIDE the step on the fp32 gradients — on the fused path one
+        norm + one scale kernel over the flat buffer with no host sync, instead
+        of torch's per-parameter foreach chain before the step."""
         defaults = dict(lr=lr, betas=betas, eps=eps, weight_decay=weight_decay)
         super().__init__(params, defaults)
+        self.max_grad_norm = max_grad_norm
         self._flat = None  # lazy per-group flat state
 
     def _build_flat(self, group):
@@ -70,6 +76,9 @@ class MasterAdamW(torch.optim.Optimizer):
         grads = [p.grad for p in fl["params"]]
         assert all(g is not None for g in grads), "missing gradient in fused AdamW step"
         torch._foreach_copy_(fl["views_g"], grads)
+        if self.max_grad_norm > 0:
+            scale = (self.max_grad_norm / (fl["g"].norm() + 1e-6)).clamp(max=1.0)
+            fl["g"].mul_(scale)
         beta1, beta2 = group["betas"]
         hip.ext().adamw_step(fl["master"], fl["m"], fl["v"], fl["g"],
                              group["lr"], beta1, beta2, group["eps"],
@@ -112,6 +121,11 @@ class MasterAdamW(torch.optim.Optimizer):
 
             # single fused multi-tensor cast instead of one cast kernel per tensor
             torch._foreach_copy_(grads, raw_grads)
+            if self.max_grad_norm > 0:
+                total = torch.linalg.vector_norm(
+                    torch.stack([torch.linalg.vector_norm(g) for g in grads]))
+                scale = (self.max_grad_norm / (total + 1e-6)).clamp(max=1.0)
+                torch._foreach_mul_(grads, scale)
 
             beta1, beta2 = group["betas"]
             lr, eps, wd = group["lr"], group["eps"], group["weight_decay"]

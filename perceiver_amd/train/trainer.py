@@ -127,7 +127,7 @@ class Trainer:
             loss.backward()
             if reducer is not None:
                 reducer.finalize()
-            if self.cfg.grad_clip:
+            if self.cfg.grad_clip and not getattr(optimizer, "max_grad_norm", 0.0):
                 torch.nn.utils.clip_grad_norm_(model.parameters(), self.cfg.grad_clip)
             optimizer.step()
             if scheduler is not None:
@@ -181,7 +181,7 @@ class Trainer:
                 loss.backward()
                 if reducer is not None:
                     reducer.finalize()
-                if self.cfg.grad_clip:
+                if self.cfg.grad_clip and not getattr(optimizer, "max_grad_norm", 0.0):
                     torch.nn.utils.clip_grad_norm_(task.parameters(), self.cfg.grad_clip)
                 optimizer.step()
                 if scheduler is not None:

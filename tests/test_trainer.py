@@ -153,3 +153,29 @@ def test_cli_fit_runs(tmp_path, monkeypatch):
     assert trainer.global_step == 2
     assert (tmp_path / "cli_run" / "config.yaml").exists()
     assert (tmp_path / "cli_run" / "checkpoints" / "last.ckpt").exists()
+
+
+def test_master_adamw_in_step_clipping_matches_torch_clip():
+    """MasterAdamW(max_grad_norm=c) must take the same step as external
+    clip_grad_norm_ + MasterAdamW without clipping (foreach path, fp32 CPU)."""
+    import copy
+
+    from perceiver_amd.train.optim import MasterAdamW
+
+    torch.manual_seed(0)
+    m1 = torch.nn.Sequential(torch.nn.Linear(8, 16), torch.nn.Linear(16, 4))
+    m2 = copy.deepcopy(m1)
+    o1 = MasterAdamW(m1.parameters(), lr=1e-2, max_grad_norm=0.5)
+    o2 = MasterAdamW(m2.parameters(), lr=1e-2)
+
+    for _ in range(3):
+        x = torch.randn(4, 8)
+        for m, o, clip in ((m1, o1, False), (m2, o2, True)):
+            o.zero_grad(set_to_none=True)
+            (m(x).square().mean() * 37).backward()
+            if clip:
+                torch.nn.utils.clip_grad_norm_(m.parameters(), 0.5)
+            o.step()
+
+    for p1, p2 in zip(m1.parameters(), m2.parameters()):
+        torch.testing.assert_close(p1, p2, rtol=1e-6, atol=1e-7)
