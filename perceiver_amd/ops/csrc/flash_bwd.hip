@@ -40,6 +40,54 @@ __global__ void delta_kernel(const unsigned short* __restrict__ dout,
     if (lane == 0) delta[row] = acc;
 }
 
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
+
+// ---- 16-column-subtiled image + ds_read_b64_tr_b16 (see flash_fwd.hip) ----
+// Replaces the separate transposed staging pass: the image is written ROW-major
+// (one 16-B write per 8-channel granule) and consumed column-major through the
+// hardware transpose read, so the 8-row packed-write pass (and its bank-conflict
+// lockstep) disappears along with the duplicate global re-read.
+template <int ROWS_TILE>
+DEVINL void stage_sub16(const unsigned short* __restrict__ src, long src_stride,
+                        int rows_valid, int d, int d_pad, char* lds, int tid) {
+    constexpr int SUBE = ROWS_TILE * 16 + 8;
+    const int gpr = d_pad / 8;
+    const int total = ROWS_TILE * gpr;
+    for (int g = tid; g < total; g += 256) {
+        int row = g / gpr;
+        int c0 = (g % gpr) * 8;
+        short8v val = {};
+        if (row < rows_valid && c0 < d) {
+            if (c0 + 8 <= d) {
+                val = *reinterpret_cast<const short8v*>(src + (long)row * src_stride + c0);
+            } else {
+#pragma unroll
+                for (int e = 0; e < 8; ++e)
+                    val[e] = (c0 + e < d) ? (short)src[(long)row * src_stride + c0 + e] : (short)0;
+            }
+        }
+        *reinterpret_cast<short8v*>(
+            lds + ((c0 / 16) * SUBE + row * 16 + (c0 % 16)) * 2) = val;
+    }
+}
+
+// B-fragment rows row0..row0+31, cols 16*sub..: per-lane 8-B loads of one 4x16
+// row-major tile per 16-lane group; the instruction's fixed cross-lane exchange
+// delivers column lo16.
+template <int ROWS_TILE>
+DEVINL bf16x8 read_bfrag_tr16(const char* lds, int sub, int row0, int hi4, int lo16) {
+    constexpr int SUBE = ROWS_TILE * 16 + 8;
+    const __bf16* base = reinterpret_cast<const __bf16*>(lds) +
+                         sub * SUBE + (row0 + hi4 * 8) * 16 + lo16 * 4;
+    auto pp = (__attribute__((address_space(3))) bf16x4*)base;
+    bf16x4 lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(pp);
+    bf16x4 hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(pp + 16);  // +4 rows
+    bf16x8 out;
+#pragma unroll
+    for (int e = 0; e < 4; ++e) { out[e] = lo[e]; out[e + 4] = hi[e]; }
+    return out;
+}
+
 // stage (rows_tile x d) tile row-major into LDS (row stride ldst_bytes), zero-pad
 template <int ROWS_TILE>
 DEVINL void stage_rm(const unsigned short* __restrict__ src, long src_stride,
@@ -139,11 +187,11 @@ __global__ void flash_dq_kernel(
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
     const int k_stride = d_pad * 2 + 16;
-    const int kt_stride = TILE * 2 + 16;   // K^T: d_pad rows x TILE keys
+    const int kt_stride = TILE * 2 + 16;   // (p-buffer row stride)
     const int v_stride = dv_pad * 2 + 16;
     char* k_lds = smem;                                  // TILE * k_stride
-    char* kt_lds = k_lds + TILE * k_stride;              // DMAX * kt_stride
-    char* v_lds = kt_lds + DMAX * kt_stride;             // TILE * v_stride
+    char* kt16_lds = k_lds + TILE * k_stride;            // (DMAX/16) * (TILE*16+8) elems
+    char* v_lds = kt16_lds + (DMAX / 16) * (TILE * 16 + 8) * 2;   // TILE * v_stride
     char* p_lds = v_lds + TILE * v_stride;               // NWAVES * QROWS * kt_stride
     char* p_mine = p_lds + wave * QROWS * kt_stride;
 
@@ -210,7 +258,7 @@ __global__ void flash_dq_kernel(
         int rows_valid = min(TILE, Lk - kv0);
         __syncthreads();
         stage_rm<TILE>(kbase + (long)kv0 * ksn, ksn, rows_valid, D, d_pad, k_lds, k_stride, tid);
-        stage_tr<TILE>(kbase + (long)kv0 * ksn, ksn, rows_valid, D, d_pad, kt_lds, kt_stride, tid);
+        stage_sub16<TILE>(kbase + (long)kv0 * ksn, ksn, rows_valid, D, d_pad, kt16_lds, tid);
         stage_rm<TILE>(vbase + (long)kv0 * vsn, vsn, rows_valid, Dv, dv_pad, v_lds, v_stride, tid);
         __syncthreads();
 
@@ -274,14 +322,13 @@ __global__ void flash_dq_kernel(
                 ds_frag[h][t32] = (bf16x8)(*reinterpret_cast<const short8v*>(
                     p_mine + (h * 16 + lo16) * kt_stride + (t32 * 32 + hi4 * 8) * 2));
 
-        // dQ += dS K : B[key k][col d] = K^T_lds[d][k]
+        // dQ += dS K : B[k=key][j=ch] via transpose reads of the subtiled K image
 #pragma unroll
         for (int cb = 0; cb < DMAX / 16; ++cb) {
             if (cb * 16 < d_pad) {
 #pragma unroll
                 for (int t32 = 0; t32 < TBLKS / 2; ++t32) {
-                    const char* src = kt_lds + (cb * 16 + lo16) * kt_stride + (t32 * 32 + hi4 * 8) * 2;
-                    bf16x8 bfrag = (bf16x8)(*reinterpret_cast<const short8v*>(src));
+                    bf16x8 bfrag = read_bfrag_tr16<TILE>(kt16_lds, cb, t32 * 32, hi4, lo16);
 #pragma unroll
                     for (int h = 0; h < QH; ++h)
                         dq_acc[h][cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -355,13 +402,13 @@ __global__ void flash_dkv_kernel(
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
     const int q_stride = d_pad * 2 + 16;      // Q row-major: TILE rows
-    const int qt_stride = TILE * 2 + 16;      // Q^T: d_pad rows
+    const int qt_stride = TILE * 2 + 16;      // (p-buffer row stride)
     const int do_stride = dv_pad * 2 + 16;    // dO row-major: TILE rows
     char* q_lds = smem;
-    char* qt_lds = q_lds + TILE * q_stride;
-    char* do_lds = qt_lds + DMAX * qt_stride;
-    char* dot_lds = do_lds + TILE * do_stride;            // DVMAX * qt_stride
-    char* p_lds = dot_lds + DVMAX * qt_stride;            // NWAVES * KROWS * qt_stride
+    char* q16_lds = q_lds + TILE * q_stride;              // (DMAX/16)*(TILE*16+8) elems
+    char* do_lds = q16_lds + (DMAX / 16) * (TILE * 16 + 8) * 2;
+    char* do16_lds = do_lds + TILE * do_stride;           // (DVMAX/16)*(TILE*16+8) elems
+    char* p_lds = do16_lds + (DVMAX / 16) * (TILE * 16 + 8) * 2;  // NWAVES*KROWS*qt_stride
     char* p_mine = p_lds + wave * KROWS * qt_stride;
 
     short8v k_frag[QH][DMAX / 32];
@@ -429,9 +476,9 @@ __global__ void flash_dkv_kernel(
         int rows_valid = min(TILE, Nq - qt0);
         __syncthreads();
         stage_rm<TILE>(qbase + (long)qt0 * qsn, qsn, rows_valid, D, d_pad, q_lds, q_stride, tid);
-        stage_tr<TILE>(qbase + (long)qt0 * qsn, qsn, rows_valid, D, d_pad, qt_lds, qt_stride, tid);
+        stage_sub16<TILE>(qbase + (long)qt0 * qsn, qsn, rows_valid, D, d_pad, q16_lds, tid);
         stage_rm<TILE>(dobase + (long)qt0 * Dv, Dv, rows_valid, Dv, dv_pad, do_lds, do_stride, tid);
-        stage_tr<TILE>(dobase + (long)qt0 * Dv, Dv, rows_valid, Dv, dv_pad, dot_lds, qt_stride, tid);
+        stage_sub16<TILE>(dobase + (long)qt0 * Dv, Dv, rows_valid, Dv, dv_pad, do16_lds, tid);
         __syncthreads();
 
         // t-outer: per 16-q-row block compute S^T and dP^T with short-lived
@@ -499,14 +546,13 @@ __global__ void flash_dkv_kernel(
                 pt_frag[h][t32] = (bf16x8)(*reinterpret_cast<const short8v*>(
                     p_mine + (h * 16 + lo16) * qt_stride + (t32 * 32 + hi4 * 8) * 2));
 
-        // dV += P^T dO : B[qrow i][ch c] = dO^T_lds[c][i] (contiguous in i)
+        // dV += P^T dO : B[k=qrow][j=ch] via transpose reads of the subtiled image
 #pragma unroll
         for (int cb = 0; cb < DVMAX / 16; ++cb) {
             if (cb * 16 < dv_pad) {
 #pragma unroll
                 for (int t32 = 0; t32 < TBLKS / 2; ++t32) {
-                    const char* src = dot_lds + (cb * 16 + lo16) * qt_stride + (t32 * 32 + hi4 * 8) * 2;
-                    bf16x8 bfrag = (bf16x8)(*reinterpret_cast<const short8v*>(src));
+                    bf16x8 bfrag = read_bfrag_tr16<TILE>(do16_lds, cb, t32 * 32, hi4, lo16);
 #pragma unroll
                     for (int h = 0; h < QH; ++h)
                         dv_acc[h][cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -534,14 +580,13 @@ __global__ void flash_dkv_kernel(
                 dst_frag[h][t32] = (bf16x8)(*reinterpret_cast<const short8v*>(
                     p_mine + (h * 16 + lo16) * qt_stride + (t32 * 32 + hi4 * 8) * 2));
 
-        // dK += dS^T Q : B[qrow i][ch d] = Q^T_lds[d][i] (contiguous in i)
+        // dK += dS^T Q : B[k=qrow][j=ch] via transpose reads of the subtiled image
 #pragma unroll
         for (int cb = 0; cb < DMAX / 16; ++cb) {
             if (cb * 16 < d_pad) {
 #pragma unroll
                 for (int t32 = 0; t32 < TBLKS / 2; ++t32) {
-                    const char* src = qt_lds + (cb * 16 + lo16) * qt_stride + (t32 * 32 + hi4 * 8) * 2;
-                    bf16x8 bfrag = (bf16x8)(*reinterpret_cast<const short8v*>(src));
+                    bf16x8 bfrag = read_bfrag_tr16<TILE>(q16_lds, cb, t32 * 32, hi4, lo16);
 #pragma unroll
                     for (int h = 0; h < QH; ++h)
                         dk_acc[h][cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -590,7 +635,7 @@ void launch_flash_bwd(const torch::Tensor& dout, const torch::Tensor& q, const t
     {   // dQ (with KV-split when the grid would underfill the chip)
         const int k_stride = d_pad * 2 + 16, kt_stride = DQ_TILE * 2 + 16, v_stride = dv_pad * 2 + 16;
         const int qblk = 16 * DQ_QH * NWAVES;
-        size_t smem = (size_t)DQ_TILE * k_stride + (size_t)DMAX * kt_stride +
+        size_t smem = (size_t)DQ_TILE * k_stride + (size_t)(DMAX / 16) * (DQ_TILE * 16 + 8) * 2 +
                       (size_t)DQ_TILE * v_stride + (size_t)NWAVES * 16 * DQ_QH * kt_stride;
         int gx = (Nq + qblk - 1) / qblk, gy = B * H;
         int nsplit = 1;
@@ -633,8 +678,8 @@ void launch_flash_bwd(const torch::Tensor& dout, const torch::Tensor& q, const t
         const int q_stride = d_pad * 2 + 16, qt_stride = DKV_TILE * 2 + 16;
         const int do_stride = dv_pad * 2 + 16;
         const int kblk = 16 * DKV_QH * NWAVES;
-        size_t smem = (size_t)DKV_TILE * q_stride + (size_t)DMAX * qt_stride +
-                      (size_t)DKV_TILE * do_stride + (size_t)DVMAX * qt_stride +
+        size_t smem = (size_t)DKV_TILE * q_stride + (size_t)(DMAX / 16) * (DKV_TILE * 16 + 8) * 2 +
+                      (size_t)DKV_TILE * do_stride + (size_t)(DVMAX / 16) * (DKV_TILE * 16 + 8) * 2 +
                       (size_t)NWAVES * 16 * DKV_QH * qt_stride;
         dim3 grid((Lk + kblk - 1) / kblk, B * H);
         hipLaunchKernelGGL((flash_dkv_kernel<DMAX, DVMAX, DKV_TILE, DKV_QH>), grid, dim3(256), smem, stream,

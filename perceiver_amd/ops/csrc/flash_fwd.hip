@@ -71,6 +71,57 @@ DEVINL void stage_tile_rowmajor(const unsigned short* __restrict__ src, long src
 }
 
 typedef __attribute__((ext_vector_type(4))) short short4x;
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
+
+// ---- 16-column-subtiled image + hardware transpose read (guide T10) ----
+// Layout: channel-subtile s holds a (KVBLK x 16) row-major bf16 block with
+// 16-elem (32 B) rows, subtiles padded by 8 elems (16 B) so consecutive
+// subtiles shift LDS banks. ds_read_b64_tr_b16's gather (lane l elem j reads
+// base + (l&15) + j*16 + (l>>4)*64) turns two reads of this image into the
+// 32x16 MFMA B-fragment — V is staged ROW-major (one 16-B write per granule,
+// no 8-row write lockstep: that pattern measured ~11% of wave cycles in bank
+// conflicts) yet consumed column-major for P@V.
+constexpr int SUB_ELEMS = 64 * 16 + 8;  // KVBLK rows x 16 cols + 16-B pad
+
+DEVINL void stage_tile_sub16(const unsigned short* __restrict__ src, long src_stride,
+                             int rows_valid, int dv, int dv_pad,
+                             char* lds, int tid) {
+    const int gpr = dv_pad / 8;
+    const int total = KVBLK * gpr;
+    for (int g = tid; g < total; g += 256) {
+        int row = g / gpr;
+        int c0 = (g % gpr) * 8;
+        short8v val = {};
+        if (row < rows_valid && c0 < dv) {
+            if (c0 + 8 <= dv) {
+                val = *reinterpret_cast<const short8v*>(src + (long)row * src_stride + c0);
+            } else {
+#pragma unroll
+                for (int e = 0; e < 8; ++e)
+                    val[e] = (c0 + e < dv) ? (short)src[(long)row * src_stride + c0 + e] : (short)0;
+            }
+        }
+        *reinterpret_cast<short8v*>(
+            lds + ((c0 / 16) * SUB_ELEMS + row * 16 + (c0 % 16)) * 2) = val;
+    }
+}
+
+// B-fragment (rows key0..key0+31, cols 16*sub..) from the subtiled image.
+// Per-lane addressing: each 16-lane group loads one 4x16 row-major tile
+// linearly (lane supplies its own 8-B chunk at lo16*4 elems) and the
+// instruction's fixed cross-lane exchange delivers column lo16 to the lane;
+// group hi4 starts at row hi4*8, the second read covers rows +4.
+DEVINL bf16x8 read_bfrag_tr16(const char* lds, int sub, int key0, int hi4, int lo16) {
+    const __bf16* base = reinterpret_cast<const __bf16*>(lds) +
+                         sub * SUB_ELEMS + (key0 + hi4 * 8) * 16 + lo16 * 4;
+    auto p = (__attribute__((address_space(3))) bf16x4*)base;
+    bf16x4 lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p);
+    bf16x4 hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p + 16);  // +4 rows
+    bf16x8 out;
+#pragma unroll
+    for (int e = 0; e < 4; ++e) { out[e] = lo[e]; out[e + 4] = hi[e]; }
+    return out;
+}
 
 // Stage a (KVBLK x Dv) tile TRANSPOSED: ldsT row = channel, col = key. Each thread
 // iteration covers 4 keys x 8 channels: 4 coalesced 16-B global reads, 8 packed
@@ -155,8 +206,8 @@ __global__ void flash_fwd_kernel(            // budget allows (not the 352 templ
     const int k_stride = d_pad * 2 + 16;
     const int vt_stride = KVBLK * 2 + 16;
     char* k_lds = smem;                                   // KVBLK * k_stride
-    char* vt_lds = k_lds + KVBLK * k_stride;              // dv_pad * vt_stride
-    char* p_lds = vt_lds + DVMAX * vt_stride;             // per wave: QROWS * vt_stride
+    char* v16_lds = k_lds + KVBLK * k_stride;             // (dv_pad/16) * SUB_ELEMS elems
+    char* p_lds = v16_lds + (DVMAX / 16) * SUB_ELEMS * 2; // per wave: QROWS * vt_stride
     char* p_mine = p_lds + wave * QROWS * vt_stride;
 
     // ---- Q fragments: lane holds A[i=lo16][k = hi4*8 + e] per 32-wide k-block ----
@@ -215,7 +266,7 @@ __global__ void flash_fwd_kernel(            // budget allows (not the 352 templ
         int rows_valid = min(KVBLK, Lk - kv0);
         __syncthreads();
         stage_tile_rowmajor(kbase + (long)kv0 * ksn, ksn, rows_valid, D, d_pad, k_lds, k_stride, tid);
-        stage_tile_transposed(vbase + (long)kv0 * vsn, vsn, rows_valid, Dv, dv_pad, vt_lds, vt_stride, tid);
+        stage_tile_sub16(vbase + (long)kv0 * vsn, vsn, rows_valid, Dv, dv_pad, v16_lds, tid);
         __syncthreads();
 
         // ---- S = Q K^T (QH x 16 rows x KVBLK keys); one B read feeds QH MFMAs ----
@@ -311,14 +362,14 @@ __global__ void flash_fwd_kernel(            // budget allows (not the 352 templ
                     p_mine + (h * 16 + lo16) * vt_stride + (kb32 * 32 + hi4 * 8) * 2));
             }
 
-        // ---- O += P V : B[k][j] = V[key k][ch j] = vt_lds[ch j][key k] ----
+        // ---- O += P V : B[k=key][j=ch] via hardware transpose reads of the
+        // row-major subtiled V image ----
 #pragma unroll
         for (int cb = 0; cb < DVMAX / 16; ++cb) {
             if (cb < dv_blocks) {
 #pragma unroll
                 for (int kb32 = 0; kb32 < KEYBLKS / 2; ++kb32) {
-                    const char* src = vt_lds + (cb * 16 + lo16) * vt_stride + (kb32 * 32 + hi4 * 8) * 2;
-                    bf16x8 bfrag = (bf16x8)(*reinterpret_cast<const short8v*>(src));
+                    bf16x8 bfrag = read_bfrag_tr16(v16_lds, cb, kb32 * 32, hi4, lo16);
 #pragma unroll
                     for (int h = 0; h < QH; ++h) {
                         o_acc[h][cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -407,7 +458,7 @@ void launch_flash_fwd(const torch::Tensor& q, const torch::Tensor& k, const torc
     const int d_pad = (D + 31) & ~31;
     int k_stride = d_pad * 2 + 16;
     int vt_stride = KVBLK * 2 + 16;
-    size_t smem = (size_t)KVBLK * k_stride + (size_t)DVMAX * vt_stride +
+    size_t smem = (size_t)KVBLK * k_stride + (size_t)(DVMAX / 16) * SUB_ELEMS * 2 +
                   (size_t)NWAVES * QROWS * vt_stride;
     int gx = (Nq + QBLK - 1) / QBLK;
     int gy = B * H;
