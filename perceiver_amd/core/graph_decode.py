@@ -52,9 +52,18 @@ class GraphedDecoder:
         tokens = gd.decode(n)                                # n graph replays
     """
 
-    def __init__(self, model, caches: List[StaticKVCache], use_graph: bool = True):
+    def __init__(self, model, caches: List[StaticKVCache], use_graph: bool = True,
+                 do_sample: bool = False, temperature: float = 1.0,
+                 top_k: Optional[int] = None):
+        """``do_sample``: in-graph ancestral sampling (temperature scaling +
+        optional top-k truncation + exponential-race Gumbel trick). torch's CUDA
+        Philox state is graph-safe, so replays draw fresh randomness with zero
+        host round-trips, same as the greedy path."""
         self.model = model
         self.caches = caches
+        self.do_sample = do_sample
+        self.temperature = temperature
+        self.top_k = top_k
         p = next(model.parameters())
         self.device, self.dtype = p.device, p.dtype
         self.use_graph = use_graph and p.device.type == "cuda"
@@ -149,10 +158,22 @@ class GraphedDecoder:
             h = m.out_norm(h)
         logits = m.output_adapter(h, txt_embedding=m.input_adapter.txt_embedding)
 
-        # greedy + bookkeeping, all on-device so the graph is self-advancing
+        # token selection + bookkeeping, all on-device: the graph is self-advancing
         self.last_logits = logits  # graph-pool tensor: valid until the next replay
         self.out_buf.index_copy_(1, self.step_idx, self.tok)
-        self.tok.copy_(logits.argmax(-1))
+        if self.do_sample:
+            scores = logits[:, -1].float()
+            if self.temperature != 1.0:
+                scores = scores / self.temperature
+            if self.top_k is not None and self.top_k > 0:
+                kth = torch.topk(scores, min(self.top_k, scores.shape[-1]))[0][..., -1, None]
+                scores = scores.masked_fill(scores < kth, float("-inf"))
+            # Gumbel-max: argmax(scores + G) ~ softmax(scores); multinomial is
+            # not capture-safe in all torch builds, exponential_/log is
+            g = torch.empty_like(scores).exponential_().log().neg_()
+            self.tok.copy_((scores + g).argmax(-1, keepdim=True))
+        else:
+            self.tok.copy_(logits.argmax(-1))
         self.ca_len.add_(1)
         self.sa_len.add_(1)
         self.step_idx.add_(1)

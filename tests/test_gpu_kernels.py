@@ -340,3 +340,34 @@ def test_fused_adamw_in_step_clipping():
     for p1, p2 in zip(m1.parameters(), m2.parameters()):
         # clip order differs (flat fp32 copy vs per-param bf16): bf16-rounding-level tolerance
         torch.testing.assert_close(p1.float(), p2.float(), rtol=0, atol=4e-2)
+
+
+@pytest.mark.gpu
+def test_rccl_reducer_single_rank():
+    """RCCL (nccl backend) process-group init + the bucketed reducer on CUDA
+    tensors: world_size 1 makes all-reduce a no-op numerically but exercises the
+    exact code path the 8-GPU bench runs (init, bucket build, async all_reduce,
+    wait, unflatten)."""
+    import os
+
+    import torch.distributed as dist
+
+    from perceiver_amd.parallel import BucketedGradReducer
+
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29612")
+    os.environ.setdefault("RANK", "0")
+    os.environ.setdefault("WORLD_SIZE", "1")
+    dist.init_process_group("nccl", rank=0, world_size=1)
+    try:
+        dev = torch.device("cuda:0")
+        m = torch.nn.Sequential(torch.nn.Linear(64, 64), torch.nn.Linear(64, 8)).to(dev, torch.bfloat16)
+        reducer = BucketedGradReducer(m, bucket_cap_mb=0.01)
+        x = torch.randn(4, 64, device=dev, dtype=torch.bfloat16)
+        loss = m(x).float().square().mean()
+        loss.backward()
+        reducer.finalize()
+        for p in m.parameters():
+            assert p.grad is not None and torch.isfinite(p.grad.float()).all()
+    finally:
+        dist.destroy_process_group()

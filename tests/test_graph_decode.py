@@ -100,3 +100,52 @@ def test_graph_decode_gpu_matches_host():
     gd.prefill(prompt, prefix_len=prefix_len)
     again = gd.decode(8)
     assert torch.equal(again, graph_toks)
+
+
+def test_graph_sampling_topk1_equals_greedy():
+    torch.manual_seed(2)
+    model = CausalLanguageModel(_cfg(abs_pos_emb=False)).eval()
+    prompt = torch.randint(0, 61, (2, 30))
+    with torch.no_grad():
+        g1 = GraphedDecoder(model, allocate_kv_cache(model, 2), use_graph=False)
+        g1.prefill(prompt, prefix_len=29)
+        greedy = g1.decode(6)
+        g2 = GraphedDecoder(model, allocate_kv_cache(model, 2), use_graph=False,
+                            do_sample=True, top_k=1)
+        g2.prefill(prompt, prefix_len=29)
+        sampled = g2.decode(6)
+    assert torch.equal(greedy, sampled)
+
+
+def test_graph_sampling_produces_varied_tokens():
+    torch.manual_seed(3)
+    model = CausalLanguageModel(_cfg(abs_pos_emb=False)).eval()
+    prompt = torch.randint(0, 61, (2, 30))
+    gd = GraphedDecoder(model, allocate_kv_cache(model, 2), use_graph=False,
+                        do_sample=True, temperature=2.0)
+    with torch.no_grad():
+        gd.prefill(prompt, prefix_len=29)
+        a = gd.decode(8)
+        gd.prefill(prompt, prefix_len=29)
+        b = gd.decode(8)
+    assert a.shape == (2, 8)
+    assert (a >= 0).all() and (a < 61).all()
+    assert not torch.equal(a, b)  # fresh randomness per decode
+
+
+@pytest.mark.gpu
+def test_graph_sampling_gpu_captures():
+    torch.manual_seed(0)
+    dev = torch.device("cuda:0")
+    model = CausalLanguageModel(_cfg()).to(dev, torch.bfloat16).eval()
+    prompt = torch.randint(0, 61, (4, 30), device=dev)
+    gd = GraphedDecoder(model, allocate_kv_cache(model, 4, device=dev, dtype=torch.bfloat16),
+                        do_sample=True, temperature=1.5, top_k=8)
+    with torch.no_grad():
+        gd.prefill(prompt, prefix_len=29)
+        a = gd.decode(8)
+        gd.prefill(prompt, prefix_len=29)
+        b = gd.decode(8)
+    assert gd._graph is not None
+    assert a.shape == (4, 8) and (a >= 0).all() and (a < 61).all()
+    assert not torch.equal(a, b), "graph replays must draw fresh randomness"
