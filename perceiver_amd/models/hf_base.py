@@ -112,6 +112,13 @@ class PerceiverCausalSequenceModel(PreTrainedModel):
     """Wraps a CausalSequenceModel backend for 🤗 inference. Subclasses set
     ``self.backend_model`` in __init__."""
 
+    @classmethod
+    def can_generate(cls) -> bool:
+        # transformers gates generation plumbing (pipeline assistant fields,
+        # generation_config) on this; our generate() is native, not
+        # GenerationMixin, so the base-class heuristic misses it
+        return True
+
     def forward(
         self,
         input_ids: torch.LongTensor,
