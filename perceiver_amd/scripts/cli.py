@@ -124,7 +124,7 @@ class CLI:
 
         trainer_cfg = dict(cfg.get("trainer", {}))
         opt_cfg = dict(cfg.get("optimizer", {}))
-        for k in ("lr", "weight_decay", "warmup_steps", "lr_schedule", "min_lr_fraction"):
+        for k in ("lr", "weight_decay", "warmup_steps", "lr_schedule", "min_lr_fraction", "optimizer"):
             if k in opt_cfg:
                 trainer_cfg[k] = opt_cfg[k]
         ckpt_path = trainer_cfg.pop("ckpt_path", None)

@@ -167,3 +167,54 @@ def convert_to_bf16_training(model: torch.nn.Module) -> torch.nn.Module:
         if isinstance(mod, (FourierPositionEncoding, FrequencyPositionEncoding)):
             mod.float()
     return model
+
+
+class Lamb(torch.optim.Optimizer):
+    """LAMB (layer-wise adaptive moments, You et al. 2020) — the optimizer the
+    reference's MLM example trains with (its CLI exposes it via torch_optimizer,
+    reference scripts/cli.py:1-47; not installable offline, so implemented here).
+    AdamW-style moments with decoupled weight decay, update scaled per parameter
+    tensor by the trust ratio ||p|| / ||update||."""
+
+    def __init__(self, params, lr: float = 1e-3, betas=(0.9, 0.999), eps: float = 1e-6,
+                 weight_decay: float = 0.01, clamp_trust: float = 10.0):
+        defaults = dict(lr=lr, betas=betas, eps=eps, weight_decay=weight_decay,
+                        clamp_trust=clamp_trust)
+        super().__init__(params, defaults)
+
+    @torch.no_grad()
+    def step(self, closure=None):
+        loss = None
+        if closure is not None:
+            with torch.enable_grad():
+                loss = closure()
+        for group in self.param_groups:
+            beta1, beta2 = group["betas"]
+            for p in group["params"]:
+                if p.grad is None:
+                    continue
+                g = p.grad.float()
+                state = self.state[p]
+                if len(state) == 0:
+                    state["step"] = 0
+                    state["exp_avg"] = torch.zeros_like(g)
+                    state["exp_avg_sq"] = torch.zeros_like(g)
+                state["step"] += 1
+                t = state["step"]
+                m, v = state["exp_avg"], state["exp_avg_sq"]
+                m.mul_(beta1).add_(g, alpha=1 - beta1)
+                v.mul_(beta2).addcmul_(g, g, value=1 - beta2)
+                m_hat = m / (1 - beta1 ** t)
+                v_hat = v / (1 - beta2 ** t)
+                update = m_hat / (v_hat.sqrt() + group["eps"])
+                if group["weight_decay"] != 0:
+                    update = update + group["weight_decay"] * p.float()
+                p_norm = p.float().norm()
+                u_norm = update.norm()
+                trust = torch.where(
+                    (p_norm > 0) & (u_norm > 0),
+                    (p_norm / u_norm).clamp(max=group["clamp_trust"]),
+                    torch.ones_like(p_norm),
+                )
+                p.add_((update * (-group["lr"] * trust)).to(p.dtype))
+        return loss

@@ -42,6 +42,7 @@ class TrainConfig:
     lr: float = 2e-4
     weight_decay: float = 0.01
     betas: tuple = (0.9, 0.999)
+    optimizer: str = "adamw"         # "adamw" | "master_adamw" | "lamb"
     precision: str = "bf16"          # "bf16" (autocast on GPU) or "fp32"
     grad_clip: Optional[float] = None
     bucket_cap_mb: float = 50.0
@@ -78,10 +79,17 @@ class Trainer:
         return torch.autocast(self.device.type, dtype=torch.bfloat16, enabled=enabled)
 
     def _make_optimizer(self, model):
-        return torch.optim.AdamW(
-            [p for p in model.parameters() if p.requires_grad],
-            lr=self.cfg.lr, weight_decay=self.cfg.weight_decay, betas=self.cfg.betas, foreach=True,
-        )
+        params = [p for p in model.parameters() if p.requires_grad]
+        kw = dict(lr=self.cfg.lr, weight_decay=self.cfg.weight_decay, betas=self.cfg.betas)
+        if self.cfg.optimizer == "lamb":
+            from perceiver_amd.train.optim import Lamb
+
+            return Lamb(params, **kw)
+        if self.cfg.optimizer == "master_adamw":
+            from perceiver_amd.train.optim import MasterAdamW
+
+            return MasterAdamW(params, max_grad_norm=self.cfg.grad_clip or 0.0, **kw)
+        return torch.optim.AdamW(params, foreach=True, **kw)
 
     def _make_scheduler(self, opt):
         if self.cfg.lr_schedule == "cosine":
