@@ -60,6 +60,12 @@ class RotaryPositionEmbedding:
         else:
             pos_enc = self.frq_pos_enc[..., :seq_len, :]
 
+        if t.is_cuda and t.dtype == torch.bfloat16 and t.dim() == 4:
+            from perceiver_amd.ops import rotary
+
+            if rotary.can_use_fused(pos_enc):
+                return rotary.fused_rotate(t, pos_enc.squeeze(1), self.rotate_dim)
+
         t_rot, t_pass = t[..., : self.rotate_dim], t[..., self.rotate_dim :]
         # rotation in the encoding dtype (fp32 tables under bf16 training), result
         # cast back to t's dtype so the bf16 fused-kernel path stays bf16

@@ -19,6 +19,7 @@ bool flash_supported_impl(long d_qk, long d_v, long needs_dropout);
 void adamw_step(torch::Tensor master, torch::Tensor m, torch::Tensor v, torch::Tensor g,
                 double lr, double beta1, double beta2, double eps, double weight_decay,
                 int64_t step);
+torch::Tensor rotary_apply(torch::Tensor t, torch::Tensor frq, int64_t rot, bool neg_sin);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("gelu_bias_fwd", &gelu_bias_fwd, "fused bias+GELU forward (bf16)");
@@ -29,4 +30,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("ln_fwd", &ln_fwd, "fused bf16 LayerNorm forward");
     m.def("ln_bwd", &ln_bwd, "fused bf16 LayerNorm backward");
     m.def("adamw_step", &adamw_step, "single-pass fused AdamW on flat fp32 state");
+    m.def("rotary_apply", &rotary_apply, "fused rotary embedding (bf16, fwd/bwd via neg_sin)");
 }

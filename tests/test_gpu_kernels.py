@@ -371,3 +371,39 @@ def test_rccl_reducer_single_rank():
             assert p.grad is not None and torch.isfinite(p.grad.float()).all()
     finally:
         dist.destroy_process_group()
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("shape,rot,right", [
+    ((2, 8, 64, 128), 64, True),     # AR self-attn class (rotate half the head)
+    ((2, 8, 33, 128), 128, True),    # full-dim rotation, odd rows
+    ((1, 4, 16, 96), 32, False),
+])
+def test_fused_rotary_matches_host(shape, rot, right):
+    from perceiver_amd.core.position import FrequencyPositionEncoding, RotaryPositionEmbedding
+
+    torch.manual_seed(0)
+    b, h, n, d = shape
+    dev = torch.device("cuda:0")
+    frq_mod = FrequencyPositionEncoding(dim=rot).to(dev)
+    pos = torch.arange(n + 5, device=dev).unsqueeze(0).expand(b, n + 5)
+    emb = RotaryPositionEmbedding(frq_mod(pos), right_align=right)
+
+    t = torch.randn(b, h, n, d, device=dev, dtype=torch.bfloat16, requires_grad=True)
+    t_host = t.detach().clone().requires_grad_(True)
+
+    out = emb.rotate(t)  # dispatches to the fused kernel on bf16 CUDA input
+    # host reference: force the eager path by upcasting the table through cat
+    from perceiver_amd.core.position import rotate_half_interleaved
+    pos_enc = emb.frq_pos_enc[..., -n:, :] if right else emb.frq_pos_enc[..., :n, :]
+    tr, tp = t_host[..., :rot], t_host[..., rot:]
+    ref = torch.cat(((tr * pos_enc.cos() + rotate_half_interleaved(tr) * pos_enc.sin())
+                     .to(t.dtype), tp), dim=-1)
+
+    assert torch.allclose(out.float(), ref.float(), atol=2e-2, rtol=2e-2), \
+        (out.float() - ref.float()).abs().max().item()
+
+    g = torch.randn_like(out)
+    out.backward(g)
+    ref.backward(g)
+    assert torch.allclose(t.grad.float(), t_host.grad.float(), atol=2e-2, rtol=2e-2)
