@@ -29,7 +29,7 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=10)
     p.add_argument("--warmup", type=int, default=3)
-    p.add_argument("--model", default="mlm", choices=["mlm", "clm-decode", "img", "flow"])
+    p.add_argument("--model", default="mlm", choices=["mlm", "clm", "clm-decode", "img", "flow"])
     p.add_argument("--batch", type=int, default=0, help="per-GPU batch size (0 = model default)")
     p.add_argument("--device", default=None, help="override device (e.g. cpu for local testing)")
     p.add_argument("--tiny", action="store_true", help="tiny config for CPU plumbing tests")
@@ -145,6 +145,67 @@ class MLMBench:
     def config_json(self):
         return {"model": "perceiver-io-mlm-201M", "global_batch": None, "seq_len": self.seq,
                 "num_latents": 512, "parallelism": None}
+
+
+class CLMTrainBench:
+    """Perceiver-AR causal-LM training (the reference's WikiText-103 bytes
+    flagship: ctx 4096, 512 latents, 512 ch, 9 layers, CA-dropout 0.5,
+    examples/training/clm/train.sh): samples/s over ranks."""
+
+    unit = "samples/s"
+    metric = "train_samples_per_s_clm_wikitext4096"
+    scaling = "weak"
+    higher_is_better = True
+
+    def __init__(self, args, device, rank):
+        from perceiver_amd.models.flagship import clm_wikitext
+        from perceiver_amd.models.text.clm import CausalLanguageModel, CausalLanguageModelConfig
+
+        if args.tiny:
+            cfg = CausalLanguageModelConfig(vocab_size=262, max_seq_len=128, max_latents=32,
+                                            num_channels=64, num_heads=4,
+                                            num_self_attention_layers=2)
+            self.batch = args.batch or 2
+        else:
+            cfg = clm_wikitext()
+            self.batch = args.batch or 24
+
+        torch.manual_seed(args.seed + rank)
+        self.model = CausalLanguageModel(cfg).to(device)
+        self.device = device
+        self.cfg = cfg
+        if device.type == "cuda":
+            from perceiver_amd.train.optim import MasterAdamW, convert_to_bf16_training
+
+            self.model = convert_to_bf16_training(self.model)
+            self.opt = MasterAdamW(self.model.parameters(), lr=2e-4, weight_decay=0.01)
+        else:
+            self.opt = torch.optim.AdamW(self.model.parameters(), lr=2e-4, foreach=True)
+        self.model.train()
+        self.prefix_len = cfg.max_seq_len - cfg.max_latents
+        g = torch.Generator(device="cpu").manual_seed(args.seed + rank)
+        self.x = torch.randint(0, cfg.vocab_size, (self.batch, cfg.max_seq_len), generator=g).to(device)
+        self.labels = self.x[:, self.prefix_len:].contiguous()
+
+    wrap_ddp = MLMBench.wrap_ddp
+
+    def step(self):
+        out = self.model(self.x, prefix_len=self.prefix_len)
+        loss = F.cross_entropy(out.logits.flatten(0, 1).float(), self.labels.flatten())
+        self.opt.zero_grad(set_to_none=True)
+        loss.backward()
+        if getattr(self, "reducer", None) is not None:
+            self.reducer.finalize()
+        self.opt.step()
+        return loss
+
+    def items_per_step(self, world):
+        return self.batch * world
+
+    def config_json(self):
+        return {"model": "perceiver-ar-30.7M-wikitext", "global_batch": None,
+                "seq_len": self.cfg.max_seq_len, "max_latents": self.cfg.max_latents,
+                "parallelism": None}
 
 
 class CLMDecodeBench:
@@ -322,7 +383,7 @@ class FlowBench(ImgBench):
                 "image_shape": list(self.cfg.encoder.image_shape), "parallelism": None}
 
 
-BENCHES = {"mlm": MLMBench, "clm-decode": CLMDecodeBench, "img": ImgBench, "flow": FlowBench}
+BENCHES = {"mlm": MLMBench, "clm": CLMTrainBench, "clm-decode": CLMDecodeBench, "img": ImgBench, "flow": FlowBench}
 
 
 def main():
