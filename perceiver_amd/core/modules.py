@@ -395,6 +395,9 @@ def activation_checkpoint_wrapper(layer: AbstractAttentionLayer, offload_to_cpu:
     HBM this is rarely needed — kept for config parity and the very large KV configs).
 
     Wraps the layer's forward; kv_cache is not supported under checkpointing.
+    ``offload_to_cpu`` additionally parks the checkpoint's saved inputs in host
+    memory (torch.autograd.graph.save_on_cpu — the fairscale offload_to_cpu
+    equivalent).
     """
     import torch.utils.checkpoint as cp
 
@@ -414,7 +417,12 @@ def activation_checkpoint_wrapper(layer: AbstractAttentionLayer, offload_to_cpu:
                 out = self.module(*args, kv_cache=None, **kwargs)
                 return out.last_hidden_state
 
-            hidden = cp.checkpoint(run, *args, use_reentrant=False)
+            if offload_to_cpu:
+                pin = torch.cuda.is_available()
+                with torch.autograd.graph.save_on_cpu(pin_memory=pin):
+                    hidden = cp.checkpoint(run, *args, use_reentrant=False)
+            else:
+                hidden = cp.checkpoint(run, *args, use_reentrant=False)
             return ModuleOutput(last_hidden_state=hidden, kv_cache=None)
 
     return _Checkpointed(layer)

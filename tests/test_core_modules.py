@@ -167,3 +167,23 @@ def test_rotary_partial_rotation_passthrough():
     out = rot.rotate(t)
     assert torch.equal(out[..., 4:], t[..., 4:])
     assert not torch.allclose(out[..., :4], t[..., :4])
+
+
+def test_activation_checkpoint_offload_cpu():
+    """offload_to_cpu path: forward+backward through a checkpointed+offloaded
+    layer produces the same gradients as the plain layer."""
+    import copy
+
+    from perceiver_amd.core.modules import SelfAttentionLayer, activation_checkpoint_wrapper
+
+    torch.manual_seed(0)
+    layer = SelfAttentionLayer(num_heads=2, num_channels=16, widening_factor=1)
+    wrapped = activation_checkpoint_wrapper(copy.deepcopy(layer), offload_to_cpu=True)
+
+    x1 = torch.randn(2, 6, 16, requires_grad=True)
+    x2 = x1.detach().clone().requires_grad_(True)
+    layer(x1).last_hidden_state.sum().backward()
+    wrapped(x2).last_hidden_state.sum().backward()
+    torch.testing.assert_close(x1.grad, x2.grad)
+    for p1, p2 in zip(layer.parameters(), wrapped.parameters()):
+        torch.testing.assert_close(p1.grad, p2.grad)
