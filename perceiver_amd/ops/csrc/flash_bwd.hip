@@ -6,9 +6,10 @@
 //   dS      = P * (dO V^T * mask/(1-p) - delta_i)
 //   dQ      = dS K          dK = dS^T Q          dV = (P*mask/(1-p))^T dO
 //
-// v3: QH 16-row A-fragment groups per wave (one B-fragment LDS read feeds QH MFMAs),
-// vectorized transposed staging (4-source-row packed b64 writes), per-template TILE
-// sizes balancing LDS occupancy against the register budget.
+// v4: QH 16-row A-fragment groups per wave (one B-fragment LDS read feeds QH
+// MFMAs); transposed operands (K^T / Q^T / dO^T) live as 16-column subtiles
+// written row-major and consumed through ds_read_b64_tr_b16 hardware transpose
+// reads; per-template TILE sizes balance LDS occupancy vs the register budget.
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
 #include <cfloat>
@@ -109,41 +110,6 @@ DEVINL void stage_rm(const unsigned short* __restrict__ src, long src_stride,
             }
         }
         *reinterpret_cast<short8v*>(lds + row * ldst_bytes + c0 * 2) = val;
-    }
-}
-
-// stage (rows_tile x d) tile TRANSPOSED: lds row = channel (d_pad rows), col =
-// source row; 4-source-row packed b64 writes.
-template <int ROWS_TILE>
-DEVINL void stage_tr(const unsigned short* __restrict__ src, long src_stride,
-                     int rows_valid, int d, int d_pad,
-                     char* ldsT, int ldst_bytes, int tid) {
-    const int gpr = d_pad / 8;
-    const int total = (ROWS_TILE / 4) * gpr;
-    for (int g = tid; g < total; g += 256) {
-        int row0 = (g / gpr) * 4;
-        int c0 = (g % gpr) * 8;
-        short8v rows[4];
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-            short8v val = {};
-            int row = row0 + r;
-            if (row < rows_valid && c0 < d) {
-                if (c0 + 8 <= d) {
-                    val = *reinterpret_cast<const short8v*>(src + (long)row * src_stride + c0);
-                } else {
-#pragma unroll
-                    for (int e = 0; e < 8; ++e)
-                        val[e] = (c0 + e < d) ? (short)src[(long)row * src_stride + c0 + e] : (short)0;
-                }
-            }
-            rows[r] = val;
-        }
-#pragma unroll
-        for (int e = 0; e < 8; ++e) {
-            short4x pack = {rows[0][e], rows[1][e], rows[2][e], rows[3][e]};
-            *reinterpret_cast<short4x*>(ldsT + (c0 + e) * ldst_bytes + row0 * 2) = pack;
-        }
     }
 }
 

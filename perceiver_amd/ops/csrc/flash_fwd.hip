@@ -12,14 +12,15 @@
 // j > Lk - Nq + i (right-aligned), optional in-kernel dropout on the probabilities
 // (counter-hash RNG regenerated in the backward).
 //
-// Structure (v2):
-//   workgroup = 4 waves x 64 lanes; each wave owns 16 q rows (64 rows/workgroup);
-//   KVBLK=64-key K/V tiles cooperatively staged in LDS — K row-major (contiguous
-//   b128 fragment reads), V transposed channels x keys with 4-key-packed b64
-//   writes; +16 B row padding keeps 16-lane b128 groups bank-conflict-free;
-//   QK^T and PV on mfma_f32_16x16x32_bf16 with fp32 accumulation; online softmax
-//   with cross-lane shfl_xor row reductions; P redistributed C-layout -> A-layout
-//   through a per-wave LDS buffer. Outputs O and logsumexp (for the backward).
+// Structure (v3):
+//   workgroup = 4 waves x 64 lanes; each wave owns QH*16 q rows; KVBLK=64-key
+//   K/V tiles cooperatively staged in LDS — K row-major (contiguous b128
+//   fragment reads, +16 B row padding for conflict-free 16-lane groups), V as
+//   16-column subtiles written row-major and consumed via ds_read_b64_tr_b16
+//   hardware transpose reads; QK^T and PV on mfma_f32_16x16x32_bf16 with fp32
+//   accumulation; online softmax (rescale-skip) with cross-lane shfl_xor row
+//   reductions; P redistributed C-layout -> A-layout through a per-wave LDS
+//   buffer; KV-split over grid.z with logsumexp merge. Outputs O + logsumexp.
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
 #include <cfloat>
@@ -121,42 +122,6 @@ DEVINL bf16x8 read_bfrag_tr16(const char* lds, int sub, int key0, int hi4, int l
 #pragma unroll
     for (int e = 0; e < 4; ++e) { out[e] = lo[e]; out[e + 4] = hi[e]; }
     return out;
-}
-
-// Stage a (KVBLK x Dv) tile TRANSPOSED: ldsT row = channel, col = key. Each thread
-// iteration covers 4 keys x 8 channels: 4 coalesced 16-B global reads, 8 packed
-// 8-B LDS writes (keys k..k+3 of one channel).
-DEVINL void stage_tile_transposed(const unsigned short* __restrict__ src, long src_stride,
-                                  int rows_valid, int dv, int dv_pad,
-                                  char* ldsT, int ldst_bytes, int tid) {
-    const int gpr = dv_pad / 8;           // 8-channel granules
-    const int total = (KVBLK / 4) * gpr;
-    for (int g = tid; g < total; g += 256) {
-        int key0 = (g / gpr) * 4;
-        int c0 = (g % gpr) * 8;
-        short8v rows[4];
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-            short8v val = {};
-            int key = key0 + r;
-            if (key < rows_valid && c0 < dv) {
-                if (c0 + 8 <= dv) {
-                    val = *reinterpret_cast<const short8v*>(src + (long)key * src_stride + c0);
-                } else {
-#pragma unroll
-                    for (int e = 0; e < 8; ++e) {
-                        val[e] = (c0 + e < dv) ? (short)src[(long)key * src_stride + c0 + e] : (short)0;
-                    }
-                }
-            }
-            rows[r] = val;
-        }
-#pragma unroll
-        for (int e = 0; e < 8; ++e) {
-            short4x pack = {rows[0][e], rows[1][e], rows[2][e], rows[3][e]};
-            *reinterpret_cast<short4x*>(ldsT + (c0 + e) * ldst_bytes + key0 * 2) = pack;
-        }
-    }
 }
 
 // QH: 16-row A-fragments per wave (2 doubles MFMA work per B-fragment LDS read;
