@@ -407,3 +407,46 @@ def test_fused_rotary_matches_host(shape, rot, right):
     out.backward(g)
     ref.backward(g)
     assert torch.allclose(t.grad.float(), t_host.grad.float(), atol=2e-2, rtol=2e-2)
+
+
+@pytest.mark.gpu
+def test_training_convergence_soak():
+    """End-to-end learning check on GPU: a mid-size MLM (pure bf16 + fused
+    master-AdamW + flash/LN/GELU kernels) must memorize a fixed synthetic
+    mapping — loss falling well below its starting point proves the backward
+    kernels produce usable gradients, not just allclose-passing ones."""
+    from perceiver_amd.models.text.common import TextEncoderConfig
+    from perceiver_amd.models.text.mlm import (
+        MaskedLanguageModel,
+        MaskedLanguageModelConfig,
+        TextDecoderConfig,
+    )
+    from perceiver_amd.train.optim import MasterAdamW, convert_to_bf16_training
+
+    torch.manual_seed(0)
+    dev = torch.device("cuda:0")
+    cfg = MaskedLanguageModelConfig(
+        encoder=TextEncoderConfig(vocab_size=262, max_seq_len=512, num_input_channels=256,
+                                  num_cross_attention_heads=8, num_self_attention_heads=8,
+                                  num_self_attention_layers_per_block=4, dropout=0.0),
+        decoder=TextDecoderConfig(vocab_size=262, max_seq_len=512, num_cross_attention_heads=8,
+                                  dropout=0.0),
+        num_latents=64, num_latent_channels=256,
+    )
+    model = convert_to_bf16_training(MaskedLanguageModel(cfg).to(dev)).train()
+    opt = MasterAdamW(model.parameters(), lr=1e-3, weight_decay=0.0)
+
+    x = torch.randint(6, 262, (8, 512), device=dev)
+    pad = torch.zeros(8, 512, dtype=torch.bool, device=dev)
+    labels = torch.randint(6, 262, (8, 512), device=dev)  # fixed -> memorizable
+
+    losses = []
+    for _ in range(60):
+        logits = model(x, pad)
+        loss = torch.nn.functional.cross_entropy(logits.flatten(0, 1).float(), labels.flatten())
+        opt.zero_grad(set_to_none=True)
+        loss.backward()
+        opt.step()
+        losses.append(float(loss))
+    assert all(torch.isfinite(torch.tensor(losses))), losses[-5:]
+    assert losses[-1] < losses[0] * 0.5, (losses[0], losses[-1])
