@@ -434,14 +434,14 @@ def test_training_convergence_soak():
         num_latents=64, num_latent_channels=256,
     )
     model = convert_to_bf16_training(MaskedLanguageModel(cfg).to(dev)).train()
-    opt = MasterAdamW(model.parameters(), lr=1e-3, weight_decay=0.0)
+    opt = MasterAdamW(model.parameters(), lr=3e-3, weight_decay=0.0)
 
     x = torch.randint(6, 262, (8, 512), device=dev)
     pad = torch.zeros(8, 512, dtype=torch.bool, device=dev)
     labels = torch.randint(6, 262, (8, 512), device=dev)  # fixed -> memorizable
 
     losses = []
-    for _ in range(60):
+    for _ in range(150):
         logits = model(x, pad)
         loss = torch.nn.functional.cross_entropy(logits.flatten(0, 1).float(), labels.flatten())
         opt.zero_grad(set_to_none=True)
