@@ -41,33 +41,15 @@ def parse_args():
     return p.parse_args()
 
 
-def load_tunableop(rank: int):
-    """Arm PyTorch TunableOp with the pre-tuned GEMM-algorithm table
-    (tuned/tunableop_gfx950.csv, produced by a one-off PYTORCH_TUNABLEOP_TUNING
-    run over every bench shape on MI355X; +6-7%% on the MLM flagship). TunableOp
-    appends the rank before the file extension, so each rank gets a private
-    copy of the canonical table. No-op if the user already set the env or the
-    table is missing; uncovered shapes fall back to hipBLASLt heuristics."""
-    import shutil
-
-    canonical = os.path.join(os.path.dirname(os.path.abspath(__file__)),
-                             "tuned", "tunableop_gfx950.csv")
-    if "PYTORCH_TUNABLEOP_ENABLED" in os.environ or not os.path.exists(canonical):
-        return
-    base = f"/tmp/tunableop_{os.getpid()}.csv"
-    shutil.copy(canonical, f"/tmp/tunableop_{os.getpid()}{rank}.csv")
-    os.environ["PYTORCH_TUNABLEOP_ENABLED"] = "1"
-    os.environ["PYTORCH_TUNABLEOP_TUNING"] = "0"
-    os.environ["PYTORCH_TUNABLEOP_FILENAME"] = base
-
-
 def setup_dist(args):
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     use_cuda = torch.cuda.is_available() if args.device is None else args.device.startswith("cuda")
     if use_cuda:
-        load_tunableop(rank)
+        from perceiver_amd.utils.tunableop import arm_tunableop
+
+        arm_tunableop(rank)
     if world_size > 1:
         backend = "nccl" if use_cuda else "gloo"
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")

@@ -61,6 +61,10 @@ class TrainConfig:
 class Trainer:
     def __init__(self, config: TrainConfig):
         self.cfg = config
+        if torch.cuda.is_available():
+            from perceiver_amd.utils.tunableop import arm_tunableop
+
+            arm_tunableop(int(os.environ.get("RANK", "0")))
         self.local_rank = init_distributed_from_env()
         self.device = torch.device("cuda", self.local_rank) if torch.cuda.is_available() else torch.device("cpu")
         self.global_step = 0
