@@ -22,7 +22,10 @@ def arm_tunableop(rank: int = 0) -> bool:
     if "PYTORCH_TUNABLEOP_ENABLED" in os.environ or not os.path.exists(canonical):
         return False
     base = f"/tmp/tunableop_{os.getpid()}.csv"
-    shutil.copy(canonical, f"/tmp/tunableop_{os.getpid()}{rank}.csv")
+    # TunableOp inserts an ordinal (device index) before the extension when it
+    # reads/writes; cover every possible ordinal on an 8-GPU node
+    for r in range(8):
+        shutil.copy(canonical, f"/tmp/tunableop_{os.getpid()}{r}.csv")
     os.environ["PYTORCH_TUNABLEOP_ENABLED"] = "1"
     os.environ["PYTORCH_TUNABLEOP_TUNING"] = "0"
     os.environ["PYTORCH_TUNABLEOP_FILENAME"] = base
