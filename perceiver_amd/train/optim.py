@@ -85,6 +85,41 @@ class MasterAdamW(torch.optim.Optimizer):
                              group["weight_decay"], fl["step"])
         torch._foreach_copy_(fl["params"], fl["views_master"])
 
+    def state_dict(self):
+        """Include the fused-path flat state (master weights, moments, step) —
+        it lives outside ``self.state`` and would otherwise silently reset on
+        resume (losing both the Adam moments and the fp32 master precision)."""
+        sd = super().state_dict()
+        if self._flat:
+            fused = {}
+            for gi, group in enumerate(self.param_groups):
+                fl = self._flat.get(id(group))
+                if fl is not None:
+                    fused[gi] = {"master": fl["master"], "m": fl["m"], "v": fl["v"],
+                                 "step": fl["step"]}
+            sd["fused_flat"] = fused
+        return sd
+
+    def load_state_dict(self, state_dict):
+        state_dict = dict(state_dict)
+        fused = state_dict.pop("fused_flat", None)
+        super().load_state_dict(state_dict)
+        if fused:
+            self._flat = {}
+            for gi, group in enumerate(self.param_groups):
+                rec = fused.get(gi, fused.get(str(gi)))
+                if rec is None:
+                    continue
+                fl = self._build_flat(group)
+                fl["master"].copy_(rec["master"].to(fl["master"].device))
+                fl["m"].copy_(rec["m"].to(fl["m"].device))
+                fl["v"].copy_(rec["v"].to(fl["v"].device))
+                fl["step"] = int(rec["step"])
+                # restore the live parameters from the fp32 masters
+                with torch.no_grad():
+                    torch._foreach_copy_(fl["params"], fl["views_master"])
+                self._flat[id(group)] = fl
+
     @torch.no_grad()
     def step(self, closure=None):
         loss = None
