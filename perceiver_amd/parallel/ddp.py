@@ -47,6 +47,11 @@ class BucketedGradReducer:
     Gradients are averaged over the world size. Parameters that do not require
     grad are skipped. Works on any torch.distributed backend (RCCL on GPU,
     gloo in CPU tests).
+
+    Gradient accumulation: call ``set_sync(False)`` for non-boundary
+    micro-batches (hooks become no-ops and gradients just accumulate locally),
+    ``set_sync(True)`` before the last micro-batch's backward; the boundary
+    backward then reduces the full accumulated gradients as usual.
     """
 
     def __init__(self, module: torch.nn.Module, bucket_cap_mb: float = 50.0,
@@ -55,6 +60,7 @@ class BucketedGradReducer:
             raise RuntimeError("torch.distributed must be initialized before BucketedGradReducer")
         self.group = process_group
         self.world_size = dist.get_world_size(process_group)
+        self.sync = True
 
         # unique params in reverse registration order (approximate backward order);
         # shared/tied params appear once
@@ -88,7 +94,13 @@ class BucketedGradReducer:
             p.register_post_accumulate_grad_hook(self._on_grad_ready) for p in params
         ]
 
+    def set_sync(self, sync: bool):
+        """False = accumulation micro-batch (no reduction); True = boundary."""
+        self.sync = sync
+
     def _on_grad_ready(self, p: torch.nn.Parameter):
+        if not self.sync:
+            return
         b, off = self._param_bucket[id(p)]
         b.flat[off: off + p.numel()].copy_(p.grad.detach().reshape(-1))
         b.ready += 1

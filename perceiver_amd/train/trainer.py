@@ -45,6 +45,7 @@ class TrainConfig:
     optimizer: str = "adamw"         # "adamw" | "master_adamw" | "lamb"
     precision: str = "bf16"          # "bf16" (autocast on GPU) or "fp32"
     grad_clip: Optional[float] = None
+    accumulate_grad_batches: int = 1
     bucket_cap_mb: float = 50.0
     lr_schedule: str = "constant"    # "constant" | "cosine" | "none"
     warmup_steps: int = 0
@@ -131,12 +132,22 @@ class Trainer:
         if get_world_size() > 1:
             reducer = BucketedGradReducer(model, bucket_cap_mb=self.cfg.bucket_cap_mb)
 
+        accum = max(1, self.cfg.accumulate_grad_batches)
+        micro = 0
         for batch in batches:
             batch = _move(batch, self.device)
+            if micro == 0:
+                optimizer.zero_grad(set_to_none=True)
+            boundary = (micro + 1) == accum
+            if reducer is not None:
+                reducer.set_sync(boundary)
             with self._autocast():
                 loss = step_fn(model, batch)
-            optimizer.zero_grad(set_to_none=True)
-            loss.backward()
+            (loss / accum).backward()
+            micro += 1
+            if not boundary:
+                continue
+            micro = 0
             if reducer is not None:
                 reducer.finalize()
             if self.cfg.grad_clip and not getattr(optimizer, "max_grad_norm", 0.0):
@@ -185,12 +196,22 @@ class Trainer:
             sampler = getattr(train_loader, "sampler", None)
             if hasattr(sampler, "set_epoch"):
                 sampler.set_epoch(self.epoch)
+            accum = max(1, self.cfg.accumulate_grad_batches)
+            micro = 0
             for batch in train_loader:
                 batch = _move(batch, self.device)
+                if micro == 0:
+                    optimizer.zero_grad(set_to_none=True)
+                boundary = (micro + 1) == accum
+                if reducer is not None:
+                    reducer.set_sync(boundary)
                 with self._autocast():
                     loss = task.training_step(batch, self.global_step)
-                optimizer.zero_grad(set_to_none=True)
-                loss.backward()
+                (loss / accum).backward()
+                micro += 1
+                if not boundary:
+                    continue
+                micro = 0
                 if reducer is not None:
                     reducer.finalize()
                 if self.cfg.grad_clip and not getattr(optimizer, "max_grad_norm", 0.0):

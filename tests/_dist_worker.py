@@ -91,7 +91,32 @@ def fsdp_worker(rank, world, out_dir):
                os.path.join(out_dir, f"rank{rank}_fsdp.pt"))
 
 
-MODES = {"reducer": reducer_worker, "trainer": trainer_worker, "fsdp": fsdp_worker}
+def accum_worker(rank, world, out_dir):
+    """Gradient accumulation under the bucketed reducer: 2 ranks x 2 micro
+    batches; the boundary reduction must yield grads equal to the mean over all
+    4 micro-batch gradients (checked against a single-process reference by the
+    test)."""
+    from perceiver_amd.parallel import BucketedGradReducer
+
+    model = make_clm_model()
+    reducer = BucketedGradReducer(model, bucket_cap_mb=0.05)
+
+    for micro in range(2):
+        torch.manual_seed(300 + rank * 2 + micro)  # distinct data per micro-batch
+        x = torch.randint(0, 50, (2, 16))
+        boundary = micro == 1
+        reducer.set_sync(boundary)
+        out = model(x, prefix_len=8)
+        loss = F.cross_entropy(out.logits.flatten(0, 1), x[:, 8:].flatten()) / 2
+        loss.backward()
+    reducer.finalize()
+
+    grads = {n: p.grad.clone() for n, p in model.named_parameters() if p.grad is not None}
+    torch.save(grads, os.path.join(out_dir, f"rank{rank}_accum.pt"))
+
+
+MODES = {"reducer": reducer_worker, "trainer": trainer_worker, "fsdp": fsdp_worker,
+         "accum": accum_worker}
 
 if __name__ == "__main__":
     name, rank, world, port, out_dir = sys.argv[1:6]

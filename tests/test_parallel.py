@@ -87,3 +87,28 @@ def test_fsdp_two_rank_cpu(tmp_path):
     assert a["losses"] == b["losses"]
     for k in a["state_dict"]:
         assert torch.equal(a["state_dict"][k], b["state_dict"][k]), k
+
+
+def test_accumulation_with_reducer_matches_full_mean(tmp_path):
+    """2 ranks x 2 accumulation micro-batches == mean gradient over all 4
+    micro-batches computed in one process."""
+    run_dist("accum", world=2, port=29519, out_dir=str(tmp_path))
+
+    sys.path.insert(0, HERE)
+    from _dist_worker import make_clm_model
+
+    model = make_clm_model()
+    for seed in range(300, 304):
+        torch.manual_seed(seed)
+        x = torch.randint(0, 50, (2, 16))
+        out = model(x, prefix_len=8)
+        # mean over 4 micro-batches == sum of (loss/2 per rank)/2 ranks
+        (F.cross_entropy(out.logits.flatten(0, 1), x[:, 8:].flatten()) / 4).backward()
+    ref = {n: p.grad for n, p in model.named_parameters() if p.grad is not None}
+
+    for r in range(2):
+        got = torch.load(tmp_path / f"rank{r}_accum.pt", weights_only=False)
+        assert set(got) == set(ref)
+        for n in ref:
+            torch.testing.assert_close(got[n], ref[n], rtol=1e-5, atol=1e-6,
+                                       msg=lambda m: f"{n}: {m}")

@@ -211,3 +211,31 @@ def test_lamb_optimizer_trains_and_trust_scales():
     update = m_hat / (g.abs() + 1e-6)
     trust = (p0.norm() / update.norm()).clamp(max=10.0)
     torch.testing.assert_close(p.detach(), p0 - 0.1 * trust * update, rtol=1e-4, atol=1e-5)
+
+
+def test_gradient_accumulation_matches_large_batch():
+    """accumulate_grad_batches=2 over half-batches must take the same optimizer
+    steps as the full batch (equal micro sizes => mean-of-means == full mean)."""
+    import copy
+
+    from perceiver_amd.train.trainer import Trainer, TrainConfig
+
+    torch.manual_seed(0)
+    model_a = torch.nn.Sequential(torch.nn.Linear(8, 16), torch.nn.Linear(16, 4))
+    model_b = copy.deepcopy(model_a)
+    xs = [torch.randn(4, 8) for _ in range(4)]
+
+    def step_fn(model, batch):
+        return model(batch).square().mean()
+
+    ta = Trainer(TrainConfig(max_steps=2, accumulate_grad_batches=2, log_every=100,
+                             lr=1e-2, lr_schedule="none", out_dir="logs/_acc_a"))
+    ta.fit_steps(model_a, xs, step_fn)
+
+    tb = Trainer(TrainConfig(max_steps=2, log_every=100, lr=1e-2, lr_schedule="none",
+                             out_dir="logs/_acc_b"))
+    tb.fit_steps(model_b, [torch.cat(xs[:2]), torch.cat(xs[2:])], step_fn)
+
+    assert ta.global_step == tb.global_step == 2
+    for p1, p2 in zip(model_a.parameters(), model_b.parameters()):
+        torch.testing.assert_close(p1, p2, rtol=1e-5, atol=1e-6)
