@@ -10,6 +10,7 @@ python -m perceiver_amd.scripts.text.clm fit \
   --data.batch_size 24 \
   --data.max_seq_len 4096 \
   --data.random_train_shift true \
+  --trainer.accumulate_grad_batches 2 \
   --optimizer.lr 2e-4 \
   --trainer.grad_clip 0.5 \
   --trainer.max_epochs 12 \
