@@ -227,6 +227,11 @@ __global__ void flash_dq_kernel(
         stage_sub16<TILE>(kbase + (long)kv0 * ksn, ksn, rows_valid, D, d_pad, kt16_lds, tid);
         stage_rm<TILE>(vbase + (long)kv0 * vsn, vsn, rows_valid, Dv, dv_pad, v_lds, v_stride, tid);
         __syncthreads();
+        // wave-uniform mask hoist (see flash_fwd.hip): interior tiles of the
+        // no-pad case skip the per-element mask chain entirely
+        const bool tile_masked =
+            (kv0 + TILE > Lk) || (padrow != nullptr) ||
+            (causal && kv0 + TILE - 1 > Lk - Nq + q0);
 
         // t-outer: per 16-key block compute S and dP with short-lived accumulators,
         // convert to dS and spill to the per-wave LDS buffer immediately
@@ -266,8 +271,9 @@ __global__ void flash_dq_kernel(
                 for (int r = 0; r < 4; ++r) {
                     int qi = q0 + h * 16 + hi4 * 4 + r;
                     int j = kv0 + t * 16 + lo16;
-                    bool masked = j >= Lk || (padrow && j < Lk && padrow[j]) ||
-                                  (causal && j > Lk - Nq + qi);
+                    bool masked = tile_masked &&
+                                  (j >= Lk || (padrow && j < Lk && padrow[j]) ||
+                                   (causal && j > Lk - Nq + qi));
                     float p = masked ? 0.f : expf(s_acc[h][r] - lse_r[h][r]);
                     float dprobs = dp_acc[h][r];
                     if (drop_p > 0.f) {
@@ -419,6 +425,12 @@ __global__ void flash_dkv_kernel(
             key_pad[h][r] = (ki2 >= Lk) || (padrow && padrow[min(ki2, Lk - 1)]);
         }
     }
+    bool any_keypad = false;
+#pragma unroll
+    for (int h = 0; h < QH; ++h)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) any_keypad |= key_pad[h][r];
+    any_keypad = __any(any_keypad);
 
     float4v dk_acc[QH][DMAX / 16];
     float4v dv_acc[QH][DVMAX / 16];
@@ -446,6 +458,11 @@ __global__ void flash_dkv_kernel(
         stage_rm<TILE>(dobase + (long)qt0 * Dv, Dv, rows_valid, Dv, dv_pad, do_lds, do_stride, tid);
         stage_sub16<TILE>(dobase + (long)qt0 * Dv, Dv, rows_valid, Dv, dv_pad, do16_lds, tid);
         __syncthreads();
+        // wave-uniform mask hoist: interior q-tiles with no padded/overhang
+        // keys skip the per-element mask chain (kernels are VALU/wait-bound)
+        const bool tile_masked =
+            any_keypad || (qt0 + TILE > Nq) ||
+            (causal && k0 + KROWS - 1 > Lk - Nq + qt0);
 
         // t-outer: per 16-q-row block compute S^T and dP^T with short-lived
         // accumulators; P^T goes to LDS now, dS^T is kept in a small register array
@@ -486,7 +503,8 @@ __global__ void flash_dkv_kernel(
                 for (int r = 0; r < 4; ++r) {
                     int ki = k0 + h * 16 + hi4 * 4 + r;
                     int qi = qt0 + t * 16 + lo16;
-                    bool masked = key_pad[h][r] || qi >= Nq || (causal && ki > Lk - Nq + qi);
+                    bool masked = tile_masked &&
+                                  (key_pad[h][r] || qi >= Nq || (causal && ki > Lk - Nq + qi));
                     float lse_i = (qi < Nq) ? lse_row[qi] : 0.f;
                     float delta_i = (qi < Nq) ? delta_row[qi] : 0.f;
                     float p = masked ? 0.f : expf(st_acc[h][r] - lse_i);

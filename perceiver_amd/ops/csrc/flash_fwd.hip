@@ -258,7 +258,13 @@ __global__ void flash_fwd_kernel(            // budget allows (not the 352 templ
         }
 
         // ---- mask + online softmax per 16-row group (P computed in place in
-        // s_acc, dropped, and written straight to the per-wave LDS buffer) ----
+        // s_acc, dropped, and written straight to the per-wave LDS buffer).
+        // Wave-uniform hoist: interior tiles of the common no-pad case need no
+        // per-element masking at all (the mask chain is ~1/4 of the softmax
+        // VALU work — the kernels measure VALU/wait-bound, profiles/) ----
+        const bool tile_masked =
+            (kv0 + KVBLK > Lk) || (padrow != nullptr) ||
+            (causal && kv0 + KVBLK - 1 > Lk - Nq + q0);
 #pragma unroll
         for (int h = 0; h < QH; ++h) {
             float rowmax[4];
@@ -266,16 +272,21 @@ __global__ void flash_fwd_kernel(            // budget allows (not the 352 templ
             for (int r = 0; r < 4; ++r) {
                 int qi = q0 + h * 16 + hi4 * 4 + r;
                 float mx = -FLT_MAX;
+                if (tile_masked) {
 #pragma unroll
-                for (int kb = 0; kb < KEYBLKS; ++kb) {
-                    int j = kv0 + kb * 16 + lo16;
-                    float sv = s_acc[h][kb][r];
-                    bool masked = j >= Lk;
-                    if (padrow && j < Lk) masked |= padrow[j];
-                    if (causal && j > Lk - Nq + qi) masked = true;
-                    sv = masked ? -FLT_MAX : sv;
-                    s_acc[h][kb][r] = sv;
-                    mx = fmaxf(mx, sv);
+                    for (int kb = 0; kb < KEYBLKS; ++kb) {
+                        int j = kv0 + kb * 16 + lo16;
+                        float sv = s_acc[h][kb][r];
+                        bool masked = j >= Lk;
+                        if (padrow && j < Lk) masked |= padrow[j];
+                        if (causal && j > Lk - Nq + qi) masked = true;
+                        sv = masked ? -FLT_MAX : sv;
+                        s_acc[h][kb][r] = sv;
+                        mx = fmaxf(mx, sv);
+                    }
+                } else {
+#pragma unroll
+                    for (int kb = 0; kb < KEYBLKS; ++kb) mx = fmaxf(mx, s_acc[h][kb][r]);
                 }
                 rowmax[r] = warp16_max(mx);
             }
