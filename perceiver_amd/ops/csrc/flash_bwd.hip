@@ -274,7 +274,7 @@ __global__ void flash_dq_kernel(
                     bool masked = tile_masked &&
                                   (j >= Lk || (padrow && j < Lk && padrow[j]) ||
                                    (causal && j > Lk - Nq + qi));
-                    float p = masked ? 0.f : exp2f(s_acc[h][r] - lse_r[h][r]);
+                    float p = masked ? 0.f : expf(s_acc[h][r] - lse_r[h][r]);
                     float dprobs = dp_acc[h][r];
                     if (drop_p > 0.f) {
                         bool kept = rng_hash(drop_seed, bh, qi, j) >= drop_thresh;
@@ -507,7 +507,7 @@ __global__ void flash_dkv_kernel(
                                   (key_pad[h][r] || qi >= Nq || (causal && ki > Lk - Nq + qi));
                     float lse_i = (qi < Nq) ? lse_row[qi] : 0.f;
                     float delta_i = (qi < Nq) ? delta_row[qi] : 0.f;
-                    float p = masked ? 0.f : exp2f(st_acc[h][r] - lse_i);
+                    float p = masked ? 0.f : expf(st_acc[h][r] - lse_i);
                     float p_eff = p;
                     float dprobs = dpt_acc[h][r];
                     if (drop_p > 0.f) {
@@ -592,8 +592,7 @@ __global__ void flash_dkv_kernel(
 #pragma unroll
             for (int cb = 0; cb < DMAX / 16; ++cb) {
                 int c = cb * 16 + lo16;
-                // q is log2e-pre-scaled (exp2-domain softmax): unscale dK
-                if (c < D) dkrow[c] = f2bf(dk_acc[h][cb][r] * 0.6931471805599453f);
+                if (c < D) dkrow[c] = f2bf(dk_acc[h][cb][r]);
             }
 #pragma unroll
             for (int cb = 0; cb < DVMAX / 16; ++cb) {
@@ -690,10 +689,6 @@ std::vector<torch::Tensor> flash_bwd(torch::Tensor dout, torch::Tensor q, torch:
                                      c10::optional<torch::Tensor> pad_mask, bool causal,
                                      double dropout_p, int64_t seed) {
     TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
-    // exp2-domain recompute (see flash_fwd): q scaled by log2(e) to match the
-    // forward's scores/lse; dq comes out in natural units directly, dK is
-    // unscaled by ln2 in the dkv epilogue.
-    q = q * 1.4426950408889634;
     dout = dout.contiguous();
     out = out.contiguous();
     if (q.stride(3) != 1) q = q.contiguous();

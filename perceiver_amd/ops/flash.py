@@ -24,9 +24,7 @@ class FlashAttention(torch.autograd.Function):
     def forward(ctx, q, k, v, pad_mask: Optional[torch.Tensor], causal: bool,
                 dropout_p: float, training: bool):
         # last-dim contiguity is enough (the kernels take batch/head/seq strides);
-        # head-transposed views and preallocated cache buffers pass zero-copy.
-        # The extension entry scales q by log2(e) and runs the whole online
-        # softmax in the exp2 domain (the hardware transcendental IS exp2).
+        # head-transposed views and preallocated cache buffers pass zero-copy
         q, k, v = _to_bf16(q), _to_bf16(k), _to_bf16(v)
         p = float(dropout_p) if training else 0.0
         seed = int(torch.randint(0, 2**62, (1,)).item()) if p > 0 else 0
