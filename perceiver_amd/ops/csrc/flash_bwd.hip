@@ -274,7 +274,7 @@ __global__ void flash_dq_kernel(
                     bool masked = tile_masked &&
                                   (j >= Lk || (padrow && j < Lk && padrow[j]) ||
                                    (causal && j > Lk - Nq + qi));
-                    float p = masked ? 0.f : expf(s_acc[h][r] - lse_r[h][r]);
+                    float p = masked ? 0.f : __expf(s_acc[h][r] - lse_r[h][r]);
                     float dprobs = dp_acc[h][r];
                     if (drop_p > 0.f) {
                         bool kept = rng_hash(drop_seed, bh, qi, j) >= drop_thresh;
@@ -507,7 +507,7 @@ __global__ void flash_dkv_kernel(
                                   (key_pad[h][r] || qi >= Nq || (causal && ki > Lk - Nq + qi));
                     float lse_i = (qi < Nq) ? lse_row[qi] : 0.f;
                     float delta_i = (qi < Nq) ? delta_row[qi] : 0.f;
-                    float p = masked ? 0.f : expf(st_acc[h][r] - lse_i);
+                    float p = masked ? 0.f : __expf(st_acc[h][r] - lse_i);
                     float p_eff = p;
                     float dprobs = dpt_acc[h][r];
                     if (drop_p > 0.f) {

@@ -300,7 +300,7 @@ __global__ void flash_fwd_kernel(            // budget allows (not the 352 templ
 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
                     float m_new = fmaxf(m_run[h][r], rowmax[r]);
-                    float alpha = expf(m_run[h][r] - m_new);
+                    float alpha = __expf(m_run[h][r] - m_new);
                     m_run[h][r] = m_new;
                     l_run[h][r] *= alpha;
 #pragma unroll
@@ -313,7 +313,7 @@ __global__ void flash_fwd_kernel(            // budget allows (not the 352 templ
                 float psum = 0.f;
 #pragma unroll
                 for (int kb = 0; kb < KEYBLKS; ++kb) {
-                    float pv = expf(s_acc[h][kb][r] - m_new);
+                    float pv = __expf(s_acc[h][kb][r] - m_new);
                     psum += pv;
                     if (drop_p > 0.f) {
                         int qi = q0 + h * 16 + hi4 * 4 + r;
@@ -406,7 +406,7 @@ __global__ void flash_merge_kernel(const float* __restrict__ o_part,
     for (int i = 0; i < nsplit; ++i) mx = fmaxf(mx, lse_part[(long)i * rows + row]);
     float wsum = 0.f;
     for (int i = lane; i < nsplit; i += 64) {
-        float wv = expf(lse_part[(long)i * rows + row] - mx);
+        float wv = __expf(lse_part[(long)i * rows + row] - mx);
         wbuf[wv_id][i] = wv;
     }
     __builtin_amdgcn_s_waitcnt(0);
