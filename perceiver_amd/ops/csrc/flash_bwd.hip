@@ -89,6 +89,36 @@ DEVINL bf16x8 read_bfrag_tr16(const char* lds, int sub, int row0, int hi4, int l
     return out;
 }
 
+// stage (rows_tile x d) tile into BOTH the row-major image (row stride
+// ldst_bytes) and the 16-col subtiled image: one global read per granule,
+// two LDS writes (replaces two separate staging passes = half the global
+// load instructions on the q/do/k operands)
+template <int ROWS_TILE>
+DEVINL void stage_rm_sub16(const unsigned short* __restrict__ src, long src_stride,
+                           int rows_valid, int d, int d_pad,
+                           char* lds_rm, int ldst_bytes, char* lds16, int tid) {
+    constexpr int SUBE = ROWS_TILE * 16 + 8;
+    const int gpr = d_pad / 8;
+    const int total = ROWS_TILE * gpr;
+    for (int g = tid; g < total; g += 256) {
+        int row = g / gpr;
+        int c0 = (g % gpr) * 8;
+        short8v val = {};
+        if (row < rows_valid && c0 < d) {
+            if (c0 + 8 <= d) {
+                val = *reinterpret_cast<const short8v*>(src + (long)row * src_stride + c0);
+            } else {
+#pragma unroll
+                for (int e = 0; e < 8; ++e)
+                    val[e] = (c0 + e < d) ? (short)src[(long)row * src_stride + c0 + e] : (short)0;
+            }
+        }
+        *reinterpret_cast<short8v*>(lds_rm + row * ldst_bytes + c0 * 2) = val;
+        *reinterpret_cast<short8v*>(
+            lds16 + ((c0 / 16) * SUBE + row * 16 + (c0 % 16)) * 2) = val;
+    }
+}
+
 // stage (rows_tile x d) tile row-major into LDS (row stride ldst_bytes), zero-pad
 template <int ROWS_TILE>
 DEVINL void stage_rm(const unsigned short* __restrict__ src, long src_stride,
@@ -223,8 +253,8 @@ __global__ void flash_dq_kernel(
     for (int kv0 = kv_begin; kv0 < kv_end; kv0 += TILE) {
         int rows_valid = min(TILE, Lk - kv0);
         __syncthreads();
-        stage_rm<TILE>(kbase + (long)kv0 * ksn, ksn, rows_valid, D, d_pad, k_lds, k_stride, tid);
-        stage_sub16<TILE>(kbase + (long)kv0 * ksn, ksn, rows_valid, D, d_pad, kt16_lds, tid);
+        stage_rm_sub16<TILE>(kbase + (long)kv0 * ksn, ksn, rows_valid, D, d_pad,
+                             k_lds, k_stride, kt16_lds, tid);
         stage_rm<TILE>(vbase + (long)kv0 * vsn, vsn, rows_valid, Dv, dv_pad, v_lds, v_stride, tid);
         __syncthreads();
         // wave-uniform mask hoist (see flash_fwd.hip): interior tiles of the
@@ -285,7 +315,7 @@ __global__ void flash_dq_kernel(
                         p_mine + (h * 16 + hi4 * 4 + r) * kt_stride + (t * 16 + lo16) * 2) = f2bf(ds);
                 }
         }
-        __builtin_amdgcn_s_waitcnt(0);
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
         bf16x8 ds_frag[QH][TBLKS / 2];
 #pragma unroll
         for (int h = 0; h < QH; ++h)
@@ -453,10 +483,10 @@ __global__ void flash_dkv_kernel(
     for (int qt0 = q_start; qt0 < Nq; qt0 += TILE) {
         int rows_valid = min(TILE, Nq - qt0);
         __syncthreads();
-        stage_rm<TILE>(qbase + (long)qt0 * qsn, qsn, rows_valid, D, d_pad, q_lds, q_stride, tid);
-        stage_sub16<TILE>(qbase + (long)qt0 * qsn, qsn, rows_valid, D, d_pad, q16_lds, tid);
-        stage_rm<TILE>(dobase + (long)qt0 * Dv, Dv, rows_valid, Dv, dv_pad, do_lds, do_stride, tid);
-        stage_sub16<TILE>(dobase + (long)qt0 * Dv, Dv, rows_valid, Dv, dv_pad, do16_lds, tid);
+        stage_rm_sub16<TILE>(qbase + (long)qt0 * qsn, qsn, rows_valid, D, d_pad,
+                             q_lds, q_stride, q16_lds, tid);
+        stage_rm_sub16<TILE>(dobase + (long)qt0 * Dv, Dv, rows_valid, Dv, dv_pad,
+                             do_lds, do_stride, do16_lds, tid);
         __syncthreads();
         // wave-uniform mask hoist: interior q-tiles with no padded/overhang
         // keys skip the per-element mask chain (kernels are VALU/wait-bound)
@@ -521,7 +551,7 @@ __global__ void flash_dkv_kernel(
                         p_mine + (h * 16 + hi4 * 4 + r) * qt_stride + (t * 16 + lo16) * 2) = f2bf(p_eff);
                 }
         }
-        __builtin_amdgcn_s_waitcnt(0);
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
         bf16x8 pt_frag[QH][TBLKS / 2];
 #pragma unroll
         for (int h = 0; h < QH; ++h)
@@ -555,7 +585,7 @@ __global__ void flash_dkv_kernel(
                     *reinterpret_cast<unsigned short*>(
                         p_mine + (h * 16 + hi4 * 4 + r) * qt_stride + (t * 16 + lo16) * 2) =
                         f2bf(ds_keep[h][t][r]);
-        __builtin_amdgcn_s_waitcnt(0);
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
         bf16x8 dst_frag[QH][TBLKS / 2];
 #pragma unroll
         for (int h = 0; h < QH; ++h)
