@@ -560,6 +560,19 @@ __global__ void flash_dkv_kernel(
                 pt_frag[h][t32] = (bf16x8)(*reinterpret_cast<const short8v*>(
                     p_mine + (h * 16 + lo16) * qt_stride + (t32 * 32 + hi4 * 8) * 2));
 
+        // second pass: dS^T through the same per-wave buffer — issued BEFORE the
+        // dV MFMAs (the P^T fragments are already in registers), so the whole
+        // write+read roundtrip latency hides under the dV matrix work
+#pragma unroll
+        for (int h = 0; h < QH; ++h)
+#pragma unroll
+            for (int t = 0; t < TBLKS; ++t)
+#pragma unroll
+                for (int r = 0; r < 4; ++r)
+                    *reinterpret_cast<unsigned short*>(
+                        p_mine + (h * 16 + hi4 * 4 + r) * qt_stride + (t * 16 + lo16) * 2) =
+                        f2bf(ds_keep[h][t][r]);
+
         // dV += P^T dO : B[k=qrow][j=ch] via transpose reads of the subtiled image
 #pragma unroll
         for (int cb = 0; cb < DVMAX / 16; ++cb) {
@@ -574,17 +587,6 @@ __global__ void flash_dkv_kernel(
                 }
             }
         }
-
-        // second pass: dS^T through the same per-wave buffer (wave-local)
-#pragma unroll
-        for (int h = 0; h < QH; ++h)
-#pragma unroll
-            for (int t = 0; t < TBLKS; ++t)
-#pragma unroll
-                for (int r = 0; r < 4; ++r)
-                    *reinterpret_cast<unsigned short*>(
-                        p_mine + (h * 16 + hi4 * 4 + r) * qt_stride + (t * 16 + lo16) * 2) =
-                        f2bf(ds_keep[h][t][r]);
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
         bf16x8 dst_frag[QH][TBLKS / 2];
 #pragma unroll
