@@ -29,4 +29,4 @@ try:
 except ImportError:  # metadata-only build
     ext_modules, cmdclass = [], {}
 
-setup(ext_modules=ext_modules, cmdclass=cmdclass)
+setup(name="perceiver-mi355x", ext_modules=ext_modules, cmdclass=cmdclass)
