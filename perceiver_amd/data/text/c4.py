@@ -36,8 +36,14 @@ class C4DataModule:
         rank: Optional[int] = None,
         world_size: Optional[int] = None,
     ):
-        from transformers import AutoTokenizer
+        from transformers import AutoTokenizer, PreTrainedTokenizerBase
 
+        if isinstance(tokenizer, PreTrainedTokenizerBase):
+            # instance accepted for offline tests, mirroring resolve_tokenizer
+            self._tokenizer_obj = tokenizer
+            tokenizer = type(tokenizer).__name__
+        else:
+            self._tokenizer_obj = None
         self.hparams = Hparams(
             tokenizer=tokenizer, max_seq_len=max_seq_len, min_seq_len=min_seq_len,
             batch_size=batch_size, shuffle_window_seed=shuffle_window_seed,
@@ -45,7 +51,8 @@ class C4DataModule:
             num_train_workers=num_train_workers, num_valid_workers=num_valid_workers,
             padding_side=padding_side, pin_memory=pin_memory, rank=rank, world_size=world_size,
         )
-        self.tokenizer = AutoTokenizer.from_pretrained(tokenizer, verbose=False)
+        self.tokenizer = (self._tokenizer_obj if self._tokenizer_obj is not None
+                          else AutoTokenizer.from_pretrained(tokenizer, verbose=False))
         self.collator = C4Collator(self.tokenizer)
         if padding_side is not None:
             self.tokenizer.padding_side = padding_side
