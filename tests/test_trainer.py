@@ -239,3 +239,33 @@ def test_gradient_accumulation_matches_large_batch():
     assert ta.global_step == tb.global_step == 2
     for p1, p2 in zip(model_a.parameters(), model_b.parameters()):
         torch.testing.assert_close(p1, p2, rtol=1e-5, atol=1e-6)
+
+
+def test_cli_validate_restores_checkpoint(tmp_path):
+    """CLI `validate` with --trainer.ckpt_path restores weights and reports
+    metrics (the reference's img_clf/valid.sh flow)."""
+    from perceiver_amd.scripts.cli import CLI
+    from perceiver_amd.scripts.vision.image_classifier import DEFAULTS, build_model, link
+
+    common = [
+        "--model.num_latents", "16",
+        "--model.num_latent_channels", "32",
+        "--model.encoder.num_frequency_bands", "4",
+        "--model.encoder.num_self_attention_layers_per_block", "1",
+        "--model.encoder.num_self_attention_blocks", "1",
+        "--model.encoder.num_cross_attention_heads", "1",
+        "--model.decoder.num_output_query_channels", "16",
+        "--trainer.out_dir", str(tmp_path / "run"),
+    ]
+    cli = CLI(LitImageClassifier, SyntheticMNISTDataModule, DEFAULTS, build_model, link,
+              argv=["fit", "--trainer.max_steps", "2"] + common, run=False)
+    cli.config["data"] = {}
+    cli.run()
+    ckpt = tmp_path / "run" / "checkpoints" / "last.ckpt"
+    assert ckpt.exists()
+
+    cli2 = CLI(LitImageClassifier, SyntheticMNISTDataModule, DEFAULTS, build_model, link,
+               argv=["validate", "--trainer.ckpt_path", str(ckpt)] + common, run=False)
+    cli2.config["data"] = {}
+    trainer = cli2.run()
+    assert trainer.global_step == 2  # restored from the checkpoint
