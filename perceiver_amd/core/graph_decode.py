@@ -103,7 +103,7 @@ class GraphedDecoder:
             # step then never touches cached rows again (no O(cache) re-rotate)
             c.pre_rotated = True
         out = self.model(prompt, prefix_len=prefix_len, kv_cache=self.caches)
-        tok = out.logits[:, -1:].argmax(-1)
+        tok = self._select(out.logits[:, -1, :].float())
 
         n0 = prompt.shape[1]
         prefix0 = prefix_len
@@ -117,6 +117,20 @@ class GraphedDecoder:
         for c in self.caches[1:]:
             c.enable_graph_append(self.sa_len)
         return tok
+
+    def _select(self, scores: torch.Tensor) -> torch.Tensor:
+        """Greedy or Gumbel-max sampled token from (batch, vocab) fp32 scores.
+        Capture-safe: torch's CUDA Philox state is graph-aware, and the
+        exponential_/log Gumbel trick avoids multinomial's sync."""
+        if not self.do_sample:
+            return scores.argmax(-1, keepdim=True)
+        if self.temperature != 1.0:
+            scores = scores / self.temperature
+        if self.top_k is not None and self.top_k > 0:
+            kth = torch.topk(scores, min(self.top_k, scores.shape[-1]))[0][..., -1, None]
+            scores = scores.masked_fill(scores < kth, float("-inf"))
+        g = torch.empty_like(scores).exponential_().log().neg_()
+        return (scores + g).argmax(-1, keepdim=True)
 
     # --------------------------------------------------------------------- step
 
@@ -161,19 +175,7 @@ class GraphedDecoder:
         # token selection + bookkeeping, all on-device: the graph is self-advancing
         self.last_logits = logits  # graph-pool tensor: valid until the next replay
         self.out_buf.index_copy_(1, self.step_idx, self.tok)
-        if self.do_sample:
-            scores = logits[:, -1].float()
-            if self.temperature != 1.0:
-                scores = scores / self.temperature
-            if self.top_k is not None and self.top_k > 0:
-                kth = torch.topk(scores, min(self.top_k, scores.shape[-1]))[0][..., -1, None]
-                scores = scores.masked_fill(scores < kth, float("-inf"))
-            # Gumbel-max: argmax(scores + G) ~ softmax(scores); multinomial is
-            # not capture-safe in all torch builds, exponential_/log is
-            g = torch.empty_like(scores).exponential_().log().neg_()
-            self.tok.copy_((scores + g).argmax(-1, keepdim=True))
-        else:
-            self.tok.copy_(logits.argmax(-1))
+        self.tok.copy_(self._select(logits[:, -1, :].float()))
         self.ca_len.add_(1)
         self.sa_len.add_(1)
         self.step_idx.add_(1)

@@ -289,6 +289,24 @@ class PerceiverCausalSequenceModel(PreTrainedModel):
             past = allocate_kv_cache(self.backend_model, input_ids.shape[0],
                                      device=p.device, dtype=p.dtype)
 
+            # hipGraph fast path: when every generated token stays within the
+            # latent-growth phase of the schedule (no prefix growth/window
+            # slide), no EOS cut-off is requested, and the sampling mode is
+            # graph-expressible, decode is one captured-graph replay per token
+            if (p.device.type == "cuda" and eos_token_id is None
+                    and not bool((attention_mask == 0).any())
+                    and top_p is None and generator is None and max_new_tokens >= 3
+                    and (seq_len - prefix_len) + max_new_tokens <= self.backend_model.max_latents
+                    and seq_len + max_new_tokens <= self.backend_model.max_seq_len):
+                from perceiver_amd.core.graph_decode import GraphedDecoder
+
+                gd = GraphedDecoder(self.backend_model, past, do_sample=do_sample,
+                                    temperature=temperature, top_k=top_k)
+                gd.prefill(input_ids, prefix_len=prefix_len)
+                first = gd.tok.clone()  # token emitted by the prefill pass
+                rest = gd.decode(max_new_tokens - 1)
+                return torch.cat([input_ids, first, rest], dim=1)
+
         for _ in range(max_new_tokens):
             model_inputs = self.prepare_inputs_for_generation(
                 input_ids, past_key_values=past, attention_mask=attention_mask,

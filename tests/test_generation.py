@@ -254,3 +254,27 @@ def test_generation_strategies_gpu():
     sampled = m.generate(input_ids=ids, num_latents=4, max_new_tokens=8,
                          do_sample=True, top_p=0.9)
     assert sampled.shape == (2, 18) and (sampled < VOCAB).all()
+
+
+@pytest.mark.gpu
+def test_generate_graph_fast_path_matches_host_loop():
+    """Eligible greedy generate() calls route through the hipGraph decoder; a
+    never-matching eos_token_id forces the host loop as reference."""
+    torch.manual_seed(0)
+    dev = torch.device("cuda:0")
+    m = PerceiverCausalLanguageModel(PerceiverCausalLanguageModelConfig(
+        CausalLanguageModelConfig(vocab_size=VOCAB, max_seq_len=SEQ, max_latents=LAT,
+                                  num_channels=32, num_heads=4,
+                                  num_self_attention_layers=2,
+                                  cross_attention_dropout=0.0))).to(dev, torch.bfloat16).eval()
+    ids = _prompt(3, 10, seed=9).to(dev)
+
+    graph = m.generate(input_ids=ids, num_latents=2, max_new_tokens=6)
+    host = m.generate(input_ids=ids, num_latents=2, max_new_tokens=6,
+                      eos_token_id=VOCAB + 7)  # unreachable: disables the graph path
+    assert graph.shape == host.shape == (3, 16)
+    assert (graph == host).float().mean() >= 0.75
+
+    sampled = m.generate(input_ids=ids, num_latents=2, max_new_tokens=6,
+                         do_sample=True, temperature=1.3, top_k=8)
+    assert sampled.shape == (3, 16) and (sampled < VOCAB).all()
