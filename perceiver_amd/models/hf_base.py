@@ -300,7 +300,12 @@ class PerceiverCausalSequenceModel(PreTrainedModel):
                     and seq_len + max_new_tokens <= self.backend_model.max_seq_len):
                 from perceiver_amd.core.graph_decode import GraphedDecoder
 
-                key = (input_ids.shape[0], do_sample, float(temperature), top_k)
+                # p.data_ptr() pins the cached graphs to the current parameter
+                # storage: model.to(device/dtype) reallocates storage, and a
+                # captured graph replaying against the old pointers would
+                # silently use stale weights (in-place load_state_dict is fine)
+                key = (input_ids.shape[0], do_sample, float(temperature), top_k,
+                       p.device, p.dtype, p.data_ptr())
                 cache = getattr(self, "_graph_decoders", None)
                 if cache is None:
                     cache = self._graph_decoders = {}
@@ -510,9 +515,13 @@ class PerceiverCausalSequenceModel(PreTrainedModel):
             kth = torch.topk(logits, min(top_k, logits.shape[-1]))[0][..., -1, None]
             logits = logits.masked_fill(logits < kth, float("-inf"))
         if top_p is not None and 0 < top_p < 1:
+            # standard nucleus rule: keep the smallest prefix of the sorted
+            # distribution whose mass exceeds top_p (shifted cumsum so the
+            # token that crosses the threshold is still kept)
             sorted_logits, sorted_idx = torch.sort(logits, descending=True)
-            probs = sorted_logits.softmax(-1).cumsum(-1)
-            remove = probs - probs.gather(-1, torch.zeros_like(sorted_idx[..., :1])) > top_p
+            cum = sorted_logits.softmax(-1).cumsum(-1)
+            remove = cum > top_p
+            remove[..., 1:] = remove[..., :-1].clone()
             remove[..., 0] = False
             scatter_mask = remove.scatter(-1, sorted_idx, remove)
             logits = logits.masked_fill(scatter_mask, float("-inf"))

@@ -77,8 +77,12 @@ class BucketedGradReducer:
         cur_bytes = 0
         for p in params:
             sz = p.numel() * p.element_size()
-            if cur and cur_bytes + sz > cap:
-                self.buckets.append(_Bucket(cur, p.device, p.dtype))
+            # close the bucket on capacity overflow and on any dtype/device
+            # change: a bucket's flat buffer is uniform, so mixing would
+            # silently cast (and all-reduce) grads in the wrong dtype/device
+            if cur and (cur_bytes + sz > cap
+                        or p.dtype != cur[0].dtype or p.device != cur[0].device):
+                self.buckets.append(_Bucket(cur, cur[0].device, cur[0].dtype))
                 cur, cur_bytes = [], 0
             cur.append(p)
             cur_bytes += sz
@@ -119,6 +123,27 @@ class BucketedGradReducer:
             b.work.wait()
             for p, off in zip(b.params, b.offsets):
                 p.grad.detach().reshape(-1).copy_(b.flat[off: off + p.numel()])
+            b.reset()
+
+    def reduce_now(self):
+        """Synchronously all-reduce whatever is in ``p.grad`` right now.
+
+        Used for a trailing partial accumulation window at epoch end, where
+        the post-accumulate hooks ran with sync=False and never filled the
+        buckets. Rare path — plain bucket-at-a-time, no overlap.
+        """
+        for b in self.buckets:
+            for p, off in zip(b.params, b.offsets):
+                g = p.grad
+                if g is None:
+                    b.flat[off: off + p.numel()].zero_()
+                else:
+                    b.flat[off: off + p.numel()].copy_(g.detach().reshape(-1))
+            b.flat.div_(self.world_size)
+            dist.all_reduce(b.flat, group=self.group)
+            for p, off in zip(b.params, b.offsets):
+                if p.grad is not None:
+                    p.grad.detach().reshape(-1).copy_(b.flat[off: off + p.numel()])
             b.reset()
 
     def remove(self):

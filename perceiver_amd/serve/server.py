@@ -15,7 +15,7 @@ import threading
 from typing import Optional
 
 import torch
-from fastapi import FastAPI
+from fastapi import FastAPI, HTTPException
 from pydantic import BaseModel, Field
 
 
@@ -54,13 +54,22 @@ def create_app(model, tokenizer) -> FastAPI:
     def generate(req: GenerateRequest):
         enc = tokenizer(req.prompt, return_tensors="pt", add_special_tokens=False)
         ids = enc["input_ids"].to(device)
+        if ids.shape[1] == 0:
+            raise HTTPException(status_code=400, detail="prompt tokenized to 0 tokens")
+        backend = model.backend_model
+        # keep the prompt inside the model's context window (last tokens win)
+        if ids.shape[1] >= backend.max_seq_len:
+            ids = ids[:, -(backend.max_seq_len - 1):]
         n_prompt = ids.shape[1]
-        max_new = min(req.max_new_tokens,
-                      model.backend_model.max_latents - 1)
+        # clamp latents into the valid range: enough that the prefix fits
+        # max_prefix_len, at most the prompt length / max_latents
+        lo = max(1, n_prompt - backend.max_prefix_len)
+        num_latents = max(lo, min(req.num_latents, n_prompt, backend.max_latents))
+        max_new = min(req.max_new_tokens, backend.max_latents - 1)
         with lock, torch.no_grad():
             out = model.generate(
                 input_ids=ids,
-                num_latents=min(req.num_latents, n_prompt),
+                num_latents=num_latents,
                 max_new_tokens=max_new,
                 do_sample=req.do_sample,
                 temperature=req.temperature,

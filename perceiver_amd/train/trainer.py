@@ -163,6 +163,17 @@ class Trainer:
             if self.cfg.max_steps and self.global_step >= self.cfg.max_steps:
                 break
 
+        if micro > 0 and not (self.cfg.max_steps and self.global_step >= self.cfg.max_steps):
+            # flush a trailing partial accumulation window (see fit())
+            if reducer is not None:
+                reducer.reduce_now()
+            if self.cfg.grad_clip and not getattr(optimizer, "max_grad_norm", 0.0):
+                torch.nn.utils.clip_grad_norm_(model.parameters(), self.cfg.grad_clip)
+            optimizer.step()
+            if scheduler is not None:
+                scheduler.step()
+            self.global_step += 1
+
         if reducer is not None:
             reducer.remove()
         return model
@@ -231,6 +242,17 @@ class Trainer:
                 if self.cfg.max_steps and self.global_step >= self.cfg.max_steps:
                     done = True
                     break
+            if micro > 0 and not done:
+                # trailing partial accumulation window at epoch end: its grads
+                # are real — flush them instead of silently discarding
+                if reducer is not None:
+                    reducer.reduce_now()
+                if self.cfg.grad_clip and not getattr(optimizer, "max_grad_norm", 0.0):
+                    torch.nn.utils.clip_grad_norm_(task.parameters(), self.cfg.grad_clip)
+                optimizer.step()
+                if scheduler is not None:
+                    scheduler.step()
+                self.global_step += 1
             self.epoch += 1
             if not done and val_loader is not None:
                 self._validate(task, val_loader, optimizer, scheduler)

@@ -278,3 +278,16 @@ def test_generate_graph_fast_path_matches_host_loop():
     sampled = m.generate(input_ids=ids, num_latents=2, max_new_tokens=6,
                          do_sample=True, temperature=1.3, top_k=8)
     assert sampled.shape == (3, 16) and (sampled < VOCAB).all()
+
+
+def test_top_p_nucleus_keeps_smallest_covering_prefix():
+    # probs [0.5, 0.3, 0.2] with top_p=0.6: standard nucleus sampling keeps
+    # tokens {0, 1} (the smallest prefix whose mass exceeds 0.6) and never
+    # token 2 (the old rule widened the distribution)
+    from perceiver_amd.models.hf_base import PerceiverCausalSequenceModel
+    probs = torch.tensor([[0.5, 0.3, 0.2]])
+    logits = probs.log()
+    g = torch.Generator().manual_seed(0)
+    draws = [int(PerceiverCausalSequenceModel._select_next(
+        logits.clone(), True, 1.0, None, 0.6, g)) for _ in range(200)]
+    assert set(draws) == {0, 1}

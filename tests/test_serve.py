@@ -58,3 +58,17 @@ def test_generate_clamps_max_new_tokens(client):
                                        "num_latents": 1})
     assert r.status_code == 200
     assert r.json()["generated_tokens"] <= 31  # max_latents - 1
+
+
+def test_generate_overlong_prompt_truncated(client):
+    # prompt longer than max_seq_len (128): server truncates to the tail
+    # instead of letting generate() raise (unhandled 500)
+    r = client.post("/generate", json={"prompt": "z" * 300, "max_new_tokens": 4,
+                                       "num_latents": 8})
+    assert r.status_code == 200
+    assert r.json()["generated_tokens"] == 4
+
+
+def test_generate_empty_prompt_is_400(client):
+    r = client.post("/generate", json={"prompt": "", "max_new_tokens": 4})
+    assert r.status_code == 400

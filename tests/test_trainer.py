@@ -269,3 +269,21 @@ def test_cli_validate_restores_checkpoint(tmp_path):
     cli2.config["data"] = {}
     trainer = cli2.run()
     assert trainer.global_step == 2  # restored from the checkpoint
+
+
+def test_trailing_partial_accumulation_window_is_flushed():
+    """3 micro-batches with accumulate_grad_batches=2: the odd trailing batch
+    must still produce an optimizer step (it was discarded)."""
+    from perceiver_amd.train.trainer import Trainer, TrainConfig
+
+    torch.manual_seed(0)
+    model = torch.nn.Linear(8, 4)
+    before = model.weight.detach().clone()
+    xs = [torch.randn(4, 8) for _ in range(3)]
+
+    t = Trainer(TrainConfig(accumulate_grad_batches=2, log_every=100, lr=1e-2,
+                            lr_schedule="none", out_dir="logs/_acc_tail"))
+    t.fit_steps(model, xs, lambda m, b: m(b).square().mean())
+
+    assert t.global_step == 2  # one full window + the flushed tail
+    assert not torch.equal(model.weight.detach(), before)
