@@ -68,7 +68,28 @@ def can_use_flash(q, k, v, dropout_p: float = 0.0, training: bool = False) -> bo
     )
     if not bf16_ok:
         return False
-    return bool(ext.flash_supported(d_qk, d_v, int(training and dropout_p > 0)))
+    ok = bool(ext.flash_supported(d_qk, d_v, int(training and dropout_p > 0)))
+    if not ok:
+        _warn_fallback_once(d_qk, d_v)
+    return ok
+
+
+_warned_shapes: set = set()
+
+
+def _warn_fallback_once(d_qk: int, d_v: int) -> None:
+    """One warning per head-dim combination when a CUDA-device attention call
+    silently drops to the eager composition — so a new config never loses the
+    fused kernels without a trace in the logs."""
+    key = (d_qk, d_v)
+    if key not in _warned_shapes:
+        _warned_shapes.add(key)
+        import warnings
+        warnings.warn(
+            f"flash kernel shape gate: head dims qk={d_qk}, v={d_v} unsupported; "
+            "attention falls back to the eager composition for this shape",
+            stacklevel=3,
+        )
 
 
 def flash_attention(q, k, v, pad_mask=None, causal: bool = False,

@@ -71,3 +71,17 @@ def test_mlm_conversion_param_count_matches():
     n_src = sum(p.numel() for p in src.parameters())
     n_tgt = sum(p.numel() for p in tgt.parameters())
     assert n_src == n_tgt, (n_src, n_tgt)
+
+
+def test_flagship_mlm_param_count_is_201_108_230():
+    """Flagship-scale architecture parity: at the reference CLI defaults
+    (256 latents x 1280ch, 26 SA layers, qk 256 / v 1280, vocab 262, seq 2048)
+    the model must have exactly the deepmind/language-perceiver parameter count
+    the reference pins (tests/masked_language_model_convert_test.py:12).
+    Built on the meta device — no 800 MB allocation."""
+    from perceiver_amd.models.flagship import mlm_flagship
+    from perceiver_amd.models.text.mlm import MaskedLanguageModel
+
+    with torch.device("meta"):
+        model = MaskedLanguageModel(mlm_flagship(num_latents=256))
+    assert sum(p.numel() for p in model.parameters()) == 201_108_230

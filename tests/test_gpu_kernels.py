@@ -183,8 +183,13 @@ def test_fused_layer_norm_matches_torch():
         out.backward(dy)
         dx, dw, db = _ext().ln_bwd(dy.bfloat16(), x.bfloat16(), w.bfloat16(), mean, rstd, True)
         assert torch.allclose(dx.float(), xf.grad, atol=5e-2, rtol=5e-2), C
-        assert torch.allclose(dw.float(), wf.grad, atol=0.5, rtol=3e-2), C
-        assert torch.allclose(db.float(), bf.grad, atol=0.5, rtol=3e-2), C
+        # dw/db are 400-term reductions of bf16 products: scale the absolute
+        # budget by the reduction magnitude instead of a flat atol=0.5 that
+        # could hide a real bug at small magnitudes
+        for got, want in ((dw.float(), wf.grad), (db.float(), bf.grad)):
+            scale = want.abs().mean().clamp(min=1.0)
+            err = (got - want).abs().max().item()
+            assert err < 0.04 * float(scale) + 0.08, (C, err, float(scale))
 
 
 def test_layer_norm_module_dispatch():
