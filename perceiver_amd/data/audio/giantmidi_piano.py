@@ -1,5 +1,10 @@
-"""GiantMIDI-Piano data module (zip download + train/valid dirs).
-Parity: reference data/audio/giantmidi_piano.py."""
+"""GiantMIDI-Piano symbolic-audio data module.
+
+Unlike Maestro (see maestro_v3.py), the GiantMIDI archive already contains
+`train/` and `valid/` directories, so loading reduces to download + extract +
+pointing `SymbolicAudioDataModule` at the extracted split directories.
+Reference behavior: data/audio/giantmidi_piano.py.
+"""
 from __future__ import annotations
 
 import os
@@ -10,12 +15,21 @@ from typing import Any, Dict
 from perceiver_amd.data.audio.symbolic import SymbolicAudioDataModule
 from perceiver_amd.data.audio.utils import download_file, extract_file
 
+_ZIP_NAME = "giantmidi-piano.zip"
+_DEFAULT_URI = f"https://martin-krasser.com/perceiver/data/midi/{_ZIP_NAME}"
+_SPLIT_ERRORS = {
+    "train": "Could not find training directory in downloaded dataset (expected=`{}`)",
+    "valid": "Could not find validation directory in downloaded dataset (expected=`{}`)",
+}
+
 
 class GiantMidiPianoDataModule(SymbolicAudioDataModule):
+    """Symbolic-audio training on the GiantMIDI-Piano transcription corpus."""
+
     def __init__(
         self,
         *args: Any,
-        dataset_uri: str = "https://martin-krasser.com/perceiver/data/midi/giantmidi-piano.zip",
+        dataset_uri: str = _DEFAULT_URI,
         dataset_dir: str = os.path.join(".cache", "giantmidi-piano"),
         **kwargs: Any,
     ):
@@ -27,20 +41,20 @@ class GiantMidiPianoDataModule(SymbolicAudioDataModule):
         return Path(self.hparams.dataset_dir) / "source"
 
     def load_source_dataset(self) -> Dict[str, Path]:
-        if self.source_dir.exists():
-            shutil.rmtree(self.source_dir)
-        self.source_dir.mkdir(parents=True, exist_ok=False)
-        download_dir = self.source_dir / "_download"
-        download_dir.mkdir(parents=True, exist_ok=False)
+        root = self.source_dir
+        if root.exists():
+            shutil.rmtree(root)
+        scratch = root / "_download"
+        scratch.mkdir(parents=True)
 
-        dataset_file = download_dir / "giantmidi-piano.zip"
-        download_file(self._dataset_uri, dataset_file)
-        extract_file(dataset_file, download_dir)
+        archive = scratch / _ZIP_NAME
+        download_file(self._dataset_uri, archive)
+        extract_file(archive, scratch)
 
-        train_dir = download_dir / "train"
-        if not train_dir.exists():
-            raise FileNotFoundError(f"Could not find training directory in downloaded dataset (expected=`{train_dir}`)")
-        valid_dir = download_dir / "valid"
-        if not valid_dir.exists():
-            raise FileNotFoundError(f"Could not find validation directory in downloaded dataset (expected=`{valid_dir}`)")
-        return {"train": train_dir, "valid": valid_dir}
+        splits: Dict[str, Path] = {}
+        for name, err in _SPLIT_ERRORS.items():
+            d = scratch / name
+            if not d.exists():
+                raise FileNotFoundError(err.format(d))
+            splits[name] = d
+        return splits

@@ -1,5 +1,12 @@
-"""Maestro V3 data module (zip download + splits from the metadata JSON).
-Parity: reference data/audio/maestro_v3.py."""
+"""Maestro V3 symbolic-audio data module.
+
+The Maestro archive ships every split in one zip plus a metadata JSON that
+assigns each MIDI file to train/validation/test. This module downloads the
+archive once, sorts the train and validation files into the split-directory
+layout `SymbolicAudioDataModule` expects, and drops the test split (the
+framework validates on the validation split only, matching the reference
+behavior at data/audio/maestro_v3.py).
+"""
 from __future__ import annotations
 
 import json
@@ -11,12 +18,29 @@ from typing import Any, Dict
 from perceiver_amd.data.audio.symbolic import SymbolicAudioDataModule
 from perceiver_amd.data.audio.utils import download_file, extract_file
 
+_ZIP_NAME = "maestro-v3.0.0-midi.zip"
+_ARCHIVE_ROOT = "maestro-v3.0.0"
+_DEFAULT_URI = f"https://martin-krasser.com/perceiver/data/midi/{_ZIP_NAME}"
+
+
+def _read_split_assignments(archive_dir: Path) -> Dict[str, str]:
+    """midi path -> split name, from the archive's metadata JSON."""
+    meta_path = archive_dir / f"{_ARCHIVE_ROOT}.json"
+    if not meta_path.exists():
+        raise FileNotFoundError(f"Could not find Maestro v3 dataset meta file (expected=`{meta_path}`)")
+    meta = json.loads(meta_path.read_text())
+    filenames = meta["midi_filename"]
+    split_of = meta["split"]
+    return {filenames[key]: split_of[key] for key in filenames}
+
 
 class MaestroV3DataModule(SymbolicAudioDataModule):
+    """Symbolic-audio training on the Maestro v3 piano-performance corpus."""
+
     def __init__(
         self,
         *args: Any,
-        dataset_uri: str = "https://martin-krasser.com/perceiver/data/midi/maestro-v3.0.0-midi.zip",
+        dataset_uri: str = _DEFAULT_URI,
         dataset_dir: str = os.path.join(".cache", "maestro-v3-midi"),
         **kwargs: Any,
     ):
@@ -28,48 +52,32 @@ class MaestroV3DataModule(SymbolicAudioDataModule):
         return Path(self.hparams.dataset_dir) / "source"
 
     def load_source_dataset(self) -> Dict[str, Path]:
-        if self.source_dir.exists():
-            shutil.rmtree(self.source_dir)
-        self.source_dir.mkdir(parents=True, exist_ok=False)
-        download_dir = self.source_dir / "_download"
-        download_dir.mkdir(parents=True, exist_ok=False)
-        split_dir = self.source_dir / "_splits"
-        train_dir = split_dir / "train"
-        train_dir.mkdir(parents=True, exist_ok=False)
-        valid_dir = split_dir / "valid"
-        valid_dir.mkdir(parents=True, exist_ok=False)
+        root = self.source_dir
+        if root.exists():
+            shutil.rmtree(root)
 
-        dataset_file = download_dir / "maestro-v3.0.0-midi.zip"
-        download_file(self._dataset_uri, dataset_file)
-        extract_file(dataset_file, download_dir)
+        scratch = root / "_download"
+        scratch.mkdir(parents=True)
+        targets = {name: root / "_splits" / name for name in ("train", "valid")}
+        for d in targets.values():
+            d.mkdir(parents=True)
 
-        dataset_dir = download_dir / "maestro-v3.0.0"
-        if not dataset_dir.exists():
+        archive = scratch / _ZIP_NAME
+        download_file(self._dataset_uri, archive)
+        extract_file(archive, scratch)
+
+        extracted = scratch / _ARCHIVE_ROOT
+        if not extracted.exists():
             raise FileNotFoundError(
-                f"Could not find Maestro v3 dataset directory in downloaded dataset (expected=`{dataset_dir}`)"
+                f"Could not find Maestro v3 dataset directory in downloaded dataset (expected=`{extracted}`)"
             )
-        self._create_dataset_splits(dataset_dir, train_dir, valid_dir)
-        shutil.rmtree(download_dir)
-        return {"train": train_dir, "valid": valid_dir}
 
-    @staticmethod
-    def _create_dataset_splits(source_dataset_dir: Path, train_dir: Path, valid_dir: Path):
-        meta_file = source_dataset_dir / "maestro-v3.0.0.json"
-        if not meta_file.exists():
-            raise FileNotFoundError(f"Could not find Maestro v3 dataset meta file (expected=`{meta_file}`)")
-        with open(meta_file) as f:
-            metadata = json.load(f)
-
-        splits = {}
-        for _id, file_path in metadata["midi_filename"].items():
-            splits[file_path] = metadata["split"][_id]
-
-        for file_path, split in splits.items():
+        for rel_path, split in _read_split_assignments(extracted).items():
             if split == "test":
                 continue
-            source_file = source_dataset_dir / file_path
-            target_dir = train_dir if split == "train" else valid_dir
-            target_file = target_dir / file_path
-            if not target_file.parent.exists():
-                target_file.parent.mkdir(parents=True, exist_ok=False)
-            shutil.move(source_file, target_file)
+            dest = targets["train" if split == "train" else "valid"] / rel_path
+            dest.parent.mkdir(parents=True, exist_ok=True)
+            shutil.move(extracted / rel_path, dest)
+
+        shutil.rmtree(scratch)
+        return dict(targets)

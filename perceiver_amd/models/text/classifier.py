@@ -1,51 +1,37 @@
-"""Text classifier (Perceiver IO): text encoder + classification decoder.
+"""Sequence-classification Perceiver IO backend.
 
-Parity: /root/reference/perceiver/model/text/classifier/backend.py:15-46.
+A byte/token sequence runs through the shared ``TextEncoder`` (the same
+encoder class the MLM trains — transfer learning loads its weights directly),
+and a small classification head decodes the latent array from one or more
+learned output queries into class logits.
+
+Behavioral contract mirrored from the reference's text classifier backend
+(/root/reference/perceiver/model/text/classifier/backend.py:15-46); the
+state-dict layout (``encoder.*`` / ``decoder.*``) is checkpoint-compatible.
 """
 from __future__ import annotations
 
-from perceiver_amd.core import (
-    ClassificationDecoderConfig,
-    ClassificationOutputAdapter,
-    PerceiverDecoder,
-    PerceiverIO,
-    PerceiverIOConfig,
-    TrainableQueryProvider,
-)
+from perceiver_amd.core import ClassificationDecoderConfig, PerceiverIO, PerceiverIOConfig
+from perceiver_amd.models.builders import classification_decoder, latent_kwargs
 from perceiver_amd.models.text.common import TextEncoder, TextEncoderConfig
 
 TextClassifierConfig = PerceiverIOConfig[TextEncoderConfig, ClassificationDecoderConfig]
 
 
 class TextClassifier(PerceiverIO):
+    """TextEncoder + classification decoder."""
+
     def __init__(self, config: TextClassifierConfig):
-        encoder = TextEncoder(
-            config.encoder,
-            num_latents=config.num_latents,
-            num_latent_channels=config.num_latent_channels,
-            activation_checkpointing=config.activation_checkpointing,
-            activation_offloading=config.activation_offloading,
+        super().__init__(
+            TextEncoder(
+                config.encoder,
+                num_latents=config.num_latents,
+                num_latent_channels=config.num_latent_channels,
+                **latent_kwargs(config),
+            ),
+            classification_decoder(config),
         )
-        output_query_provider = TrainableQueryProvider(
-            num_queries=config.decoder.num_output_queries,
-            num_query_channels=config.decoder.num_output_query_channels,
-            init_scale=config.decoder.init_scale,
-        )
-        output_adapter = ClassificationOutputAdapter(
-            num_classes=config.decoder.num_classes,
-            num_output_query_channels=config.decoder.num_output_query_channels,
-        )
-        decoder = PerceiverDecoder(
-            output_adapter=output_adapter,
-            output_query_provider=output_query_provider,
-            num_latent_channels=config.num_latent_channels,
-            activation_checkpointing=config.activation_checkpointing,
-            activation_offloading=config.activation_offloading,
-            **config.decoder.base_kwargs(),
-        )
-        super().__init__(encoder, decoder)
         self.config = config
 
     def forward(self, x, pad_mask=None):
-        latents = self.encoder(x, pad_mask=pad_mask)
-        return self.decoder(latents)
+        return self.decoder(self.encoder(x, pad_mask=pad_mask))

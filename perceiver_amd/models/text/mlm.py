@@ -1,7 +1,19 @@
-"""Masked language model (Perceiver IO): encoder + learned-output-query decoder with
-tied or untied token logits head.
+"""Masked-language-model Perceiver IO backend.
 
-Parity: /root/reference/perceiver/model/text/mlm/backend.py:18-89.
+The encoder compresses the (masked) token sequence into the latent array; the
+decoder queries it with ``max_seq_len`` learned output queries and projects to
+vocabulary logits. Two heads exist, selected by
+``decoder.num_output_query_channels``:
+
+* ``None`` → tied head: output queries live in the encoder's input-channel
+  space and logits come from the shared token embedding
+  (``TiedTokenOutputAdapter``, x @ emb.Tᵀ) — the deepmind/language-perceiver
+  layout.
+* an int → untied head: a separate Linear projects query channels to vocab.
+
+Behavioral contract mirrored from the reference MLM backend
+(/root/reference/perceiver/model/text/mlm/backend.py:18-89); state-dict keys
+are checkpoint-compatible.
 """
 from __future__ import annotations
 
@@ -17,8 +29,8 @@ from perceiver_amd.core import (
     PerceiverIO,
     PerceiverIOConfig,
     TiedTokenOutputAdapter,
-    TrainableQueryProvider,
 )
+from perceiver_amd.models.builders import assemble_decoder, latent_kwargs, learned_queries
 from perceiver_amd.models.text.common import TextEncoder, TextEncoderConfig
 
 
@@ -36,7 +48,7 @@ MaskedLanguageModelConfig = PerceiverIOConfig[TextEncoderConfig, TextDecoderConf
 
 
 class TokenOutputAdapter(OutputAdapter):
-    """Untied logits head (separate Linear to vocab)."""
+    """Untied logits head: its own Linear from query channels to vocab."""
 
     def __init__(self, vocab_size: int, num_output_query_channels: int):
         super().__init__()
@@ -46,52 +58,46 @@ class TokenOutputAdapter(OutputAdapter):
         return self.linear(x).squeeze(dim=1)
 
 
+def _token_logits_decoder(config: MaskedLanguageModelConfig) -> PerceiverDecoder:
+    dc = config.decoder
+    tied = dc.num_output_query_channels is None
+    query_channels = config.encoder.num_input_channels if tied else dc.num_output_query_channels
+    head = (TiedTokenOutputAdapter(vocab_size=dc.vocab_size) if tied
+            else TokenOutputAdapter(vocab_size=dc.vocab_size,
+                                    num_output_query_channels=dc.num_output_query_channels))
+    queries = learned_queries(dc.max_seq_len, query_channels, dc.init_scale)
+    return assemble_decoder(head, queries, config, dc)
+
+
 class MaskedLanguageModel(PerceiverIO):
+    """TextEncoder + token-logits decoder (tied or untied)."""
+
     def __init__(self, config: MaskedLanguageModelConfig):
-        encoder = TextEncoder(
-            config.encoder,
-            num_latents=config.num_latents,
-            num_latent_channels=config.num_latent_channels,
-            activation_checkpointing=config.activation_checkpointing,
-            activation_offloading=config.activation_offloading,
+        super().__init__(
+            TextEncoder(
+                config.encoder,
+                num_latents=config.num_latents,
+                num_latent_channels=config.num_latent_channels,
+                **latent_kwargs(config),
+            ),
+            _token_logits_decoder(config),
         )
-        if config.decoder.num_output_query_channels is None:
-            # tied head: output queries live in encoder-input-channel space
-            output_query_provider = TrainableQueryProvider(
-                num_queries=config.decoder.max_seq_len,
-                num_query_channels=config.encoder.num_input_channels,
-                init_scale=config.decoder.init_scale,
-            )
-            output_adapter = TiedTokenOutputAdapter(vocab_size=config.decoder.vocab_size)
-        else:
-            output_query_provider = TrainableQueryProvider(
-                num_queries=config.decoder.max_seq_len,
-                num_query_channels=config.decoder.num_output_query_channels,
-                init_scale=config.decoder.init_scale,
-            )
-            output_adapter = TokenOutputAdapter(
-                vocab_size=config.decoder.vocab_size,
-                num_output_query_channels=config.decoder.num_output_query_channels,
-            )
-        decoder = PerceiverDecoder(
-            output_adapter=output_adapter,
-            output_query_provider=output_query_provider,
-            num_latent_channels=config.num_latent_channels,
-            activation_checkpointing=config.activation_checkpointing,
-            activation_offloading=config.activation_offloading,
-            **config.decoder.base_kwargs(),
-        )
-        super().__init__(encoder, decoder)
         self.config = config
 
+    @property
+    def _tied(self) -> bool:
+        return isinstance(self.decoder.output_adapter, TiedTokenOutputAdapter)
+
     def forward(self, x_masked, pad_mask=None):
-        _, n = x_masked.shape
-        x_latent = self.encoder(x_masked, pad_mask)
-        if isinstance(self.decoder.output_adapter, TiedTokenOutputAdapter):
-            x_logits = self.decoder(x_latent, txt_embedding=self.encoder.input_adapter.txt_embedding)
+        seq_len = x_masked.shape[1]
+        latents = self.encoder(x_masked, pad_mask)
+        if self._tied:
+            # the tied adapter receives the token embedding at call time
+            logits = self.decoder(latents, txt_embedding=self.encoder.input_adapter.txt_embedding)
         else:
-            x_logits = self.decoder(x_latent)
-        return x_logits[:, :n, :]
+            logits = self.decoder(latents)
+        # the decoder always produces max_seq_len rows; crop to the input length
+        return logits[:, :seq_len, :]
 
 
 # alias kept for API parity with the reference

@@ -422,6 +422,18 @@ __global__ void flash_merge_kernel(const float* __restrict__ o_part,
     if (lane == 0) lsep[row] = mx + logf(fmaxf(wsum, 1e-37f));
 }
 
+void launch_merge(float* o_part_p, float* lse_part_p, torch::Tensor& out, torch::Tensor& lse,
+                  long rows, int nsplit, int Dv) {
+    auto stream = at::cuda::getCurrentCUDAStream();
+    int wpb = 4;
+    long blocks = (rows + wpb - 1) / wpb;
+    hipLaunchKernelGGL(flash_merge_kernel, dim3(blocks), dim3(64 * wpb), 0, stream,
+                       o_part_p, lse_part_p,
+                       reinterpret_cast<unsigned short*>(out.data_ptr()),
+                       lse.data_ptr<float>(), rows, nsplit, Dv);
+    HIP_CHECK_LAST();
+}
+
 template <int DMAX, int DVMAX, int QH>
 void launch_flash_fwd(const torch::Tensor& q, const torch::Tensor& k, const torch::Tensor& v,
                       const c10::optional<torch::Tensor>& pad_mask, bool causal,
@@ -496,6 +508,24 @@ bool flash_supported_impl(long d_qk, long d_v, long needs_dropout) {
     return d_qk <= 352 && d_v <= 352;
 }
 
+// Deep-pipelined forward (flash_fwd_pipe.hip) for the regular head-dim regimes
+bool flash_fwd_pipe_applicable(long D, long Dv, long Nq);
+void flash_fwd_pipe_launch(const torch::Tensor& q, const torch::Tensor& k,
+                           const torch::Tensor& v, const bool* padp, bool causal,
+                           float drop_p, unsigned long long drop_seed,
+                           torch::Tensor& out, torch::Tensor& lse,
+                           float* o_part_p, float* lse_part_p, long kv_chunk, int nsplit);
+
+namespace {
+bool pipe_enabled() {
+    static int v = [] {
+        const char* e = getenv("PERCEIVER_NO_PIPE");
+        return (e && e[0] == '1') ? 0 : 1;
+    }();
+    return v != 0;
+}
+}  // namespace
+
 std::vector<torch::Tensor> flash_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                                      c10::optional<torch::Tensor> pad_mask, bool causal,
                                      double dropout_p, int64_t seed) {
@@ -526,6 +556,41 @@ std::vector<torch::Tensor> flash_fwd(torch::Tensor q, torch::Tensor k, torch::Te
 
     float dp = (float)dropout_p;
     unsigned long long sd = (unsigned long long)seed;
+
+    if (pipe_enabled() && flash_fwd_pipe_applicable(D, Dv, q.size(2))) {
+        int B = q.size(0), H = q.size(1), Nq = q.size(2), Lk = k.size(2);
+        constexpr int PIPE_QBLK = 128;
+        constexpr int KVB = 64;
+        long gx = (Nq + PIPE_QBLK - 1) / PIPE_QBLK;
+        long gy = (long)B * H;
+        int nsplit = 1;
+        long kv_chunk = Lk;
+        if (gx * gy < 512 && Lk > 4 * KVB) {
+            int want = (int)(512 / (gx * gy)) + 1;
+            int max_split = (Lk + 4 * KVB - 1) / (4 * KVB);
+            nsplit = std::min({want, max_split, 32});
+            long tiles = (Lk + KVB - 1) / KVB;
+            long tiles_per = (tiles + nsplit - 1) / nsplit;
+            kv_chunk = tiles_per * KVB;
+            nsplit = (int)((Lk + kv_chunk - 1) / kv_chunk);
+        }
+        torch::Tensor o_part, lse_part;
+        float* o_part_p = nullptr;
+        float* lse_part_p = nullptr;
+        long rows = (long)B * H * Nq;
+        if (nsplit > 1) {
+            o_part = torch::empty({(long)nsplit, rows, (long)Dv}, q.options().dtype(torch::kFloat32));
+            lse_part = torch::empty({(long)nsplit, rows}, q.options().dtype(torch::kFloat32));
+            o_part_p = o_part.data_ptr<float>();
+            lse_part_p = lse_part.data_ptr<float>();
+        }
+        const bool* padp = (pm.has_value() && pm->defined()) ? pm->data_ptr<bool>() : nullptr;
+        flash_fwd_pipe_launch(q, k, v, padp, causal, dp, sd, out, lse,
+                              o_part_p, lse_part_p, kv_chunk, nsplit);
+        if (nsplit > 1) launch_merge(o_part_p, lse_part_p, out, lse, rows, nsplit, Dv);
+        return {out, lse};
+    }
+
     if (D <= 32 && Dv <= 160)       launch_flash_fwd<32, 160, 2>(q, k, v, pm, causal, dp, sd, out, lse);
     else if (D <= 64 && Dv <= 64)   launch_flash_fwd<64, 64, 2>(q, k, v, pm, causal, dp, sd, out, lse);
     else if (D <= 128 && Dv <= 128) launch_flash_fwd<128, 128, 2>(q, k, v, pm, causal, dp, sd, out, lse);

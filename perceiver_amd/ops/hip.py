@@ -27,11 +27,12 @@ def _load():
         import importlib.util
         import pathlib
 
-        so = None
         here = pathlib.Path(__file__).parent
-        for cand in sorted(here.glob("_perceiver_hip*.so")):
-            so = cand
-            break
+        # canonical artifact first (what ops.build writes); fall back to any
+        # interpreter-tagged name a setup.py build may have produced
+        so = here / "_perceiver_hip.so"
+        if not so.exists():
+            so = next(iter(sorted(here.glob("_perceiver_hip*.so"))), None)
         if so is None:
             raise ImportError("no _perceiver_hip*.so found (run python -m perceiver_amd.ops.build)")
         spec = importlib.util.spec_from_file_location("_perceiver_hip", so)
