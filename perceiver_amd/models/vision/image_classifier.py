@@ -1,29 +1,35 @@
-"""Image classifier (Perceiver IO): Fourier-position-encoded pixels -> latents ->
-single-output-query classification decoder.
+"""Image-classification Perceiver IO backend.
 
-Parity: /root/reference/perceiver/model/vision/image_classifier/backend.py:21-96.
-On GPU the Fourier-encode + concat prologue fuses into the cross-attention K/V
-loader instead of materializing the (B, H*W, C) adapted input.
+Images enter as raw channels-last pixels: the input adapter flattens the
+spatial grid and appends a fixed Fourier position code per pixel (sin/cos over
+``num_frequency_bands`` frequencies per spatial dim, plus the raw coordinate),
+giving a (B, H*W, C_pixel + C_pos) K/V array of ~50k rows at 224². The encoder
+cross-attends a few hundred latents into that array; a single learned query
+decodes the class logits.
+
+Behavioral contract mirrored from the reference image-classifier backend
+(/root/reference/perceiver/model/vision/image_classifier/backend.py:21-96),
+including its default of widening the cross-attention qk channels to the
+adapter width when unset. State-dict layout is checkpoint-compatible. On GPU
+the Fourier-encode + concat prologue is the K2 fusion target.
 """
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Tuple
 
 import torch
 
 from perceiver_amd.core import (
     ClassificationDecoderConfig,
-    ClassificationOutputAdapter,
     EncoderConfig,
     FourierPositionEncoding,
     InputAdapter,
-    PerceiverDecoder,
     PerceiverEncoder,
     PerceiverIO,
     PerceiverIOConfig,
-    TrainableQueryProvider,
 )
+from perceiver_amd.models.builders import classification_decoder, latent_kwargs
 
 
 @dataclass
@@ -39,66 +45,52 @@ ImageClassifierConfig = PerceiverIOConfig[ImageEncoderConfig, ClassificationDeco
 
 
 class ImageInputAdapter(InputAdapter):
-    """Flatten H x W pixels and concat Fourier position encodings."""
+    """Pixel-flattening adapter with concatenated Fourier position codes."""
 
     def __init__(self, image_shape: Tuple[int, ...], num_frequency_bands: int):
-        *spatial_shape, num_image_channels = image_shape
-        position_encoding = FourierPositionEncoding(
-            input_shape=tuple(spatial_shape), num_frequency_bands=num_frequency_bands
-        )
-        super().__init__(num_input_channels=num_image_channels + position_encoding.num_position_encoding_channels())
+        spatial, channels = tuple(image_shape[:-1]), image_shape[-1]
+        pos = FourierPositionEncoding(input_shape=spatial, num_frequency_bands=num_frequency_bands)
+        super().__init__(num_input_channels=channels + pos.num_position_encoding_channels())
         self.image_shape = tuple(image_shape)
-        self.position_encoding = position_encoding
+        self.position_encoding = pos
 
     def forward(self, x):
-        b, *d = x.shape
-        if tuple(d) != self.image_shape:
-            raise ValueError(f"Input vision shape {tuple(d)} different from required shape {self.image_shape}")
-        x = x.flatten(1, len(self.image_shape) - 1)
-        return torch.cat([x, self.position_encoding(b).to(x.dtype)], dim=-1)
+        got = tuple(x.shape[1:])
+        if got != self.image_shape:
+            raise ValueError(f"Input vision shape {got} different from required shape {self.image_shape}")
+        flat = x.flatten(1, -2)  # (B, H*W, C) from channels-last input
+        codes = self.position_encoding(x.shape[0]).to(flat.dtype)
+        return torch.cat([flat, codes], dim=-1)
+
+
+def _image_encoder(config: ImageClassifierConfig) -> PerceiverEncoder:
+    adapter = ImageInputAdapter(
+        image_shape=config.encoder.image_shape,
+        num_frequency_bands=config.encoder.num_frequency_bands,
+    )
+    kwargs = config.encoder.base_kwargs()
+    # reference default: qk width follows the (position-encoding-heavy) adapter
+    # channels unless the config pins it
+    if kwargs.get("num_cross_attention_qk_channels") is None:
+        kwargs["num_cross_attention_qk_channels"] = adapter.num_input_channels
+    return PerceiverEncoder(
+        input_adapter=adapter,
+        num_latents=config.num_latents,
+        num_latent_channels=config.num_latent_channels,
+        **latent_kwargs(config),
+        **kwargs,
+    )
 
 
 class ImageClassifier(PerceiverIO):
-    def __init__(self, config: ImageClassifierConfig):
-        input_adapter = ImageInputAdapter(
-            image_shape=config.encoder.image_shape,
-            num_frequency_bands=config.encoder.num_frequency_bands,
-        )
-        encoder_kwargs = config.encoder.base_kwargs()
-        if encoder_kwargs["num_cross_attention_qk_channels"] is None:
-            encoder_kwargs["num_cross_attention_qk_channels"] = input_adapter.num_input_channels
+    """Fourier-encoded pixel encoder + single-query classification decoder."""
 
-        encoder = PerceiverEncoder(
-            input_adapter=input_adapter,
-            num_latents=config.num_latents,
-            num_latent_channels=config.num_latent_channels,
-            activation_checkpointing=config.activation_checkpointing,
-            activation_offloading=config.activation_offloading,
-            **encoder_kwargs,
-        )
-        output_query_provider = TrainableQueryProvider(
-            num_queries=1,
-            num_query_channels=config.decoder.num_output_query_channels,
-            init_scale=config.decoder.init_scale,
-        )
-        output_adapter = ClassificationOutputAdapter(
-            num_classes=config.decoder.num_classes,
-            num_output_query_channels=config.decoder.num_output_query_channels,
-        )
-        decoder = PerceiverDecoder(
-            output_adapter=output_adapter,
-            output_query_provider=output_query_provider,
-            num_latent_channels=config.num_latent_channels,
-            activation_checkpointing=config.activation_checkpointing,
-            activation_offloading=config.activation_offloading,
-            **config.decoder.base_kwargs(),
-        )
-        super().__init__(encoder, decoder)
+    def __init__(self, config: ImageClassifierConfig):
+        super().__init__(_image_encoder(config), classification_decoder(config, num_queries=1))
         self.config = config
 
     def forward(self, x, pad_mask=None):
-        latents = self.encoder(x, pad_mask=pad_mask)
-        return self.decoder(latents)
+        return self.decoder(self.encoder(x, pad_mask=pad_mask))
 
 
 # alias kept for API parity with the reference

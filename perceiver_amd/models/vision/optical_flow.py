@@ -1,8 +1,19 @@
-"""Optical flow (Perceiver IO): temporal-concat patch projection + Fourier encodings;
-the decoder queries the latents with the *adapted input* (identity query provider)
-and maps to per-pixel (dx, dy) flow.
+"""Optical-flow Perceiver IO backend.
 
-Parity: /root/reference/perceiver/model/vision/optical_flow/backend.py:22-137.
+Input is a pair of frames expanded to per-pixel 3×3-neighborhood features,
+shape (B, 2, C_patch, H, W). The adapter concatenates the two frames along
+channels, projects each pixel with a small Linear, flattens the grid and
+appends Fourier position codes — producing the ~182k-row K/V array at
+368×496. The decoder is the unusual part: its query array is not learned but
+IS the adapted input (the encoder is asked to return it), so every pixel
+queries the latents and the output adapter maps each decoded pixel to a
+(dx, dy) flow vector, divided by ``rescale_factor`` (training targets are
+pre-scaled by the data pipeline).
+
+Behavioral contract mirrored from the reference optical-flow backend
+(/root/reference/perceiver/model/vision/optical_flow/backend.py:22-137),
+including widening both qk and v cross-attention channels to the adapter
+width when unset. State-dict layout is checkpoint-compatible.
 """
 from __future__ import annotations
 
@@ -18,12 +29,12 @@ from perceiver_amd.core import (
     FourierPositionEncoding,
     InputAdapter,
     OutputAdapter,
-    PerceiverDecoder,
     PerceiverEncoder,
     PerceiverIO,
     PerceiverIOConfig,
     QueryProvider,
 )
+from perceiver_amd.models.builders import assemble_decoder, latent_kwargs
 
 
 @dataclass
@@ -53,8 +64,11 @@ OpticalFlowConfig = PerceiverIOConfig[OpticalFlowEncoderConfig, OpticalFlowDecod
 
 
 class OpticalFlowInputAdapter(InputAdapter):
-    """(b, t, c, h, w) -> concat temporal frames along channels -> Linear patch
-    projection -> flatten + Fourier position-encoding concat."""
+    """Frame-pair patch projection plus Fourier position codes.
+
+    The temporal axis is folded into channels (both frames of a pixel feed one
+    projection), so the Linear sees ``2 * num_patch_input_channels`` inputs.
+    """
 
     def __init__(
         self,
@@ -63,21 +77,22 @@ class OpticalFlowInputAdapter(InputAdapter):
         num_patch_hidden_channels: int,
         num_frequency_bands: int,
     ):
-        position_encoding = FourierPositionEncoding(input_shape=image_shape, num_frequency_bands=num_frequency_bands)
-        super().__init__(num_patch_hidden_channels + position_encoding.num_position_encoding_channels())
-        self.linear = nn.Linear(num_patch_input_channels * 2, num_patch_hidden_channels)
-        self.position_encoding = position_encoding
+        pos = FourierPositionEncoding(input_shape=image_shape, num_frequency_bands=num_frequency_bands)
+        super().__init__(num_patch_hidden_channels + pos.num_position_encoding_channels())
+        self.linear = nn.Linear(2 * num_patch_input_channels, num_patch_hidden_channels)
+        self.position_encoding = pos
 
     def forward(self, x):
         b, t, c, h, w = x.shape
-        # b t c h w -> b h w (t c)
-        x = x.permute(0, 3, 4, 1, 2).reshape(b, h, w, t * c)
-        x = self.linear(x)
-        x = x.flatten(1, 2)
-        return torch.cat([x, self.position_encoding(b).to(x.dtype)], dim=-1)
+        per_pixel = x.permute(0, 3, 4, 1, 2).reshape(b, h * w, t * c)
+        projected = self.linear(per_pixel)
+        codes = self.position_encoding(b).to(projected.dtype)
+        return torch.cat([projected, codes], dim=-1)
 
 
 class OpticalFlowOutputAdapter(OutputAdapter):
+    """Per-pixel Linear to (dx, dy), rescaled and reshaped to the image grid."""
+
     def __init__(
         self,
         image_shape: Tuple[int, int],
@@ -91,13 +106,13 @@ class OpticalFlowOutputAdapter(OutputAdapter):
         self.linear = nn.Linear(num_output_query_channels, num_output_image_channels)
 
     def forward(self, x):
-        x = self.linear(x) / self.rescale_factor
-        b, _, c = x.shape
-        return x.view(b, self.image_shape[0], self.image_shape[1], c)
+        flow = self.linear(x) * (1.0 / self.rescale_factor)
+        h, w = self.image_shape
+        return flow.view(flow.shape[0], h, w, flow.shape[-1])
 
 
 class OpticalFlowQueryProvider(nn.Module, QueryProvider):
-    """Identity query provider: the decoder queries with the adapted input."""
+    """Identity provider: the decoder's queries are the adapted input rows."""
 
     def __init__(self, num_query_channels: int):
         super().__init__()
@@ -112,45 +127,43 @@ class OpticalFlowQueryProvider(nn.Module, QueryProvider):
         return x
 
 
+def _flow_encoder(config: OpticalFlowConfig, adapter: OpticalFlowInputAdapter) -> PerceiverEncoder:
+    kwargs = config.encoder.base_kwargs()
+    # both attention widths track the adapter channels unless pinned
+    for key in ("num_cross_attention_qk_channels", "num_cross_attention_v_channels"):
+        if kwargs.get(key) is None:
+            kwargs[key] = adapter.num_input_channels
+    return PerceiverEncoder(
+        input_adapter=adapter,
+        num_latents=config.num_latents,
+        num_latent_channels=config.num_latent_channels,
+        **latent_kwargs(config),
+        **kwargs,
+    )
+
+
 class OpticalFlow(PerceiverIO):
+    """Patch-projection encoder + adapted-input-query flow decoder."""
+
     def __init__(self, config: OpticalFlowConfig):
-        input_adapter = OpticalFlowInputAdapter(
+        adapter = OpticalFlowInputAdapter(
             image_shape=config.encoder.image_shape,
             num_patch_input_channels=config.encoder.num_patch_input_channels,
             num_patch_hidden_channels=config.encoder.num_patch_hidden_channels,
             num_frequency_bands=config.encoder.num_frequency_bands,
         )
-        encoder_kwargs = config.encoder.base_kwargs()
-        if encoder_kwargs["num_cross_attention_qk_channels"] is None:
-            encoder_kwargs["num_cross_attention_qk_channels"] = input_adapter.num_input_channels
-        if encoder_kwargs["num_cross_attention_v_channels"] is None:
-            encoder_kwargs["num_cross_attention_v_channels"] = input_adapter.num_input_channels
-
-        encoder = PerceiverEncoder(
-            input_adapter=input_adapter,
-            num_latents=config.num_latents,
-            num_latent_channels=config.num_latent_channels,
-            activation_checkpointing=config.activation_checkpointing,
-            activation_offloading=config.activation_offloading,
-            **encoder_kwargs,
-        )
-        output_adapter = OpticalFlowOutputAdapter(
+        head = OpticalFlowOutputAdapter(
             image_shape=config.decoder.image_shape,
-            num_output_query_channels=input_adapter.num_input_channels,
+            num_output_query_channels=adapter.num_input_channels,
             rescale_factor=config.decoder.rescale_factor,
         )
-        output_query_provider = OpticalFlowQueryProvider(num_query_channels=input_adapter.num_input_channels)
-        decoder = PerceiverDecoder(
-            output_adapter=output_adapter,
-            output_query_provider=output_query_provider,
-            num_latent_channels=config.num_latent_channels,
-            activation_checkpointing=config.activation_checkpointing,
-            activation_offloading=config.activation_offloading,
-            **config.decoder.base_kwargs(),
+        queries = OpticalFlowQueryProvider(num_query_channels=adapter.num_input_channels)
+        super().__init__(
+            _flow_encoder(config, adapter),
+            assemble_decoder(head, queries, config, config.decoder),
         )
-        super().__init__(encoder, decoder)
         self.config = config
 
     def forward(self, x: torch.Tensor):
-        x_latent, x_adapted = self.encoder(x, return_adapted_input=True)
-        return self.decoder(x_latent, x_adapted=x_adapted)
+        latents, adapted = self.encoder(x, return_adapted_input=True)
+        return self.decoder(latents, x_adapted=adapted)
