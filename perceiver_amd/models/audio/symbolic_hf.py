@@ -24,6 +24,7 @@ from perceiver_amd.models.hf_base import (
     PerceiverCausalSequenceModel,
     PerceiverCausalSequenceModelOutput,
 )
+from perceiver_amd.models.hf_registry import BackendConfigMixin
 
 
 class ReturnType(enum.Enum):
@@ -31,18 +32,14 @@ class ReturnType(enum.Enum):
     AUDIO = 1
 
 
-class PerceiverSymbolicAudioModelConfig(PretrainedConfig):
+class PerceiverSymbolicAudioModelConfig(BackendConfigMixin, PretrainedConfig):
     model_type = "perceiver-ar-symbolic-audio-model"
+    backend_config_class = SymbolicAudioModelConfig
 
-    def __init__(self, backend_config: Optional[SymbolicAudioModelConfig] = None, **kwargs):
-        if backend_config is None:
-            backend_config = SymbolicAudioModelConfig()
-        self.model_config = asdict(backend_config)
-        super().__init__(**kwargs)
-
-    @property
-    def backend_config(self) -> SymbolicAudioModelConfig:
-        return SymbolicAudioModelConfig.create(**self.model_config)
+    def __init__(self, backend_config=None, **kwargs):
+        # explicit __init__: transformers 5.x wraps configs without one
+        # in a kwargs-only guard that would swallow backend_config
+        super().__init__(backend_config, **kwargs)
 
 
 class PerceiverSymbolicAudioModel(PerceiverCausalSequenceModel):
@@ -50,22 +47,16 @@ class PerceiverSymbolicAudioModel(PerceiverCausalSequenceModel):
 
     def __init__(self, config: PerceiverSymbolicAudioModelConfig, **kwargs):
         super().__init__(config)
-        if "backend_model" in kwargs:
-            self.backend_model = kwargs["backend_model"]  # zero-copy wrap
-        else:
-            self.backend_model = SymbolicAudioModel(config.backend_config)
+        self.backend_model = kwargs.get("backend_model") or SymbolicAudioModel(config.backend_config)
         self.post_init()
 
     @staticmethod
     def from_checkpoint(ckpt_path):
+        from perceiver_amd.models.hf_registry import wrap_lit_checkpoint
         from perceiver_amd.train.lit import LitSymbolicAudioModel
 
-        model = LitSymbolicAudioModel.load_from_checkpoint(ckpt_path).model
-        hgf_config = PerceiverSymbolicAudioModelConfig(model.config)
-        hgf_config.is_decoder = True
-        hgf_model = PerceiverSymbolicAudioModel(hgf_config)
-        hgf_model.backend_model.load_state_dict(model.state_dict())
-        return hgf_model
+        return wrap_lit_checkpoint(LitSymbolicAudioModel, PerceiverSymbolicAudioModel,
+                                   ckpt_path, is_decoder=True)
 
 
 class SymbolicAudioPipeline(Pipeline):

@@ -1,46 +1,62 @@
-"""🤗 wrapper for the text classifier: AutoModelForSequenceClassification
-registration (sentiment pipeline) + checkpoint conversion with id2label.
-Parity: reference text/classifier/huggingface.py."""
+"""🤗 face of the text classifier.
+
+Registers the model with ``AutoModelForSequenceClassification`` so the
+sentiment-analysis pipeline works out of the box, and converts training
+checkpoints (optionally with IMDb's NEGATIVE/POSITIVE label map) into
+``save_pretrained`` directories.
+
+Behavioral contract mirrored from the reference
+(/root/reference/perceiver/model/text/classifier/huggingface.py).
+"""
 from __future__ import annotations
 
-from dataclasses import asdict
 from typing import Optional
 
 import torch
 from transformers import (
     AutoConfig,
     AutoModelForSequenceClassification,
-    AutoTokenizer,
     PretrainedConfig,
     PreTrainedModel,
 )
 from transformers.modeling_outputs import SequenceClassifierOutput
 
 from perceiver_amd.core import ClassificationDecoderConfig
+from perceiver_amd.models.hf_registry import (
+    BackendConfigMixin,
+    save_with_tokenizer,
+    wrap_lit_checkpoint,
+)
 from perceiver_amd.models.text.classifier import TextClassifier, TextClassifierConfig
 from perceiver_amd.models.text.common import TextEncoderConfig
 
+IMDB_LABELS = {0: "NEGATIVE", 1: "POSITIVE"}
 
-class PerceiverTextClassifierConfig(PretrainedConfig):
+
+class PerceiverTextClassifierConfig(BackendConfigMixin, PretrainedConfig):
     model_type = "perceiver-io-text-classifier"
+    backend_config_class = TextClassifierConfig
 
-    def __init__(self, backend_config: Optional[TextClassifierConfig] = None, **kwargs):
-        if backend_config is None:
-            backend_config = TextClassifierConfig(
-                TextEncoderConfig(), ClassificationDecoderConfig(), num_latents=512, num_latent_channels=512
-            )
-        self.model_config = asdict(backend_config)
-        super().__init__(**kwargs)
+    def __init__(self, backend_config=None, **kwargs):
+        # explicit __init__: transformers 5.x wraps configs without one
+        # in a kwargs-only guard that would swallow backend_config
+        super().__init__(backend_config, **kwargs)
 
-    @property
-    def backend_config(self) -> TextClassifierConfig:
-        model_config = self.model_config.copy()
-        encoder_config = model_config.pop("encoder")
-        decoder_config = model_config.pop("decoder")
+    @classmethod
+    def default_backend_config(cls):
         return TextClassifierConfig(
-            encoder=TextEncoderConfig(**encoder_config),
-            decoder=ClassificationDecoderConfig(**decoder_config),
-            **model_config,
+            TextEncoderConfig(), ClassificationDecoderConfig(),
+            num_latents=512, num_latent_channels=512,
+        )
+
+    @classmethod
+    def decode_backend_config(cls, model_config):
+        # the generic PerceiverIOConfig nests two dataclasses: rebuild them
+        flat = dict(model_config)
+        return TextClassifierConfig(
+            encoder=TextEncoderConfig(**flat.pop("encoder")),
+            decoder=ClassificationDecoderConfig(**flat.pop("decoder")),
+            **flat,
         )
 
 
@@ -56,12 +72,8 @@ class PerceiverTextClassifier(PreTrainedModel):
     def from_checkpoint(ckpt_path):
         from perceiver_amd.train.lit import LitTextClassifier
 
-        model = LitTextClassifier.load_from_checkpoint(ckpt_path).model
-        hgf_config = PerceiverTextClassifierConfig(model.config)
-        hgf_config.is_decoder = False
-        hgf_model = PerceiverTextClassifier(hgf_config)
-        hgf_model.backend_model.load_state_dict(model.state_dict())
-        return hgf_model
+        return wrap_lit_checkpoint(LitTextClassifier, PerceiverTextClassifier,
+                                   ckpt_path, is_decoder=False)
 
     def forward(self, input_ids: torch.LongTensor,
                 attention_mask: Optional[torch.FloatTensor] = None,
@@ -69,8 +81,7 @@ class PerceiverTextClassifier(PreTrainedModel):
         if labels is not None:
             raise ValueError("Loss computation from labels not supported yet")
         pad_mask = None if attention_mask is None else ~attention_mask.type(torch.bool)
-        logits = self.backend_model(input_ids, pad_mask=pad_mask)
-        return SequenceClassifierOutput(logits=logits)
+        return SequenceClassifierOutput(logits=self.backend_model(input_ids, pad_mask=pad_mask))
 
 
 AutoConfig.register(PerceiverTextClassifierConfig.model_type, PerceiverTextClassifierConfig)
@@ -78,21 +89,18 @@ AutoModelForSequenceClassification.register(PerceiverTextClassifierConfig, Perce
 
 
 def convert_checkpoint(save_dir, ckpt_url, tokenizer_name, id2label=None, label2id=None, **kwargs):
-    tokenizer = AutoTokenizer.from_pretrained(tokenizer_name, verbose=False)
-    tokenizer.save_pretrained(save_dir, **kwargs)
-
-    model = PerceiverTextClassifier.from_checkpoint(ckpt_url)
-    model.config.tokenizer_class = tokenizer.__class__.__name__
+    overrides = {}
     if id2label is not None:
-        model.config.id2label = id2label
+        overrides["id2label"] = id2label
     if label2id is not None:
-        model.config.label2id = label2id
-    model.save_pretrained(save_dir, **kwargs)
+        overrides["label2id"] = label2id
+    save_with_tokenizer(PerceiverTextClassifier.from_checkpoint(ckpt_url),
+                        tokenizer_name, save_dir, config_overrides=overrides, **kwargs)
 
 
 def convert_imdb_classifier_checkpoint(save_dir, ckpt_url, tokenizer_name, **kwargs):
     convert_checkpoint(
         save_dir=save_dir, ckpt_url=ckpt_url, tokenizer_name=tokenizer_name,
-        id2label={0: "NEGATIVE", 1: "POSITIVE"}, label2id={"NEGATIVE": 0, "POSITIVE": 1},
+        id2label=dict(IMDB_LABELS), label2id={v: k for k, v in IMDB_LABELS.items()},
         **kwargs,
     )

@@ -1,28 +1,35 @@
-"""🤗 wrapper for the causal LM: AutoModelForCausalLM registration + checkpoint
-conversion. Parity: reference text/clm/huggingface.py."""
+"""🤗 face of the causal language model.
+
+Registers the Perceiver-AR CLM with ``AutoModelForCausalLM`` (and therefore
+the text-generation pipeline), and converts training checkpoints into
+``save_pretrained`` directories with a LEFT-padding tokenizer — Perceiver-AR
+requires left padding so the latent window stays right-aligned.
+
+Behavioral contract mirrored from the reference
+(/root/reference/perceiver/model/text/clm/huggingface.py); the generation
+machinery itself lives in models/hf_base.py.
+"""
 from __future__ import annotations
 
-from dataclasses import asdict
-from typing import Optional
-
-from transformers import AutoConfig, AutoModelForCausalLM, AutoTokenizer, PretrainedConfig
+from transformers import AutoConfig, AutoModelForCausalLM, PretrainedConfig
 
 from perceiver_amd.models.hf_base import PerceiverCausalSequenceModel
+from perceiver_amd.models.hf_registry import (
+    BackendConfigMixin,
+    save_with_tokenizer,
+    wrap_lit_checkpoint,
+)
 from perceiver_amd.models.text.clm import CausalLanguageModel, CausalLanguageModelConfig
 
 
-class PerceiverCausalLanguageModelConfig(PretrainedConfig):
+class PerceiverCausalLanguageModelConfig(BackendConfigMixin, PretrainedConfig):
     model_type = "perceiver-ar-causal-language-model"
+    backend_config_class = CausalLanguageModelConfig
 
-    def __init__(self, backend_config: Optional[CausalLanguageModelConfig] = None, **kwargs):
-        if backend_config is None:
-            backend_config = CausalLanguageModelConfig()
-        self.model_config = asdict(backend_config)
-        super().__init__(**kwargs)
-
-    @property
-    def backend_config(self) -> CausalLanguageModelConfig:
-        return CausalLanguageModelConfig.create(**self.model_config)
+    def __init__(self, backend_config=None, **kwargs):
+        # explicit __init__: transformers 5.x wraps configs without one
+        # in a kwargs-only guard that would swallow backend_config
+        super().__init__(backend_config, **kwargs)
 
 
 class PerceiverCausalLanguageModel(PerceiverCausalSequenceModel):
@@ -30,22 +37,17 @@ class PerceiverCausalLanguageModel(PerceiverCausalSequenceModel):
 
     def __init__(self, config: PerceiverCausalLanguageModelConfig, **kwargs):
         super().__init__(config)
-        if "backend_model" in kwargs:
-            self.backend_model = kwargs["backend_model"]  # zero-copy wrap (in-training generation)
-        else:
-            self.backend_model = CausalLanguageModel(config.backend_config)
+        # "backend_model" kwarg = zero-copy wrap of an existing (training)
+        # model for in-training sample generation
+        self.backend_model = kwargs.get("backend_model") or CausalLanguageModel(config.backend_config)
         self.post_init()
 
     @staticmethod
     def from_checkpoint(ckpt_path):
         from perceiver_amd.train.lit import LitCausalLanguageModel
 
-        model = LitCausalLanguageModel.load_from_checkpoint(ckpt_path).model
-        hgf_config = PerceiverCausalLanguageModelConfig(model.config)
-        hgf_config.is_decoder = True
-        hgf_model = PerceiverCausalLanguageModel(hgf_config)
-        hgf_model.backend_model.load_state_dict(model.state_dict())
-        return hgf_model
+        return wrap_lit_checkpoint(LitCausalLanguageModel, PerceiverCausalLanguageModel,
+                                   ckpt_path, is_decoder=True)
 
 
 AutoConfig.register(PerceiverCausalLanguageModelConfig.model_type, PerceiverCausalLanguageModelConfig)
@@ -53,10 +55,6 @@ AutoModelForCausalLM.register(PerceiverCausalLanguageModelConfig, PerceiverCausa
 
 
 def convert_checkpoint(save_dir, ckpt_url, tokenizer_name, **kwargs):
-    """LitCausalLanguageModel .ckpt -> persistent PerceiverCausalLanguageModel dir
-    (saves a left-padding tokenizer alongside)."""
-    tokenizer = AutoTokenizer.from_pretrained(tokenizer_name, padding_side="left", verbose=False)
-    tokenizer.save_pretrained(save_dir, **kwargs)
-    model = PerceiverCausalLanguageModel.from_checkpoint(ckpt_url)
-    model.config.tokenizer_class = tokenizer.__class__.__name__
-    model.save_pretrained(save_dir, **kwargs)
+    """Training .ckpt -> persistent 🤗 dir (left-padding tokenizer included)."""
+    save_with_tokenizer(PerceiverCausalLanguageModel.from_checkpoint(ckpt_url),
+                        tokenizer_name, save_dir, padding_side="left", **kwargs)

@@ -19,13 +19,12 @@ from perceiver_amd.models.hf_base import (
     copy_params,
     copy_self_attention_block_params,
 )
+from perceiver_amd.core import PerceiverDecoder, PerceiverEncoder
 from perceiver_amd.models.vision.optical_flow import (
     OpticalFlow,
     OpticalFlowConfig,
     OpticalFlowDecoderConfig,
     OpticalFlowEncoderConfig,
-    PerceiverDecoder,
-    PerceiverEncoder,
 )
 
 

@@ -119,6 +119,60 @@ DEVINL void stage_rm_sub16(const unsigned short* __restrict__ src, long src_stri
     }
 }
 
+// ---- T14 async-STAGE split (exact-template fast path) ----
+// issue: clamped-row unconditional vector loads into registers — the loads go
+// out early and their vmcnt waits land at the write call one compute phase
+// later, hiding HBM latency under the MFMA work (the synchronous stage_* path
+// above pays ~2k cycles of serial staging per tile). Only valid when the
+// runtime dims match the template exactly (every granule full-width).
+template <int ROWS_TILE, int DD, int NG>
+DEVINL void issue_tile(short8v (&st)[NG], const unsigned short* __restrict__ src,
+                       long sstride, int row_limit, int tid) {
+    constexpr int GPR = DD / 8;
+#pragma unroll
+    for (int i = 0; i < NG; ++i) {
+        int g = tid + i * 256;
+        int row = g / GPR, c0 = (g % GPR) * 8;
+        int rowc = min(row, row_limit - 1);
+        st[i] = *reinterpret_cast<const short8v*>(src + (long)rowc * sstride + c0);
+    }
+}
+
+// write halves: spill the staged registers into the row-major (and subtiled)
+// images, zero-filling clamp-duplicated tail rows
+template <int ROWS_TILE, int DD, int NG>
+DEVINL void write_rm_sub16(const short8v (&st)[NG], char* lds_rm, int ldst_bytes,
+                           char* lds16, int rows_valid, int tid) {
+    constexpr int GPR = DD / 8;
+    constexpr int SUBE = ROWS_TILE * 16 + 8;
+    const bool tail = rows_valid < ROWS_TILE;
+#pragma unroll
+    for (int i = 0; i < NG; ++i) {
+        int g = tid + i * 256;
+        int row = g / GPR, c0 = (g % GPR) * 8;
+        short8v val = st[i];
+        if (tail && row >= rows_valid) val = short8v{};
+        *reinterpret_cast<short8v*>(lds_rm + row * ldst_bytes + c0 * 2) = val;
+        *reinterpret_cast<short8v*>(
+            lds16 + ((c0 / 16) * SUBE + row * 16 + (c0 % 16)) * 2) = val;
+    }
+}
+
+template <int ROWS_TILE, int DD, int NG>
+DEVINL void write_rm(const short8v (&st)[NG], char* lds_rm, int ldst_bytes,
+                     int rows_valid, int tid) {
+    constexpr int GPR = DD / 8;
+    const bool tail = rows_valid < ROWS_TILE;
+#pragma unroll
+    for (int i = 0; i < NG; ++i) {
+        int g = tid + i * 256;
+        int row = g / GPR, c0 = (g % GPR) * 8;
+        short8v val = st[i];
+        if (tail && row >= rows_valid) val = short8v{};
+        *reinterpret_cast<short8v*>(lds_rm + row * ldst_bytes + c0 * 2) = val;
+    }
+}
+
 // stage (rows_tile x d) tile row-major into LDS (row stride ldst_bytes), zero-pad
 template <int ROWS_TILE>
 DEVINL void stage_rm(const unsigned short* __restrict__ src, long src_stride,
@@ -250,12 +304,35 @@ __global__ void flash_dq_kernel(
     }
     const unsigned int drop_thresh = (unsigned int)(drop_p * 4294967296.0);
 
+    // T14 split staging on exact template matches: K/V loads for tile t+1 fly
+    // during tile t's compute instead of serializing between the barriers
+    constexpr bool kFast = (DMAX % 32 == 0) && (DVMAX % 32 == 0) && DMAX <= 64 &&
+                           ((TILE * DMAX) % 2048 == 0) && ((TILE * DVMAX) % 2048 == 0);
+    constexpr int NG_K = kFast ? (TILE * DMAX) / 2048 : 1;
+    constexpr int NG_V = kFast ? (TILE * DVMAX) / 2048 : 1;
+    short8v st_k[NG_K], st_v[NG_V];
+    const bool fast = kFast && D == DMAX && Dv == DVMAX;
+    const int n_tiles = (kv_end > kv_begin) ? (kv_end - kv_begin + TILE - 1) / TILE : 0;
+    const int kv_last = kv_begin + (n_tiles > 0 ? (n_tiles - 1) * TILE : 0);
+    if (fast && n_tiles > 0) {
+        issue_tile<TILE, DMAX>(st_k, kbase + (long)kv_begin * ksn, ksn, Lk - kv_begin, tid);
+        issue_tile<TILE, DVMAX>(st_v, vbase + (long)kv_begin * vsn, vsn, Lk - kv_begin, tid);
+    }
+
     for (int kv0 = kv_begin; kv0 < kv_end; kv0 += TILE) {
         int rows_valid = min(TILE, Lk - kv0);
         __syncthreads();
-        stage_rm_sub16<TILE>(kbase + (long)kv0 * ksn, ksn, rows_valid, D, d_pad,
-                             k_lds, k_stride, kt16_lds, tid);
-        stage_rm<TILE>(vbase + (long)kv0 * vsn, vsn, rows_valid, Dv, dv_pad, v_lds, v_stride, tid);
+        if (fast) {
+            write_rm_sub16<TILE, DMAX>(st_k, k_lds, k_stride, kt16_lds, rows_valid, tid);
+            write_rm<TILE, DVMAX>(st_v, v_lds, v_stride, rows_valid, tid);
+            int kv_n = min(kv0 + TILE, kv_last);
+            issue_tile<TILE, DMAX>(st_k, kbase + (long)kv_n * ksn, ksn, Lk - kv_n, tid);
+            issue_tile<TILE, DVMAX>(st_v, vbase + (long)kv_n * vsn, vsn, Lk - kv_n, tid);
+        } else {
+            stage_rm_sub16<TILE>(kbase + (long)kv0 * ksn, ksn, rows_valid, D, d_pad,
+                                 k_lds, k_stride, kt16_lds, tid);
+            stage_rm<TILE>(vbase + (long)kv0 * vsn, vsn, rows_valid, Dv, dv_pad, v_lds, v_stride, tid);
+        }
         __syncthreads();
         // wave-uniform mask hoist (see flash_fwd.hip): interior tiles of the
         // no-pad case skip the per-element mask chain entirely
@@ -370,7 +447,7 @@ __global__ void flash_dq_kernel(
 // grid.x over KV blocks (QH*16 keys per wave), grid.y = B*H.
 // Loops over TILE-row Q tiles; stages Q row-major + Q^T, dO row-major + dO^T.
 template <int DMAX, int DVMAX, int TILE, int QH>
-__launch_bounds__(256)
+__launch_bounds__(256, DMAX <= 64 ? 2 : 1)
 __global__ void flash_dkv_kernel(
     const unsigned short* __restrict__ qp, const unsigned short* __restrict__ kp,
     const unsigned short* __restrict__ vp, const unsigned short* __restrict__ dop,
@@ -480,13 +557,36 @@ __global__ void flash_dkv_kernel(
     }
     const unsigned int drop_thresh = (unsigned int)(drop_p * 4294967296.0);
 
+    // T14 split staging (see flash_dq_kernel): Q/dO tile t+1 loads fly under
+    // tile t's MFMA work on exact template matches
+    constexpr bool kFast = (DMAX % 32 == 0) && (DVMAX % 32 == 0) && DMAX <= 64 &&
+                           ((TILE * DMAX) % 2048 == 0) && ((TILE * DVMAX) % 2048 == 0);
+    constexpr int NG_Q = kFast ? (TILE * DMAX) / 2048 : 1;
+    constexpr int NG_DO = kFast ? (TILE * DVMAX) / 2048 : 1;
+    short8v st_q[NG_Q], st_do[NG_DO];
+    const bool fast = kFast && D == DMAX && Dv == DVMAX;
+    const int nq_tiles = (Nq > q_start) ? (Nq - q_start + TILE - 1) / TILE : 0;
+    const int qt_last = q_start + (nq_tiles > 0 ? (nq_tiles - 1) * TILE : 0);
+    if (fast && nq_tiles > 0) {
+        issue_tile<TILE, DMAX>(st_q, qbase + (long)q_start * qsn, qsn, Nq - q_start, tid);
+        issue_tile<TILE, DVMAX>(st_do, dobase + (long)q_start * Dv, Dv, Nq - q_start, tid);
+    }
+
     for (int qt0 = q_start; qt0 < Nq; qt0 += TILE) {
         int rows_valid = min(TILE, Nq - qt0);
         __syncthreads();
-        stage_rm_sub16<TILE>(qbase + (long)qt0 * qsn, qsn, rows_valid, D, d_pad,
-                             q_lds, q_stride, q16_lds, tid);
-        stage_rm_sub16<TILE>(dobase + (long)qt0 * Dv, Dv, rows_valid, Dv, dv_pad,
-                             do_lds, do_stride, do16_lds, tid);
+        if (fast) {
+            write_rm_sub16<TILE, DMAX>(st_q, q_lds, q_stride, q16_lds, rows_valid, tid);
+            write_rm_sub16<TILE, DVMAX>(st_do, do_lds, do_stride, do16_lds, rows_valid, tid);
+            int qt_n = min(qt0 + TILE, qt_last);
+            issue_tile<TILE, DMAX>(st_q, qbase + (long)qt_n * qsn, qsn, Nq - qt_n, tid);
+            issue_tile<TILE, DVMAX>(st_do, dobase + (long)qt_n * Dv, Dv, Nq - qt_n, tid);
+        } else {
+            stage_rm_sub16<TILE>(qbase + (long)qt0 * qsn, qsn, rows_valid, D, d_pad,
+                                 q_lds, q_stride, q16_lds, tid);
+            stage_rm_sub16<TILE>(dobase + (long)qt0 * Dv, Dv, rows_valid, Dv, dv_pad,
+                                 do_lds, do_stride, do16_lds, tid);
+        }
         __syncthreads();
         // wave-uniform mask hoist: interior q-tiles with no padded/overhang
         // keys skip the per-element mask chain (kernels are VALU/wait-bound)
