@@ -1,10 +1,16 @@
-"""Batch collators for text tasks. Batch contract (parity with reference
-data/text/collator.py): calling a collator returns ``(labels, input_ids, pad_mask)``
-with ``pad_mask=True`` at padding positions."""
+"""Batch collators for the text tasks.
+
+Every collator is a callable over tokenized examples producing the batch
+contract the task models train on: ``(labels, input_ids, pad_mask)`` with
+``pad_mask == True`` at padding positions (the attention-mask inversion
+happens here, once). Masking behavior mirrors the reference's
+data/text/collator.py: whole-word masking selects words at ``mask_prob`` and
+applies the BERT-style 80/10/10 mask/random/keep split per WORD (every token
+of a selected word receives the same treatment).
+"""
 from __future__ import annotations
 
-from collections import defaultdict
-from typing import Optional
+from typing import Dict, List, Optional
 
 import numpy as np
 import torch
@@ -18,34 +24,49 @@ from transformers.utils import PaddingStrategy
 
 
 class Collator:
+    """Base: subclasses implement ``collate`` returning a 🤗 batch dict."""
+
     def collate(self, examples):
         raise NotImplementedError
 
     def __call__(self, examples):
-        result = self.collate(examples)
-        return result["labels"], result["input_ids"], ~result["attention_mask"].type(torch.bool)
+        batch = self.collate(examples)
+        pad_mask = ~batch["attention_mask"].type(torch.bool)
+        return batch["labels"], batch["input_ids"], pad_mask
 
 
 class RandomTruncateCollator(Collator):
-    """Randomly right-truncates each batch down to >= min_seq_len tokens."""
+    """Wraps a collator and randomly right-truncates each batch.
+
+    The surviving length is uniform over [min_seq_len, seq_len); used to make
+    the MLM robust to variable-length inference inputs.
+    """
 
     def __init__(self, collator: Collator, min_seq_len: int):
-        self.collator = collator
+        self.inner = collator
         self.min_seq_len = min_seq_len
 
+    @property
+    def collator(self):  # underlying collator (API parity)
+        return self.inner
+
     def collate(self, examples):
-        result = self.collator.collate(examples)
-        seq_len = result["input_ids"].shape[1]
-        if seq_len <= self.min_seq_len:
-            return result
-        drop = int(torch.randint(1, seq_len - self.min_seq_len + 1, size=(1,)))
-        for key in ("labels", "input_ids", "attention_mask"):
-            result[key] = result[key][:, :-drop]
-        return result
+        batch = self.inner.collate(examples)
+        seq_len = batch["input_ids"].shape[1]
+        if seq_len > self.min_seq_len:
+            drop = int(torch.randint(1, seq_len - self.min_seq_len + 1, size=(1,)))
+            for key in ("labels", "input_ids", "attention_mask"):
+                batch[key] = batch[key][:, :-drop]
+        return batch
 
 
 class DefaultCollator(Collator):
-    """Pads/truncates via tokenizer.prepare_for_model; passes through label keys."""
+    """Pad/truncate through ``tokenizer.prepare_for_model``.
+
+    ``add_special_tokens`` stays off — the chunked training records already
+    carry their special tokens, and re-adding would duplicate them. Label
+    keys (classification targets, CLM label_ids) pass through.
+    """
 
     label_keys = ("label", "labels")
 
@@ -55,33 +76,48 @@ class DefaultCollator(Collator):
         self.max_seq_len = max_seq_len
 
     def collate(self, examples):
-        cur_length = max(len(e["input_ids"]) for e in examples)
-        max_length = min(cur_length, self.max_seq_len) if self.max_seq_len else cur_length
-        return self.collator([self._prepare(e, max_length) for e in examples])
+        longest = max(len(e["input_ids"]) for e in examples)
+        target = min(longest, self.max_seq_len) if self.max_seq_len else longest
+        return self.collator([self._prepare(e, target) for e in examples])
 
-    def _prepare(self, example, max_length):
-        prepared = self._prepare_sequence(example["input_ids"], max_length)
-        if "label_ids" in example:
-            prepared["label_ids"] = self._prepare_sequence(example["label_ids"], max_length)["input_ids"]
-        for key in self.label_keys:
-            if key in example:
-                prepared[key] = example[key]
-        return prepared
-
-    def _prepare_sequence(self, sequence, max_length):
+    def _pad_or_truncate(self, sequence, target_len):
         return self.tokenizer.prepare_for_model(
             sequence,
             add_special_tokens=False,
             return_token_type_ids=False,
             padding=False if self.tokenizer.pad_token is None else PaddingStrategy.MAX_LENGTH,
-            max_length=max_length,
+            max_length=target_len,
             truncation=True,
         )
 
+    def _prepare(self, example, target_len):
+        out = self._pad_or_truncate(example["input_ids"], target_len)
+        if "label_ids" in example:
+            out["label_ids"] = self._pad_or_truncate(example["label_ids"], target_len)["input_ids"]
+        for key in self.label_keys:
+            if key in example:
+                out[key] = example[key]
+        return out
+
+
+def word_spans(word_ids: List[Optional[int]]) -> List[List[int]]:
+    """Token-index groups per word: consecutive equal non-None ids form one
+    word; None (special tokens) breaks and never joins a word."""
+    spans: List[List[int]] = []
+    previous = object()
+    for idx, wid in enumerate(word_ids):
+        if wid is None:
+            previous = None
+            continue
+        if wid != previous:
+            spans.append([])
+            previous = wid
+        spans[-1].append(idx)
+    return spans
+
 
 class WordMaskingCollator(Collator):
-    """Whole-word masking with the 80/10/10 mask/random/keep split applied per word
-    (all tokens of a selected word get the same treatment)."""
+    """Whole-word masking (80/10/10 applied per word, not per token)."""
 
     def __init__(self, tokenizer: PreTrainedTokenizerFast, mask_prob: float = 0.15):
         self.collator = DataCollatorWithPadding(tokenizer)
@@ -97,38 +133,32 @@ class WordMaskingCollator(Collator):
             self.mask_words_1(example)
         return examples
 
-    def mask_words_1(self, example):
-        # mutates its argument (used both dynamically per batch and for static masking)
-        word_ids = example.pop("word_ids")
+    def mask_words_1(self, example: Dict) -> Dict:
+        """Mutates ``example`` in place (the same routine serves dynamic
+        per-batch masking and one-shot static masking)."""
+        spans = word_spans(example.pop("word_ids"))
         input_ids = example["input_ids"]
         labels = [-100] * len(input_ids)
 
-        mapping = defaultdict(list)
-        current_word_index = -1
-        current_word_id = None
-        for idx, word_id in enumerate(word_ids):
-            if word_id is not None:
-                if word_id != current_word_id:
-                    current_word_id = word_id
-                    current_word_index += 1
-                mapping[current_word_index].append(idx)
-
-        mask = np.random.binomial(1, self.mask_prob, len(mapping))
-        for word_index in np.where(mask)[0]:
-            rand_nr = np.random.rand(2)
-            for idx in mapping[word_index]:
+        selected = np.random.binomial(1, self.mask_prob, len(spans))
+        for word_index in np.where(selected)[0]:
+            u_mask, u_rand = np.random.rand(2)
+            for idx in spans[word_index]:
                 labels[idx] = input_ids[idx]
-                if rand_nr[0] < 0.8:
-                    input_ids[idx] = self.mask_token_id        # 80%: mask token(s)
-                elif rand_nr[1] < 0.5:
-                    input_ids[idx] = np.random.randint(self.vocab_size)  # 10%: random
-                # else 10%: unchanged
+                if u_mask < 0.8:
+                    input_ids[idx] = self.mask_token_id    # 80%: mask token(s)
+                elif u_rand < 0.5:
+                    # 10%: an independent random token per position
+                    input_ids[idx] = np.random.randint(self.vocab_size)
+                # remaining 10%: tokens kept unchanged
 
         example["labels"] = labels
         return example
 
 
 class TokenMaskingCollator(Collator):
+    """Per-token masking via 🤗's standard MLM collator."""
+
     def __init__(self, tokenizer: PreTrainedTokenizerFast, mask_prob: float = 0.15):
         self.collator = DataCollatorForLanguageModeling(tokenizer, mlm_probability=mask_prob)
 

@@ -1,33 +1,46 @@
-"""Support for the (non-fast) PerceiverTokenizer: whitespace-boundary word ids for
-whole-word masking (parity with reference data/text/utils.py:12-39)."""
+"""Word-boundary recovery for the byte-level PerceiverTokenizer.
+
+The PerceiverTokenizer has no fast (Rust) implementation, so 🤗 provides no
+``word_ids()`` for it — but whole-word masking needs word membership per
+token. Byte tokens make this recoverable from the ids alone: a "word" is a
+maximal run of non-whitespace tokens, and the whitespace run *preceding* a
+word is grouped with that word (so masking a word also masks its leading
+spaces, matching the reference's data/text/utils.py:12-39 semantics).
+Special tokens map to None and always break words.
+"""
 from __future__ import annotations
 
 import string
 
 
 class PerceiverTokenizerUtil:
+    """Derives word ids from PerceiverTokenizer output for whole-word masking."""
+
     def __init__(self, tokenizer):
         self.tokenizer = tokenizer
-        self.whitespace_ids = set(tokenizer(string.whitespace, add_special_tokens=False).input_ids)
+        encoded = tokenizer(string.whitespace, add_special_tokens=False).input_ids
+        self.whitespace_ids = set(encoded)
 
     def word_ids(self, token_ids):
-        """Word ids from token ids using whitespace boundaries. Whitespaces preceding
-        a word share its word id; special tokens get None; distinct words are
-        guaranteed distinct ids (not necessarily consecutive)."""
-        word_ids = []
-        curr_id = 0
-        special_mask = self.tokenizer.get_special_tokens_mask(token_ids, already_has_special_tokens=True)
-        regular_token = True
-        for i, token_id in enumerate(token_ids):
-            if special_mask[i]:
-                word_ids.append(None)
-                curr_id += 1
-            elif token_id in self.whitespace_ids:
-                if regular_token:
-                    regular_token = False
-                    curr_id += 1
-                word_ids.append(curr_id)
+        """One id per token; None for specials. Distinct words get distinct
+        (not necessarily consecutive) ids."""
+        specials = self.tokenizer.get_special_tokens_mask(
+            token_ids, already_has_special_tokens=True)
+        ids = []
+        word = 0
+        in_word = True  # True while consuming non-whitespace tokens
+        for tok, is_special in zip(token_ids, specials):
+            if is_special:
+                ids.append(None)
+                word += 1
+                continue
+            if tok in self.whitespace_ids:
+                if in_word:
+                    # first whitespace after a word opens the NEXT word's id:
+                    # leading spaces share the id of the word that follows
+                    in_word = False
+                    word += 1
             else:
-                regular_token = True
-                word_ids.append(curr_id)
-        return word_ids
+                in_word = True
+            ids.append(word)
+        return ids

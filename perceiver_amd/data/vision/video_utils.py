@@ -1,6 +1,11 @@
-"""Video frame IO for the optical-flow pipeline (parity: reference
-data/vision/video_utils.py). cv2 is imported lazily; environments without OpenCV can
-use the rest of the vision stack."""
+"""Video frame IO for the optical-flow pipeline.
+
+Thin OpenCV shims: decode a video into RGB frames (and consecutive-frame
+pairs, the flow model's input unit), and encode predicted-flow renderings back
+to mp4. cv2 is imported lazily so environments without OpenCV can still use
+the rest of the vision stack. Behavior mirrors the reference's
+data/vision/video_utils.py.
+"""
 from __future__ import annotations
 
 import os
@@ -18,36 +23,39 @@ def _cv2():
 
 
 def read_video_frames(video_path: str) -> List[np.ndarray]:
+    """Decode every frame of ``video_path`` as an RGB ndarray."""
     cv2 = _cv2()
     if not os.path.exists(video_path):
         raise ValueError(f"Invalid video path supplied. Path '{video_path}' does not exist.")
-    frames = []
-    cap = cv2.VideoCapture(video_path)
+    capture = cv2.VideoCapture(video_path)
+    decoded: List[np.ndarray] = []
     try:
-        while cap.isOpened():
-            ret, frame = cap.read()
-            if not ret:
+        while capture.isOpened():
+            ok, bgr = capture.read()
+            if not ok:
                 break
-            frames.append(cv2.cvtColor(frame, cv2.COLOR_BGR2RGB))
+            decoded.append(cv2.cvtColor(bgr, cv2.COLOR_BGR2RGB))
     finally:
-        if cap is not None:
-            cap.release()
-    return frames
+        if capture is not None:
+            capture.release()
+    return decoded
 
 
 def read_video_frame_pairs(video_path: str) -> List[Tuple[np.ndarray, np.ndarray]]:
+    """Consecutive frame pairs (t, t+1) — one optical-flow model input each."""
     frames = read_video_frames(video_path)
-    return list(zip(frames, frames[1:]))
+    return [(a, b) for a, b in zip(frames, frames[1:])]
 
 
 def write_video(video_path: str, frames: List[np.ndarray], fps: int) -> None:
+    """Encode RGB ``frames`` to an mp4 at ``fps``."""
     cv2 = _cv2()
-    _, ext = os.path.splitext(video_path)
-    if ext != ".mp4":
+    if os.path.splitext(video_path)[1] != ".mp4":
         raise ValueError("Invalid video path supplied. Only files of type 'mp4' are supported.")
-    frame_shape = frames[0].shape
-    video = cv2.VideoWriter(video_path, cv2.VideoWriter_fourcc(*"mp4v"), fps,
-                            (frame_shape[1], frame_shape[0]))
-    for frame in frames:
-        video.write(cv2.cvtColor(frame, cv2.COLOR_RGB2BGR))
-    video.release()
+    height, width = frames[0].shape[:2]
+    sink = cv2.VideoWriter(video_path, cv2.VideoWriter_fourcc(*"mp4v"), fps, (width, height))
+    try:
+        for rgb in frames:
+            sink.write(cv2.cvtColor(rgb, cv2.COLOR_RGB2BGR))
+    finally:
+        sink.release()

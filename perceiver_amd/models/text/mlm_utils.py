@@ -1,31 +1,42 @@
-"""MaskFiller: top-k predictions for <mask> positions (parity: reference
-text/mlm/utils.py)."""
+"""Mask-filling helper for qualitative MLM evaluation.
+
+Turns ``<mask>`` placeholders in plain text into the tokenizer's mask token,
+runs the model once, and decodes the top-k replacement candidates per masked
+position — the table logged at validation end mirrors the reference's
+text/mlm/utils.py behavior.
+"""
 from __future__ import annotations
 
 import torch
 
 
 class MaskFiller:
+    """Fill ``<mask>`` slots in a batch of strings with model predictions."""
+
     def __init__(self, preprocessor):
         self.preprocessor = preprocessor
 
+    @property
+    def _tokenizer(self):
+        return self.preprocessor.tokenizer
+
     def fill(self, model, masked_text_batch, num_predictions, device="cpu"):
-        masked_text_batch = [
-            ms.replace("<mask>", self.preprocessor.tokenizer.mask_token) for ms in masked_text_batch
-        ]
-        xs, ms = self.preprocessor.preprocess_batch(masked_text_batch)
-        xs = xs.to(device)
-        ms = ms.to(device)
+        """Returns (normalized inputs, per-sample list of k filled strings)."""
+        texts = [s.replace("<mask>", self._tokenizer.mask_token) for s in masked_text_batch]
+        token_ids, pad_mask = self.preprocessor.preprocess_batch(texts)
+        token_ids = token_ids.to(device)
 
         with torch.no_grad():
-            x_logits = model(xs, ms)
+            logits = model(token_ids, pad_mask.to(device))
 
-        pred_mask = xs == self.preprocessor.tokenizer.mask_token_id
-        pred_ids = torch.topk(x_logits[pred_mask, :], k=num_predictions, dim=1).indices
+        at_mask = token_ids == self._tokenizer.mask_token_id
+        candidates = torch.topk(logits[at_mask, :], k=num_predictions, dim=1).indices
 
-        results = []
-        for i in range(num_predictions):
-            xs[pred_mask] = pred_ids[:, i]
-            results.append(self.preprocessor.tokenizer.batch_decode(xs, skip_special_tokens=True))
+        # decode once per rank, substituting every masked position in place
+        decoded_per_rank = []
+        for rank in range(num_predictions):
+            token_ids[at_mask] = candidates[:, rank]
+            decoded_per_rank.append(self._tokenizer.batch_decode(token_ids, skip_special_tokens=True))
 
-        return masked_text_batch, list(map(list, zip(*results)))
+        per_sample = [list(ranks) for ranks in zip(*decoded_per_rank)]
+        return texts, per_sample
