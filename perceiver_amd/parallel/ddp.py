@@ -1,13 +1,22 @@
 """Native data-parallel gradient reduction over RCCL/xGMI.
 
-Replaces Lightning/torch-DDP's implicit reducer with an explicit
-bucketed all-reduce: parameters are grouped (in reverse registration order, which
-approximates backward completion order) into flat buckets; a bucket's all-reduce
-launches asynchronously as soon as its last gradient materializes, overlapping the
-remaining backward compute. On ROCm the "nccl" backend IS RCCL and the collectives
-ride xGMI; each MI355X has 7 point-to-point xGMI links (~153 GB/s each), so ring
-all-reduce is per-link bound — bucket sizes default to 50 MiB so several buckets
-pipeline across the links while backward proceeds.
+Replaces Lightning/torch-DDP's implicit reducer with an
+explicit bucketed all-reduce: parameters are grouped (in reverse registration
+order, approximating backward completion order) into flat buckets, and a
+bucket's all-reduce launches asynchronously as soon as its last gradient
+materializes, overlapping the remaining backward compute. On ROCm the "nccl"
+backend IS RCCL and the collectives ride xGMI; each MI355X has 7
+point-to-point xGMI links (~153 GB/s each), so ring all-reduce is per-link
+bound — bucket sizes default to 50 MiB so several buckets pipeline across the
+links while backward proceeds.
+
+Gradients live as VIEWS into the bucket's flat buffer (torch DDP's
+``gradient_as_bucket_view``): ``p.grad`` is installed as a slice of the flat
+tensor up front, autograd accumulates straight into the bucket, and the
+all-reduce result is in place — no per-step gradient copy in either
+direction. The one contract this imposes: zero gradients through
+``reducer.zero_grad()`` (``optimizer.zero_grad(set_to_none=True)`` would
+detach the views); the Trainer does this automatically.
 """
 from __future__ import annotations
 
@@ -18,17 +27,24 @@ import torch.distributed as dist
 
 
 class _Bucket:
-    def __init__(self, params: List[torch.nn.Parameter], device, dtype):
+    """A contiguous gradient buffer for a group of same-dtype/device params."""
+
+    def __init__(self, params: List[torch.nn.Parameter]):
         self.params = params
-        self.numel = sum(p.numel() for p in params)
-        self.flat = torch.zeros(self.numel, device=device, dtype=dtype)
-        self.offsets = []
-        off = 0
+        self.offsets: List[int] = []
+        total = 0
         for p in params:
-            self.offsets.append(off)
-            off += p.numel()
+            self.offsets.append(total)
+            total += p.numel()
+        self.numel = total
+        self.flat = torch.zeros(total, device=params[0].device, dtype=params[0].dtype)
         self.ready = 0
         self.work: Optional[dist.Work] = None
+
+    def install_views(self):
+        """Point every parameter's .grad at its slice of the flat buffer."""
+        for p, off in zip(self.params, self.offsets):
+            p.grad = self.flat[off: off + p.numel()].view_as(p)
 
     def reset(self):
         self.ready = 0
@@ -40,18 +56,19 @@ class BucketedGradReducer:
 
     Usage per step::
 
+        reducer.zero_grad()      # zeros the buckets, keeps the grad views
         loss.backward()          # hooks fire, buckets all-reduce asynchronously
-        reducer.finalize()       # wait + write averaged grads back
+        reducer.finalize()       # wait for the in-flight reductions
         optimizer.step()
 
-    Gradients are averaged over the world size. Parameters that do not require
-    grad are skipped. Works on any torch.distributed backend (RCCL on GPU,
-    gloo in CPU tests).
+    Gradients are averaged over the world size. Parameters that do not
+    require grad are skipped. Works on any torch.distributed backend (RCCL on
+    GPU, gloo in CPU tests).
 
     Gradient accumulation: call ``set_sync(False)`` for non-boundary
-    micro-batches (hooks become no-ops and gradients just accumulate locally),
-    ``set_sync(True)`` before the last micro-batch's backward; the boundary
-    backward then reduces the full accumulated gradients as usual.
+    micro-batches (hooks become no-ops; grads accumulate locally in the
+    buckets), ``set_sync(True)`` before the boundary micro-batch's backward;
+    that backward then reduces the fully-accumulated buckets.
     """
 
     def __init__(self, module: torch.nn.Module, bucket_cap_mb: float = 50.0,
@@ -62,8 +79,8 @@ class BucketedGradReducer:
         self.world_size = dist.get_world_size(process_group)
         self.sync = True
 
-        # unique params in reverse registration order (approximate backward order);
-        # shared/tied params appear once
+        # unique params in reverse registration order (approximate backward
+        # order); shared/tied params appear once
         seen = set()
         params: List[torch.nn.Parameter] = []
         for p in reversed(list(module.parameters())):
@@ -78,25 +95,35 @@ class BucketedGradReducer:
         for p in params:
             sz = p.numel() * p.element_size()
             # close the bucket on capacity overflow and on any dtype/device
-            # change: a bucket's flat buffer is uniform, so mixing would
-            # silently cast (and all-reduce) grads in the wrong dtype/device
+            # change: a flat buffer is uniform, so mixing would silently cast
+            # (and all-reduce) grads in the wrong dtype/device
             if cur and (cur_bytes + sz > cap
                         or p.dtype != cur[0].dtype or p.device != cur[0].device):
-                self.buckets.append(_Bucket(cur, cur[0].device, cur[0].dtype))
+                self.buckets.append(_Bucket(cur))
                 cur, cur_bytes = [], 0
             cur.append(p)
             cur_bytes += sz
         if cur:
-            self.buckets.append(_Bucket(cur, cur[0].device, cur[0].dtype))
+            self.buckets.append(_Bucket(cur))
 
         self._param_bucket = {}
         for b in self.buckets:
-            for p, off in zip(b.params, b.offsets):
-                self._param_bucket[id(p)] = (b, off)
+            b.install_views()
+            for p in b.params:
+                self._param_bucket[id(p)] = b
 
         self._hooks = [
             p.register_post_accumulate_grad_hook(self._on_grad_ready) for p in params
         ]
+
+    def zero_grad(self):
+        """Zero all bucketed gradients, re-installing views if anything
+        (e.g. a stray ``zero_grad(set_to_none=True)``) detached them."""
+        for b in self.buckets:
+            b.flat.zero_()
+            first = b.params[0]
+            if first.grad is None or first.grad.data_ptr() != b.flat.data_ptr():
+                b.install_views()
 
     def set_sync(self, sync: bool):
         """False = accumulation micro-batch (no reduction); True = boundary."""
@@ -105,15 +132,14 @@ class BucketedGradReducer:
     def _on_grad_ready(self, p: torch.nn.Parameter):
         if not self.sync:
             return
-        b, off = self._param_bucket[id(p)]
-        b.flat[off: off + p.numel()].copy_(p.grad.detach().reshape(-1))
+        b = self._param_bucket[id(p)]
         b.ready += 1
         if b.ready == len(b.params):
             b.flat.div_(self.world_size)
             b.work = dist.all_reduce(b.flat, group=self.group, async_op=True)
 
     def finalize(self):
-        """Wait for all bucket reductions and write averaged gradients back."""
+        """Wait for all bucket reductions; grads are views, nothing to copy."""
         for b in self.buckets:
             if b.ready != len(b.params):
                 raise RuntimeError(
@@ -121,29 +147,19 @@ class BucketedGradReducer:
                     "(a parameter did not receive a gradient this step)"
                 )
             b.work.wait()
-            for p, off in zip(b.params, b.offsets):
-                p.grad.detach().reshape(-1).copy_(b.flat[off: off + p.numel()])
             b.reset()
 
     def reduce_now(self):
-        """Synchronously all-reduce whatever is in ``p.grad`` right now.
+        """Synchronously all-reduce the current bucket contents.
 
         Used for a trailing partial accumulation window at epoch end, where
-        the post-accumulate hooks ran with sync=False and never filled the
-        buckets. Rare path — plain bucket-at-a-time, no overlap.
+        the hooks ran with sync=False and never launched reductions. The
+        grads are already in the buckets (views), so this is just the
+        collectives.
         """
         for b in self.buckets:
-            for p, off in zip(b.params, b.offsets):
-                g = p.grad
-                if g is None:
-                    b.flat[off: off + p.numel()].zero_()
-                else:
-                    b.flat[off: off + p.numel()].copy_(g.detach().reshape(-1))
             b.flat.div_(self.world_size)
             dist.all_reduce(b.flat, group=self.group)
-            for p, off in zip(b.params, b.offsets):
-                if p.grad is not None:
-                    p.grad.detach().reshape(-1).copy_(b.flat[off: off + p.numel()])
             b.reset()
 
     def remove(self):

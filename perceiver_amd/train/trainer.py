@@ -137,7 +137,11 @@ class Trainer:
         for batch in batches:
             batch = _move(batch, self.device)
             if micro == 0:
-                optimizer.zero_grad(set_to_none=True)
+                # the reducer's grads are bucket views: zero through it
+                if reducer is not None:
+                    reducer.zero_grad()
+                else:
+                    optimizer.zero_grad(set_to_none=True)
             boundary = (micro + 1) == accum
             if reducer is not None:
                 reducer.set_sync(boundary)
@@ -212,7 +216,11 @@ class Trainer:
             for batch in train_loader:
                 batch = _move(batch, self.device)
                 if micro == 0:
-                    optimizer.zero_grad(set_to_none=True)
+                    # the reducer's grads are bucket views: zero through it
+                    if reducer is not None:
+                        reducer.zero_grad()
+                    else:
+                        optimizer.zero_grad(set_to_none=True)
                 boundary = (micro + 1) == accum
                 if reducer is not None:
                     reducer.set_sync(boundary)
