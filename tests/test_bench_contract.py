@@ -47,3 +47,14 @@ def test_bench_json_contract_two_rank_gloo():
     assert rec["n_gpus"] == 2
     assert rec["config"]["parallelism"] == "dp2"
     assert rec["value"] > 0
+
+
+def test_bench_json_contract_eight_rank_gloo():
+    """The exact code path the driver times at N=8 (one rank per GPU over
+    RCCL) executed end-to-end here over 8 gloo CPU ranks — reducer, metric
+    aggregation and the JSON contract all exercised at the full rank count."""
+    rec = _run_bench(["--tiny", "--device", "cpu", "--steps", "1", "--warmup", "0"],
+                     nproc=8, timeout=600)
+    assert rec["n_gpus"] == 8
+    assert rec["config"]["parallelism"] == "dp8"
+    assert rec["value"] > 0
