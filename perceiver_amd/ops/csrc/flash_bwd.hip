@@ -447,7 +447,7 @@ __global__ void flash_dq_kernel(
 // grid.x over KV blocks (QH*16 keys per wave), grid.y = B*H.
 // Loops over TILE-row Q tiles; stages Q row-major + Q^T, dO row-major + dO^T.
 template <int DMAX, int DVMAX, int TILE, int QH>
-__launch_bounds__(256, DMAX <= 64 ? 2 : 1)
+__launch_bounds__(256, DMAX <= 128 ? 2 : 1)
 __global__ void flash_dkv_kernel(
     const unsigned short* __restrict__ qp, const unsigned short* __restrict__ kp,
     const unsigned short* __restrict__ vp, const unsigned short* __restrict__ dop,
@@ -559,7 +559,7 @@ __global__ void flash_dkv_kernel(
 
     // T14 split staging (see flash_dq_kernel): Q/dO tile t+1 loads fly under
     // tile t's MFMA work on exact template matches
-    constexpr bool kFast = (DMAX % 32 == 0) && (DVMAX % 32 == 0) && DMAX <= 64 &&
+    constexpr bool kFast = (DMAX % 32 == 0) && (DVMAX % 32 == 0) && DMAX <= 128 &&
                            ((TILE * DMAX) % 2048 == 0) && ((TILE * DVMAX) % 2048 == 0);
     constexpr int NG_Q = kFast ? (TILE * DMAX) / 2048 : 1;
     constexpr int NG_DO = kFast ? (TILE * DVMAX) / 2048 : 1;
@@ -853,7 +853,9 @@ std::vector<torch::Tensor> flash_bwd(torch::Tensor dout, torch::Tensor q, torch:
     unsigned long long sd = (unsigned long long)seed;
     // <DMAX, DVMAX, DQ_TILE, DQ_QH, DKV_TILE, DKV_QH> — tiles/QH balance LDS
     // occupancy against the 256-VGPR budget per template
-    if (D <= 32 && Dv <= 160)
+    if (D <= 32 && Dv <= 96)
+        launch_flash_bwd<32, 96, 64, 2, 64, 1>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
+    else if (D <= 32 && Dv <= 160)
         launch_flash_bwd<32, 160, 64, 2, 64, 1>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
     else if (D <= 64 && Dv <= 64)
         launch_flash_bwd<64, 64, 64, 2, 64, 1>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);

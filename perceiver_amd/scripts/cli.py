@@ -29,8 +29,18 @@ def _set_path(tree: Dict, path: str, value: Any) -> None:
     keys = path.split(".")
     node = tree
     for k in keys[:-1]:
-        node = node.setdefault(k, {})
-    node[keys[-1]] = value
+        nxt = node.get(k)
+        if not isinstance(nxt, dict):
+            # LightningCLI style writes a class name at the section key
+            # (--data=ImdbDataModule) before section fields; keep it aside
+            node[k] = {"_class_name": nxt} if isinstance(nxt, str) else {}
+            nxt = node[k]
+        node = nxt
+    last = keys[-1]
+    if isinstance(node.get(last), dict) and not isinstance(value, dict):
+        node[last]["_class_name"] = value
+    else:
+        node[last] = value
 
 
 def _merge(base: Dict, override: Dict) -> Dict:
@@ -48,6 +58,62 @@ def _parse_value(raw: str) -> Any:
         return yaml.safe_load(raw)
     except yaml.YAMLError:
         return raw
+
+
+def _strip_class_paths(node: Any) -> Any:
+    """Flatten LightningCLI/jsonargparse ``{class_path, init_args}`` nodes.
+
+    The reference's saved run configs (and its docs' examples) wrap
+    instantiable sections in class_path/init_args; this CLI binds by section
+    name, so the wrapper collapses to its init_args (recursively).
+    """
+    if isinstance(node, dict):
+        if "class_path" in node and set(node) <= {"class_path", "init_args"}:
+            return _strip_class_paths(node.get("init_args", {}))
+        return {k: _strip_class_paths(v) for k, v in node.items()}
+    if isinstance(node, list):
+        return [_strip_class_paths(v) for v in node]
+    return node
+
+
+# Lightning Trainer flag -> native TrainConfig field (None = accepted, ignored)
+_LIGHTNING_TRAINER_KEYS = {
+    "gradient_clip_val": "grad_clip",
+    "default_root_dir": "out_dir",
+    "accumulate_grad_batches": "accumulate_grad_batches",
+    "max_steps": "max_steps",
+    "max_epochs": "max_epochs",
+    "precision": "precision",
+    "devices": None,      # process count comes from the torchrun launch
+    "accelerator": None,
+    "strategy": None,     # DDP is explicit here (BucketedGradReducer)
+    "num_sanity_val_steps": None,
+    "logger": None,
+    "callbacks": None,
+    "log_every_n_steps": "log_every",
+    "val_check_interval": "val_every_steps",
+}
+
+_PRECISION_ALIASES = {
+    "16": "fp16", "16-mixed": "fp16", "fp16": "fp16",
+    "bf16": "bf16", "bf16-mixed": "bf16",
+    "32": "fp32", "32-true": "fp32", "fp32": "fp32",
+}
+
+
+def translate_lightning_trainer(trainer_cfg: Dict) -> Dict:
+    """Map the reference's Lightning trainer flags onto TrainConfig fields,
+    so e.g. ``--trainer.precision 16 --trainer.gradient_clip_val 0.5``
+    (examples/training/*/train.sh) run unchanged."""
+    out: Dict = {}
+    for key, value in trainer_cfg.items():
+        target = _LIGHTNING_TRAINER_KEYS.get(key, key)
+        if target is None:
+            continue
+        if target == "precision":
+            value = _PRECISION_ALIASES.get(str(value), value)
+        out[target] = value
+    return out
 
 
 def parse_cli_config(argv, defaults: Dict) -> (str, Dict):
@@ -74,11 +140,15 @@ def parse_cli_config(argv, defaults: Dict) -> (str, Dict):
             i += 2
         if key == "config":
             with open(raw) as f:
-                tree = _merge(tree, yaml.safe_load(f) or {})
+                loaded = _strip_class_paths(yaml.safe_load(f) or {})
+            tree = _merge(tree, loaded)
         else:
             _set_path(tree, key, _parse_value(raw))
 
-    return subcommand, _merge(defaults, tree)
+    tree = _merge(defaults, _strip_class_paths(tree))
+    if "trainer" in tree:
+        tree["trainer"] = translate_lightning_trainer(tree["trainer"])
+    return subcommand, tree
 
 
 def build_dataclass(cls, cfg: Dict):

@@ -43,7 +43,7 @@ class TrainConfig:
     weight_decay: float = 0.01
     betas: tuple = (0.9, 0.999)
     optimizer: str = "adamw"         # "adamw" | "master_adamw" | "lamb"
-    precision: str = "bf16"          # "bf16" (autocast on GPU) or "fp32"
+    precision: str = "bf16"          # "bf16" | "fp16" (autocast+GradScaler) | "fp32"
     grad_clip: Optional[float] = None
     accumulate_grad_batches: int = 1
     bucket_cap_mb: float = 50.0
@@ -77,11 +77,24 @@ class Trainer:
             self._log_file = open(os.path.join(self.cfg.out_dir, "metrics.jsonl"), "a")
         if config.seed is not None:
             torch.manual_seed(config.seed + get_rank())
+        # fp16 needs loss scaling; for bf16/fp32 the scaler is a no-op passthrough
+        self.scaler = torch.amp.GradScaler(
+            "cuda", enabled=(self.cfg.precision == "fp16" and self.device.type == "cuda"))
 
     # ------------------------------------------------------------------ helpers
     def _autocast(self):
-        enabled = self.cfg.precision == "bf16" and self.device.type == "cuda"
-        return torch.autocast(self.device.type, dtype=torch.bfloat16, enabled=enabled)
+        dtype = {"bf16": torch.bfloat16, "fp16": torch.float16}.get(self.cfg.precision)
+        enabled = dtype is not None and self.device.type == "cuda"
+        return torch.autocast(self.device.type, dtype=dtype or torch.bfloat16, enabled=enabled)
+
+    def _optimizer_step(self, optimizer, model_params):
+        """Clip (through the scaler when fp16) and step."""
+        if self.scaler.is_enabled():
+            self.scaler.unscale_(optimizer)
+        if self.cfg.grad_clip and not getattr(optimizer, "max_grad_norm", 0.0):
+            torch.nn.utils.clip_grad_norm_(model_params, self.cfg.grad_clip)
+        self.scaler.step(optimizer)
+        self.scaler.update()
 
     def _make_optimizer(self, model):
         params = [p for p in model.parameters() if p.requires_grad]
@@ -147,16 +160,14 @@ class Trainer:
                 reducer.set_sync(boundary)
             with self._autocast():
                 loss = step_fn(model, batch)
-            (loss / accum).backward()
+            self.scaler.scale(loss / accum).backward()
             micro += 1
             if not boundary:
                 continue
             micro = 0
             if reducer is not None:
                 reducer.finalize()
-            if self.cfg.grad_clip and not getattr(optimizer, "max_grad_norm", 0.0):
-                torch.nn.utils.clip_grad_norm_(model.parameters(), self.cfg.grad_clip)
-            optimizer.step()
+            self._optimizer_step(optimizer, model.parameters())
             if scheduler is not None:
                 scheduler.step()
             self.global_step += 1
@@ -171,9 +182,7 @@ class Trainer:
             # flush a trailing partial accumulation window (see fit())
             if reducer is not None:
                 reducer.reduce_now()
-            if self.cfg.grad_clip and not getattr(optimizer, "max_grad_norm", 0.0):
-                torch.nn.utils.clip_grad_norm_(model.parameters(), self.cfg.grad_clip)
-            optimizer.step()
+            self._optimizer_step(optimizer, model.parameters())
             if scheduler is not None:
                 scheduler.step()
             self.global_step += 1
@@ -226,16 +235,14 @@ class Trainer:
                     reducer.set_sync(boundary)
                 with self._autocast():
                     loss = task.training_step(batch, self.global_step)
-                (loss / accum).backward()
+                self.scaler.scale(loss / accum).backward()
                 micro += 1
                 if not boundary:
                     continue
                 micro = 0
                 if reducer is not None:
                     reducer.finalize()
-                if self.cfg.grad_clip and not getattr(optimizer, "max_grad_norm", 0.0):
-                    torch.nn.utils.clip_grad_norm_(task.parameters(), self.cfg.grad_clip)
-                optimizer.step()
+                self._optimizer_step(optimizer, task.parameters())
                 if scheduler is not None:
                     scheduler.step()
                 self.global_step += 1
@@ -255,9 +262,7 @@ class Trainer:
                 # are real — flush them instead of silently discarding
                 if reducer is not None:
                     reducer.reduce_now()
-                if self.cfg.grad_clip and not getattr(optimizer, "max_grad_norm", 0.0):
-                    torch.nn.utils.clip_grad_norm_(task.parameters(), self.cfg.grad_clip)
-                optimizer.step()
+                self._optimizer_step(optimizer, task.parameters())
                 if scheduler is not None:
                     scheduler.step()
                 self.global_step += 1
