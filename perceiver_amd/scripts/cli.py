@@ -55,9 +55,16 @@ def _merge(base: Dict, override: Dict) -> Dict:
 
 def _parse_value(raw: str) -> Any:
     try:
-        return yaml.safe_load(raw)
+        value = yaml.safe_load(raw)
     except yaml.YAMLError:
         return raw
+    if isinstance(value, str):
+        # YAML 1.1 misses bare scientific notation ("1e-5" stays a string)
+        try:
+            return float(value) if any(c in value for c in ".eE") and value[0] in "0123456789+-." else value
+        except ValueError:
+            return value
+    return value
 
 
 def _strip_class_paths(node: Any) -> Any:
@@ -148,6 +155,23 @@ def parse_cli_config(argv, defaults: Dict) -> (str, Dict):
     tree = _merge(defaults, _strip_class_paths(tree))
     if "trainer" in tree:
         tree["trainer"] = translate_lightning_trainer(tree["trainer"])
+    # reference-style --lr_scheduler.* section: fold into the optimizer section
+    # (warmup + schedule selection live on TrainConfig here)
+    lrs = tree.pop("lr_scheduler", None)
+    if isinstance(lrs, dict):
+        opt = tree.setdefault("optimizer", {})
+        if "warmup_steps" in lrs:
+            opt.setdefault("warmup_steps", lrs["warmup_steps"])
+        sched_name = str(lrs.get("_class_name", ""))
+        if "Cosine" in sched_name:
+            opt.setdefault("lr_schedule", "cosine")
+        elif "Constant" in sched_name:
+            opt.setdefault("lr_schedule", "constant")
+    # --optimizer=AdamW / Lamb class selection
+    opt = tree.get("optimizer")
+    if isinstance(opt, dict) and "_class_name" in opt:
+        name = str(opt.pop("_class_name")).lower()
+        opt.setdefault("optimizer", "lamb" if "lamb" in name else "adamw")
     return subcommand, tree
 
 
@@ -161,6 +185,19 @@ def build_dataclass(cls, cfg: Dict):
         if f.name in cfg:
             kwargs[f.name] = cfg[f.name]
     return cls(**kwargs)
+
+
+def _resolve_datamodule(name: str):
+    """LightningCLI-style class selection (--data=ImdbDataModule): look the
+    name up across the data packages."""
+    import importlib
+
+    for pkg in ("perceiver_amd.data.text", "perceiver_amd.data.vision",
+                "perceiver_amd.data.audio"):
+        module = importlib.import_module(pkg)
+        if hasattr(module, name):
+            return getattr(module, name)
+    raise SystemExit(f"unknown data module class: {name}")
 
 
 class CLI:
@@ -188,7 +225,12 @@ class CLI:
 
     def run(self):
         cfg = self.config
-        datamodule = self.datamodule_cls(**cfg.get("data", {}))
+        data_cfg = dict(cfg.get("data", {}))
+        class_name = data_cfg.pop("_class_name", None)
+        dm_cls = self.datamodule_cls if class_name is None else _resolve_datamodule(class_name)
+        if isinstance(cfg.get("model"), dict):
+            cfg["model"].pop("_class_name", None)
+        datamodule = dm_cls(**data_cfg)
         if self.link is not None:
             self.link(cfg, datamodule)
 
