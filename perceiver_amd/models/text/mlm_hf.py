@@ -1,9 +1,16 @@
-"""🤗 wrapper for the masked LM: AutoModelForMaskedLM registration, checkpoint and
-deepmind/language-perceiver conversion. Parity: reference text/mlm/huggingface.py
-and text/common/huggingface.py."""
+"""🤗 face of the masked language model.
+
+Registers the model with ``AutoModelForMaskedLM`` (fill-mask pipeline) and
+carries the conversion utilities in both directions: training checkpoints →
+``save_pretrained`` directories, and the published
+``deepmind/language-perceiver`` weights → this framework's module layout
+(the parity target pinned at 201,108,230 parameters).
+
+Behavioral contract mirrored from the reference's text/mlm/huggingface.py
+and text/common/huggingface.py.
+"""
 from __future__ import annotations
 
-from dataclasses import asdict
 from typing import Optional
 
 import torch
@@ -17,6 +24,11 @@ from perceiver_amd.models.hf_base import (
     copy_params,
     copy_self_attention_block_params,
 )
+from perceiver_amd.models.hf_registry import (
+    BackendConfigMixin,
+    save_with_tokenizer,
+    wrap_lit_checkpoint,
+)
 from perceiver_amd.models.text.common import TextEncoder, TextEncoderConfig
 from perceiver_amd.models.text.mlm import (
     MaskedLanguageModel,
@@ -26,25 +38,29 @@ from perceiver_amd.models.text.mlm import (
 )
 
 
-class PerceiverMaskedLanguageModelConfig(PretrainedConfig):
+class PerceiverMaskedLanguageModelConfig(BackendConfigMixin, PretrainedConfig):
     model_type = "perceiver-io-masked-language-model"
+    backend_config_class = MaskedLanguageModelConfig
 
-    def __init__(self, backend_config: Optional[MaskedLanguageModelConfig] = None, **kwargs):
-        if backend_config is None:
-            backend_config = MaskedLanguageModelConfig(
-                TextEncoderConfig(), TextDecoderConfig(), num_latents=512, num_latent_channels=512
-            )
-        self.model_config = asdict(backend_config)
-        super().__init__(**kwargs)
+    def __init__(self, backend_config=None, **kwargs):
+        # explicit __init__: transformers 5.x wraps configs without one
+        # in a kwargs-only guard that would swallow backend_config
+        super().__init__(backend_config, **kwargs)
 
-    @property
-    def backend_config(self) -> MaskedLanguageModelConfig:
-        model_config = self.model_config.copy()
-        encoder_config = model_config.pop("encoder")
-        decoder_config = model_config.pop("decoder")
+    @classmethod
+    def default_backend_config(cls):
         return MaskedLanguageModelConfig(
-            encoder=TextEncoderConfig(**encoder_config),
-            decoder=TextDecoderConfig(**decoder_config), **model_config
+            TextEncoderConfig(), TextDecoderConfig(),
+            num_latents=512, num_latent_channels=512,
+        )
+
+    @classmethod
+    def decode_backend_config(cls, model_config):
+        flat = dict(model_config)
+        return MaskedLanguageModelConfig(
+            encoder=TextEncoderConfig(**flat.pop("encoder")),
+            decoder=TextDecoderConfig(**flat.pop("decoder")),
+            **flat,
         )
 
 
@@ -60,12 +76,8 @@ class PerceiverMaskedLanguageModel(PreTrainedModel):
     def from_checkpoint(ckpt_path):
         from perceiver_amd.train.lit import LitMaskedLanguageModel
 
-        model = LitMaskedLanguageModel.load_from_checkpoint(ckpt_path).model
-        hgf_config = PerceiverMaskedLanguageModelConfig(model.config)
-        hgf_config.is_decoder = False
-        hgf_model = PerceiverMaskedLanguageModel(hgf_config)
-        hgf_model.backend_model.load_state_dict(model.state_dict())
-        return hgf_model
+        return wrap_lit_checkpoint(LitMaskedLanguageModel, PerceiverMaskedLanguageModel,
+                                   ckpt_path, is_decoder=False)
 
     def forward(self, input_ids: torch.LongTensor,
                 attention_mask: Optional[torch.FloatTensor] = None,
@@ -73,8 +85,7 @@ class PerceiverMaskedLanguageModel(PreTrainedModel):
         if labels is not None:
             raise ValueError("Loss computation from labels not supported yet")
         pad_mask = None if attention_mask is None else ~attention_mask.type(torch.bool)
-        logits = self.backend_model(input_ids, pad_mask=pad_mask)
-        return MaskedLMOutput(logits=logits)
+        return MaskedLMOutput(logits=self.backend_model(input_ids, pad_mask=pad_mask))
 
 
 AutoConfig.register(PerceiverMaskedLanguageModelConfig.model_type, PerceiverMaskedLanguageModelConfig)
@@ -83,6 +94,7 @@ AutoModelForMaskedLM.register(PerceiverMaskedLanguageModelConfig, PerceiverMaske
 
 # ------------------------------------------------------------------ conversion
 def copy_text_encoder_params(src, tgt: TextEncoder):
+    """transformers PerceiverModel -> TextEncoder (latents, CA, SA, embeddings)."""
     copy_cross_attention_layer_params(src.encoder.cross_attention, tgt.cross_attn_1, query_residual=True)
     copy_self_attention_block_params(src.encoder.self_attends, tgt.self_attn_1)
     copy_latent_provider_params(src, tgt)
@@ -91,6 +103,7 @@ def copy_text_encoder_params(src, tgt: TextEncoder):
 
 
 def copy_text_decoder_params(src, tgt: PerceiverDecoder):
+    """transformers PerceiverForMaskedLM decoder -> tied-logits PerceiverDecoder."""
     copy_cross_attention_layer_params(
         src.perceiver.decoder.decoding_cross_attention, tgt.cross_attn, query_residual=False
     )
@@ -99,68 +112,61 @@ def copy_text_decoder_params(src, tgt: PerceiverDecoder):
     copy_param(src.embedding_decoder.bias, tgt.output_adapter.bias)
 
 
-def convert_checkpoint(save_dir, ckpt_url, tokenizer_name, **kwargs):
-    """LitMaskedLanguageModel .ckpt -> persistent PerceiverMaskedLanguageModel dir."""
-    tokenizer = AutoTokenizer.from_pretrained(tokenizer_name, verbose=False)
-    tokenizer.save_pretrained(save_dir, **kwargs)
-    model = PerceiverMaskedLanguageModel.from_checkpoint(ckpt_url)
-    model.config.tokenizer_class = tokenizer.__class__.__name__
-    model.save_pretrained(save_dir, **kwargs)
-
-
 def convert_config(config) -> MaskedLanguageModelConfig:
-    """transformers PerceiverConfig -> MaskedLanguageModelConfig."""
+    """transformers PerceiverConfig -> the backend dataclass tree.
+
+    Field-by-field translation of the deepmind checkpoint hyperparameters;
+    the decoder is the tied-head variant with no cross-attention residual
+    (matching the published architecture).
+    """
     assert config.hidden_act == "gelu"
     # transformers 5.x guards global access to tie_word_embeddings on some configs
-    tie = config.to_dict().get("tie_word_embeddings", True)
-    assert tie
+    assert config.to_dict().get("tie_word_embeddings", True)
 
-    encoder_config = TextEncoderConfig(
+    shared = dict(
         vocab_size=config.vocab_size,
         max_seq_len=config.max_position_embeddings,
-        num_input_channels=config.d_model,
         num_cross_attention_qk_channels=config.qk_channels,
-        num_cross_attention_v_channels=config.v_channels,
         num_cross_attention_heads=config.num_cross_attention_heads,
+        cross_attention_widening_factor=config.cross_attention_widening_factor,
+        dropout=config.attention_probs_dropout_prob,
+        init_scale=config.initializer_range,
+    )
+    encoder = TextEncoderConfig(
+        num_input_channels=config.d_model,
+        num_cross_attention_v_channels=config.v_channels,
         num_self_attention_qk_channels=config.qk_channels,
         num_self_attention_v_channels=config.v_channels,
         num_self_attention_heads=config.num_self_attention_heads,
         num_self_attention_layers_per_block=config.num_self_attends_per_block,
         num_self_attention_blocks=config.num_blocks,
-        cross_attention_widening_factor=config.cross_attention_widening_factor,
         self_attention_widening_factor=config.self_attention_widening_factor,
-        dropout=config.attention_probs_dropout_prob,
-        init_scale=config.initializer_range,
+        **shared,
     )
-    decoder_config = TextDecoderConfig(
-        vocab_size=config.vocab_size,
-        max_seq_len=config.max_position_embeddings,
-        num_cross_attention_qk_channels=config.qk_channels,
+    decoder = TextDecoderConfig(
         num_cross_attention_v_channels=config.d_model,
-        num_cross_attention_heads=config.num_cross_attention_heads,
-        cross_attention_widening_factor=config.cross_attention_widening_factor,
         cross_attention_residual=False,
-        dropout=config.attention_probs_dropout_prob,
-        init_scale=config.initializer_range,
+        **shared,
     )
     return MaskedLanguageModelConfig(
-        encoder_config, decoder_config,
+        encoder, decoder,
         num_latents=config.num_latents, num_latent_channels=config.d_latents,
     )
 
 
+def convert_checkpoint(save_dir, ckpt_url, tokenizer_name, **kwargs):
+    """Training .ckpt -> persistent 🤗 directory with tokenizer."""
+    save_with_tokenizer(PerceiverMaskedLanguageModel.from_checkpoint(ckpt_url),
+                        tokenizer_name, save_dir, **kwargs)
+
+
 def convert_model(save_dir, source_repo_id="deepmind/language-perceiver", **kwargs):
-    """transformers PerceiverForMaskedLM -> persistent PerceiverMaskedLanguageModel."""
+    """Published transformers PerceiverForMaskedLM -> this framework's format."""
     import transformers
 
-    src_model = transformers.PerceiverForMaskedLM.from_pretrained(source_repo_id)
-    tgt_config = PerceiverMaskedLanguageModelConfig(convert_config(src_model.config))
-    tgt_model = PerceiverMaskedLanguageModel(tgt_config)
-
-    copy_text_encoder_params(src_model.perceiver, tgt_model.backend_model.encoder)
-    copy_text_decoder_params(src_model, tgt_model.backend_model.decoder)
-
-    src_tokenizer = AutoTokenizer.from_pretrained(source_repo_id, verbose=False)
-    src_tokenizer.save_pretrained(save_dir, **kwargs)
-    tgt_model.config.tokenizer_class = src_tokenizer.__class__.__name__
-    tgt_model.save_pretrained(save_dir, **kwargs)
+    source = transformers.PerceiverForMaskedLM.from_pretrained(source_repo_id)
+    target = PerceiverMaskedLanguageModel(
+        PerceiverMaskedLanguageModelConfig(convert_config(source.config)))
+    copy_text_encoder_params(source.perceiver, target.backend_model.encoder)
+    copy_text_decoder_params(source, target.backend_model.decoder)
+    save_with_tokenizer(target, source_repo_id, save_dir, **kwargs)

@@ -7,7 +7,6 @@ options.
 """
 from __future__ import annotations
 
-from dataclasses import asdict
 from typing import Any, List, Optional, Union
 
 import numpy as np
@@ -23,6 +22,7 @@ from transformers.modeling_outputs import ImageClassifierOutput
 
 from perceiver_amd.core import ClassificationDecoderConfig
 from perceiver_amd.data.vision.common import Normalize, center_crop_resize, to_tensor
+from perceiver_amd.models.hf_registry import BackendConfigMixin, wrap_lit_checkpoint
 from perceiver_amd.models.hf_base import (
     copy_classification_decoder_params,
     copy_cross_attention_layer_params,
@@ -37,26 +37,29 @@ from perceiver_amd.models.vision.image_classifier import (
 )
 
 
-class PerceiverImageClassifierConfig(PretrainedConfig):
+class PerceiverImageClassifierConfig(BackendConfigMixin, PretrainedConfig):
     model_type = "perceiver-io-image-classifier"
+    backend_config_class = ImageClassifierConfig
 
-    def __init__(self, backend_config: Optional[ImageClassifierConfig] = None, **kwargs):
-        if backend_config is None:
-            backend_config = ImageClassifierConfig(
-                ImageEncoderConfig(), ClassificationDecoderConfig(), num_latents=512, num_latent_channels=512
-            )
-        self.model_config = asdict(backend_config)
-        super().__init__(**kwargs)
+    def __init__(self, backend_config=None, **kwargs):
+        # explicit __init__: transformers 5.x wraps configs without one
+        # in a kwargs-only guard that would swallow backend_config
+        super().__init__(backend_config, **kwargs)
 
-    @property
-    def backend_config(self) -> ImageClassifierConfig:
-        model_config = self.model_config.copy()
-        encoder_config = model_config.pop("encoder")
-        decoder_config = model_config.pop("decoder")
+    @classmethod
+    def default_backend_config(cls):
+        return ImageClassifierConfig(
+            ImageEncoderConfig(), ClassificationDecoderConfig(),
+            num_latents=512, num_latent_channels=512,
+        )
+
+    @classmethod
+    def decode_backend_config(cls, model_config):
+        flat = dict(model_config)
         config = ImageClassifierConfig(
-            encoder=ImageEncoderConfig(**encoder_config),
-            decoder=ClassificationDecoderConfig(**decoder_config),
-            **model_config,
+            encoder=ImageEncoderConfig(**flat.pop("encoder")),
+            decoder=ClassificationDecoderConfig(**flat.pop("decoder")),
+            **flat,
         )
         config.encoder.image_shape = tuple(config.encoder.image_shape)
         return config
@@ -139,12 +142,8 @@ class PerceiverImageClassifier(PreTrainedModel):
     def from_checkpoint(ckpt_path):
         from perceiver_amd.train.lit import LitImageClassifier
 
-        model = LitImageClassifier.load_from_checkpoint(ckpt_path).model
-        hgf_config = PerceiverImageClassifierConfig(model.config)
-        hgf_config.is_decoder = False
-        hgf_model = PerceiverImageClassifier(hgf_config)
-        hgf_model.backend_model.load_state_dict(model.state_dict())
-        return hgf_model
+        return wrap_lit_checkpoint(LitImageClassifier, PerceiverImageClassifier,
+                                   ckpt_path, is_decoder=False)
 
     def forward(self, inputs: Optional[torch.Tensor] = None,
                 labels: Optional[torch.LongTensor] = None,
