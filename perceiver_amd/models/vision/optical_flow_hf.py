@@ -3,7 +3,7 @@ tiling) -> micro-batched forward -> distance-weighted recombination, optional HS
 render. Parity: reference vision/optical_flow/huggingface.py."""
 from __future__ import annotations
 
-from dataclasses import asdict, dataclass
+from dataclasses import dataclass
 from typing import Optional, Tuple, Union
 
 import numpy as np
@@ -13,6 +13,7 @@ from transformers.modeling_outputs import ModelOutput
 from transformers.pipelines import PIPELINE_REGISTRY
 
 from perceiver_amd.data.vision.optical_flow import OpticalFlowProcessor, render_optical_flow
+from perceiver_amd.models.hf_registry import BackendConfigMixin
 from perceiver_amd.models.hf_base import (
     copy_cross_attention_layer_params,
     copy_latent_provider_params,
@@ -28,27 +29,29 @@ from perceiver_amd.models.vision.optical_flow import (
 )
 
 
-class OpticalFlowPerceiverConfig(PretrainedConfig):
+class OpticalFlowPerceiverConfig(BackendConfigMixin, PretrainedConfig):
     model_type = "perceiver-io-optical-flow"
+    backend_config_class = OpticalFlowConfig
 
-    def __init__(self, backend_config: Optional[OpticalFlowConfig] = None, **kwargs):
-        if backend_config is None:
-            backend_config = OpticalFlowConfig(
-                OpticalFlowEncoderConfig(), OpticalFlowDecoderConfig(),
-                num_latents=512, num_latent_channels=512,
-            )
-        self.model_config = asdict(backend_config)
-        super().__init__(**kwargs)
+    def __init__(self, backend_config=None, **kwargs):
+        # explicit __init__: transformers 5.x wraps configs without one
+        # in a kwargs-only guard that would swallow backend_config
+        super().__init__(backend_config, **kwargs)
 
-    @property
-    def backend_config(self) -> OpticalFlowConfig:
-        model_config = self.model_config.copy()
-        encoder_config = model_config.pop("encoder")
-        decoder_config = model_config.pop("decoder")
+    @classmethod
+    def default_backend_config(cls):
+        return OpticalFlowConfig(
+            OpticalFlowEncoderConfig(), OpticalFlowDecoderConfig(),
+            num_latents=512, num_latent_channels=512,
+        )
+
+    @classmethod
+    def decode_backend_config(cls, model_config):
+        flat = dict(model_config)
         cfg = OpticalFlowConfig(
-            encoder=OpticalFlowEncoderConfig(**encoder_config),
-            decoder=OpticalFlowDecoderConfig(**decoder_config),
-            **model_config,
+            encoder=OpticalFlowEncoderConfig(**flat.pop("encoder")),
+            decoder=OpticalFlowDecoderConfig(**flat.pop("decoder")),
+            **flat,
         )
         cfg.encoder.image_shape = tuple(cfg.encoder.image_shape)
         cfg.decoder.image_shape = tuple(cfg.decoder.image_shape)
@@ -130,22 +133,20 @@ PIPELINE_REGISTRY.register_pipeline(
 
 # ------------------------------------------------------------------ conversion
 def convert_config(config) -> OpticalFlowConfig:
-    """transformers PerceiverConfig (optical-flow) -> OpticalFlowConfig."""
+    """transformers PerceiverConfig (optical-flow) -> the backend dataclass tree."""
     assert config.hidden_act == "gelu"
-    encoder_config = OpticalFlowEncoderConfig(
+    shared = dict(
         num_cross_attention_heads=config.num_cross_attention_heads,
+        dropout=config.attention_probs_dropout_prob,
+        init_scale=config.initializer_range,
+    )
+    encoder = OpticalFlowEncoderConfig(
         num_self_attention_heads=config.num_self_attention_heads,
         num_self_attention_layers_per_block=config.num_self_attends_per_block,
         num_self_attention_blocks=config.num_blocks,
-        dropout=config.attention_probs_dropout_prob,
-        init_scale=config.initializer_range,
+        **shared,
     )
-    decoder_config = OpticalFlowDecoderConfig(
-        num_cross_attention_heads=config.num_cross_attention_heads,
-        dropout=config.attention_probs_dropout_prob,
-        init_scale=config.initializer_range,
-    )
-    return OpticalFlowConfig(encoder_config, decoder_config,
+    return OpticalFlowConfig(encoder, OpticalFlowDecoderConfig(**shared),
                              num_latents=config.num_latents, num_latent_channels=config.d_latents)
 
 
