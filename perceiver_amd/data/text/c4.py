@@ -91,41 +91,50 @@ class C4DataModule:
                                   buffer_size=self.hparams.shuffle_window_size)
         return split_dataset_by_node(dataset, rank=self.rank, world_size=self.world_size)
 
-    def _create_pipeline(self, dataset, min_seq_len=None):
+    def _tokenize_fn(self):
         def tokenize(examples):
             return self.tokenizer(
                 examples["text"], padding=False, truncation=False, max_length=None,
                 add_special_tokens=False, return_token_type_ids=False,
                 return_attention_mask=False,
             )
+        return tokenize
 
-        def concat(examples):
-            for example in examples:
-                yield from example
-                yield self.tokenizer.eos_token_id
+    def _next_chunk_len(self, min_seq_len):
+        """Chunk length incl. the shift token: fixed, or random in
+        [min, max] for variable-length training."""
+        if min_seq_len is None:
+            return self.hparams.max_seq_len + 1
+        return int(torch.randint(min_seq_len, self.hparams.max_seq_len + 1, size=(1,))) + 1
 
-        def chunk_len():
-            if min_seq_len is None:
-                return self.hparams.max_seq_len + 1
-            return int(torch.randint(min_seq_len, self.hparams.max_seq_len + 1, size=(1,))) + 1
+    def _chunk_fn(self, min_seq_len):
+        eos = self.tokenizer.eos_token_id
 
         def chunk(examples):
-            chs, ch = [], []
-            ch_len = chunk_len()
-            for token_id in concat(examples["input_ids"]):
-                ch.append(token_id)
-                if len(ch) == ch_len:
-                    chs.append(ch)
-                    ch = []
-                    ch_len = chunk_len()
-            if not chs:
+            # join the documents of this map-batch into one EOS-separated
+            # stream and cut it into (variable-length) training chunks; a
+            # trailing partial chunk is dropped with its batch
+            chunks = []
+            current = []
+            want = self._next_chunk_len(min_seq_len)
+            for doc in examples["input_ids"]:
+                for token_id in (*doc, eos):
+                    current.append(token_id)
+                    if len(current) == want:
+                        chunks.append(current)
+                        current = []
+                        want = self._next_chunk_len(min_seq_len)
+            if not chunks:
                 return []
-            examples["input_ids"] = chs
+            examples["input_ids"] = chunks
             return examples
+        return chunk
 
-        return dataset.map(tokenize, batched=True,
-                           remove_columns=["text", "timestamp", "url"]).map(
-            chunk, batched=True, batch_size=self.hparams.concat_batch_size)
+    def _create_pipeline(self, dataset, min_seq_len=None):
+        tokenized = dataset.map(self._tokenize_fn(), batched=True,
+                                remove_columns=["text", "timestamp", "url"])
+        return tokenized.map(self._chunk_fn(min_seq_len), batched=True,
+                             batch_size=self.hparams.concat_batch_size)
 
     def setup(self, stage=None):
         self.ds_train = self._create_pipeline(self._create_dataset("train"),
@@ -146,12 +155,15 @@ class C4DataModule:
 
 
 class C4Collator(Collator):
+    """Pad the chunked streams, then shift by one for next-token prediction."""
+
     def __init__(self, tokenizer):
         self.tokenizer = tokenizer
 
     def collate(self, examples):
         batch = self.tokenizer.pad(examples, return_attention_mask=True, return_tensors="pt")
-        batch["labels"] = batch["input_ids"][..., 1:]
-        batch["input_ids"] = batch["input_ids"][..., :-1]
+        window = batch["input_ids"]
+        batch["labels"] = window[..., 1:]
+        batch["input_ids"] = window[..., :-1]
         batch["attention_mask"] = batch["attention_mask"][..., :-1]
         return batch
