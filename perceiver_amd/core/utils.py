@@ -57,8 +57,24 @@ class Residual(nn.Module):
 
     def forward(self, *args, **kwargs):
         output = self.module(*args, **kwargs)
-        output.last_hidden_state = self.dropout(output.last_hidden_state) + args[0]
+        hidden = output.last_hidden_state
+        shortcut = args[0]
+        p = self.dropout.p
+        if _fused_dropout_add_ok(hidden, shortcut, p, self.training):
+            from perceiver_amd.ops.dropadd import dropout_add
+
+            output.last_hidden_state = dropout_add(hidden, shortcut, p)
+        else:
+            output.last_hidden_state = self.dropout(hidden) + shortcut
         return output
+
+
+def _fused_dropout_add_ok(hidden, shortcut, p, training) -> bool:
+    if not (training and p > 0.0 and hidden.is_cuda):
+        return False
+    from perceiver_amd.ops.dropadd import can_use_dropout_add
+
+    return can_use_dropout_add(hidden, shortcut, p, training)
 
 
 def init_parameters(module: nn.Module, init_scale: float) -> None:

@@ -20,6 +20,8 @@ void adamw_step(torch::Tensor master, torch::Tensor m, torch::Tensor v, torch::T
                 double lr, double beta1, double beta2, double eps, double weight_decay,
                 int64_t step);
 torch::Tensor rotary_apply(torch::Tensor t, torch::Tensor frq, int64_t rot, bool neg_sin);
+torch::Tensor dropout_add_fwd(torch::Tensor x, torch::Tensor res, double p, int64_t seed);
+torch::Tensor dropout_add_bwd(torch::Tensor dy, double p, int64_t seed);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("gelu_bias_fwd", &gelu_bias_fwd, "fused bias+GELU forward (bf16)");
@@ -31,4 +33,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("ln_bwd", &ln_bwd, "fused bf16 LayerNorm backward");
     m.def("adamw_step", &adamw_step, "single-pass fused AdamW on flat fp32 state");
     m.def("rotary_apply", &rotary_apply, "fused rotary embedding (bf16, fwd/bwd via neg_sin)");
+    m.def("dropout_add_fwd", &dropout_add_fwd, "fused residual dropout-add forward (bf16)");
+    m.def("dropout_add_bwd", &dropout_add_bwd, "fused residual dropout-add backward (bf16)");
 }
