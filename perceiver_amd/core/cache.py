@@ -91,16 +91,24 @@ def cache_len(entry: KVCacheEntry) -> int:
     return entry[0].shape[1]
 
 
-def allocate_kv_cache(model, batch: int, device=None, dtype=torch.float32) -> List[StaticKVCache]:
+def allocate_kv_cache(model, batch: int, device=None, dtype=torch.float32,
+                      ca_capacity=None) -> List[StaticKVCache]:
     """Preallocate the [cross-attention, *self-attention] cache list for a
-    PerceiverAR-family ``model`` (capacities: max_seq_len / max_latents)."""
+    PerceiverAR-family ``model``.
+
+    ``ca_capacity`` bounds the cross-attention cache below ``max_seq_len`` —
+    the graph decoder reads the FULL capacity (masked) every step, so sizing
+    the cache to the actual sequence need (bucketed, see hf_base.generate)
+    cuts the dead-row KV traffic of short decodes.
+    """
     def chans(layer):
         if not hasattr(layer, "num_qk_channels"):  # activation-checkpoint wrapper
             layer = layer.module
         return layer.num_qk_channels, layer.num_v_channels
 
+    cap = model.max_seq_len if ca_capacity is None else min(int(ca_capacity), model.max_seq_len)
     qk, vc = chans(model.cross_attention)
-    caches = [StaticKVCache(batch, model.max_seq_len, qk, vc, device=device, dtype=dtype)]
+    caches = [StaticKVCache(batch, cap, qk, vc, device=device, dtype=dtype)]
     for layer in model.self_attention:
         qk, vc = chans(layer)
         caches.append(StaticKVCache(batch, model.max_latents, qk, vc, device=device, dtype=dtype))

@@ -300,12 +300,23 @@ class PerceiverCausalSequenceModel(PreTrainedModel):
                     and seq_len + max_new_tokens <= self.backend_model.max_seq_len):
                 from perceiver_amd.core.graph_decode import GraphedDecoder
 
+                # bucketed cache capacity: the graphed step reads the FULL
+                # (masked) CA cache every token, so size it to the request —
+                # the next power-of-two bucket >= seq_len + max_new_tokens
+                # (>= 2048) instead of always max_seq_len (8192-ctx models
+                # were paying 8192 dead rows for 1k-token requests)
+                need = seq_len + max_new_tokens
+                bucket = 2048
+                while bucket < need:
+                    bucket *= 2
+                bucket = min(bucket, self.backend_model.max_seq_len)
+
                 # p.data_ptr() pins the cached graphs to the current parameter
                 # storage: model.to(device/dtype) reallocates storage, and a
                 # captured graph replaying against the old pointers would
                 # silently use stale weights (in-place load_state_dict is fine)
                 key = (input_ids.shape[0], do_sample, float(temperature), top_k,
-                       p.device, p.dtype, p.data_ptr())
+                       bucket, p.device, p.dtype, p.data_ptr())
                 cache = getattr(self, "_graph_decoders", None)
                 if cache is None:
                     cache = self._graph_decoders = {}
@@ -316,7 +327,8 @@ class PerceiverCausalSequenceModel(PreTrainedModel):
                     gd = cache[key] = GraphedDecoder(
                         self.backend_model,
                         allocate_kv_cache(self.backend_model, input_ids.shape[0],
-                                          device=p.device, dtype=p.dtype),
+                                          device=p.device, dtype=p.dtype,
+                                          ca_capacity=bucket),
                         do_sample=do_sample, temperature=temperature, top_k=top_k)
                 gd.prefill(input_ids, prefix_len=prefix_len)
                 first = gd.tok.clone()  # token emitted by the prefill pass

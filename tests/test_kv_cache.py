@@ -160,3 +160,22 @@ def test_causal_sequence_model_cached_decode_matches_full_with_left_padding():
     incremental = torch.cat(outs, dim=1)
 
     assert torch.allclose(full, incremental, atol=1e-4)
+
+
+def test_allocate_kv_cache_bucketed_capacity():
+    """ca_capacity bounds the cross-attention cache (bucketed decode graphs);
+    self-attention caches keep max_latents."""
+    import torch
+
+    from perceiver_amd.core.cache import allocate_kv_cache
+    from perceiver_amd.models.text.clm import CausalLanguageModel, CausalLanguageModelConfig
+
+    model = CausalLanguageModel(CausalLanguageModelConfig(
+        vocab_size=32, max_seq_len=8192, max_latents=64, num_channels=32,
+        num_heads=4, num_self_attention_layers=2))
+    caches = allocate_kv_cache(model, 2, ca_capacity=2048)
+    assert caches[0].capacity == 2048
+    assert all(c.capacity == 64 for c in caches[1:])
+    # never above max_seq_len
+    caches = allocate_kv_cache(model, 2, ca_capacity=1 << 20)
+    assert caches[0].capacity == 8192
