@@ -46,11 +46,14 @@ __global__ void ln_fwd_kernel(const unsigned short* __restrict__ x,
                 sumsq += f * f;
             }
         } else if (c0 < C) {
-            for (int e = 0; e < C - c0; ++e) {
-                float f = bf2f(xrow[c0 + e]);
-                vals[i][e] = f;
-                sum += f;
-                sumsq += f * f;
+#pragma unroll
+            for (int e = 0; e < 8; ++e) {   // static index: a runtime bound
+                if (c0 + e < C) {           // would push vals[] to scratch
+                    float f = bf2f(xrow[c0 + e]);
+                    vals[i][e] = f;
+                    sum += f;
+                    sumsq += f * f;
+                }
             }
         }
     }
@@ -77,9 +80,12 @@ __global__ void ln_fwd_kernel(const unsigned short* __restrict__ x,
             }
             *reinterpret_cast<short8v*>(yrow + c0) = o;
         } else if (c0 < C) {
-            for (int e = 0; e < C - c0; ++e) {
-                float bb = b ? bf2f(b[c0 + e]) : 0.f;
-                yrow[c0 + e] = f2bf((vals[i][e] - mean) * rstd * bf2f(w[c0 + e]) + bb);
+#pragma unroll
+            for (int e = 0; e < 8; ++e) {
+                if (c0 + e < C) {
+                    float bb = b ? bf2f(b[c0 + e]) : 0.f;
+                    yrow[c0 + e] = f2bf((vals[i][e] - mean) * rstd * bf2f(w[c0 + e]) + bb);
+                }
             }
         }
     }
@@ -122,13 +128,16 @@ __global__ void ln_bwd_dx_kernel(const unsigned short* __restrict__ dy,
                 s2 += gg * xhat;
             }
         } else if (c0 < C) {
-            for (int e = 0; e < C - c0; ++e) {
-                float gg = bf2f(dyrow[c0 + e]) * bf2f(w[c0 + e]);
-                float xhat = (bf2f(xrow[c0 + e]) - mean) * rstd;
-                g[i][e] = gg;
-                xh[i][e] = xhat;
-                s1 += gg;
-                s2 += gg * xhat;
+#pragma unroll
+            for (int e = 0; e < 8; ++e) {
+                if (c0 + e < C) {
+                    float gg = bf2f(dyrow[c0 + e]) * bf2f(w[c0 + e]);
+                    float xhat = (bf2f(xrow[c0 + e]) - mean) * rstd;
+                    g[i][e] = gg;
+                    xh[i][e] = xhat;
+                    s1 += gg;
+                    s2 += gg * xhat;
+                }
             }
         }
     }
@@ -145,8 +154,10 @@ __global__ void ln_bwd_dx_kernel(const unsigned short* __restrict__ dy,
                 o[e] = (short)f2bf(rstd * (g[i][e] - s1 - xh[i][e] * s2));
             *reinterpret_cast<short8v*>(dxrow + c0) = o;
         } else if (c0 < C) {
-            for (int e = 0; e < C - c0; ++e)
-                dxrow[c0 + e] = f2bf(rstd * (g[i][e] - s1 - xh[i][e] * s2));
+#pragma unroll
+            for (int e = 0; e < 8; ++e)
+                if (c0 + e < C)
+                    dxrow[c0 + e] = f2bf(rstd * (g[i][e] - s1 - xh[i][e] * s2));
         }
     }
 }
@@ -192,13 +203,13 @@ __global__ void ln_bwd_fused_kernel(const unsigned short* __restrict__ dy,
         unsigned short* dxrow = dx + row * C;
         float mean = mean_in[row], rstd = rstd_in[row];
 
-        float g[CHUNKS][8], xh[CHUNKS][8], dyl[CHUNKS][8];
+        float g[CHUNKS][8], xh[CHUNKS][8];
         float s1 = 0.f, s2 = 0.f;
 #pragma unroll
         for (int i = 0; i < CHUNKS; ++i) {
             int c0 = lane * 8 + i * 512;
 #pragma unroll
-            for (int e = 0; e < 8; ++e) { g[i][e] = 0.f; xh[i][e] = 0.f; dyl[i][e] = 0.f; }
+            for (int e = 0; e < 8; ++e) { g[i][e] = 0.f; xh[i][e] = 0.f; }
             if (c0 + 8 <= C) {
                 short8v dyv = *reinterpret_cast<const short8v*>(dyrow + c0);
                 short8v xv = *reinterpret_cast<const short8v*>(xrow + c0);
@@ -208,22 +219,27 @@ __global__ void ln_bwd_fused_kernel(const unsigned short* __restrict__ dy,
                     float dv = bf2f((unsigned short)dyv[e]);
                     float gg = dv * bf2f((unsigned short)wv[e]);
                     float xhat = (bf2f((unsigned short)xv[e]) - mean) * rstd;
-                    dyl[i][e] = dv;
                     g[i][e] = gg;
                     xh[i][e] = xhat;
                     s1 += gg;
                     s2 += gg * xhat;
+                    dwacc[i][e] += dv * xhat;
+                    dbacc[i][e] += dv;
                 }
             } else if (c0 < C) {
-                for (int e = 0; e < C - c0; ++e) {
-                    float dv = bf2f(dyrow[c0 + e]);
-                    float gg = dv * bf2f(w[c0 + e]);
-                    float xhat = (bf2f(xrow[c0 + e]) - mean) * rstd;
-                    dyl[i][e] = dv;
-                    g[i][e] = gg;
-                    xh[i][e] = xhat;
-                    s1 += gg;
-                    s2 += gg * xhat;
+#pragma unroll
+                for (int e = 0; e < 8; ++e) {
+                    if (c0 + e < C) {   // static index: runtime bounds would
+                        float dv = bf2f(dyrow[c0 + e]);  // scratch the arrays
+                        float gg = dv * bf2f(w[c0 + e]);
+                        float xhat = (bf2f(xrow[c0 + e]) - mean) * rstd;
+                        g[i][e] = gg;
+                        xh[i][e] = xhat;
+                        s1 += gg;
+                        s2 += gg * xhat;
+                        dwacc[i][e] += dv * xhat;
+                        dbacc[i][e] += dv;
+                    }
                 }
             }
         }
@@ -236,18 +252,14 @@ __global__ void ln_bwd_fused_kernel(const unsigned short* __restrict__ dy,
             if (c0 + 8 <= C) {
                 short8v o;
 #pragma unroll
-                for (int e = 0; e < 8; ++e) {
+                for (int e = 0; e < 8; ++e)
                     o[e] = (short)f2bf(rstd * (g[i][e] - s1 - xh[i][e] * s2));
-                    dwacc[i][e] += dyl[i][e] * xh[i][e];
-                    dbacc[i][e] += dyl[i][e];
-                }
                 *reinterpret_cast<short8v*>(dxrow + c0) = o;
             } else if (c0 < C) {
-                for (int e = 0; e < C - c0; ++e) {
-                    dxrow[c0 + e] = f2bf(rstd * (g[i][e] - s1 - xh[i][e] * s2));
-                    dwacc[i][e] += dyl[i][e] * xh[i][e];
-                    dbacc[i][e] += dyl[i][e];
-                }
+#pragma unroll
+                for (int e = 0; e < 8; ++e)
+                    if (c0 + e < C)
+                        dxrow[c0 + e] = f2bf(rstd * (g[i][e] - s1 - xh[i][e] * s2));
             }
         }
     }
