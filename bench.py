@@ -313,17 +313,29 @@ class ImgBench(MLMBench):
         self.model = ImageClassifier(cfg).to(device)
         self.device = device
         self.cfg = cfg
-        self.opt = torch.optim.AdamW(self.model.parameters(), lr=2e-4, foreach=True)
+        self._setup_precision_and_opt()
         self.model.train()
         shape = cfg.encoder.image_shape
         g = torch.Generator(device="cpu").manual_seed(args.seed + rank)
         self.x = torch.randn(self.batch, *shape, generator=g).to(device)
         self.labels = torch.randint(0, cfg.decoder.num_classes, (self.batch,), generator=g).to(device)
 
+    def _setup_precision_and_opt(self):
+        # same scheme as MLMBench: pure-bf16 weights + fp32-master AdamW on GPU
+        # (fused LN/dropout paths key on bf16 weights; no autocast cast traffic)
+        self.pure_bf16 = self.device.type == "cuda"
+        if self.pure_bf16:
+            from perceiver_amd.train.optim import MasterAdamW, convert_to_bf16_training
+
+            self.model = convert_to_bf16_training(self.model)
+            self.opt = MasterAdamW(self.model.parameters(), lr=2e-4, weight_decay=0.01)
+        else:
+            self.opt = torch.optim.AdamW(self.model.parameters(), lr=2e-4, foreach=True)
+
     def step(self):
-        with torch.autocast(self.device.type, dtype=torch.bfloat16, enabled=self.device.type == "cuda"):
-            logits = self.model(self.x)
-            loss = F.cross_entropy(logits.float(), self.labels)
+        x = self.x.to(torch.bfloat16) if self.pure_bf16 else self.x
+        logits = self.model(x)
+        loss = F.cross_entropy(logits.float(), self.labels)
         self.opt.zero_grad(set_to_none=True)
         loss.backward()
         if getattr(self, "reducer", None) is not None:
@@ -355,7 +367,7 @@ class FlowBench(ImgBench):
                                                  num_patch_hidden_channels=8, num_frequency_bands=2,
                                                  num_cross_attention_heads=1, num_self_attention_heads=2,
                                                  num_self_attention_layers_per_block=2),
-                decoder=OpticalFlowDecoderConfig(image_shape=(16, 24)),
+                decoder=OpticalFlowDecoderConfig(image_shape=(16, 24), num_cross_attention_heads=1),
                 num_latents=16, num_latent_channels=32,
             )
         self.batch = args.batch or (1 if not args.tiny else 2)
@@ -363,7 +375,7 @@ class FlowBench(ImgBench):
         self.model = OpticalFlow(cfg).to(device)
         self.device = device
         self.cfg = cfg
-        self.opt = torch.optim.AdamW(self.model.parameters(), lr=2e-4, foreach=True)
+        self._setup_precision_and_opt()
         self.model.train()
         h, w = cfg.encoder.image_shape
         c = cfg.encoder.num_patch_input_channels
@@ -372,9 +384,9 @@ class FlowBench(ImgBench):
         self.target = torch.randn(self.batch, h, w, 2, generator=g).to(device)
 
     def step(self):
-        with torch.autocast(self.device.type, dtype=torch.bfloat16, enabled=self.device.type == "cuda"):
-            flow = self.model(self.x)
-            loss = F.mse_loss(flow.float(), self.target)
+        x = self.x.to(torch.bfloat16) if self.pure_bf16 else self.x
+        flow = self.model(x)
+        loss = F.mse_loss(flow.float(), self.target)
         self.opt.zero_grad(set_to_none=True)
         loss.backward()
         if getattr(self, "reducer", None) is not None:

@@ -863,6 +863,11 @@ std::vector<torch::Tensor> flash_bwd(torch::Tensor dout, torch::Tensor q, torch:
         launch_flash_bwd<128, 128, 64, 2, 32, 1>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
     else if (D <= 160 && Dv <= 160)
         launch_flash_bwd<160, 160, 64, 1, 32, 1>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
+    else if (D <= 288 && Dv <= 288)
+        // the img/flow cross-attention class (D = 261, pad 288): dedicated
+        // instantiation — the 352 template allocated for 352-wide register
+        // arrays and spilled 19 B/lane on dkv
+        launch_flash_bwd<288, 288, 32, 1, 32, 1>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
     else
         launch_flash_bwd<352, 352, 32, 1, 32, 1>(dout, q, k, v, lse, delta, pm, causal, dp, sd, dq, dk, dv);
 

@@ -595,6 +595,7 @@ std::vector<torch::Tensor> flash_fwd(torch::Tensor q, torch::Tensor k, torch::Te
     else if (D <= 64 && Dv <= 64)   launch_flash_fwd<64, 64, 2>(q, k, v, pm, causal, dp, sd, out, lse);
     else if (D <= 128 && Dv <= 128) launch_flash_fwd<128, 128, 2>(q, k, v, pm, causal, dp, sd, out, lse);
     else if (D <= 160 && Dv <= 160) launch_flash_fwd<160, 160, 1>(q, k, v, pm, causal, dp, sd, out, lse);
+    else if (D <= 288 && Dv <= 288) launch_flash_fwd<288, 288, 1>(q, k, v, pm, causal, dp, sd, out, lse);
     else                            launch_flash_fwd<352, 352, 1>(q, k, v, pm, causal, dp, sd, out, lse);
 
     return {out, lse};
