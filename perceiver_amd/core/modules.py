@@ -24,6 +24,7 @@ import torch
 import torch.nn as nn
 
 from perceiver_amd.core.position import RotaryPositionEmbedding
+from perceiver_amd.ops.linear import PerceiverLinear
 from perceiver_amd.core.utils import ModuleOutput, Residual, init_parameters
 from perceiver_amd.core.adapter import (
     InputAdapter,
@@ -96,17 +97,17 @@ class MultiHeadAttention(nn.Module):
             # qkv_proj.weight rows are [q | k | v] in the reference's per-matrix
             # order (modules.py:50-55); the load hook accepts legacy checkpoints
             # with split q_proj/k_proj/v_proj keys.
-            self.qkv_proj = nn.Linear(
+            self.qkv_proj = PerceiverLinear(
                 num_kv_input_channels, 2 * num_qk_channels + num_v_channels, bias=qkv_bias
             )
             self._register_load_state_dict_pre_hook(
                 MultiHeadAttention._merge_sd_hook, with_module=True
             )
         else:
-            self.q_proj = nn.Linear(num_q_input_channels, num_qk_channels, bias=qkv_bias)
-            self.k_proj = nn.Linear(num_kv_input_channels, num_qk_channels, bias=qkv_bias)
-            self.v_proj = nn.Linear(num_kv_input_channels, num_v_channels, bias=qkv_bias)
-        self.o_proj = nn.Linear(num_v_channels, num_output_channels, bias=out_bias)
+            self.q_proj = PerceiverLinear(num_q_input_channels, num_qk_channels, bias=qkv_bias)
+            self.k_proj = PerceiverLinear(num_kv_input_channels, num_qk_channels, bias=qkv_bias)
+            self.v_proj = PerceiverLinear(num_kv_input_channels, num_v_channels, bias=qkv_bias)
+        self.o_proj = PerceiverLinear(num_v_channels, num_output_channels, bias=out_bias)
         # kept as a module for state-dict/layout parity; the dispatch path applies
         # dropout functionally inside the attention core
         self.dropout = nn.Dropout(dropout)
@@ -511,9 +512,9 @@ class MLP(nn.Sequential):
     def __init__(self, num_channels: int, widening_factor: int, bias: bool = True):
         super().__init__(
             LayerNorm(num_channels),
-            nn.Linear(num_channels, widening_factor * num_channels, bias=bias),
+            PerceiverLinear(num_channels, widening_factor * num_channels, bias=bias),
             nn.GELU(),
-            nn.Linear(widening_factor * num_channels, num_channels, bias=bias),
+            PerceiverLinear(widening_factor * num_channels, num_channels, bias=bias),
         )
 
     def forward(self, x: torch.Tensor) -> ModuleOutput:

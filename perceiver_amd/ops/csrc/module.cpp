@@ -22,6 +22,7 @@ void adamw_step(torch::Tensor master, torch::Tensor m, torch::Tensor v, torch::T
 torch::Tensor rotary_apply(torch::Tensor t, torch::Tensor frq, int64_t rot, bool neg_sin);
 torch::Tensor dropout_add_fwd(torch::Tensor x, torch::Tensor res, double p, int64_t seed);
 torch::Tensor dropout_add_bwd(torch::Tensor dy, double p, int64_t seed);
+torch::Tensor colsum_bf16(torch::Tensor x);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("gelu_bias_fwd", &gelu_bias_fwd, "fused bias+GELU forward (bf16)");
@@ -35,4 +36,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("rotary_apply", &rotary_apply, "fused rotary embedding (bf16, fwd/bwd via neg_sin)");
     m.def("dropout_add_fwd", &dropout_add_fwd, "fused residual dropout-add forward (bf16)");
     m.def("dropout_add_bwd", &dropout_add_bwd, "fused residual dropout-add backward (bf16)");
+    m.def("colsum_bf16", &colsum_bf16, "coalesced bf16 column sum (bias gradients)");
 }
