@@ -54,6 +54,7 @@ class TrainConfig:
     val_every_steps: Optional[int] = None
     ckpt_every_steps: Optional[int] = None
     out_dir: str = "logs/run"
+    tensorboard: bool = True         # scalar event files under out_dir/tb
     save_top_k: int = 1
     monitor: str = "val_loss"
     seed: Optional[int] = None
@@ -72,9 +73,14 @@ class Trainer:
         self.epoch = 0
         self._best_monitor = math.inf
         self._log_file = None
+        self._tb = None
         if is_main_process():
             os.makedirs(os.path.join(self.cfg.out_dir, "checkpoints"), exist_ok=True)
             self._log_file = open(os.path.join(self.cfg.out_dir, "metrics.jsonl"), "a")
+            if self.cfg.tensorboard:
+                from perceiver_amd.utils.tensorboard import ScalarWriter
+
+                self._tb = ScalarWriter(os.path.join(self.cfg.out_dir, "tb"))
         if config.seed is not None:
             torch.manual_seed(config.seed + get_rank())
         # fp16 needs loss scaling; for bf16/fp32 the scaler is a no-op passthrough
@@ -119,10 +125,16 @@ class Trainer:
         return None
 
     def log_metrics(self, metrics: dict, step: Optional[int] = None):
+        at = self.global_step if step is None else step
         if self._log_file is not None:
-            rec = {"step": self.global_step if step is None else step, "time": time.time(), **metrics}
+            rec = {"step": at, "time": time.time(), **metrics}
             self._log_file.write(json.dumps(rec) + "\n")
             self._log_file.flush()
+        if self._tb is not None:
+            for key, value in metrics.items():
+                if isinstance(value, (int, float)):
+                    self._tb.add_scalar(key, value, at)
+            self._tb.flush()
 
     def _reduce_mean(self, value: torch.Tensor) -> torch.Tensor:
         value = value.detach()
