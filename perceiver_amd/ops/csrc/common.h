@@ -61,3 +61,72 @@ DEVINL float gelu_grad_f(float x) {
             TORCH_CHECK(false, "HIP kernel launch failed: ", hipGetErrorString(e)); \
         }                                                                           \
     } while (0)
+
+// ---- T14 async-STAGE split helpers (shared by the attention kernels) ----
+// issue: clamped-row unconditional vector loads into registers; the vmcnt
+// waits land at the write call one compute phase later, hiding HBM latency
+// under MFMA work. Valid only when the runtime dims match the template
+// exactly (every granule full-width; callers pad odd dims to the template).
+template <int ROWS_TILE, int DD, int NG>
+DEVINL void issue_tile(short8v (&st)[NG], const unsigned short* __restrict__ src,
+                       long sstride, int row_limit, int tid) {
+    constexpr int GPR = DD / 8;
+#pragma unroll
+    for (int i = 0; i < NG; ++i) {
+        int g = tid + i * 256;
+        int row = g / GPR, c0 = (g % GPR) * 8;
+        int rowc = row < row_limit ? row : row_limit - 1;
+        st[i] = *reinterpret_cast<const short8v*>(src + (long)rowc * sstride + c0);
+    }
+}
+
+// write halves: spill staged registers into the row-major and/or 16-column
+// subtiled LDS images, zero-filling clamp-duplicated tail rows
+template <int ROWS_TILE, int DD, int NG>
+DEVINL void write_rm(const short8v (&st)[NG], char* lds_rm, int ldst_bytes,
+                     int rows_valid, int tid) {
+    constexpr int GPR = DD / 8;
+    const bool tail = rows_valid < ROWS_TILE;
+#pragma unroll
+    for (int i = 0; i < NG; ++i) {
+        int g = tid + i * 256;
+        int row = g / GPR, c0 = (g % GPR) * 8;
+        short8v val = st[i];
+        if (tail && row >= rows_valid) val = short8v{};
+        *reinterpret_cast<short8v*>(lds_rm + row * ldst_bytes + c0 * 2) = val;
+    }
+}
+
+template <int ROWS_TILE, int DD, int NG>
+DEVINL void write_sub16(const short8v (&st)[NG], char* lds16, int rows_valid, int tid) {
+    constexpr int GPR = DD / 8;
+    constexpr int SUBE = ROWS_TILE * 16 + 8;
+    const bool tail = rows_valid < ROWS_TILE;
+#pragma unroll
+    for (int i = 0; i < NG; ++i) {
+        int g = tid + i * 256;
+        int row = g / GPR, c0 = (g % GPR) * 8;
+        short8v val = st[i];
+        if (tail && row >= rows_valid) val = short8v{};
+        *reinterpret_cast<short8v*>(
+            lds16 + ((c0 / 16) * SUBE + row * 16 + (c0 % 16)) * 2) = val;
+    }
+}
+
+template <int ROWS_TILE, int DD, int NG>
+DEVINL void write_rm_sub16_c(const short8v (&st)[NG], char* lds_rm, int ldst_bytes,
+                             char* lds16, int rows_valid, int tid) {
+    constexpr int GPR = DD / 8;
+    constexpr int SUBE = ROWS_TILE * 16 + 8;
+    const bool tail = rows_valid < ROWS_TILE;
+#pragma unroll
+    for (int i = 0; i < NG; ++i) {
+        int g = tid + i * 256;
+        int row = g / GPR, c0 = (g % GPR) * 8;
+        short8v val = st[i];
+        if (tail && row >= rows_valid) val = short8v{};
+        *reinterpret_cast<short8v*>(lds_rm + row * ldst_bytes + c0 * 2) = val;
+        *reinterpret_cast<short8v*>(
+            lds16 + ((c0 / 16) * SUBE + row * 16 + (c0 % 16)) * 2) = val;
+    }
+}

@@ -119,60 +119,6 @@ DEVINL void stage_rm_sub16(const unsigned short* __restrict__ src, long src_stri
     }
 }
 
-// ---- T14 async-STAGE split (exact-template fast path) ----
-// issue: clamped-row unconditional vector loads into registers — the loads go
-// out early and their vmcnt waits land at the write call one compute phase
-// later, hiding HBM latency under the MFMA work (the synchronous stage_* path
-// above pays ~2k cycles of serial staging per tile). Only valid when the
-// runtime dims match the template exactly (every granule full-width).
-template <int ROWS_TILE, int DD, int NG>
-DEVINL void issue_tile(short8v (&st)[NG], const unsigned short* __restrict__ src,
-                       long sstride, int row_limit, int tid) {
-    constexpr int GPR = DD / 8;
-#pragma unroll
-    for (int i = 0; i < NG; ++i) {
-        int g = tid + i * 256;
-        int row = g / GPR, c0 = (g % GPR) * 8;
-        int rowc = min(row, row_limit - 1);
-        st[i] = *reinterpret_cast<const short8v*>(src + (long)rowc * sstride + c0);
-    }
-}
-
-// write halves: spill the staged registers into the row-major (and subtiled)
-// images, zero-filling clamp-duplicated tail rows
-template <int ROWS_TILE, int DD, int NG>
-DEVINL void write_rm_sub16(const short8v (&st)[NG], char* lds_rm, int ldst_bytes,
-                           char* lds16, int rows_valid, int tid) {
-    constexpr int GPR = DD / 8;
-    constexpr int SUBE = ROWS_TILE * 16 + 8;
-    const bool tail = rows_valid < ROWS_TILE;
-#pragma unroll
-    for (int i = 0; i < NG; ++i) {
-        int g = tid + i * 256;
-        int row = g / GPR, c0 = (g % GPR) * 8;
-        short8v val = st[i];
-        if (tail && row >= rows_valid) val = short8v{};
-        *reinterpret_cast<short8v*>(lds_rm + row * ldst_bytes + c0 * 2) = val;
-        *reinterpret_cast<short8v*>(
-            lds16 + ((c0 / 16) * SUBE + row * 16 + (c0 % 16)) * 2) = val;
-    }
-}
-
-template <int ROWS_TILE, int DD, int NG>
-DEVINL void write_rm(const short8v (&st)[NG], char* lds_rm, int ldst_bytes,
-                     int rows_valid, int tid) {
-    constexpr int GPR = DD / 8;
-    const bool tail = rows_valid < ROWS_TILE;
-#pragma unroll
-    for (int i = 0; i < NG; ++i) {
-        int g = tid + i * 256;
-        int row = g / GPR, c0 = (g % GPR) * 8;
-        short8v val = st[i];
-        if (tail && row >= rows_valid) val = short8v{};
-        *reinterpret_cast<short8v*>(lds_rm + row * ldst_bytes + c0 * 2) = val;
-    }
-}
-
 // stage (rows_tile x d) tile row-major into LDS (row stride ldst_bytes), zero-pad
 template <int ROWS_TILE>
 DEVINL void stage_rm(const unsigned short* __restrict__ src, long src_stride,
@@ -306,7 +252,8 @@ __global__ void flash_dq_kernel(
 
     // T14 split staging on exact template matches: K/V loads for tile t+1 fly
     // during tile t's compute instead of serializing between the barriers
-    constexpr bool kFast = (DMAX % 32 == 0) && (DVMAX % 32 == 0) && DMAX <= 64 &&
+    constexpr bool kFast = (DMAX % 32 == 0) && (DVMAX % 32 == 0) &&
+                           (DMAX <= 64 || DMAX == 288) &&
                            ((TILE * DMAX) % 2048 == 0) && ((TILE * DVMAX) % 2048 == 0);
     constexpr int NG_K = kFast ? (TILE * DMAX) / 2048 : 1;
     constexpr int NG_V = kFast ? (TILE * DVMAX) / 2048 : 1;
@@ -323,7 +270,7 @@ __global__ void flash_dq_kernel(
         int rows_valid = min(TILE, Lk - kv0);
         __syncthreads();
         if (fast) {
-            write_rm_sub16<TILE, DMAX>(st_k, k_lds, k_stride, kt16_lds, rows_valid, tid);
+            write_rm_sub16_c<TILE, DMAX>(st_k, k_lds, k_stride, kt16_lds, rows_valid, tid);
             write_rm<TILE, DVMAX>(st_v, v_lds, v_stride, rows_valid, tid);
             int kv_n = min(kv0 + TILE, kv_last);
             issue_tile<TILE, DMAX>(st_k, kbase + (long)kv_n * ksn, ksn, Lk - kv_n, tid);
@@ -576,8 +523,8 @@ __global__ void flash_dkv_kernel(
         int rows_valid = min(TILE, Nq - qt0);
         __syncthreads();
         if (fast) {
-            write_rm_sub16<TILE, DMAX>(st_q, q_lds, q_stride, q16_lds, rows_valid, tid);
-            write_rm_sub16<TILE, DVMAX>(st_do, do_lds, do_stride, do16_lds, rows_valid, tid);
+            write_rm_sub16_c<TILE, DMAX>(st_q, q_lds, q_stride, q16_lds, rows_valid, tid);
+            write_rm_sub16_c<TILE, DVMAX>(st_do, do_lds, do_stride, do16_lds, rows_valid, tid);
             int qt_n = min(qt0 + TILE, qt_last);
             issue_tile<TILE, DMAX>(st_q, qbase + (long)qt_n * qsn, qsn, Nq - qt_n, tid);
             issue_tile<TILE, DVMAX>(st_do, dobase + (long)qt_n * Dv, Dv, Nq - qt_n, tid);
@@ -828,6 +775,20 @@ std::vector<torch::Tensor> flash_bwd(torch::Tensor dout, torch::Tensor q, torch:
     if (v.stride(3) != 1) v = v.contiguous();
     int B = q.size(0), H = q.size(1), Nq = q.size(2), D = q.size(3);
     int Lk = k.size(2), Dv = v.size(3);
+
+    // mirror the forward's pad-to-288 shim (see flash_fwd.hip): zero-padded
+    // channels are exact — padded dout/out columns contribute 0 to delta, and
+    // the padded dq/dk/dv columns are sliced off
+    if (D > 160 && (D != 288 || Dv != 288) && D <= 288 && Dv <= 288) {
+        namespace F = torch::nn::functional;
+        auto res = flash_bwd(F::pad(dout, F::PadFuncOptions({0, 288 - Dv})),
+                             F::pad(q, F::PadFuncOptions({0, 288 - D})),
+                             F::pad(k, F::PadFuncOptions({0, 288 - D})),
+                             F::pad(v, F::PadFuncOptions({0, 288 - Dv})),
+                             F::pad(out, F::PadFuncOptions({0, 288 - Dv})),
+                             lse, pad_mask, causal, dropout_p, seed);
+        return {res[0].narrow(-1, 0, D), res[1].narrow(-1, 0, D), res[2].narrow(-1, 0, Dv)};
+    }
 
     auto delta = torch::empty({B, H, Nq}, q.options().dtype(torch::kFloat32));
     {
