@@ -506,7 +506,8 @@ __global__ void flash_dkv_kernel(
 
     // T14 split staging (see flash_dq_kernel): Q/dO tile t+1 loads fly under
     // tile t's MFMA work on exact template matches
-    constexpr bool kFast = (DMAX % 32 == 0) && (DVMAX % 32 == 0) && DMAX <= 128 &&
+    constexpr bool kFast = (DMAX % 32 == 0) && (DVMAX % 32 == 0) &&
+                           (DMAX <= 128 || DMAX == 288) &&
                            ((TILE * DMAX) % 2048 == 0) && ((TILE * DVMAX) % 2048 == 0);
     constexpr int NG_Q = kFast ? (TILE * DMAX) / 2048 : 1;
     constexpr int NG_DO = kFast ? (TILE * DVMAX) / 2048 : 1;
@@ -745,6 +746,15 @@ void launch_flash_bwd(const torch::Tensor& dout, const torch::Tensor& q, const t
                       (size_t)DKV_TILE * do_stride + (size_t)(DVMAX / 16) * (DKV_TILE * 16 + 8) * 2 +
                       (size_t)NWAVES * 16 * DKV_QH * qt_stride;
         dim3 grid((Lk + kblk - 1) / kblk, B * H);
+        if (smem > 65536) {
+            static bool raised = [] {
+                (void)hipFuncSetAttribute(
+                    reinterpret_cast<const void*>(&flash_dkv_kernel<DMAX, DVMAX, DKV_TILE, DKV_QH>),
+                    hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+                return true;
+            }();
+            (void)raised;
+        }
         hipLaunchKernelGGL((flash_dkv_kernel<DMAX, DVMAX, DKV_TILE, DKV_QH>), grid, dim3(256), smem, stream,
                            reinterpret_cast<const unsigned short*>(q.data_ptr()),
                            reinterpret_cast<const unsigned short*>(k.data_ptr()),

@@ -246,11 +246,33 @@ __global__ void flash_fwd_pipe_kernel(
     }
     __syncthreads();
 
+    // O += P(tile) @ V(tile) from the wave-private transposed-P image and the
+    // given V buffer — called one tile LATE (T15-style single-tile software
+    // pipeline) so these MFMAs sit beside the CURRENT tile's softmax VALU in
+    // the instruction stream, hiding it on the separate pipes
+    auto pv_accumulate = [&](const char* v_buf) {
+#pragma unroll
+        for (int kb32 = 0; kb32 < KEYBLKS / 2; ++kb32) {
+            bf16x8 a_frag[QH];
+#pragma unroll
+            for (int h = 0; h < QH; ++h)
+                a_frag[h] = read_frag_tr16(p_mine, h, kb32 * 32, hi4, lo16);
+#pragma unroll
+            for (int cb = 0; cb < CBLOCKS; ++cb) {
+                bf16x8 bfrag = read_frag_tr16(v_buf, cb, kb32 * 32, hi4, lo16);
+#pragma unroll
+                for (int h = 0; h < QH; ++h) {
+                    o_acc[h][cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        a_frag[h], bfrag, o_acc[h][cb], 0, 0, 0);
+                }
+            }
+        }
+    };
+
     for (int ti = 0; ti < n_tiles; ++ti) {
         const int kv0 = kv_begin + ti * KVBLK;
         const int buf = ti & 1;
         char* k_cur = k_lds + buf * KVBLK * K_STRIDE;
-        char* v_cur = v_lds + buf * NSUB * SUB_ELEMS * 2;
         const float* bias_cur = bias_lds + buf * KVBLK;
 
         // ---- S = Q K^T ----
