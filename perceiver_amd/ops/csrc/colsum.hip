@@ -2,42 +2,66 @@
 // db[c] = sum_r dy[r][c]. torch's generic reduce picks a per-output-column
 // configuration that runs ~18x off the bandwidth roofline on the tall-skinny
 // shapes Perceiver produces (e.g. 401k x 261 for the image-classifier K/V
-// projections, 11.7 ms/step of an 85 ms step). This kernel streams the matrix
-// row-major coalesced, accumulates per-thread fp32 partials in registers and
-// combines with one global atomicAdd per column per block.
+// projections). This kernel reads whole rows coalesced at 16 B/lane,
+// accumulates fp32 partials in registers over a block-strided row range,
+// combines the block through LDS and publishes ONE global atomicAdd per
+// column per block (a few hundred blocks -> atomic traffic is negligible).
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
 #include "common.h"
 
 namespace {
 
-constexpr int TPB = 256;
+constexpr int LANES = 64;   // x: one wave covers 64 granules = 512 columns
+constexpr int ROWS = 4;     // y: rows in flight per block
 
 __global__ void colsum_kernel(const unsigned short* __restrict__ x,
                               float* __restrict__ out, long rows, int C) {
-    // each block owns a row range; thread t covers columns t, t+TPB, ...
-    // (<= 8 register accumulators at C <= 2048)
-    // statically-indexed accumulators with runtime guards (a runtime loop
-    // bound would push acc[] to scratch — guide rule 20)
-    float acc[8];
+    // lane x covers granules x, x+64, x+128, x+192 (8 columns each, C<=2048);
+    // statically-indexed accumulators (runtime bounds would scratch them)
+    float acc[4][8];
 #pragma unroll
-    for (int i = 0; i < 8; ++i) acc[i] = 0.f;
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+        for (int e = 0; e < 8; ++e) acc[j][e] = 0.f;
 
-    long r0 = (long)blockIdx.x * blockDim.y + threadIdx.y;
-    for (long r = r0; r < rows; r += (long)gridDim.x * blockDim.y) {
+    const int gpr = (C + 7) / 8;  // granules per row
+    for (long r = (long)blockIdx.x * ROWS + threadIdx.y; r < rows;
+         r += (long)gridDim.x * ROWS) {
         const unsigned short* row = x + r * C;
 #pragma unroll
-        for (int i = 0; i < 8; ++i) {
-            int c = threadIdx.x + i * TPB;
-            if (c < C) acc[i] += bf2f(row[c]);
+        for (int j = 0; j < 4; ++j) {
+            int g = threadIdx.x + j * LANES;
+            int c0 = g * 8;
+            if (g < gpr) {
+                if (c0 + 8 <= C) {
+                    short8v v = *reinterpret_cast<const short8v*>(row + c0);
+#pragma unroll
+                    for (int e = 0; e < 8; ++e) acc[j][e] += bf2f((unsigned short)v[e]);
+                } else {
+#pragma unroll
+                    for (int e = 0; e < 8; ++e)
+                        if (c0 + e < C) acc[j][e] += bf2f(row[c0 + e]);
+                }
+            }
         }
     }
-    // combine the block's row-parallel (threadIdx.y) partials via global
-    // atomics — fp32, pre-zeroed by the launcher
+
+    // combine the ROWS per-column partials through LDS, one atomic per column
+    __shared__ float red[2048];
+    for (int i = threadIdx.y * LANES + threadIdx.x; i < 2048; i += LANES * ROWS)
+        red[i] = 0.f;
+    __syncthreads();
 #pragma unroll
-    for (int i = 0; i < 8; ++i) {
-        int c = threadIdx.x + i * TPB;
-        if (c < C) atomicAdd(&out[c], acc[i]);
+    for (int j = 0; j < 4; ++j) {
+        int c0 = (threadIdx.x + j * LANES) * 8;
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+            if (c0 + e < C) atomicAdd(&red[c0 + e], acc[j][e]);
+    }
+    __syncthreads();
+    if (threadIdx.y == 0) {
+        for (int c = threadIdx.x; c < C; c += LANES) atomicAdd(&out[c], red[c]);
     }
 }
 
@@ -51,9 +75,8 @@ torch::Tensor colsum_bf16(torch::Tensor x) {
     TORCH_CHECK(C <= 2048, "colsum_bf16: C must be <= 2048");
     auto out = torch::zeros({(long)C}, x.options().dtype(torch::kFloat32));
     if (rows == 0) return out;
-    // 4 rows per block (dim.y), enough blocks to fill the chip
-    dim3 block(TPB, 4);
-    long nblocks = std::min((rows + 3) / 4, (long)2048);
+    dim3 block(LANES, ROWS);
+    long nblocks = std::min((rows + ROWS - 1) / ROWS, (long)512);
     hipLaunchKernelGGL(colsum_kernel, dim3(nblocks), block, 0,
                        at::cuda::getCurrentCUDAStream(),
                        reinterpret_cast<const unsigned short*>(x.data_ptr()),

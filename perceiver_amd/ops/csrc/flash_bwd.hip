@@ -401,6 +401,8 @@ __global__ void flash_dkv_kernel(
     const float* __restrict__ lsep, const float* __restrict__ deltap,
     const bool* __restrict__ pad,
     unsigned short* __restrict__ dkp, unsigned short* __restrict__ dvp,
+    float* __restrict__ dk_part, float* __restrict__ dv_part,  // (S,B*H,Lk,D[v]) when gridDim.z>1
+    long q_chunk,
     long qsb, long qsh, long qsn, long ksb, long ksh, long ksn,
     long vsb, long vsh, long vsn,
     int B, int H, int Nq, int Lk, int D, int Dv, int causal,
@@ -502,6 +504,15 @@ __global__ void flash_dkv_kernel(
         q_start = max(0, j_lo - (Lk - Nq));
         q_start = (q_start / TILE) * TILE;
     }
+    // Q-split over gridDim.z for grids that underfill the chip (e.g. the
+    // flow/img DECODER cross-attention backward: 50k-182k queries against a
+    // few hundred latent keys = a handful of key blocks); fp32 partials
+    // summed by the launcher
+    int q_end = Nq;
+    if (gridDim.z > 1) {
+        q_start = max(q_start, (int)((long)blockIdx.z * q_chunk));
+        q_end = (int)min((long)Nq, (long)(blockIdx.z + 1) * q_chunk);
+    }
     const unsigned int drop_thresh = (unsigned int)(drop_p * 4294967296.0);
 
     // T14 split staging (see flash_dq_kernel): Q/dO tile t+1 loads fly under
@@ -513,14 +524,14 @@ __global__ void flash_dkv_kernel(
     constexpr int NG_DO = kFast ? (TILE * DVMAX) / 2048 : 1;
     short8v st_q[NG_Q], st_do[NG_DO];
     const bool fast = kFast && D == DMAX && Dv == DVMAX;
-    const int nq_tiles = (Nq > q_start) ? (Nq - q_start + TILE - 1) / TILE : 0;
+    const int nq_tiles = (q_end > q_start) ? (q_end - q_start + TILE - 1) / TILE : 0;
     const int qt_last = q_start + (nq_tiles > 0 ? (nq_tiles - 1) * TILE : 0);
     if (fast && nq_tiles > 0) {
         issue_tile<TILE, DMAX>(st_q, qbase + (long)q_start * qsn, qsn, Nq - q_start, tid);
         issue_tile<TILE, DVMAX>(st_do, dobase + (long)q_start * Dv, Dv, Nq - q_start, tid);
     }
 
-    for (int qt0 = q_start; qt0 < Nq; qt0 += TILE) {
+    for (int qt0 = q_start; qt0 < q_end; qt0 += TILE) {
         int rows_valid = min(TILE, Nq - qt0);
         __syncthreads();
         if (fast) {
@@ -660,24 +671,41 @@ __global__ void flash_dkv_kernel(
         }
     }
 
-    // store dK/dV (C layout: key row h*16 + hi4*4+r, col lo16+16cb)
+    // store dK/dV (C layout: key row h*16 + hi4*4+r, col lo16+16cb);
+    // fp32 partial slabs when Q-split (summed by the launcher)
 #pragma unroll
     for (int h = 0; h < QH; ++h)
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
             int ki = k0 + h * 16 + hi4 * 4 + r;
             if (ki >= Lk) continue;
-            unsigned short* dkrow = dkp + ((long)bh * Lk + ki) * D;
-            unsigned short* dvrow = dvp + ((long)bh * Lk + ki) * Dv;
+            if (gridDim.z > 1) {
+                long base = ((long)blockIdx.z * B * H + bh) * Lk + ki;
+                float* dkrow = dk_part + base * D;
+                float* dvrow = dv_part + base * Dv;
 #pragma unroll
-            for (int cb = 0; cb < DMAX / 16; ++cb) {
-                int c = cb * 16 + lo16;
-                if (c < D) dkrow[c] = f2bf(dk_acc[h][cb][r]);
-            }
+                for (int cb = 0; cb < DMAX / 16; ++cb) {
+                    int c = cb * 16 + lo16;
+                    if (c < D) dkrow[c] = dk_acc[h][cb][r];
+                }
 #pragma unroll
-            for (int cb = 0; cb < DVMAX / 16; ++cb) {
-                int c = cb * 16 + lo16;
-                if (c < Dv) dvrow[c] = f2bf(dv_acc[h][cb][r]);
+                for (int cb = 0; cb < DVMAX / 16; ++cb) {
+                    int c = cb * 16 + lo16;
+                    if (c < Dv) dvrow[c] = dv_acc[h][cb][r];
+                }
+            } else {
+                unsigned short* dkrow = dkp + ((long)bh * Lk + ki) * D;
+                unsigned short* dvrow = dvp + ((long)bh * Lk + ki) * Dv;
+#pragma unroll
+                for (int cb = 0; cb < DMAX / 16; ++cb) {
+                    int c = cb * 16 + lo16;
+                    if (c < D) dkrow[c] = f2bf(dk_acc[h][cb][r]);
+                }
+#pragma unroll
+                for (int cb = 0; cb < DVMAX / 16; ++cb) {
+                    int c = cb * 16 + lo16;
+                    if (c < Dv) dvrow[c] = f2bf(dv_acc[h][cb][r]);
+                }
             }
         }
 }
@@ -738,14 +766,37 @@ void launch_flash_bwd(const torch::Tensor& dout, const torch::Tensor& q, const t
             dq.copy_(dq_part.sum(0).view_as(dq));
         }
     }
-    {   // dK/dV
+    {   // dK/dV (with Q-split when the key grid underfills the chip)
         const int q_stride = d_pad * 2 + 16, qt_stride = DKV_TILE * 2 + 16;
         const int do_stride = dv_pad * 2 + 16;
         const int kblk = 16 * DKV_QH * NWAVES;
         size_t smem = (size_t)DKV_TILE * q_stride + (size_t)(DMAX / 16) * (DKV_TILE * 16 + 8) * 2 +
                       (size_t)DKV_TILE * do_stride + (size_t)(DVMAX / 16) * (DKV_TILE * 16 + 8) * 2 +
                       (size_t)NWAVES * 16 * DKV_QH * qt_stride;
-        dim3 grid((Lk + kblk - 1) / kblk, B * H);
+        int gx = (Lk + kblk - 1) / kblk, gy = B * H;
+        int nsplit = 1;
+        long q_chunk = Nq;
+        if (!causal && (long)gx * gy < 512 && Nq > 4 * DKV_TILE) {
+            int want = (int)(512 / ((long)gx * gy)) + 1;
+            int max_split = (Nq + 4 * DKV_TILE - 1) / (4 * DKV_TILE);
+            nsplit = std::min({want, max_split, 32});
+            long tiles = (Nq + DKV_TILE - 1) / DKV_TILE;
+            long tiles_per = (tiles + nsplit - 1) / nsplit;
+            q_chunk = tiles_per * DKV_TILE;
+            nsplit = (int)((Nq + q_chunk - 1) / q_chunk);
+        }
+        torch::Tensor dk_part, dv_part;
+        float* dk_part_p = nullptr;
+        float* dv_part_p = nullptr;
+        if (nsplit > 1) {
+            dk_part = torch::zeros({(long)nsplit, (long)B * H, (long)Lk, (long)D},
+                                   q.options().dtype(torch::kFloat32));
+            dv_part = torch::zeros({(long)nsplit, (long)B * H, (long)Lk, (long)Dv},
+                                   q.options().dtype(torch::kFloat32));
+            dk_part_p = dk_part.data_ptr<float>();
+            dv_part_p = dv_part.data_ptr<float>();
+        }
+        dim3 grid(gx, gy, nsplit);
         if (smem > 65536) {
             static bool raised = [] {
                 (void)hipFuncSetAttribute(
@@ -763,11 +814,16 @@ void launch_flash_bwd(const torch::Tensor& dout, const torch::Tensor& q, const t
                            lse.data_ptr<float>(), delta.data_ptr<float>(), padp,
                            reinterpret_cast<unsigned short*>(dk.data_ptr()),
                            reinterpret_cast<unsigned short*>(dv.data_ptr()),
+                           dk_part_p, dv_part_p, q_chunk,
                            q.stride(0), q.stride(1), q.stride(2),
                            k.stride(0), k.stride(1), k.stride(2),
                            v.stride(0), v.stride(1), v.stride(2),
                            B, H, Nq, Lk, D, Dv, (int)causal, drop_p, drop_seed);
         HIP_CHECK_LAST();
+        if (nsplit > 1) {
+            dk.copy_(dk_part.sum(0).view_as(dk));
+            dv.copy_(dv_part.sum(0).view_as(dv));
+        }
     }
 }
 

@@ -295,6 +295,16 @@ __global__ void flash_fwd_pipe_kernel(
             }
         }
 
+        // ---- previous tile's PV: independent of s_acc, so the scheduler
+        // interleaves its MFMAs with the softmax VALU below. The rescale in
+        // the softmax touches o_acc only after these MFMAs in program order —
+        // exactly the T13-safe ordering (P(t-1) fully applied at its own
+        // scale before tile t's max can rescale O). p_mine is rewritten only
+        // below the reads (same wave; LDS ops complete in order). ----
+        if (ti > 0) {
+            pv_accumulate(v_lds + ((ti - 1) & 1) * NSUB * SUB_ELEMS * 2);
+        }
+
         // ---- mask + online softmax; P packed transposed into LDS ----
         // per-key bias covers oob + pad; the causal chain only runs on tiles
         // the right-aligned mask actually cuts (wave-uniform hoist)
@@ -372,26 +382,6 @@ __global__ void flash_fwd_pipe_kernel(
             for (int r = 0; r < 4; ++r) l_run[h][r] += warp16_sum(psum_r[r]);
         }
 
-        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");  // P writes visible to own reads
-
-        // ---- O += P V via transpose reads ----
-#pragma unroll
-        for (int kb32 = 0; kb32 < KEYBLKS / 2; ++kb32) {
-            bf16x8 a_frag[QH];
-#pragma unroll
-            for (int h = 0; h < QH; ++h)
-                a_frag[h] = read_frag_tr16(p_mine, h, kb32 * 32, hi4, lo16);
-#pragma unroll
-            for (int cb = 0; cb < CBLOCKS; ++cb) {
-                bf16x8 bfrag = read_frag_tr16(v_cur, cb, kb32 * 32, hi4, lo16);
-#pragma unroll
-                for (int h = 0; h < QH; ++h) {
-                    o_acc[h][cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        a_frag[h], bfrag, o_acc[h][cb], 0, 0, 0);
-                }
-            }
-        }
-
         __syncthreads();  // B1: all waves done with tile ti's LDS buffers
         if (ti + 1 < n_tiles) {
             stage_write(st, k_lds + ((ti + 1) & 1) * KVBLK * K_STRIDE,
@@ -402,6 +392,11 @@ __global__ void flash_fwd_pipe_kernel(
                         min(kv0 + 2 * KVBLK, kv_last), Lk, tid);
             __syncthreads();  // B2: tile ti+1 visible
         }
+    }
+
+    // drain the software pipeline: the last tile's PV
+    if (n_tiles > 0) {
+        pv_accumulate(v_lds + ((n_tiles - 1) & 1) * NSUB * SUB_ELEMS * 2);
     }
 
     // ---- epilogue: O /= l, staged through LDS, stored as dwordx4 rows ----

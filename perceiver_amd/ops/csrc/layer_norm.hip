@@ -162,122 +162,83 @@ __global__ void ln_bwd_dx_kernel(const unsigned short* __restrict__ dy,
     }
 }
 
-// Fused dx + dw/db backward: one pass over (dy, x) computes dx AND accumulates
-// the weight/bias gradients — the separate dwdb kernel re-read both tensors in
-// full (an extra ~84 MB round trip per flagship LN). Each wave grid-strides
-// over rows keeping per-column partials in registers; the block's 8 waves
-// combine through LDS atomics and one wave publishes with global fp32
-// atomicAdds (dw/db zeroed by the launcher).
-constexpr int BWD_WAVES = 8;
+// dW/db reduction, bandwidth-shaped: whole rows read coalesced at 16 B/lane,
+// fp32 per-column partials in registers over a block-strided row range,
+// block-combined through LDS, ONE global atomicAdd per column per block
+// (dw/db buffers zeroed by the launcher). Replaces both the old per-column
+// dwdb kernel (~3x off roofline) and its torch partial-sum epilogue.
+constexpr int DW_LANES = 64;
+constexpr int DW_ROWS = 4;
 
-template <int CHUNKS>
-__launch_bounds__(64 * BWD_WAVES)
-__global__ void ln_bwd_fused_kernel(const unsigned short* __restrict__ dy,
-                                    const unsigned short* __restrict__ x,
-                                    const unsigned short* __restrict__ w,
-                                    const float* __restrict__ mean_in,
-                                    const float* __restrict__ rstd_in,
-                                    unsigned short* __restrict__ dx,
-                                    float* __restrict__ dw_accum,
-                                    float* __restrict__ db_accum,
-                                    long rows, int C) {
-    const int lane = threadIdx.x % 64;
-    const int wave = threadIdx.x / 64;
-    __shared__ float red[2][CHUNKS * 512];
-    for (int i = threadIdx.x; i < CHUNKS * 512; i += 64 * BWD_WAVES) {
-        red[0][i] = 0.f;
-        red[1][i] = 0.f;
-    }
-    __syncthreads();
+__global__ void ln_bwd_dwdb_fast_kernel(const unsigned short* __restrict__ dy,
+                                        const unsigned short* __restrict__ x,
+                                        const float* __restrict__ mean_in,
+                                        const float* __restrict__ rstd_in,
+                                        float* __restrict__ dw_accum,
+                                        float* __restrict__ db_accum,
+                                        long rows, int C) {
+    float dw[4][8], db[4][8];
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+        for (int e = 0; e < 8; ++e) { dw[j][e] = 0.f; db[j][e] = 0.f; }
 
-    float dwacc[CHUNKS][8], dbacc[CHUNKS][8];
+    const int gpr = (C + 7) / 8;
+    for (long r = (long)blockIdx.x * DW_ROWS + threadIdx.y; r < rows;
+         r += (long)gridDim.x * DW_ROWS) {
+        const unsigned short* dyrow = dy + r * C;
+        const unsigned short* xrow = x + r * C;
+        float mean = mean_in[r], rstd = rstd_in[r];
 #pragma unroll
-    for (int i = 0; i < CHUNKS; ++i)
+        for (int j = 0; j < 4; ++j) {
+            int g = threadIdx.x + j * DW_LANES;
+            int c0 = g * 8;
+            if (g < gpr) {
+                if (c0 + 8 <= C) {
+                    short8v dv = *reinterpret_cast<const short8v*>(dyrow + c0);
+                    short8v xv = *reinterpret_cast<const short8v*>(xrow + c0);
 #pragma unroll
-        for (int e = 0; e < 8; ++e) { dwacc[i][e] = 0.f; dbacc[i][e] = 0.f; }
-
-    for (long row = (long)blockIdx.x * BWD_WAVES + wave; row < rows;
-         row += (long)gridDim.x * BWD_WAVES) {
-        const unsigned short* dyrow = dy + row * C;
-        const unsigned short* xrow = x + row * C;
-        unsigned short* dxrow = dx + row * C;
-        float mean = mean_in[row], rstd = rstd_in[row];
-
-        float g[CHUNKS][8], xh[CHUNKS][8];
-        float s1 = 0.f, s2 = 0.f;
+                    for (int e = 0; e < 8; ++e) {
+                        float d = bf2f((unsigned short)dv[e]);
+                        db[j][e] += d;
+                        dw[j][e] += d * (bf2f((unsigned short)xv[e]) - mean) * rstd;
+                    }
+                } else {
 #pragma unroll
-        for (int i = 0; i < CHUNKS; ++i) {
-            int c0 = lane * 8 + i * 512;
-#pragma unroll
-            for (int e = 0; e < 8; ++e) { g[i][e] = 0.f; xh[i][e] = 0.f; }
-            if (c0 + 8 <= C) {
-                short8v dyv = *reinterpret_cast<const short8v*>(dyrow + c0);
-                short8v xv = *reinterpret_cast<const short8v*>(xrow + c0);
-                short8v wv = *reinterpret_cast<const short8v*>(w + c0);
-#pragma unroll
-                for (int e = 0; e < 8; ++e) {
-                    float dv = bf2f((unsigned short)dyv[e]);
-                    float gg = dv * bf2f((unsigned short)wv[e]);
-                    float xhat = (bf2f((unsigned short)xv[e]) - mean) * rstd;
-                    g[i][e] = gg;
-                    xh[i][e] = xhat;
-                    s1 += gg;
-                    s2 += gg * xhat;
-                    dwacc[i][e] += dv * xhat;
-                    dbacc[i][e] += dv;
-                }
-            } else if (c0 < C) {
-#pragma unroll
-                for (int e = 0; e < 8; ++e) {
-                    if (c0 + e < C) {   // static index: runtime bounds would
-                        float dv = bf2f(dyrow[c0 + e]);  // scratch the arrays
-                        float gg = dv * bf2f(w[c0 + e]);
-                        float xhat = (bf2f(xrow[c0 + e]) - mean) * rstd;
-                        g[i][e] = gg;
-                        xh[i][e] = xhat;
-                        s1 += gg;
-                        s2 += gg * xhat;
-                        dwacc[i][e] += dv * xhat;
-                        dbacc[i][e] += dv;
+                    for (int e = 0; e < 8; ++e) {
+                        if (c0 + e < C) {
+                            float d = bf2f(dyrow[c0 + e]);
+                            db[j][e] += d;
+                            dw[j][e] += d * (bf2f(xrow[c0 + e]) - mean) * rstd;
+                        }
                     }
                 }
             }
         }
-        s1 = wave_sum(s1) / C;
-        s2 = wave_sum(s2) / C;
-
-#pragma unroll
-        for (int i = 0; i < CHUNKS; ++i) {
-            int c0 = lane * 8 + i * 512;
-            if (c0 + 8 <= C) {
-                short8v o;
-#pragma unroll
-                for (int e = 0; e < 8; ++e)
-                    o[e] = (short)f2bf(rstd * (g[i][e] - s1 - xh[i][e] * s2));
-                *reinterpret_cast<short8v*>(dxrow + c0) = o;
-            } else if (c0 < C) {
-#pragma unroll
-                for (int e = 0; e < 8; ++e)
-                    if (c0 + e < C)
-                        dxrow[c0 + e] = f2bf(rstd * (g[i][e] - s1 - xh[i][e] * s2));
-            }
-        }
     }
 
-    // cross-wave combine in LDS, then one global atomic per column per block
-#pragma unroll
-    for (int i = 0; i < CHUNKS; ++i) {
-        int c0 = lane * 8 + i * 512;
-#pragma unroll
-        for (int e = 0; e < 8; ++e) {
-            atomicAdd(&red[0][c0 + e], dwacc[i][e]);
-            atomicAdd(&red[1][c0 + e], dbacc[i][e]);
-        }
+    __shared__ float red[2][2048];
+    for (int i = threadIdx.y * DW_LANES + threadIdx.x; i < 2048; i += DW_LANES * DW_ROWS) {
+        red[0][i] = 0.f;
+        red[1][i] = 0.f;
     }
     __syncthreads();
-    for (int c = threadIdx.x; c < C; c += 64 * BWD_WAVES) {
-        atomicAdd(&dw_accum[c], red[0][c]);
-        atomicAdd(&db_accum[c], red[1][c]);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+        int c0 = (threadIdx.x + j * DW_LANES) * 8;
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+            if (c0 + e < C) {
+                atomicAdd(&red[0][c0 + e], dw[j][e]);
+                atomicAdd(&red[1][c0 + e], db[j][e]);
+            }
+    }
+    __syncthreads();
+    if (threadIdx.y == 0) {
+        for (int c = threadIdx.x; c < C; c += DW_LANES) {
+            atomicAdd(&dw_accum[c], red[0][c]);
+            atomicAdd(&db_accum[c], red[1][c]);
+        }
     }
 }
 
@@ -340,25 +301,6 @@ std::vector<torch::Tensor> ln_fwd(torch::Tensor x, torch::Tensor w, c10::optiona
     return {y, mean, rstd};
 }
 
-namespace {
-template <int CHUNKS>
-void launch_ln_bwd_fused(const torch::Tensor& dy, const torch::Tensor& x, const torch::Tensor& w,
-                         const torch::Tensor& mean, const torch::Tensor& rstd,
-                         torch::Tensor& dx, torch::Tensor& dwf, torch::Tensor& dbf,
-                         long rows, int C) {
-    long blocks = std::min((rows + BWD_WAVES - 1) / BWD_WAVES, (long)768);
-    hipLaunchKernelGGL((ln_bwd_fused_kernel<CHUNKS>), dim3(blocks), dim3(64 * BWD_WAVES), 0,
-                       at::cuda::getCurrentCUDAStream(),
-                       reinterpret_cast<const unsigned short*>(dy.data_ptr()),
-                       reinterpret_cast<const unsigned short*>(x.data_ptr()),
-                       reinterpret_cast<const unsigned short*>(w.data_ptr()),
-                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                       reinterpret_cast<unsigned short*>(dx.data_ptr()),
-                       dwf.data_ptr<float>(), dbf.data_ptr<float>(), rows, C);
-    HIP_CHECK_LAST();
-}
-}  // namespace
-
 std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor w,
                                   torch::Tensor mean, torch::Tensor rstd, bool needs_dwdb) {
     dy = dy.contiguous(); x = x.contiguous();
@@ -368,22 +310,26 @@ std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor x, torch::Tens
     auto dx = torch::empty_like(x);
     if (rows == 0) return {dx, torch::zeros_like(w), torch::zeros_like(w)};
     int chunks = (C + 511) / 512;
+    if (chunks == 1)      launch_ln_bwd_dx<1>(dy, x, wc, mean, rstd, dx, rows, C);
+    else if (chunks == 2) launch_ln_bwd_dx<2>(dy, x, wc, mean, rstd, dx, rows, C);
+    else if (chunks == 3) launch_ln_bwd_dx<3>(dy, x, wc, mean, rstd, dx, rows, C);
+    else                  launch_ln_bwd_dx<4>(dy, x, wc, mean, rstd, dx, rows, C);
 
     torch::Tensor dw, db;
     if (needs_dwdb) {
         auto dwf = torch::zeros({(long)C}, x.options().dtype(torch::kFloat32));
         auto dbf = torch::zeros_like(dwf);
-        if (chunks == 1)      launch_ln_bwd_fused<1>(dy, x, wc, mean, rstd, dx, dwf, dbf, rows, C);
-        else if (chunks == 2) launch_ln_bwd_fused<2>(dy, x, wc, mean, rstd, dx, dwf, dbf, rows, C);
-        else if (chunks == 3) launch_ln_bwd_fused<3>(dy, x, wc, mean, rstd, dx, dwf, dbf, rows, C);
-        else                  launch_ln_bwd_fused<4>(dy, x, wc, mean, rstd, dx, dwf, dbf, rows, C);
+        dim3 block(DW_LANES, DW_ROWS);
+        long nblocks = std::min((rows + DW_ROWS - 1) / DW_ROWS, (long)512);
+        hipLaunchKernelGGL(ln_bwd_dwdb_fast_kernel, dim3(nblocks), block, 0,
+                           at::cuda::getCurrentCUDAStream(),
+                           reinterpret_cast<const unsigned short*>(dy.data_ptr()),
+                           reinterpret_cast<const unsigned short*>(x.data_ptr()),
+                           mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                           dwf.data_ptr<float>(), dbf.data_ptr<float>(), rows, C);
+        HIP_CHECK_LAST();
         dw = dwf.to(x.scalar_type());
         db = dbf.to(x.scalar_type());
-    } else {
-        if (chunks == 1)      launch_ln_bwd_dx<1>(dy, x, wc, mean, rstd, dx, rows, C);
-        else if (chunks == 2) launch_ln_bwd_dx<2>(dy, x, wc, mean, rstd, dx, rows, C);
-        else if (chunks == 3) launch_ln_bwd_dx<3>(dy, x, wc, mean, rstd, dx, rows, C);
-        else                  launch_ln_bwd_dx<4>(dy, x, wc, mean, rstd, dx, rows, C);
     }
     return {dx, dw, db};
 }
