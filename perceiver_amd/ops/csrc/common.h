@@ -43,14 +43,30 @@ DEVINL unsigned int rng_hash(unsigned long long seed, int bh, int i, int j) {
     return (unsigned int)z;
 }
 
-// exact-erf GELU, matches torch.nn.GELU default
-DEVINL float gelu_f(float x) { return 0.5f * x * (1.0f + erff(x * 0.70710678118654752440f)); }
+// erf via the Abramowitz-Stegun 7.1.26 rational approximation (|err| <=
+// 1.5e-7 absolute — below bf16 output resolution, so GELU results match the
+// exact-erf torch.nn.GELU after rounding). libm's float-exact erff costs
+// several times more VALU and made the fused GELU kernels compute-bound.
+DEVINL float erf_fast(float x) {
+    const float ax = fabsf(x);
+    const float t = __frcp_rn(1.0f + 0.3275911f * ax);
+    float poly = 1.061405429f;
+    poly = poly * t - 1.453152027f;
+    poly = poly * t + 1.421413741f;
+    poly = poly * t - 0.284496736f;
+    poly = poly * t + 0.254829592f;
+    float e = 1.0f - poly * t * __expf(-ax * ax);
+    return copysignf(e, x);
+}
+
+// GELU matching torch.nn.GELU default (erf form) to bf16 output precision
+DEVINL float gelu_f(float x) { return 0.5f * x * (1.0f + erf_fast(x * 0.70710678118654752440f)); }
 
 DEVINL float gelu_grad_f(float x) {
     const float kInvSqrt2 = 0.70710678118654752440f;
     const float kInvSqrt2Pi = 0.39894228040143267794f;
-    float cdf = 0.5f * (1.0f + erff(x * kInvSqrt2));
-    float pdf = kInvSqrt2Pi * expf(-0.5f * x * x);
+    float cdf = 0.5f * (1.0f + erf_fast(x * kInvSqrt2));
+    float pdf = kInvSqrt2Pi * __expf(-0.5f * x * x);
     return cdf + x * pdf;
 }
 

@@ -20,11 +20,14 @@ __global__ void gelu_bias_fwd_kernel_bf16(
     long stride = (long)gridDim.x * blockDim.x * 8;
     for (long i = i0; i + 8 <= total; i += stride) {
         short8v xv = *reinterpret_cast<const short8v*>(x + i);
+        // one modulo per granule: i and C are both multiples of 8, so the
+        // 8-element span never wraps the bias row
+        const int c0 = bias ? (int)(i % C) : 0;
         short8v yv;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
             float v = bf2f((unsigned short)xv[j]);
-            if (bias) v += bf2f(bias[(i + j) % C]);
+            if (bias) v += bf2f(bias[c0 + j]);
             yv[j] = (short)f2bf(gelu_f(v));
         }
         *reinterpret_cast<short8v*>(y + i) = yv;
@@ -42,11 +45,12 @@ __global__ void gelu_bias_bwd_kernel_bf16(
     for (long i = i0; i + 8 <= total; i += stride) {
         short8v xv = *reinterpret_cast<const short8v*>(x + i);
         short8v gv = *reinterpret_cast<const short8v*>(dy + i);
+        const int c0 = bias ? (int)(i % C) : 0;
         short8v ov;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
             float v = bf2f((unsigned short)xv[j]);
-            if (bias) v += bf2f(bias[(i + j) % C]);
+            if (bias) v += bf2f(bias[c0 + j]);
             float g = bf2f((unsigned short)gv[j]);
             ov[j] = (short)f2bf(g * gelu_grad_f(v));
         }
