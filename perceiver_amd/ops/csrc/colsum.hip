@@ -26,6 +26,7 @@ __global__ void colsum_kernel(const unsigned short* __restrict__ x,
         for (int e = 0; e < 8; ++e) acc[j][e] = 0.f;
 
     const int gpr = (C + 7) / 8;  // granules per row
+#pragma unroll 2
     for (long r = (long)blockIdx.x * ROWS + threadIdx.y; r < rows;
          r += (long)gridDim.x * ROWS) {
         const unsigned short* row = x + r * C;
@@ -76,7 +77,7 @@ torch::Tensor colsum_bf16(torch::Tensor x) {
     auto out = torch::zeros({(long)C}, x.options().dtype(torch::kFloat32));
     if (rows == 0) return out;
     dim3 block(LANES, ROWS);
-    long nblocks = std::min((rows + ROWS - 1) / ROWS, (long)512);
+    long nblocks = std::min((rows + ROWS - 1) / ROWS, (long)1024);
     hipLaunchKernelGGL(colsum_kernel, dim3(nblocks), block, 0,
                        at::cuda::getCurrentCUDAStream(),
                        reinterpret_cast<const unsigned short*>(x.data_ptr()),

@@ -25,12 +25,17 @@ constexpr int NWAVES = 4;
 // ---------------------------------------------------------------- delta kernel
 __global__ void delta_kernel(const unsigned short* __restrict__ dout,
                              const unsigned short* __restrict__ out,
-                             float* __restrict__ delta, long rows, int dv) {
+                             float* __restrict__ delta,
+                             long dsb, long dsh, long dsn,  // dout strides
+                             long osb, long osh, long osn,  // out strides
+                             int H, int Nq, long rows, int dv) {
     long row = (long)blockIdx.x * (blockDim.x / 64) + threadIdx.x / 64;
     int lane = threadIdx.x % 64;
     if (row >= rows) return;
-    const unsigned short* a = dout + row * dv;
-    const unsigned short* b = out + row * dv;
+    long bb = row / ((long)H * Nq);
+    long rem = row % ((long)H * Nq);
+    const unsigned short* a = dout + bb * dsb + (rem / Nq) * dsh + (rem % Nq) * dsn;
+    const unsigned short* b = out + bb * osb + (rem / Nq) * osh + (rem % Nq) * osn;
     float acc = 0.f;
     for (int c = lane * 2; c + 1 < dv; c += 128) {
         acc += bf2f(a[c]) * bf2f(b[c]) + bf2f(a[c + 1]) * bf2f(b[c + 1]);
@@ -157,7 +162,7 @@ __global__ void flash_dq_kernel(
     float* __restrict__ dq_part,  // (S,B,H,Nq,D) fp32 when gridDim.z > 1
     long kv_chunk,
     long qsb, long qsh, long qsn, long ksb, long ksh, long ksn,
-    long vsb, long vsh, long vsn,
+    long vsb, long vsh, long vsn, long dsb, long dsh, long dsn,
     int B, int H, int Nq, int Lk, int D, int Dv, int causal,
     float drop_p, unsigned long long drop_seed) {
     constexpr int TBLKS = TILE / 16;
@@ -176,7 +181,7 @@ __global__ void flash_dq_kernel(
     const unsigned short* qbase = qp + (long)b * qsb + (long)hh * qsh;
     const unsigned short* kbase = kp + (long)b * ksb + (long)hh * ksh;
     const unsigned short* vbase = vp + (long)b * vsb + (long)hh * vsh;
-    const unsigned short* dobase = dop + (long)bh * Nq * Dv;
+    const unsigned short* dobase = dop + (long)b * dsb + (long)hh * dsh;
     const float* lse_row = lsep + (long)bh * Nq;
     const float* delta_row = deltap + (long)bh * Nq;
     const bool* padrow = pad ? pad + (long)b * Lk : nullptr;
@@ -200,7 +205,7 @@ __global__ void flash_dq_kernel(
         bool valid = qi < Nq;
         int qc = valid ? qi : Nq - 1;
         const unsigned short* qrow = qbase + (long)qc * qsn;
-        const unsigned short* dorow = dobase + (long)qc * Dv;
+        const unsigned short* dorow = dobase + (long)qc * dsn;
 #pragma unroll
         for (int kb = 0; kb < DMAX / 32; ++kb) {
             short8v val = {};
@@ -404,7 +409,7 @@ __global__ void flash_dkv_kernel(
     float* __restrict__ dk_part, float* __restrict__ dv_part,  // (S,B*H,Lk,D[v]) when gridDim.z>1
     long q_chunk,
     long qsb, long qsh, long qsn, long ksb, long ksh, long ksn,
-    long vsb, long vsh, long vsn,
+    long vsb, long vsh, long vsn, long dsb, long dsh, long dsn,
     int B, int H, int Nq, int Lk, int D, int Dv, int causal,
     float drop_p, unsigned long long drop_seed) {
     constexpr int TBLKS = TILE / 16;
@@ -423,7 +428,7 @@ __global__ void flash_dkv_kernel(
     const unsigned short* qbase = qp + (long)b * qsb + (long)hh * qsh;
     const unsigned short* kbase = kp + (long)b * ksb + (long)hh * ksh;
     const unsigned short* vbase = vp + (long)b * vsb + (long)hh * vsh;
-    const unsigned short* dobase = dop + (long)bh * Nq * Dv;
+    const unsigned short* dobase = dop + (long)b * dsb + (long)hh * dsh;
     const float* lse_row = lsep + (long)bh * Nq;
     const float* delta_row = deltap + (long)bh * Nq;
     const bool* padrow = pad ? pad + (long)b * Lk : nullptr;
@@ -528,7 +533,7 @@ __global__ void flash_dkv_kernel(
     const int qt_last = q_start + (nq_tiles > 0 ? (nq_tiles - 1) * TILE : 0);
     if (fast && nq_tiles > 0) {
         issue_tile<TILE, DMAX>(st_q, qbase + (long)q_start * qsn, qsn, Nq - q_start, tid);
-        issue_tile<TILE, DVMAX>(st_do, dobase + (long)q_start * Dv, Dv, Nq - q_start, tid);
+        issue_tile<TILE, DVMAX>(st_do, dobase + (long)q_start * dsn, dsn, Nq - q_start, tid);
     }
 
     for (int qt0 = q_start; qt0 < q_end; qt0 += TILE) {
@@ -539,11 +544,11 @@ __global__ void flash_dkv_kernel(
             write_rm_sub16_c<TILE, DVMAX>(st_do, do_lds, do_stride, do16_lds, rows_valid, tid);
             int qt_n = min(qt0 + TILE, qt_last);
             issue_tile<TILE, DMAX>(st_q, qbase + (long)qt_n * qsn, qsn, Nq - qt_n, tid);
-            issue_tile<TILE, DVMAX>(st_do, dobase + (long)qt_n * Dv, Dv, Nq - qt_n, tid);
+            issue_tile<TILE, DVMAX>(st_do, dobase + (long)qt_n * dsn, dsn, Nq - qt_n, tid);
         } else {
             stage_rm_sub16<TILE>(qbase + (long)qt0 * qsn, qsn, rows_valid, D, d_pad,
                                  q_lds, q_stride, q16_lds, tid);
-            stage_rm_sub16<TILE>(dobase + (long)qt0 * Dv, Dv, rows_valid, Dv, dv_pad,
+            stage_rm_sub16<TILE>(dobase + (long)qt0 * dsn, dsn, rows_valid, Dv, dv_pad,
                                  do_lds, do_stride, do16_lds, tid);
         }
         __syncthreads();
@@ -760,6 +765,7 @@ void launch_flash_bwd(const torch::Tensor& dout, const torch::Tensor& q, const t
                            q.stride(0), q.stride(1), q.stride(2),
                            k.stride(0), k.stride(1), k.stride(2),
                            v.stride(0), v.stride(1), v.stride(2),
+                           dout.stride(0), dout.stride(1), dout.stride(2),
                            B, H, Nq, Lk, D, Dv, (int)causal, drop_p, drop_seed);
         HIP_CHECK_LAST();
         if (nsplit > 1) {
@@ -818,6 +824,7 @@ void launch_flash_bwd(const torch::Tensor& dout, const torch::Tensor& q, const t
                            q.stride(0), q.stride(1), q.stride(2),
                            k.stride(0), k.stride(1), k.stride(2),
                            v.stride(0), v.stride(1), v.stride(2),
+                           dout.stride(0), dout.stride(1), dout.stride(2),
                            B, H, Nq, Lk, D, Dv, (int)causal, drop_p, drop_seed);
         HIP_CHECK_LAST();
         if (nsplit > 1) {
@@ -834,8 +841,10 @@ std::vector<torch::Tensor> flash_bwd(torch::Tensor dout, torch::Tensor q, torch:
                                      c10::optional<torch::Tensor> pad_mask, bool causal,
                                      double dropout_p, int64_t seed) {
     TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
-    dout = dout.contiguous();
-    out = out.contiguous();
+    // last-dim contiguity suffices; the merged-heads (B,N,H,Dv) layouts the
+    // forward now produces (and the matching grads) pass through zero-copy
+    if (dout.stride(3) != 1) dout = dout.contiguous();
+    if (out.stride(3) != 1) out = out.contiguous();
     if (q.stride(3) != 1) q = q.contiguous();
     if (k.stride(3) != 1) k = k.contiguous();
     if (v.stride(3) != 1) v = v.contiguous();
@@ -865,7 +874,10 @@ std::vector<torch::Tensor> flash_bwd(torch::Tensor dout, torch::Tensor q, torch:
                            at::cuda::getCurrentCUDAStream(),
                            reinterpret_cast<const unsigned short*>(dout.data_ptr()),
                            reinterpret_cast<const unsigned short*>(out.data_ptr()),
-                           delta.data_ptr<float>(), rows, Dv);
+                           delta.data_ptr<float>(),
+                           dout.stride(0), dout.stride(1), dout.stride(2),
+                           out.stride(0), out.stride(1), out.stride(2),
+                           H, Nq, rows, Dv);
         HIP_CHECK_LAST();
     }
 

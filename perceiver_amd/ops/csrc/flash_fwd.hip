@@ -141,6 +141,7 @@ __global__ void flash_fwd_kernel(            // budget allows (not the 352 templ
     long qsb, long qsh, long qsn,           // q strides (elements; last dim contiguous)
     long ksb, long ksh, long ksn,           // k strides
     long vsb, long vsh, long vsn,           // v strides
+    long osb, long osh, long osn,           // out strides (merged-heads layout)
     int B, int H, int Nq, int Lk, int D, int Dv, int causal,
     float drop_p, unsigned long long drop_seed) {
     const int d_pad = (D + 31) & ~31;
@@ -402,7 +403,7 @@ __global__ void flash_fwd_kernel(            // budget allows (not the 352 templ
                 }
                 if (lo16 == 0) lse_part[row] = lse_v;
             } else {
-                unsigned short* orow = op + ((long)bh * Nq + qi) * Dv;
+                unsigned short* orow = op + (long)b * osb + (long)hh * osh + (long)qi * osn;
 #pragma unroll
                 for (int cb = 0; cb < DVMAX / 16; ++cb) {
                     int c = cb * 16 + lo16;
@@ -419,6 +420,7 @@ __global__ void flash_merge_kernel(const float* __restrict__ o_part,
                                    const float* __restrict__ lse_part,
                                    unsigned short* __restrict__ op,
                                    float* __restrict__ lsep,
+                                   long osb, long osh, long osn, int H, int Nq,
                                    long rows, int nsplit, int dv) {
     long row = (long)blockIdx.x * (blockDim.x / 64) + threadIdx.x / 64;
     int lane = threadIdx.x % 64;
@@ -438,11 +440,14 @@ __global__ void flash_merge_kernel(const float* __restrict__ o_part,
     for (int i = 0; i < nsplit; ++i) wsum += wbuf[wv_id][i];
     float inv = (wsum > 0.f) ? 1.0f / wsum : 0.f;
 
+    long b = row / ((long)H * Nq);
+    long rem = row % ((long)H * Nq);
+    unsigned short* orow = op + b * osb + (rem / Nq) * osh + (rem % Nq) * osn;
     for (int c = lane; c < dv; c += 64) {
         float acc = 0.f;
         for (int i = 0; i < nsplit; ++i)
             acc += wbuf[wv_id][i] * o_part[((long)i * rows + row) * dv + c];
-        op[row * dv + c] = f2bf(acc * inv);
+        orow[c] = f2bf(acc * inv);
     }
     if (lane == 0) lsep[row] = mx + logf(fmaxf(wsum, 1e-37f));
 }
@@ -455,7 +460,9 @@ void launch_merge(float* o_part_p, float* lse_part_p, torch::Tensor& out, torch:
     hipLaunchKernelGGL(flash_merge_kernel, dim3(blocks), dim3(64 * wpb), 0, stream,
                        o_part_p, lse_part_p,
                        reinterpret_cast<unsigned short*>(out.data_ptr()),
-                       lse.data_ptr<float>(), rows, nsplit, Dv);
+                       lse.data_ptr<float>(),
+                       out.stride(0), out.stride(1), out.stride(2),
+                       (int)out.size(1), (int)out.size(2), rows, nsplit, Dv);
     HIP_CHECK_LAST();
 }
 
@@ -513,16 +520,11 @@ void launch_flash_fwd(const torch::Tensor& q, const torch::Tensor& k, const torc
                        q.stride(0), q.stride(1), q.stride(2),
                        k.stride(0), k.stride(1), k.stride(2),
                        v.stride(0), v.stride(1), v.stride(2),
+                       out.stride(0), out.stride(1), out.stride(2),
                        B, H, Nq, Lk, D, Dv, (int)causal, drop_p, drop_seed);
     HIP_CHECK_LAST();
     if (nsplit > 1) {
-        int wpb = 4;
-        long blocks = (rows + wpb - 1) / wpb;
-        hipLaunchKernelGGL(flash_merge_kernel, dim3(blocks), dim3(64 * wpb), 0, stream,
-                           o_part_p, lse_part_p,
-                           reinterpret_cast<unsigned short*>(out.data_ptr()),
-                           lse.data_ptr<float>(), rows, nsplit, Dv);
-        HIP_CHECK_LAST();
+        launch_merge(o_part_p, lse_part_p, out, lse, rows, nsplit, Dv);
     }
 }
 
@@ -578,7 +580,11 @@ std::vector<torch::Tensor> flash_fwd(torch::Tensor q, torch::Tensor k, torch::Te
         return {res[0].narrow(-1, 0, Dv), res[1]};
     }
 
-    auto out = torch::empty({q.size(0), q.size(1), q.size(2), (long)Dv}, q.options());
+    // out lives in merged-heads (B, N, H, Dv) memory and is returned as the
+    // (B, H, N, Dv) permuted view: the module's transpose+reshape after
+    // attention then costs nothing (it was a 42 MB copy per layer on MLM)
+    auto out = torch::empty({q.size(0), q.size(2), q.size(1), (long)Dv}, q.options())
+                   .permute({0, 2, 1, 3});
     auto lse = torch::empty({q.size(0), q.size(1), q.size(2)}, q.options().dtype(torch::kFloat32));
     if (q.numel() == 0 || k.numel() == 0) {
         // degenerate shapes fall back to the eager path upstream; just return zeros

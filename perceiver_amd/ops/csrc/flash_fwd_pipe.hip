@@ -164,6 +164,8 @@ __global__ void flash_fwd_pipe_kernel(
     long qsb, long qsh, long qsn,
     long ksb, long ksh, long ksn,
     long vsb, long vsh, long vsn,
+    long osb, long osh, long osn,  // out strides: flash returns the merged-
+                                   // heads (B,N,H,Dv) memory layout as a view
     int B, int H, int Nq, int Lk, int causal,
     float drop_p, unsigned long long drop_seed) {
     constexpr int KGR = DMAX / 32;          // 16-B staging granules per thread
@@ -464,7 +466,8 @@ __global__ void flash_fwd_pipe_kernel(
             int qi = q0 + row;
             short8v val = *reinterpret_cast<const short8v*>(o_mine + row * OROW + c0 * 2);
             if (all_valid || qi < Nq) {
-                *reinterpret_cast<short8v*>(op + ((long)bh * Nq + qi) * DVMAX + c0) = val;
+                *reinterpret_cast<short8v*>(
+                    op + (long)b * osb + (long)hh * osh + (long)qi * osn + c0) = val;
             }
         }
     }
@@ -522,6 +525,7 @@ void launch_pipe(const torch::Tensor& q, const torch::Tensor& k, const torch::Te
                        q.stride(0), q.stride(1), q.stride(2),
                        k.stride(0), k.stride(1), k.stride(2),
                        v.stride(0), v.stride(1), v.stride(2),
+                       out.stride(0), out.stride(1), out.stride(2),
                        B, H, Nq, Lk, (int)causal, drop_p, drop_seed);
     HIP_CHECK_LAST();
 }

@@ -184,6 +184,7 @@ __global__ void ln_bwd_dwdb_fast_kernel(const unsigned short* __restrict__ dy,
         for (int e = 0; e < 8; ++e) { dw[j][e] = 0.f; db[j][e] = 0.f; }
 
     const int gpr = (C + 7) / 8;
+#pragma unroll 2
     for (long r = (long)blockIdx.x * DW_ROWS + threadIdx.y; r < rows;
          r += (long)gridDim.x * DW_ROWS) {
         const unsigned short* dyrow = dy + r * C;
@@ -317,10 +318,13 @@ std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor x, torch::Tens
 
     torch::Tensor dw, db;
     if (needs_dwdb) {
-        auto dwf = torch::zeros({(long)C}, x.options().dtype(torch::kFloat32));
-        auto dbf = torch::zeros_like(dwf);
+        // one zeroed allocation for both accumulators (a separate FillFunctor
+        // launch per tiny tensor measured ~1 ms/step across the LN/bias sites)
+        auto acc = torch::zeros({2, (long)C}, x.options().dtype(torch::kFloat32));
+        auto dwf = acc[0];
+        auto dbf = acc[1];
         dim3 block(DW_LANES, DW_ROWS);
-        long nblocks = std::min((rows + DW_ROWS - 1) / DW_ROWS, (long)512);
+        long nblocks = std::min((rows + DW_ROWS - 1) / DW_ROWS, (long)1024);
         hipLaunchKernelGGL(ln_bwd_dwdb_fast_kernel, dim3(nblocks), block, 0,
                            at::cuda::getCurrentCUDAStream(),
                            reinterpret_cast<const unsigned short*>(dy.data_ptr()),
