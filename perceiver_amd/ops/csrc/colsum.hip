@@ -1,11 +1,13 @@
-// Column sum over a tall bf16 matrix — the bias-gradient reduction
-// db[c] = sum_r dy[r][c]. torch's generic reduce picks a per-output-column
-// configuration that runs ~18x off the bandwidth roofline on the tall-skinny
-// shapes Perceiver produces (e.g. 401k x 261 for the image-classifier K/V
-// projections). This kernel reads whole rows coalesced at 16 B/lane,
-// accumulates fp32 partials in registers over a block-strided row range,
-// combines the block through LDS and publishes ONE global atomicAdd per
-// column per block (a few hundred blocks -> atomic traffic is negligible).
+// Column sums over tall bf16 matrices — bias gradients (db[c] = sum_r dy[r][c])
+// and the LN dw/db reductions share this machinery.
+//
+// torch's generic reduce runs ~18x off the bandwidth roofline on these
+// tall-skinny shapes; a first atomic-publishing version here was atomic-bound
+// (blocks x C fp32 atomicAdds measured slower than the data read). Final
+// shape: each block streams a strided row range at 16 B/lane into register
+// partials, combines its waves through LDS, writes ONE partial row per block
+// (plain stores, no zero-init needed), and a small finalize kernel reduces
+// the (nblocks, C) partial matrix column-wise.
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
 #include "common.h"
@@ -15,9 +17,9 @@ namespace {
 constexpr int LANES = 64;   // x: one wave covers 64 granules = 512 columns
 constexpr int ROWS = 4;     // y: rows in flight per block
 
-__global__ void colsum_kernel(const unsigned short* __restrict__ x,
-                              float* __restrict__ out, long rows, int C) {
-    // lane x covers granules x, x+64, x+128, x+192 (8 columns each, C<=2048);
+__global__ void colsum_partial_kernel(const unsigned short* __restrict__ x,
+                                      float* __restrict__ partial,
+                                      long rows, int C) {
     // statically-indexed accumulators (runtime bounds would scratch them)
     float acc[4][8];
 #pragma unroll
@@ -25,7 +27,7 @@ __global__ void colsum_kernel(const unsigned short* __restrict__ x,
 #pragma unroll
         for (int e = 0; e < 8; ++e) acc[j][e] = 0.f;
 
-    const int gpr = (C + 7) / 8;  // granules per row
+    const int gpr = (C + 7) / 8;
 #pragma unroll 2
     for (long r = (long)blockIdx.x * ROWS + threadIdx.y; r < rows;
          r += (long)gridDim.x * ROWS) {
@@ -48,7 +50,6 @@ __global__ void colsum_kernel(const unsigned short* __restrict__ x,
         }
     }
 
-    // combine the ROWS per-column partials through LDS, one atomic per column
     __shared__ float red[2048];
     for (int i = threadIdx.y * LANES + threadIdx.x; i < 2048; i += LANES * ROWS)
         red[i] = 0.f;
@@ -58,15 +59,46 @@ __global__ void colsum_kernel(const unsigned short* __restrict__ x,
         int c0 = (threadIdx.x + j * LANES) * 8;
 #pragma unroll
         for (int e = 0; e < 8; ++e)
-            if (c0 + e < C) atomicAdd(&red[c0 + e], acc[j][e]);
+            if (c0 + e < C) atomicAdd(&red[c0 + e], acc[j][e]);  // LDS only
     }
     __syncthreads();
-    if (threadIdx.y == 0) {
-        for (int c = threadIdx.x; c < C; c += LANES) atomicAdd(&out[c], red[c]);
+    float* mine = partial + (long)blockIdx.x * C;
+    for (int c = threadIdx.y * LANES + threadIdx.x; c < C; c += LANES * ROWS)
+        mine[c] = red[c];
+}
+
+// out[c] = sum_p partial[p][c]; lanes own consecutive columns (coalesced),
+// the y waves split the partial rows and combine through LDS
+__global__ void colsum_finalize_kernel(const float* __restrict__ partial,
+                                       float* __restrict__ out,
+                                       int nparts, int C) {
+    int c = blockIdx.x * LANES + threadIdx.x;
+    float acc = 0.f;
+    if (c < C) {
+        for (int p = threadIdx.y; p < nparts; p += ROWS)
+            acc += partial[(long)p * C + c];
+    }
+    __shared__ float red[ROWS][LANES];
+    red[threadIdx.y][threadIdx.x] = acc;
+    __syncthreads();
+    if (threadIdx.y == 0 && c < C) {
+        float total = red[0][threadIdx.x];
+#pragma unroll
+        for (int y = 1; y < ROWS; ++y) total += red[y][threadIdx.x];
+        out[c] = total;
     }
 }
 
 }  // namespace
+
+void colsum_reduce_partials(const torch::Tensor& partial, torch::Tensor& out,
+                            int nparts, int C) {
+    dim3 block(LANES, ROWS);
+    hipLaunchKernelGGL(colsum_finalize_kernel, dim3((C + LANES - 1) / LANES), block, 0,
+                       at::cuda::getCurrentCUDAStream(),
+                       partial.data_ptr<float>(), out.data_ptr<float>(), nparts, C);
+    HIP_CHECK_LAST();
+}
 
 torch::Tensor colsum_bf16(torch::Tensor x) {
     TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16 && x.dim() == 2);
@@ -74,14 +106,16 @@ torch::Tensor colsum_bf16(torch::Tensor x) {
     long rows = x.size(0);
     int C = x.size(1);
     TORCH_CHECK(C <= 2048, "colsum_bf16: C must be <= 2048");
-    auto out = torch::zeros({(long)C}, x.options().dtype(torch::kFloat32));
-    if (rows == 0) return out;
+    if (rows == 0) return torch::zeros({(long)C}, x.options().dtype(torch::kFloat32));
+    long nblocks = std::min((rows + ROWS - 1) / ROWS, (long)768);
+    auto partial = torch::empty({nblocks, (long)C}, x.options().dtype(torch::kFloat32));
+    auto out = torch::empty({(long)C}, x.options().dtype(torch::kFloat32));
     dim3 block(LANES, ROWS);
-    long nblocks = std::min((rows + ROWS - 1) / ROWS, (long)1024);
-    hipLaunchKernelGGL(colsum_kernel, dim3(nblocks), block, 0,
+    hipLaunchKernelGGL(colsum_partial_kernel, dim3(nblocks), block, 0,
                        at::cuda::getCurrentCUDAStream(),
                        reinterpret_cast<const unsigned short*>(x.data_ptr()),
-                       out.data_ptr<float>(), rows, C);
+                       partial.data_ptr<float>(), rows, C);
     HIP_CHECK_LAST();
+    colsum_reduce_partials(partial, out, (int)nblocks, C);
     return out;
 }

@@ -7,6 +7,9 @@
 #include <ATen/cuda/CUDAContext.h>
 #include "common.h"
 
+void colsum_reduce_partials(const torch::Tensor& partial, torch::Tensor& out,
+                            int nparts, int C);
+
 namespace {
 
 DEVINL float wave_sum(float x) {
@@ -230,17 +233,19 @@ __global__ void ln_bwd_dwdb_fast_kernel(const unsigned short* __restrict__ dy,
 #pragma unroll
         for (int e = 0; e < 8; ++e)
             if (c0 + e < C) {
-                atomicAdd(&red[0][c0 + e], dw[j][e]);
+                atomicAdd(&red[0][c0 + e], dw[j][e]);  // LDS only
                 atomicAdd(&red[1][c0 + e], db[j][e]);
             }
     }
     __syncthreads();
-    if (threadIdx.y == 0) {
-        for (int c = threadIdx.x; c < C; c += DW_LANES) {
-            atomicAdd(&dw_accum[c], red[0][c]);
-            atomicAdd(&db_accum[c], red[1][c]);
-        }
+    // one partial row per block ([dw | db]); a finalize pass (colsum.hip)
+    // reduces them — global fp32 atomics measured slower than the data read
+    float* mine = dw_accum + (long)blockIdx.x * 2 * C;
+    for (int c = threadIdx.y * DW_LANES + threadIdx.x; c < C; c += DW_LANES * DW_ROWS) {
+        mine[c] = red[0][c];
+        mine[C + c] = red[1][c];
     }
+    (void)db_accum;
 }
 
 template <int CHUNKS>
@@ -318,22 +323,20 @@ std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor x, torch::Tens
 
     torch::Tensor dw, db;
     if (needs_dwdb) {
-        // one zeroed allocation for both accumulators (a separate FillFunctor
-        // launch per tiny tensor measured ~1 ms/step across the LN/bias sites)
-        auto acc = torch::zeros({2, (long)C}, x.options().dtype(torch::kFloat32));
-        auto dwf = acc[0];
-        auto dbf = acc[1];
         dim3 block(DW_LANES, DW_ROWS);
-        long nblocks = std::min((rows + DW_ROWS - 1) / DW_ROWS, (long)1024);
+        long nblocks = std::min((rows + DW_ROWS - 1) / DW_ROWS, (long)768);
+        auto partial = torch::empty({nblocks, 2 * (long)C}, x.options().dtype(torch::kFloat32));
         hipLaunchKernelGGL(ln_bwd_dwdb_fast_kernel, dim3(nblocks), block, 0,
                            at::cuda::getCurrentCUDAStream(),
                            reinterpret_cast<const unsigned short*>(dy.data_ptr()),
                            reinterpret_cast<const unsigned short*>(x.data_ptr()),
                            mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                           dwf.data_ptr<float>(), dbf.data_ptr<float>(), rows, C);
+                           partial.data_ptr<float>(), nullptr, rows, C);
         HIP_CHECK_LAST();
-        dw = dwf.to(x.scalar_type());
-        db = dbf.to(x.scalar_type());
+        auto acc = torch::empty({2, (long)C}, x.options().dtype(torch::kFloat32));
+        colsum_reduce_partials(partial, acc, (int)nblocks, 2 * C);
+        dw = acc[0].to(x.scalar_type());
+        db = acc[1].to(x.scalar_type());
     }
     return {dx, dw, db};
 }
