@@ -193,8 +193,12 @@ __global__ void flash_dq_kernel(
     char* k_lds = smem;                                  // TILE * k_stride
     char* kt16_lds = k_lds + TILE * k_stride;            // (DMAX/16) * (TILE*16+8) elems
     char* v_lds = kt16_lds + (DMAX / 16) * (TILE * 16 + 8) * 2;   // TILE * v_stride
-    char* p_lds = v_lds + TILE * v_stride;               // NWAVES * QROWS * kt_stride
-    char* p_mine = p_lds + wave * QROWS * kt_stride;
+    // per-wave transposed dS image ([key][16 q-rows] subtile per h): packed
+    // ushort2 writes + ds_read_b64_tr_b16 A-fragment reads (the row-major
+    // image needed 32 scalar b16 stores per tile per wave)
+    constexpr int SUBE_T = TILE * 16 + 8;
+    char* p_lds = v_lds + TILE * v_stride;               // NWAVES * QH * SUBE_T * 2
+    char* p_mine = p_lds + wave * QH * SUBE_T * 2;
 
     short8v q_frag[QH][DMAX / 32];
     short8v do_frag[QH][DVMAX / 32];
@@ -325,7 +329,8 @@ __global__ void flash_dq_kernel(
                 }
             }
 #pragma unroll
-            for (int h = 0; h < QH; ++h)
+            for (int h = 0; h < QH; ++h) {
+                float ds_pack[4];
 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
                     int qi = q0 + h * 16 + hi4 * 4 + r;
@@ -339,10 +344,18 @@ __global__ void flash_dq_kernel(
                         bool kept = rng_hash(drop_seed, bh, qi, j) >= drop_thresh;
                         dprobs = kept ? dprobs / (1.0f - drop_p) : 0.f;
                     }
-                    float ds = p * (dprobs - delta_r[h][r]);
-                    *reinterpret_cast<unsigned short*>(
-                        p_mine + (h * 16 + hi4 * 4 + r) * kt_stride + (t * 16 + lo16) * 2) = f2bf(ds);
+                    ds_pack[r] = p * (dprobs - delta_r[h][r]);
                 }
+                unsigned short* dst = reinterpret_cast<unsigned short*>(
+                    p_mine + (h * SUBE_T + (t * 16 + lo16) * 16 + hi4 * 4) * 2);
+#pragma unroll
+                for (int rp = 0; rp < 2; ++rp) {
+                    ushort2 pk;
+                    pk.x = f2bf(ds_pack[2 * rp]);
+                    pk.y = f2bf(ds_pack[2 * rp + 1]);
+                    *reinterpret_cast<ushort2*>(dst + 2 * rp) = pk;
+                }
+            }
         }
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
         bf16x8 ds_frag[QH][TBLKS / 2];
@@ -350,8 +363,8 @@ __global__ void flash_dq_kernel(
         for (int h = 0; h < QH; ++h)
 #pragma unroll
             for (int t32 = 0; t32 < TBLKS / 2; ++t32)
-                ds_frag[h][t32] = (bf16x8)(*reinterpret_cast<const short8v*>(
-                    p_mine + (h * 16 + lo16) * kt_stride + (t32 * 32 + hi4 * 8) * 2));
+                ds_frag[h][t32] = read_bfrag_tr16<TILE>(
+                    p_mine + h * SUBE_T * 2, 0, t32 * 32, hi4, lo16);
 
         // dQ += dS K : B[k=key][j=ch] via transpose reads of the subtiled K image
 #pragma unroll
@@ -441,8 +454,11 @@ __global__ void flash_dkv_kernel(
     char* q16_lds = q_lds + TILE * q_stride;              // (DMAX/16)*(TILE*16+8) elems
     char* do_lds = q16_lds + (DMAX / 16) * (TILE * 16 + 8) * 2;
     char* do16_lds = do_lds + TILE * do_stride;           // (DVMAX/16)*(TILE*16+8) elems
-    char* p_lds = do16_lds + (DVMAX / 16) * (TILE * 16 + 8) * 2;  // NWAVES*KROWS*qt_stride
-    char* p_mine = p_lds + wave * KROWS * qt_stride;
+    // per-wave transposed P/dS image ([q][16 key-rows] subtile per h) —
+    // packed ushort2 writes + tr16 A-fragment reads, see the dq kernel
+    constexpr int SUBE_T = TILE * 16 + 8;
+    char* p_lds = do16_lds + (DVMAX / 16) * (TILE * 16 + 8) * 2;  // NWAVES*QH*SUBE_T*2
+    char* p_mine = p_lds + wave * QH * SUBE_T * 2;
 
     short8v k_frag[QH][DMAX / 32];
     short8v v_frag[QH][DVMAX / 32];
@@ -592,7 +608,8 @@ __global__ void flash_dkv_kernel(
                 }
             }
 #pragma unroll
-            for (int h = 0; h < QH; ++h)
+            for (int h = 0; h < QH; ++h) {
+                float p_pack[4];
 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
                     int ki = k0 + h * 16 + hi4 * 4 + r;
@@ -611,9 +628,18 @@ __global__ void flash_dkv_kernel(
                         dprobs = kept ? dprobs * inv_keep : 0.f;
                     }
                     ds_keep[h][t][r] = p * (dprobs - delta_i);
-                    *reinterpret_cast<unsigned short*>(
-                        p_mine + (h * 16 + hi4 * 4 + r) * qt_stride + (t * 16 + lo16) * 2) = f2bf(p_eff);
+                    p_pack[r] = p_eff;
                 }
+                unsigned short* dst = reinterpret_cast<unsigned short*>(
+                    p_mine + (h * SUBE_T + (t * 16 + lo16) * 16 + hi4 * 4) * 2);
+#pragma unroll
+                for (int rp = 0; rp < 2; ++rp) {
+                    ushort2 pk;
+                    pk.x = f2bf(p_pack[2 * rp]);
+                    pk.y = f2bf(p_pack[2 * rp + 1]);
+                    *reinterpret_cast<ushort2*>(dst + 2 * rp) = pk;
+                }
+            }
         }
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
         bf16x8 pt_frag[QH][TBLKS / 2];
@@ -621,8 +647,8 @@ __global__ void flash_dkv_kernel(
         for (int h = 0; h < QH; ++h)
 #pragma unroll
             for (int t32 = 0; t32 < TBLKS / 2; ++t32)
-                pt_frag[h][t32] = (bf16x8)(*reinterpret_cast<const short8v*>(
-                    p_mine + (h * 16 + lo16) * qt_stride + (t32 * 32 + hi4 * 8) * 2));
+                pt_frag[h][t32] = read_bfrag_tr16<TILE>(
+                    p_mine + h * SUBE_T * 2, 0, t32 * 32, hi4, lo16);
 
         // second pass: dS^T through the same per-wave buffer — issued BEFORE the
         // dV MFMAs (the P^T fragments are already in registers), so the whole
@@ -630,12 +656,17 @@ __global__ void flash_dkv_kernel(
 #pragma unroll
         for (int h = 0; h < QH; ++h)
 #pragma unroll
-            for (int t = 0; t < TBLKS; ++t)
+            for (int t = 0; t < TBLKS; ++t) {
+                unsigned short* dst = reinterpret_cast<unsigned short*>(
+                    p_mine + (h * SUBE_T + (t * 16 + lo16) * 16 + hi4 * 4) * 2);
 #pragma unroll
-                for (int r = 0; r < 4; ++r)
-                    *reinterpret_cast<unsigned short*>(
-                        p_mine + (h * 16 + hi4 * 4 + r) * qt_stride + (t * 16 + lo16) * 2) =
-                        f2bf(ds_keep[h][t][r]);
+                for (int rp = 0; rp < 2; ++rp) {
+                    ushort2 pk;
+                    pk.x = f2bf(ds_keep[h][t][2 * rp]);
+                    pk.y = f2bf(ds_keep[h][t][2 * rp + 1]);
+                    *reinterpret_cast<ushort2*>(dst + 2 * rp) = pk;
+                }
+            }
 
         // dV += P^T dO : B[k=qrow][j=ch] via transpose reads of the subtiled image
 #pragma unroll
@@ -657,8 +688,8 @@ __global__ void flash_dkv_kernel(
         for (int h = 0; h < QH; ++h)
 #pragma unroll
             for (int t32 = 0; t32 < TBLKS / 2; ++t32)
-                dst_frag[h][t32] = (bf16x8)(*reinterpret_cast<const short8v*>(
-                    p_mine + (h * 16 + lo16) * qt_stride + (t32 * 32 + hi4 * 8) * 2));
+                dst_frag[h][t32] = read_bfrag_tr16<TILE>(
+                    p_mine + h * SUBE_T * 2, 0, t32 * 32, hi4, lo16);
 
         // dK += dS^T Q : B[k=qrow][j=ch] via transpose reads of the subtiled image
 #pragma unroll
@@ -733,7 +764,7 @@ void launch_flash_bwd(const torch::Tensor& dout, const torch::Tensor& q, const t
         const int k_stride = d_pad * 2 + 16, kt_stride = DQ_TILE * 2 + 16, v_stride = dv_pad * 2 + 16;
         const int qblk = 16 * DQ_QH * NWAVES;
         size_t smem = (size_t)DQ_TILE * k_stride + (size_t)(DMAX / 16) * (DQ_TILE * 16 + 8) * 2 +
-                      (size_t)DQ_TILE * v_stride + (size_t)NWAVES * 16 * DQ_QH * kt_stride;
+                      (size_t)DQ_TILE * v_stride + (size_t)NWAVES * DQ_QH * (DQ_TILE * 16 + 8) * 2;
         int gx = (Nq + qblk - 1) / qblk, gy = B * H;
         int nsplit = 1;
         long kv_chunk = Lk;
@@ -778,7 +809,7 @@ void launch_flash_bwd(const torch::Tensor& dout, const torch::Tensor& q, const t
         const int kblk = 16 * DKV_QH * NWAVES;
         size_t smem = (size_t)DKV_TILE * q_stride + (size_t)(DMAX / 16) * (DKV_TILE * 16 + 8) * 2 +
                       (size_t)DKV_TILE * do_stride + (size_t)(DVMAX / 16) * (DKV_TILE * 16 + 8) * 2 +
-                      (size_t)NWAVES * 16 * DKV_QH * qt_stride;
+                      (size_t)NWAVES * DKV_QH * (DKV_TILE * 16 + 8) * 2;
         int gx = (Lk + kblk - 1) / kblk, gy = B * H;
         int nsplit = 1;
         long q_chunk = Nq;
