@@ -170,28 +170,31 @@ __global__ void gemm_bt_kernel(const unsigned short* __restrict__ xp,   // (M,K)
     // peel the last triple so the tail below always knows its drain counts
     const int kt_main = (kt_total / 3) * 3;
     const int kt_peel = (kt_total % 3 == 0) ? kt_main - 3 : kt_main;
+    // MF_HALF==1 phases read the same tile their half-0 sibling already
+    // forced, so their wait is vmcnt(12) = never stalls (outstanding is at
+    // most two tiles); a 6 there was measured to serialize the prefetch lead.
     int kt = 0;
     for (; kt < kt_peel; kt += 3) {
         GEMM_PHASE(kt + 0, kt + 2, 0, 6)
-        GEMM_PHASE(kt + 0, kt + 2, 1, 6)
+        GEMM_PHASE(kt + 0, kt + 2, 1, 12)
         GEMM_PHASE(kt + 1, kt + 3, 0, 6)
-        GEMM_PHASE(kt + 1, kt + 3, 1, 6)
+        GEMM_PHASE(kt + 1, kt + 3, 1, 12)
         GEMM_PHASE(kt + 2, kt + 4, 0, 6)
-        GEMM_PHASE(kt + 2, kt + 4, 1, 6)
+        GEMM_PHASE(kt + 2, kt + 4, 1, 12)
     }
     // tail: 1..3 tiles left, nothing further to prefetch past kt_total
     switch (kt_total - kt) {
         case 3:
             GEMM_PHASE(kt + 0, kt + 2, 0, 6)
-            GEMM_PHASE(kt + 0, kt + 2, 1, 6)
+            GEMM_PHASE(kt + 0, kt + 2, 1, 12)
             GEMM_PHASE(kt + 1, kt_total, 0, 6)
-            GEMM_PHASE(kt + 1, kt_total, 1, 6)
+            GEMM_PHASE(kt + 1, kt_total, 1, 12)
             GEMM_PHASE(kt + 2, kt_total, 0, 0)
             GEMM_PHASE(kt + 2, kt_total, 1, 0)
             break;
         case 2:
             GEMM_PHASE(kt + 0, kt_total, 0, 6)
-            GEMM_PHASE(kt + 0, kt_total, 1, 6)
+            GEMM_PHASE(kt + 0, kt_total, 1, 12)
             GEMM_PHASE(kt + 1, kt_total, 0, 0)
             GEMM_PHASE(kt + 1, kt_total, 1, 0)
             break;
@@ -203,6 +206,15 @@ __global__ void gemm_bt_kernel(const unsigned short* __restrict__ xp,   // (M,K)
 #undef GEMM_PHASE
 
     // ---- epilogue: stage C through LDS (ring is dead), bias, 16-B stores ----
+    // bias is added in fp32 BEFORE the bf16 round, matching hipBLASLt's
+    // epilogue exactly (rounding C first then adding bias flips ~25% of
+    // outputs by one ulp vs F.linear)
+    float bvals[4] = {0.f, 0.f, 0.f, 0.f};
+    if (bias != nullptr) {
+#pragma unroll
+        for (int nf = 0; nf < 4; ++nf)
+            bvals[nf] = bf2f(bias[n0 + bcol0 + nf * 16 + lo16]);
+    }
     __builtin_amdgcn_s_barrier();
     char* cmine = smem + wid * (WM * WN * 2);  // 8 KB per wave
 #pragma unroll
@@ -214,23 +226,16 @@ __global__ void gemm_bt_kernel(const unsigned short* __restrict__ xp,   // (M,K)
                 int row = mf * 16 + hi4 * 4 + r;
                 int col = nf * 16 + lo16;
                 *reinterpret_cast<unsigned short*>(cmine + (row * WN + col) * 2) =
-                    f2bf(acc[mf][nf][r]);
+                    f2bf(acc[mf][nf][r] + bvals[nf]);
             }
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");  // own-wave LDS ordering
-    const bool has_bias = bias != nullptr;
 #pragma unroll
     for (int i = 0; i < (WM * WN * 2) / (64 * 16); ++i) {
         int g = lane + i * 64;
         int row = g / (WN / 8);
         int c0 = (g % (WN / 8)) * 8;
         short8v v = *reinterpret_cast<const short8v*>(cmine + (row * WN + c0) * 2);
-        int gcol = n0 + bcol0 + c0;
-        if (has_bias) {
-#pragma unroll
-            for (int e = 0; e < 8; ++e)
-                v[e] = (short)f2bf(bf2f((unsigned short)v[e]) + bf2f(bias[gcol + e]));
-        }
-        *reinterpret_cast<short8v*>(yp + ((long)m0 + arow0 + row) * N + gcol) = v;
+        *reinterpret_cast<short8v*>(yp + ((long)m0 + arow0 + row) * N + n0 + bcol0 + c0) = v;
     }
 }
 

@@ -1,13 +1,21 @@
-"""Linear with a fast bias-gradient path.
+"""Linear with fast bias-gradient and custom-GEMM paths.
 
 For tall activations (hundreds of thousands of rows in the big-KV encoder
 cross-attentions), torch's generic reduce computes the bias gradient far off
 the bandwidth roofline. ``PerceiverLinear`` keeps nn.Linear's state-dict and
 forward exactly, but its backward computes ``db`` with the coalesced
-column-sum kernel (ops/csrc/colsum.hip); ``dx``/``dw`` stay on hipBLASLt
-through torch.matmul (the same GEMMs torch's own backward runs).
+column-sum kernel (ops/csrc/colsum.hip), and the forward GEMM plus the
+data-gradient GEMM route through the deep-pipeline CDNA4 kernel
+(ops/csrc/gemm_bt.hip) when the shape matches its tiling — both operands of
+``y = x @ w^T`` are K-contiguous in nn.Linear's native layout, and the dgrad
+``dx = dy @ w`` reuses the same kernel with a (cheap, 2-4 MB) transposed
+weight copy. ``dw`` stays on hipBLASLt through torch.matmul.
+
+Set PERCEIVER_NO_CUSTOM_GEMM=1 to keep every GEMM on hipBLASLt (A/B lever).
 """
 from __future__ import annotations
+
+import os
 
 import torch
 import torch.nn as nn
@@ -15,11 +23,24 @@ import torch.nn.functional as F
 
 from perceiver_amd.ops import hip
 
+_NO_CUSTOM_GEMM = os.environ.get("PERCEIVER_NO_CUSTOM_GEMM", "") == "1"
+
+
+def _gemm_bt_ok(M: int, N: int, K: int) -> bool:
+    if _NO_CUSTOM_GEMM:
+        return False
+    return bool(hip.ext().gemm_bt_applicable(M, N, K))
+
 
 class _ColsumLinearFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias):
         ctx.save_for_backward(x, weight)
+        lead = x.shape[:-1]
+        M, K, N = x.numel() // x.shape[-1], x.shape[-1], weight.shape[0]
+        if _gemm_bt_ok(M, N, K):
+            x2 = x.reshape(M, K)
+            return hip.ext().gemm_bt(x2, weight, bias).view(*lead, N)
         return F.linear(x, weight, bias)
 
     @staticmethod
@@ -28,7 +49,15 @@ class _ColsumLinearFn(torch.autograd.Function):
         dy2 = dy.reshape(-1, dy.shape[-1]).contiguous()
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
-            dx = dy.matmul(weight)
+            M, N = dy2.shape
+            K = weight.shape[1]
+            if _gemm_bt_ok(M, K, N):
+                # dx = dy @ w == dy @ (w^T)^T: feed the kernel the transposed
+                # weight so its B^T orientation matches
+                dx = hip.ext().gemm_bt(dy2, weight.t().contiguous(), None)
+                dx = dx.view(*dy.shape[:-1], K)
+            else:
+                dx = dy.matmul(weight)
         if ctx.needs_input_grad[1]:
             dw = dy2.t().matmul(x.reshape(-1, x.shape[-1]))
         if ctx.needs_input_grad[2]:

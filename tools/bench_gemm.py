@@ -3,6 +3,11 @@ flagship projection shapes. Run on a GPU box:
     python tools/bench_gemm.py
 Prints TF/s for both paths per shape.
 """
+import pathlib
+import sys
+
+sys.path.insert(0, str(pathlib.Path(__file__).resolve().parents[1]))
+
 import torch
 import torch.nn.functional as F
 
