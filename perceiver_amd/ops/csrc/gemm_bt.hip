@@ -13,10 +13,10 @@
 //     (48 KB/buffer = 144 KB), two K-tiles of prefetch lead, COUNTED
 //     s_waitcnt vmcnt — the main loop never drains to vmcnt(0), so HBM
 //     traffic stays in flight across the raw s_barriers;
-//   - st_16x32 XOR swizzle (byte ^= ((byte>>9)&1)<<5 per 1 KB subtile)
-//     applied on the glds SOURCE addresses and repeated on the ds_read
-//     offsets — glds placement is forced lane-linear, so the swizzle has to
-//     ride on which global granule each lane fetches;
+//   - a conflict-free XOR granule swizzle applied on the glds SOURCE
+//     addresses and repeated on the ds_read offsets — glds placement is
+//     forced lane-linear, so the swizzle has to ride on which global granule
+//     each lane fetches (see swz() for the bank derivation);
 //   - phase = [vmcnt][s_barrier][issue next tile's glds][ds_read 12 frags]
 //     [setprio(1) 16x MFMA setprio(0)]; two phases per K-tile (mf halves);
 //   - C staged through LDS after the loop (buffers are dead by then) and
@@ -47,10 +47,16 @@ constexpr int BUF_BYTES = A_BYTES + B_BYTES;   // 48 KB; x3 ring = 144 KB
 constexpr int GLDS_A = A_BYTES / (8 * 1024);   // glds per wave per tile (A): 4
 constexpr int GLDS_B = B_BYTES / (8 * 1024);   // 2
 
-// st_16x32 swizzle: XOR byte-bit-5 with bit-9. Involution; moves whole 32-B
-// pairs, so 16-B glds granules stay granules.
+// Granule swizzle matched to the fragment-read pattern. Reads fetch 16 B at
+// byte (row*128 + c*16) with a quarter-wave covering rows r0..r0+15 at fixed
+// c; the bank group of a 16-B granule is (byte>>4) & 15, which unswizzled
+// depends on row parity only -> 8-way conflicts (PMC measured 2e8 conflict
+// cycles; hipBLASLt has zero). XORing the granule-within-row bits (4..6)
+// with (row>>1)&7 makes (row&1, c^((row>>1)&7)) injective over 16 rows: all
+// 16 lanes of a quarter-wave hit distinct bank groups. Involution, moves
+// whole granules, and stays inside each 128-B row.
 DEVINL int swz(int byte_off) {
-    return byte_off ^ (((byte_off >> 9) & 1) << 5);
+    return byte_off ^ (((byte_off >> 8) & 7) << 4);
 }
 
 // one wave-level glds: every lane deposits 16 B at lds_base + lane*16;
@@ -157,9 +163,11 @@ __global__ void gemm_bt_kernel(const unsigned short* __restrict__ xp,   // (M,K)
             _Pragma("unroll") for (int kk = 0; kk < 2; ++kk)                           \
                 bfrag[nf][kk] = read_frag(bbuf, bcol0 + nf * 16 + lo16, kk);           \
         __builtin_amdgcn_s_setprio(1);                                                 \
-        _Pragma("unroll") for (int mf = 0; mf < 2; ++mf)                               \
-            _Pragma("unroll") for (int nf = 0; nf < 4; ++nf)                           \
-                _Pragma("unroll") for (int kk = 0; kk < 2; ++kk)                       \
+        /* kk outermost: the 8 accumulators between dependent kk pairs keep */         \
+        /* the MFMA pipe free of read-after-write stalls */                            \
+        _Pragma("unroll") for (int kk = 0; kk < 2; ++kk)                               \
+            _Pragma("unroll") for (int mf = 0; mf < 2; ++mf)                           \
+                _Pragma("unroll") for (int nf = 0; nf < 4; ++nf)                       \
                     acc[(MF_HALF)*2 + mf][nf] =                                        \
                         __builtin_amdgcn_mfma_f32_16x16x32_bf16(                       \
                             afrag[mf][kk], bfrag[nf][kk],                              \
