@@ -140,17 +140,22 @@ __global__ void gemm_bt_kernel(const unsigned short* __restrict__ xp,   // (M,K)
     issue_tile(0);
     issue_tile(1);
 
-    // One phase: half the wave's C rows x full K-tile. VM is the counted
-    // vmcnt: 6 = one tile (6 glds/wave) still in flight, 0 = full drain
-    // (only ever used for the final tiles). KT_NEXT < kt_total gates the
-    // prefetch. The barrier directly after the vmcnt makes every wave's glds
-    // for this phase's tile visible; the prefetch issues after the barrier,
-    // which is what makes overwriting the ring slot safe (its last readers
-    // finished before they could reach this barrier).
-#define GEMM_PHASE(KT, KT_NEXT, MF_HALF, VM)                                           \
+    // One phase: half the wave's C rows x full K-tile. VM in GEMM_SYNC is
+    // the counted vmcnt: 6 = one tile (6 glds/wave) still in flight, 0 =
+    // full drain (only for the final tiles). KT_NEXT < kt_total gates the
+    // prefetch, which issues right after the sync barrier.
+    // Sync happens ONCE per K-tile (before the half-0 phase): the vmcnt
+    // forces this tile's glds landed, the barrier makes every wave's visible
+    // AND proves the previous tile's readers are done before the prefetch
+    // below overwrites its ring slot. The half-1 phase rides with no sync at
+    // all: its tile was forced by its half-0 sibling, and the next half-0
+    // barrier orders any later overwrite against its reads.
+#define GEMM_SYNC(VM)                                                                  \
+    asm volatile("s_waitcnt vmcnt(" #VM ")" ::: "memory");                             \
+    __builtin_amdgcn_s_barrier();
+
+#define GEMM_PHASE(KT, KT_NEXT, MF_HALF)                                               \
     {                                                                                  \
-        asm volatile("s_waitcnt vmcnt(" #VM ")" ::: "memory");                         \
-        __builtin_amdgcn_s_barrier();                                                  \
         if ((MF_HALF) == 0 && (KT_NEXT) < kt_total) issue_tile(KT_NEXT);               \
         const char* abuf = smem + ((KT) % 3) * BUF_BYTES;                              \
         const char* bbuf = abuf + A_BYTES;                                             \
@@ -183,35 +188,45 @@ __global__ void gemm_bt_kernel(const unsigned short* __restrict__ xp,   // (M,K)
     // most two tiles); a 6 there was measured to serialize the prefetch lead.
     int kt = 0;
     for (; kt < kt_peel; kt += 3) {
-        GEMM_PHASE(kt + 0, kt + 2, 0, 6)
-        GEMM_PHASE(kt + 0, kt + 2, 1, 12)
-        GEMM_PHASE(kt + 1, kt + 3, 0, 6)
-        GEMM_PHASE(kt + 1, kt + 3, 1, 12)
-        GEMM_PHASE(kt + 2, kt + 4, 0, 6)
-        GEMM_PHASE(kt + 2, kt + 4, 1, 12)
+        GEMM_SYNC(6)
+        GEMM_PHASE(kt + 0, kt + 2, 0)
+        GEMM_PHASE(kt + 0, kt + 2, 1)
+        GEMM_SYNC(6)
+        GEMM_PHASE(kt + 1, kt + 3, 0)
+        GEMM_PHASE(kt + 1, kt + 3, 1)
+        GEMM_SYNC(6)
+        GEMM_PHASE(kt + 2, kt + 4, 0)
+        GEMM_PHASE(kt + 2, kt + 4, 1)
     }
     // tail: 1..3 tiles left, nothing further to prefetch past kt_total
     switch (kt_total - kt) {
         case 3:
-            GEMM_PHASE(kt + 0, kt + 2, 0, 6)
-            GEMM_PHASE(kt + 0, kt + 2, 1, 12)
-            GEMM_PHASE(kt + 1, kt_total, 0, 6)
-            GEMM_PHASE(kt + 1, kt_total, 1, 12)
-            GEMM_PHASE(kt + 2, kt_total, 0, 0)
-            GEMM_PHASE(kt + 2, kt_total, 1, 0)
+            GEMM_SYNC(6)
+            GEMM_PHASE(kt + 0, kt + 2, 0)
+            GEMM_PHASE(kt + 0, kt + 2, 1)
+            GEMM_SYNC(6)
+            GEMM_PHASE(kt + 1, kt_total, 0)
+            GEMM_PHASE(kt + 1, kt_total, 1)
+            GEMM_SYNC(0)
+            GEMM_PHASE(kt + 2, kt_total, 0)
+            GEMM_PHASE(kt + 2, kt_total, 1)
             break;
         case 2:
-            GEMM_PHASE(kt + 0, kt_total, 0, 6)
-            GEMM_PHASE(kt + 0, kt_total, 1, 12)
-            GEMM_PHASE(kt + 1, kt_total, 0, 0)
-            GEMM_PHASE(kt + 1, kt_total, 1, 0)
+            GEMM_SYNC(6)
+            GEMM_PHASE(kt + 0, kt_total, 0)
+            GEMM_PHASE(kt + 0, kt_total, 1)
+            GEMM_SYNC(0)
+            GEMM_PHASE(kt + 1, kt_total, 0)
+            GEMM_PHASE(kt + 1, kt_total, 1)
             break;
         default:
-            GEMM_PHASE(kt + 0, kt_total, 0, 0)
-            GEMM_PHASE(kt + 0, kt_total, 1, 0)
+            GEMM_SYNC(0)
+            GEMM_PHASE(kt + 0, kt_total, 0)
+            GEMM_PHASE(kt + 0, kt_total, 1)
             break;
     }
 #undef GEMM_PHASE
+#undef GEMM_SYNC
 
     // ---- epilogue: stage C through LDS (ring is dead), bias, 16-B stores ----
     // bias is added in fp32 BEFORE the bf16 round, matching hipBLASLt's
