@@ -60,6 +60,23 @@ def main():
         else:
             print(f"M={M} N={N} K={K}: hipBLASLt {t_ref:.3f} ms ({tf_ref:.0f} TF/s) | gemm_bt n/a")
 
+    # dgrad comparison: dx = dy @ w (NN for hipBLASLt; gemm_bt runs B^T on a
+    # transposed weight copy, amortized here exactly as the autograd path does)
+    print("-- dgrad (dx = dy @ w) --")
+    for M, N, K in [(16384, 1280, 1280), (16384, 1792, 1280), (65536, 768, 768)]:
+        dy = torch.randn(M, N, device=dev, dtype=torch.bfloat16)
+        w = torch.randn(N, K, device=dev, dtype=torch.bfloat16) * 0.05
+        flops = 2.0 * M * N * K
+        t_nn = bench(lambda: dy.matmul(w))
+        if ext.gemm_bt_applicable(M, K, N):
+            t_bt = bench(lambda: ext.gemm_bt(dy, w.t().contiguous(), None))
+            ref = (dy.float() @ w.float()).to(torch.bfloat16)
+            got = ext.gemm_bt(dy, w.t().contiguous(), None)
+            match = (got == ref).float().mean().item()
+            print(f"M={M} N={N} K={K}: torch-NN {t_nn:.3f} ms ({flops/(t_nn*1e9):.0f} TF/s) | "
+                  f"gemm_bt+T {t_bt:.3f} ms ({flops/(t_bt*1e9):.0f} TF/s) | "
+                  f"speedup {t_nn/t_bt:.2f}x | exact-match {match:.4f}")
+
 
 if __name__ == "__main__":
     main()
