@@ -518,12 +518,17 @@ class MLP(nn.Sequential):
         )
 
     def forward(self, x: torch.Tensor) -> ModuleOutput:
-        from perceiver_amd.ops.gelu import GeluBias, can_fuse_gelu_bias
+        from perceiver_amd.ops.gelu import (GeluBias, LinearGeluBias, can_fuse_gelu_bias,
+                                            can_fuse_linear_gelu)
 
         if can_fuse_gelu_bias(x):
             h = self[0](x)
-            h = torch.nn.functional.linear(h, self[1].weight)  # bias folded into GELU
-            h = GeluBias.apply(h.contiguous(), self[1].bias)
+            if self[1].bias is not None and can_fuse_linear_gelu(h, self[1].weight):
+                # one kernel: GEMM + bias + GELU (epilogue-fused)
+                h = LinearGeluBias.apply(h, self[1].weight, self[1].bias)
+            else:
+                h = torch.nn.functional.linear(h, self[1].weight)  # bias folds into GELU
+                h = GeluBias.apply(h.contiguous(), self[1].bias)
             h = self[3](h)
             return ModuleOutput(last_hidden_state=h)
         return ModuleOutput(last_hidden_state=super().forward(x))

@@ -24,6 +24,8 @@ torch::Tensor dropout_add_fwd(torch::Tensor x, torch::Tensor res, double p, int6
 torch::Tensor dropout_add_bwd(torch::Tensor dy, double p, int64_t seed);
 torch::Tensor colsum_bf16(torch::Tensor x);
 torch::Tensor gemm_bt_bf16(torch::Tensor x, torch::Tensor w, c10::optional<torch::Tensor> bias);
+std::vector<torch::Tensor> gemm_bt_gelu_bf16(torch::Tensor x, torch::Tensor w,
+                                             c10::optional<torch::Tensor> bias);
 bool gemm_bt_applicable(long M, long N, long K);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -40,6 +42,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("dropout_add_bwd", &dropout_add_bwd, "fused residual dropout-add backward (bf16)");
     m.def("colsum_bf16", &colsum_bf16, "coalesced bf16 column sum (bias gradients)");
     m.def("gemm_bt", &gemm_bt_bf16, "deep-pipeline bf16 GEMM: x @ w^T (+ bias)");
+    m.def("gemm_bt_gelu", &gemm_bt_gelu_bf16,
+          "deep-pipeline bf16 GEMM with fused GELU epilogue: returns (pre, gelu(pre+bias))");
     m.def("gemm_bt_applicable", [](int64_t M, int64_t N, int64_t K) {
         return gemm_bt_applicable(M, N, K);
     }, "shape gate for the custom GEMM");

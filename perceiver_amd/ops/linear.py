@@ -24,11 +24,15 @@ import torch.nn.functional as F
 from perceiver_amd.ops import hip
 
 _NO_CUSTOM_GEMM = os.environ.get("PERCEIVER_NO_CUSTOM_GEMM", "") == "1"
+# Forward GEMMs stay on hipBLASLt by default: the NT (B^T) layout is its
+# strong path (~815 TF/s tuned) and gemm_bt measures ~0.85x of it. The dgrad
+# NN layout is hipBLASLt's WEAK path; gemm_bt + a cheap transposed-weight
+# copy beats it there, so that routes through the custom kernel by default.
+_GEMM_FWD = os.environ.get("PERCEIVER_GEMM_FWD", "0") == "1" and not _NO_CUSTOM_GEMM
+_GEMM_DGRAD = os.environ.get("PERCEIVER_GEMM_DGRAD", "1") == "1" and not _NO_CUSTOM_GEMM
 
 
 def _gemm_bt_ok(M: int, N: int, K: int) -> bool:
-    if _NO_CUSTOM_GEMM:
-        return False
     return bool(hip.ext().gemm_bt_applicable(M, N, K))
 
 
@@ -38,7 +42,7 @@ class _ColsumLinearFn(torch.autograd.Function):
         ctx.save_for_backward(x, weight)
         lead = x.shape[:-1]
         M, K, N = x.numel() // x.shape[-1], x.shape[-1], weight.shape[0]
-        if _gemm_bt_ok(M, N, K):
+        if _GEMM_FWD and _gemm_bt_ok(M, N, K):
             x2 = x.reshape(M, K)
             return hip.ext().gemm_bt(x2, weight, bias).view(*lead, N)
         return F.linear(x, weight, bias)
@@ -51,7 +55,7 @@ class _ColsumLinearFn(torch.autograd.Function):
         if ctx.needs_input_grad[0]:
             M, N = dy2.shape
             K = weight.shape[1]
-            if _gemm_bt_ok(M, K, N):
+            if _GEMM_DGRAD and _gemm_bt_ok(M, K, N):
                 # dx = dy @ w == dy @ (w^T)^T: feed the kernel the transposed
                 # weight so its B^T orientation matches
                 dx = hip.ext().gemm_bt(dy2, weight.t().contiguous(), None)
