@@ -181,27 +181,45 @@ __global__ void gemm_bt_kernel(const unsigned short* __restrict__ xp,   // (M,K)
     asm volatile("s_waitcnt vmcnt(%0)" ::"i"(VM) : "memory");                          \
     __builtin_amdgcn_s_barrier();
 
-    // Whole K-tile, split by kk (32-deep half) so only one kk's fragments
-    // are live at a time — NF=5 with both kks live spilled 29 VGPRs. Each
-    // accumulator is touched once per kk cluster (4*NF independent MFMAs in
-    // between), so the matrix pipe sees no read-after-write stalls.
+    // Whole K-tile. NF=4 reads all fragments up front and runs one 32-MFMA
+    // cluster (measured fastest; fits in 224 VGPRs). NF=5 cannot hold both
+    // kk fragment sets without spilling (29 VGPRs), so it splits by kk with
+    // only one set live. Accumulators repeat at distance >= 4*NF, so the
+    // matrix pipe sees no read-after-write stalls either way.
 #define GEMM_TILE(KT, KT_NEXT)                                                         \
     {                                                                                  \
         if ((KT_NEXT) < kt_total) issue_tile(KT_NEXT);                                 \
         const char* abuf = smem + ((KT) % 3) * BUF_BYTES;                              \
         const char* bbuf = abuf + A_BYTES;                                             \
-        _Pragma("unroll") for (int kk = 0; kk < 2; ++kk) {                             \
-            bf16x8 afrag[4], bfrag[NF];                                                \
+        if constexpr (NF == 4) {                                                       \
+            bf16x8 afrag[4][2], bfrag[NF][2];                                          \
             _Pragma("unroll") for (int nf = 0; nf < NF; ++nf)                          \
-                bfrag[nf] = read_frag(bbuf, bcol0 + nf * 16 + lo16, kk);               \
+                _Pragma("unroll") for (int kk = 0; kk < 2; ++kk)                       \
+                    bfrag[nf][kk] = read_frag(bbuf, bcol0 + nf * 16 + lo16, kk);       \
             _Pragma("unroll") for (int mf = 0; mf < 4; ++mf)                           \
-                afrag[mf] = read_frag(abuf, arow0 + mf * 16 + lo16, kk);               \
+                _Pragma("unroll") for (int kk = 0; kk < 2; ++kk)                       \
+                    afrag[mf][kk] = read_frag(abuf, arow0 + mf * 16 + lo16, kk);       \
             __builtin_amdgcn_s_setprio(1);                                             \
-            _Pragma("unroll") for (int mf = 0; mf < 4; ++mf)                           \
-                _Pragma("unroll") for (int nf = 0; nf < NF; ++nf)                      \
-                    acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(             \
-                        afrag[mf], bfrag[nf], acc[mf][nf], 0, 0, 0);                   \
+            _Pragma("unroll") for (int kk = 0; kk < 2; ++kk)                           \
+                _Pragma("unroll") for (int mf = 0; mf < 4; ++mf)                       \
+                    _Pragma("unroll") for (int nf = 0; nf < NF; ++nf)                  \
+                        acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(         \
+                            afrag[mf][kk], bfrag[nf][kk], acc[mf][nf], 0, 0, 0);       \
             __builtin_amdgcn_s_setprio(0);                                             \
+        } else {                                                                       \
+            _Pragma("unroll") for (int kk = 0; kk < 2; ++kk) {                         \
+                bf16x8 afrag[4], bfrag[NF];                                            \
+                _Pragma("unroll") for (int nf = 0; nf < NF; ++nf)                      \
+                    bfrag[nf] = read_frag(bbuf, bcol0 + nf * 16 + lo16, kk);           \
+                _Pragma("unroll") for (int mf = 0; mf < 4; ++mf)                       \
+                    afrag[mf] = read_frag(abuf, arow0 + mf * 16 + lo16, kk);           \
+                __builtin_amdgcn_s_setprio(1);                                         \
+                _Pragma("unroll") for (int mf = 0; mf < 4; ++mf)                       \
+                    _Pragma("unroll") for (int nf = 0; nf < NF; ++nf)                  \
+                        acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(         \
+                            afrag[mf], bfrag[nf], acc[mf][nf], 0, 0, 0);               \
+                __builtin_amdgcn_s_setprio(0);                                         \
+            }                                                                          \
         }                                                                              \
     }
 
@@ -287,8 +305,16 @@ __global__ void gemm_bt_kernel(const unsigned short* __restrict__ xp,   // (M,K)
 }
 
 // best fill on a 256-CU chip at one block/CU: fewer idle slots in the last
-// round wins; tie goes to the wider tile (fewer blocks, more B reuse)
+// round wins; tie goes to the wider tile (fewer blocks, more B reuse).
+// PERCEIVER_GEMM_BN=128|160 forces a tile for A/B measurement.
 int pick_bn(long M, long N) {
+    static const int forced = [] {
+        const char* e = getenv("PERCEIVER_GEMM_BN");
+        return e ? atoi(e) : 0;
+    }();
+    if (forced == 128 || forced == 160) {
+        if (N % forced == 0) return forced;
+    }
     auto waste = [&](long bn) -> long {
         if (N % bn != 0) return 1L << 60;
         long g = (M / BM) * (N / bn);
