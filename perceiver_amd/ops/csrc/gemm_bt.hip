@@ -90,20 +90,33 @@ DEVINL void stage_a(const unsigned short* __restrict__ gbase, long ld,
     }
 }
 
-// B-tile staging at 4-B lanes (256-B pieces): TBN*BK*2/256 pieces, uniform
-// per wave for both BN=128 (8/wave) and BN=160 (10/wave)
+// B-tile staging: 16-B lanes when the tile splits into uniform per-wave 1-KB
+// pieces (BN=128: 2/wave — 4-B granules measured ~20% slower on the small-N
+// shapes), else 4-B lanes in 256-B pieces (BN=160: 10/wave). Uniformity
+// keeps the per-wave vmcnt counts exact.
 template <int B_BYTES>
 DEVINL void stage_b(const unsigned short* __restrict__ gbase, long ld,
                     char* lds_base, int wid, int lane) {
-    constexpr int PIECES = B_BYTES / 256;
-    constexpr int PER_WAVE = PIECES / 8;
+    if constexpr (B_BYTES % (8 * 1024) == 0) {
+        constexpr int PER_WAVE = B_BYTES / (8 * 1024);
 #pragma unroll
-    for (int i = 0; i < PER_WAVE; ++i) {
-        int piece = wid * PER_WAVE + i;
-        int o = piece * 256 + lane * 4;
-        int so = swz(o);
-        glds4(gbase + (long)(so >> 7) * ld + ((so & 127) >> 1),
-              lds_base + piece * 256);
+        for (int i = 0; i < PER_WAVE; ++i) {
+            int piece = wid * PER_WAVE + i;
+            int o = piece * 1024 + lane * 16;
+            int so = swz(o);
+            glds16(gbase + (long)(so >> 7) * ld + ((so & 127) >> 1),
+                   lds_base + piece * 1024);
+        }
+    } else {
+        constexpr int PER_WAVE = B_BYTES / 256 / 8;
+#pragma unroll
+        for (int i = 0; i < PER_WAVE; ++i) {
+            int piece = wid * PER_WAVE + i;
+            int o = piece * 256 + lane * 4;
+            int so = swz(o);
+            glds4(gbase + (long)(so >> 7) * ld + ((so & 127) >> 1),
+                  lds_base + piece * 256);
+        }
     }
 }
 
@@ -119,8 +132,9 @@ __global__ void gemm_bt_kernel(const unsigned short* __restrict__ xp,   // (M,K)
     constexpr int WN = TBN / 2;                   // wave cols: 64 or 80
     constexpr int B_BYTES = TBN * BK * 2;         // 16 KB or 20 KB
     constexpr int BUF_BYTES = A_BYTES + B_BYTES;  // 48 KB or 52 KB
-    // per-wave glds per K-tile: A 4 x 16 B + B (8 or 10) x 4 B
-    constexpr int VM_TILE = GLDS_A + B_BYTES / 256 / 8;
+    // per-wave glds per K-tile: A 4 x 16 B + B (2 x 16 B, or 10 x 4 B)
+    constexpr int VM_TILE =
+        GLDS_A + (B_BYTES % (8 * 1024) == 0 ? B_BYTES / (8 * 1024) : B_BYTES / 256 / 8);
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
     const int tid = threadIdx.x;

@@ -57,7 +57,9 @@ class LinearGeluBias(torch.autograd.Function):
         dpre = hip.ext().gelu_bias_bwd(pre, bias, dy2)
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
-            dx = dpre.matmul(weight).view(*dy.shape[:-1], weight.shape[1])
+            from perceiver_amd.ops.linear import dgrad_matmul
+
+            dx = dgrad_matmul(dpre, weight, dy.shape[:-1])
         if ctx.needs_input_grad[1]:
             dw = dpre.t().matmul(x2)
         if ctx.needs_input_grad[2]:

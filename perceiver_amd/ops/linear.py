@@ -36,6 +36,19 @@ def _gemm_bt_ok(M: int, N: int, K: int) -> bool:
     return bool(hip.ext().gemm_bt_applicable(M, N, K))
 
 
+def dgrad_matmul(d2, weight, lead_shape):
+    """dx = d @ w for a 2-D upstream grad d2, reshaped to (*lead_shape, K).
+    Routes through the custom B^T GEMM with a (cheap) transposed-weight copy
+    when the inner dim is large enough to win (measured: 1.07x at 1280,
+    parity at 1792, loses below ~1024)."""
+    K = weight.shape[1]
+    if _GEMM_DGRAD and d2.shape[1] >= 1024 and _gemm_bt_ok(d2.shape[0], K, d2.shape[1]):
+        dx = hip.ext().gemm_bt(d2, weight.t().contiguous(), None)
+    else:
+        dx = d2.matmul(weight)
+    return dx.view(*lead_shape, K)
+
+
 class _ColsumLinearFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias):
@@ -53,15 +66,7 @@ class _ColsumLinearFn(torch.autograd.Function):
         dy2 = dy.reshape(-1, dy.shape[-1]).contiguous()
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
-            M, N = dy2.shape
-            K = weight.shape[1]
-            if _GEMM_DGRAD and _gemm_bt_ok(M, K, N):
-                # dx = dy @ w == dy @ (w^T)^T: feed the kernel the transposed
-                # weight so its B^T orientation matches
-                dx = hip.ext().gemm_bt(dy2, weight.t().contiguous(), None)
-                dx = dx.view(*dy.shape[:-1], K)
-            else:
-                dx = dy.matmul(weight)
+            dx = dgrad_matmul(dy2, weight, dy.shape[:-1])
         if ctx.needs_input_grad[1]:
             dw = dy2.t().matmul(x.reshape(-1, x.shape[-1]))
         if ctx.needs_input_grad[2]:

@@ -77,8 +77,14 @@ def test_perceiver_linear_dgrad_routes_custom():
 
 
 def test_linear_gelu_fused_matches_reference():
-    """Fused GEMM+bias+GELU (forward and backward) vs fp32 autograd."""
-    from perceiver_amd.ops.gelu import LinearGeluBias
+    """Fused GEMM+bias+GELU vs the unfused bf16 path (F.linear + GeluBias).
+
+    The unfused path is the numerics contract: both recompute gelu' from the
+    bf16-rounded pre-activation, so they must agree to accumulation-order
+    noise. (A pure fp32 reference is only a sanity bound on the output — the
+    bf16 pre-activation shifts gelu' systematically, and column sums of that
+    shift are visible in dw for the unfused path exactly as much.)"""
+    from perceiver_amd.ops.gelu import GeluBias, LinearGeluBias
 
     torch.manual_seed(2)
     dev = torch.device("cuda")
@@ -90,16 +96,23 @@ def test_linear_gelu_fused_matches_reference():
     dy = torch.randn_like(y)
     y.backward(dy)
 
-    xr = x.detach().float().requires_grad_(True)
-    wr = w.detach().float().requires_grad_(True)
-    br = b.detach().float().requires_grad_(True)
-    yr = torch.nn.functional.gelu(torch.nn.functional.linear(xr, wr, br))
-    yr.backward(dy.float())
+    xr = x.detach().clone().requires_grad_(True)
+    wr = w.detach().clone().requires_grad_(True)
+    br = b.detach().clone().requires_grad_(True)
+    hr = torch.nn.functional.linear(xr, wr)
+    yr = GeluBias.apply(hr.contiguous(), br)
+    yr.backward(dy)
 
-    for got, ref, tol in [(y, yr, 2e-2), (x.grad, xr.grad, 3e-2),
-                          (w.grad, wr.grad, 3e-2), (b.grad, br.grad, 3e-2)]:
-        err = (got.float() - ref).abs()
-        scale = ref.abs().clamp_min(1.0)
+    # output sanity vs fp32 as well
+    yf = torch.nn.functional.gelu(
+        torch.nn.functional.linear(x.detach().float(), w.detach().float(), b.detach().float()))
+    err_f = ((y.float() - yf).abs() / yf.abs().clamp_min(1.0)).max().item()
+    assert err_f < 2e-2, err_f
+
+    for got, ref, tol in [(y, yr, 1e-2), (x.grad, xr.grad, 1e-2),
+                          (w.grad, wr.grad, 2e-2), (b.grad, br.grad, 2e-2)]:
+        err = (got.float() - ref.float()).abs()
+        scale = ref.float().abs().clamp_min(1.0)
         assert (err / scale).max().item() < tol, (err / scale).max().item()
 
 
