@@ -74,6 +74,10 @@ def can_fuse_gelu_bias(x: torch.Tensor) -> bool:
 
 
 def can_fuse_linear_gelu(x: torch.Tensor, weight: torch.Tensor) -> bool:
+    from perceiver_amd.ops.linear import _NO_CUSTOM_GEMM
+
+    if _NO_CUSTOM_GEMM:
+        return False
     if not (x.is_cuda and x.dtype == torch.bfloat16 and weight.dtype == torch.bfloat16
             and hip.is_available()):
         return False
