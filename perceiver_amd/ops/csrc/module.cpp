@@ -27,6 +27,7 @@ torch::Tensor gemm_bt_bf16(torch::Tensor x, torch::Tensor w, c10::optional<torch
 std::vector<torch::Tensor> gemm_bt_gelu_bf16(torch::Tensor x, torch::Tensor w,
                                              c10::optional<torch::Tensor> bias);
 bool gemm_bt_applicable(long M, long N, long K);
+torch::Tensor transpose_bf16(torch::Tensor x);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("gelu_bias_fwd", &gelu_bias_fwd, "fused bias+GELU forward (bf16)");
@@ -42,6 +43,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("dropout_add_bwd", &dropout_add_bwd, "fused residual dropout-add backward (bf16)");
     m.def("colsum_bf16", &colsum_bf16, "coalesced bf16 column sum (bias gradients)");
     m.def("gemm_bt", &gemm_bt_bf16, "deep-pipeline bf16 GEMM: x @ w^T (+ bias)");
+    m.def("transpose_bf16", &transpose_bf16, "LDS-tiled bf16 2-D transpose");
     m.def("gemm_bt_gelu", &gemm_bt_gelu_bf16,
           "deep-pipeline bf16 GEMM with fused GELU epilogue: returns (pre, gelu(pre+bias))");
     m.def("gemm_bt_applicable", [](int64_t M, int64_t N, int64_t K) {

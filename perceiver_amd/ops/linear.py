@@ -43,10 +43,22 @@ def dgrad_matmul(d2, weight, lead_shape):
     parity at 1792, loses below ~1024)."""
     K = weight.shape[1]
     if _GEMM_DGRAD and d2.shape[1] >= 1024 and _gemm_bt_ok(d2.shape[0], K, d2.shape[1]):
-        dx = hip.ext().gemm_bt(d2, weight.t().contiguous(), None)
+        dx = hip.ext().gemm_bt(d2, _transposed(weight), None)
     else:
         dx = d2.matmul(weight)
     return dx.view(*lead_shape, K)
+
+
+def _transposed(weight: torch.Tensor) -> torch.Tensor:
+    """Transposed copy via the LDS-tiled kernel, cached per tensor version —
+    weight-shared blocks hit the same weight several times per backward, and
+    the version bump from the optimizer's in-place update invalidates."""
+    cached = getattr(weight, "_perceiver_t_cache", None)
+    if cached is not None and cached[0] == weight._version:
+        return cached[1]
+    wt = hip.ext().transpose_bf16(weight.detach())
+    weight._perceiver_t_cache = (weight._version, wt)
+    return wt
 
 
 class _ColsumLinearFn(torch.autograd.Function):

@@ -133,6 +133,17 @@ def test_gemm_bt_bn160_path():
         assert match > 0.98, (M, N, K, match)
 
 
+def test_transpose_bf16():
+    from perceiver_amd.ops import hip as hip_ops
+
+    ext = hip_ops.ext()
+    torch.manual_seed(4)
+    dev = torch.device("cuda")
+    for N, K in [(1280, 1280), (1792, 1280), (100, 37), (64, 64), (2816, 1280)]:
+        x = torch.randn(N, K, device=dev, dtype=torch.bfloat16)
+        assert torch.equal(ext.transpose_bf16(x), x.t().contiguous())
+
+
 def test_gemm_bt_rejects_bad_shapes():
     from perceiver_amd.ops import hip as hip_ops
 
