@@ -24,12 +24,14 @@ import torch.nn.functional as F
 from perceiver_amd.ops import hip
 
 _NO_CUSTOM_GEMM = os.environ.get("PERCEIVER_NO_CUSTOM_GEMM", "") == "1"
-# Forward GEMMs stay on hipBLASLt by default: the NT (B^T) layout is its
-# strong path (~815 TF/s tuned) and gemm_bt measures ~0.85x of it. The dgrad
-# NN layout is hipBLASLt's WEAK path; gemm_bt + a cheap transposed-weight
-# copy beats it there, so that routes through the custom kernel by default.
+# Plain GEMMs stay on hipBLASLt by default: with the shipped TunableOp table
+# it runs ~850-1100 TF/s on the flagship shapes and the custom kernel
+# measures ~0.9x of that (same-box A/B: routing dgrad through gemm_bt +
+# transpose cost 1.3 ms/step on the MLM flagship). The custom kernel earns
+# its keep where hipBLASLt cannot follow: the fused GEMM+bias+GELU epilogue
+# (ops/gelu.py LinearGeluBias), which stays on by default.
 _GEMM_FWD = os.environ.get("PERCEIVER_GEMM_FWD", "0") == "1" and not _NO_CUSTOM_GEMM
-_GEMM_DGRAD = os.environ.get("PERCEIVER_GEMM_DGRAD", "1") == "1" and not _NO_CUSTOM_GEMM
+_GEMM_DGRAD = os.environ.get("PERCEIVER_GEMM_DGRAD", "0") == "1" and not _NO_CUSTOM_GEMM
 
 
 def _gemm_bt_ok(M: int, N: int, K: int) -> bool:
