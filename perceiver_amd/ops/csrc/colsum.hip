@@ -28,7 +28,7 @@ __global__ void colsum_partial_kernel(const unsigned short* __restrict__ x,
         for (int e = 0; e < 8; ++e) acc[j][e] = 0.f;
 
     const int gpr = (C + 7) / 8;
-#pragma unroll 2
+#pragma unroll 4
     for (long r = (long)blockIdx.x * ROWS + threadIdx.y; r < rows;
          r += (long)gridDim.x * ROWS) {
         const unsigned short* row = x + r * C;
@@ -111,7 +111,11 @@ torch::Tensor colsum_bf16(torch::Tensor x) {
     int C = x.size(1);
     TORCH_CHECK(C <= 2048, "colsum_bf16: C must be <= 2048");
     if (rows == 0) return torch::zeros({(long)C}, x.options().dtype(torch::kFloat32));
-    long nblocks = std::min((rows + ROWS - 1) / ROWS, (long)768);
+    static const long kRedBlocks = [] {
+        const char* e = getenv("PERCEIVER_RED_BLOCKS");
+        return e ? atol(e) : 1024L;
+    }();
+    long nblocks = std::min((rows + ROWS - 1) / ROWS, kRedBlocks);
     auto partial = torch::empty({nblocks, (long)C}, x.options().dtype(torch::kFloat32));
     auto out = torch::empty({(long)C}, x.options().dtype(torch::kFloat32));
     dim3 block(LANES, ROWS);

@@ -187,7 +187,7 @@ __global__ void ln_bwd_dwdb_fast_kernel(const unsigned short* __restrict__ dy,
         for (int e = 0; e < 8; ++e) { dw[j][e] = 0.f; db[j][e] = 0.f; }
 
     const int gpr = (C + 7) / 8;
-#pragma unroll 2
+#pragma unroll 4
     for (long r = (long)blockIdx.x * DW_ROWS + threadIdx.y; r < rows;
          r += (long)gridDim.x * DW_ROWS) {
         const unsigned short* dyrow = dy + r * C;
@@ -324,7 +324,11 @@ std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor x, torch::Tens
     torch::Tensor dw, db;
     if (needs_dwdb) {
         dim3 block(DW_LANES, DW_ROWS);
-        long nblocks = std::min((rows + DW_ROWS - 1) / DW_ROWS, (long)768);
+        static const long kRedBlocks = [] {
+            const char* e = getenv("PERCEIVER_RED_BLOCKS");
+            return e ? atol(e) : 1024L;
+        }();
+        long nblocks = std::min((rows + DW_ROWS - 1) / DW_ROWS, kRedBlocks);
         auto partial = torch::empty({nblocks, 2 * (long)C}, x.options().dtype(torch::kFloat32));
         hipLaunchKernelGGL(ln_bwd_dwdb_fast_kernel, dim3(nblocks), block, 0,
                            at::cuda::getCurrentCUDAStream(),
