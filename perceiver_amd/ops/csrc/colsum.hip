@@ -62,29 +62,31 @@ __global__ void colsum_partial_kernel(const unsigned short* __restrict__ x,
             if (c0 + e < C) atomicAdd(&red[c0 + e], acc[j][e]);  // LDS only
     }
     __syncthreads();
-    float* mine = partial + (long)blockIdx.x * C;
+    // partial slab layout [c>>6][part][c&63]: the finalize then streams each
+    // 64-column group CONTIGUOUSLY (the row-major [part][C] layout made it
+    // read 4-B columns at a 4*C stride — 16-32x DRAM amplification, and the
+    // whole reason more partial blocks measured SLOWER)
     for (int c = threadIdx.y * LANES + threadIdx.x; c < C; c += LANES * ROWS)
-        mine[c] = red[c];
+        partial[((long)(c >> 6) * gridDim.x + blockIdx.x) * LANES + (c & 63)] = red[c];
 }
 
-// out[c] = sum_p partial[p][c]; lanes own consecutive columns (coalesced),
-// 16 y-waves split the partial rows (unrolled so several L2 loads stay in
-// flight — a 4-wave serial version measured latency-bound at ~48 us/call)
+// out[cb*64+lane] = sum_p partial[cb][p][lane]: one block per 64-column
+// group; each (p) row of the group is a contiguous 256-B burst, y-waves
+// split the parts so the streams stay deep
 constexpr int FIN_ROWS = 16;
 
 __global__ void colsum_finalize_kernel(const float* __restrict__ partial,
                                        float* __restrict__ out,
                                        int nparts, int C) {
-    int c = blockIdx.x * LANES + threadIdx.x;
+    const float* grp = partial + (long)blockIdx.x * nparts * LANES;
     float acc = 0.f;
-    if (c < C) {
 #pragma unroll 8
-        for (int p = threadIdx.y; p < nparts; p += FIN_ROWS)
-            acc += partial[(long)p * C + c];
-    }
+    for (int p = threadIdx.y; p < nparts; p += FIN_ROWS)
+        acc += grp[(long)p * LANES + threadIdx.x];
     __shared__ float red[FIN_ROWS][LANES];
     red[threadIdx.y][threadIdx.x] = acc;
     __syncthreads();
+    int c = blockIdx.x * LANES + threadIdx.x;
     if (threadIdx.y == 0 && c < C) {
         float total = red[0][threadIdx.x];
 #pragma unroll
@@ -116,7 +118,9 @@ torch::Tensor colsum_bf16(torch::Tensor x) {
         return e ? atol(e) : 1024L;
     }();
     long nblocks = std::min((rows + ROWS - 1) / ROWS, kRedBlocks);
-    auto partial = torch::empty({nblocks, (long)C}, x.options().dtype(torch::kFloat32));
+    long cgroups = (C + LANES - 1) / LANES;
+    auto partial = torch::empty({cgroups * nblocks, (long)LANES},
+                                x.options().dtype(torch::kFloat32));
     auto out = torch::empty({(long)C}, x.options().dtype(torch::kFloat32));
     dim3 block(LANES, ROWS);
     hipLaunchKernelGGL(colsum_partial_kernel, dim3(nblocks), block, 0,

@@ -238,12 +238,17 @@ __global__ void ln_bwd_dwdb_fast_kernel(const unsigned short* __restrict__ dy,
             }
     }
     __syncthreads();
-    // one partial row per block ([dw | db]); a finalize pass (colsum.hip)
-    // reduces them — global fp32 atomics measured slower than the data read
-    float* mine = dw_accum + (long)blockIdx.x * 2 * C;
+    // partial slab in the finalize-friendly [c>>6][part][c&63] tiling over
+    // the concatenated [dw | db] columns (see colsum.hip for why: the
+    // row-major slab made the finalize read 4-B columns at a huge stride);
+    // a finalize pass (colsum.hip) reduces them — global fp32 atomics
+    // measured slower than the data read
     for (int c = threadIdx.y * DW_LANES + threadIdx.x; c < C; c += DW_LANES * DW_ROWS) {
-        mine[c] = red[0][c];
-        mine[C + c] = red[1][c];
+        dw_accum[((long)(c >> 6) * gridDim.x + blockIdx.x) * DW_LANES + (c & 63)] =
+            red[0][c];
+        int c2 = C + c;
+        dw_accum[((long)(c2 >> 6) * gridDim.x + blockIdx.x) * DW_LANES + (c2 & 63)] =
+            red[1][c];
     }
     (void)db_accum;
 }
@@ -329,7 +334,9 @@ std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor x, torch::Tens
             return e ? atol(e) : 1024L;
         }();
         long nblocks = std::min((rows + DW_ROWS - 1) / DW_ROWS, kRedBlocks);
-        auto partial = torch::empty({nblocks, 2 * (long)C}, x.options().dtype(torch::kFloat32));
+        long cgroups = (2 * (long)C + DW_LANES - 1) / DW_LANES;
+        auto partial = torch::empty({cgroups * nblocks, (long)DW_LANES},
+                                    x.options().dtype(torch::kFloat32));
         hipLaunchKernelGGL(ln_bwd_dwdb_fast_kernel, dim3(nblocks), block, 0,
                            at::cuda::getCurrentCUDAStream(),
                            reinterpret_cast<const unsigned short*>(dy.data_ptr()),
