@@ -2,7 +2,8 @@
 // Memory-bound: one wave64 per row, ushort8 vector loads, fp32 accumulation,
 // cross-lane shfl_xor reductions; saves per-row mean/rstd for the backward.
 // Register caching uses compile-time CHUNKS (static indexing — guide §5.4 rule 20);
-// dW/db come from a per-block partial-reduction kernel + a final torch sum.
+// the backward computes dx AND the dW/db per-block partials in one fused pass
+// (tiled partial slab + the colsum finalize kernel).
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
 #include "common.h"
@@ -11,6 +12,8 @@ void colsum_reduce_partials(const torch::Tensor& partial, torch::Tensor& out,
                             int nparts, int C);
 
 namespace {
+
+constexpr int DW_LANES = 64;  // partial-slab tile width (shared with colsum)
 
 DEVINL float wave_sum(float x) {
 #pragma unroll
@@ -165,92 +168,125 @@ __global__ void ln_bwd_dx_kernel(const unsigned short* __restrict__ dy,
     }
 }
 
-// dW/db reduction, bandwidth-shaped: whole rows read coalesced at 16 B/lane,
-// fp32 per-column partials in registers over a block-strided row range,
-// block-combined through LDS, ONE global atomicAdd per column per block
-// (dw/db buffers zeroed by the launcher). Replaces both the old per-column
-// dwdb kernel (~3x off roofline) and its torch partial-sum epilogue.
-constexpr int DW_LANES = 64;
-constexpr int DW_ROWS = 4;
+// Fused backward: dx AND the dw/db per-block partials in ONE pass over
+// dy/x. The separate dwdb kernel re-read the same 2*rows*C bf16 and paid a
+// per-row mean/rstd load chain — ~55 us/call on the MLM shapes where the
+// dx-only kernel runs the identical traffic in ~21 us. One wave per row,
+// grid-stride row loop, per-lane fp32 dw/db accumulators, LDS combine,
+// tiled partial slab (same [c>>6][part][c&63] layout + finalize as colsum).
+template <int CHUNKS>
+__launch_bounds__(256, CHUNKS >= 3 ? 2 : 4)  // wide rows need >128 VGPRs
+__global__ void ln_bwd_fused_kernel(const unsigned short* __restrict__ dy,
+                                    const unsigned short* __restrict__ x,
+                                    const unsigned short* __restrict__ w,
+                                    const float* __restrict__ mean_in,
+                                    const float* __restrict__ rstd_in,
+                                    unsigned short* __restrict__ dx,
+                                    float* __restrict__ partial,
+                                    long rows, int C) {
+    const int lane = threadIdx.x % 64;
+    const int wpb = blockDim.x / 64;
+    const long wave0 = (long)blockIdx.x * wpb + threadIdx.x / 64;
+    const long wstride = (long)gridDim.x * wpb;
 
-__global__ void ln_bwd_dwdb_fast_kernel(const unsigned short* __restrict__ dy,
-                                        const unsigned short* __restrict__ x,
-                                        const float* __restrict__ mean_in,
-                                        const float* __restrict__ rstd_in,
-                                        float* __restrict__ dw_accum,
-                                        float* __restrict__ db_accum,
-                                        long rows, int C) {
-    float dw[4][8], db[4][8];
+    float dwacc[CHUNKS][8], dbacc[CHUNKS][8];
 #pragma unroll
-    for (int j = 0; j < 4; ++j)
+    for (int i = 0; i < CHUNKS; ++i)
 #pragma unroll
-        for (int e = 0; e < 8; ++e) { dw[j][e] = 0.f; db[j][e] = 0.f; }
+        for (int e = 0; e < 8; ++e) { dwacc[i][e] = 0.f; dbacc[i][e] = 0.f; }
 
-    const int gpr = (C + 7) / 8;
-#pragma unroll 4
-    for (long r = (long)blockIdx.x * DW_ROWS + threadIdx.y; r < rows;
-         r += (long)gridDim.x * DW_ROWS) {
-        const unsigned short* dyrow = dy + r * C;
-        const unsigned short* xrow = x + r * C;
-        float mean = mean_in[r], rstd = rstd_in[r];
+    for (long row = wave0; row < rows; row += wstride) {
+        const unsigned short* dyrow = dy + row * C;
+        const unsigned short* xrow = x + row * C;
+        unsigned short* dxrow = dx + row * C;
+        float mean = mean_in[row], rstd = rstd_in[row];
+
+        float g[CHUNKS][8], xh[CHUNKS][8];
+        float s1 = 0.f, s2 = 0.f;
 #pragma unroll
-        for (int j = 0; j < 4; ++j) {
-            int g = threadIdx.x + j * DW_LANES;
-            int c0 = g * 8;
-            if (g < gpr) {
-                if (c0 + 8 <= C) {
-                    short8v dv = *reinterpret_cast<const short8v*>(dyrow + c0);
-                    short8v xv = *reinterpret_cast<const short8v*>(xrow + c0);
+        for (int i = 0; i < CHUNKS; ++i) {
+            int c0 = lane * 8 + i * 512;
 #pragma unroll
-                    for (int e = 0; e < 8; ++e) {
-                        float d = bf2f((unsigned short)dv[e]);
-                        db[j][e] += d;
-                        dw[j][e] += d * (bf2f((unsigned short)xv[e]) - mean) * rstd;
-                    }
-                } else {
+            for (int e = 0; e < 8; ++e) { g[i][e] = 0.f; xh[i][e] = 0.f; }
+            if (c0 + 8 <= C) {
+                short8v dyv = *reinterpret_cast<const short8v*>(dyrow + c0);
+                short8v xv = *reinterpret_cast<const short8v*>(xrow + c0);
+                short8v wv = *reinterpret_cast<const short8v*>(w + c0);
 #pragma unroll
-                    for (int e = 0; e < 8; ++e) {
-                        if (c0 + e < C) {
-                            float d = bf2f(dyrow[c0 + e]);
-                            db[j][e] += d;
-                            dw[j][e] += d * (bf2f(xrow[c0 + e]) - mean) * rstd;
-                        }
+                for (int e = 0; e < 8; ++e) {
+                    float d = bf2f((unsigned short)dyv[e]);
+                    float gg = d * bf2f((unsigned short)wv[e]);
+                    float xhat = (bf2f((unsigned short)xv[e]) - mean) * rstd;
+                    g[i][e] = gg;
+                    xh[i][e] = xhat;
+                    s1 += gg;
+                    s2 += gg * xhat;
+                    dwacc[i][e] += d * xhat;
+                    dbacc[i][e] += d;
+                }
+            } else if (c0 < C) {
+#pragma unroll
+                for (int e = 0; e < 8; ++e) {
+                    if (c0 + e < C) {
+                        float d = bf2f(dyrow[c0 + e]);
+                        float gg = d * bf2f(w[c0 + e]);
+                        float xhat = (bf2f(xrow[c0 + e]) - mean) * rstd;
+                        g[i][e] = gg;
+                        xh[i][e] = xhat;
+                        s1 += gg;
+                        s2 += gg * xhat;
+                        dwacc[i][e] += d * xhat;
+                        dbacc[i][e] += d;
                     }
                 }
             }
         }
+        s1 = wave_sum(s1) / C;
+        s2 = wave_sum(s2) / C;
+#pragma unroll
+        for (int i = 0; i < CHUNKS; ++i) {
+            int c0 = lane * 8 + i * 512;
+            if (c0 + 8 <= C) {
+                short8v o;
+#pragma unroll
+                for (int e = 0; e < 8; ++e)
+                    o[e] = (short)f2bf(rstd * (g[i][e] - s1 - xh[i][e] * s2));
+                *reinterpret_cast<short8v*>(dxrow + c0) = o;
+            } else if (c0 < C) {
+#pragma unroll
+                for (int e = 0; e < 8; ++e)
+                    if (c0 + e < C)
+                        dxrow[c0 + e] = f2bf(rstd * (g[i][e] - s1 - xh[i][e] * s2));
+            }
+        }
     }
 
+    // combine the block's waves (all cover the same column range) and emit
+    // one tiled partial row (layout shared with the colsum finalize)
     __shared__ float red[2][2048];
-    for (int i = threadIdx.y * DW_LANES + threadIdx.x; i < 2048; i += DW_LANES * DW_ROWS) {
+    for (int i = threadIdx.x; i < 2048; i += blockDim.x) {
         red[0][i] = 0.f;
         red[1][i] = 0.f;
     }
     __syncthreads();
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
-        int c0 = (threadIdx.x + j * DW_LANES) * 8;
+    for (int i = 0; i < CHUNKS; ++i) {
+        int c0 = lane * 8 + i * 512;
 #pragma unroll
         for (int e = 0; e < 8; ++e)
             if (c0 + e < C) {
-                atomicAdd(&red[0][c0 + e], dw[j][e]);  // LDS only
-                atomicAdd(&red[1][c0 + e], db[j][e]);
+                atomicAdd(&red[0][c0 + e], dwacc[i][e]);  // LDS only
+                atomicAdd(&red[1][c0 + e], dbacc[i][e]);
             }
     }
     __syncthreads();
-    // partial slab in the finalize-friendly [c>>6][part][c&63] tiling over
-    // the concatenated [dw | db] columns (see colsum.hip for why: the
-    // row-major slab made the finalize read 4-B columns at a huge stride);
-    // a finalize pass (colsum.hip) reduces them — global fp32 atomics
-    // measured slower than the data read
-    for (int c = threadIdx.y * DW_LANES + threadIdx.x; c < C; c += DW_LANES * DW_ROWS) {
-        dw_accum[((long)(c >> 6) * gridDim.x + blockIdx.x) * DW_LANES + (c & 63)] =
+    for (int c = threadIdx.x; c < C; c += blockDim.x) {
+        partial[((long)(c >> 6) * gridDim.x + blockIdx.x) * DW_LANES + (c & 63)] =
             red[0][c];
         int c2 = C + c;
-        dw_accum[((long)(c2 >> 6) * gridDim.x + blockIdx.x) * DW_LANES + (c2 & 63)] =
+        partial[((long)(c2 >> 6) * gridDim.x + blockIdx.x) * DW_LANES + (c2 & 63)] =
             red[1][c];
     }
-    (void)db_accum;
 }
 
 template <int CHUNKS>
@@ -282,6 +318,24 @@ void launch_ln_bwd_dx(const torch::Tensor& dy, const torch::Tensor& x, const tor
                        mean.data_ptr<float>(), rstd.data_ptr<float>(),
                        reinterpret_cast<unsigned short*>(dx.data_ptr()), rows, C);
     HIP_CHECK_LAST();
+}
+
+template <int CHUNKS>
+long launch_ln_bwd_fused(const torch::Tensor& dy, const torch::Tensor& x,
+                         const torch::Tensor& w, const torch::Tensor& mean,
+                         const torch::Tensor& rstd, torch::Tensor& dx,
+                         torch::Tensor& partial, long rows, int C, long nblocks) {
+    int wpb = 4;
+    hipLaunchKernelGGL((ln_bwd_fused_kernel<CHUNKS>), dim3(nblocks), dim3(64 * wpb), 0,
+                       at::cuda::getCurrentCUDAStream(),
+                       reinterpret_cast<const unsigned short*>(dy.data_ptr()),
+                       reinterpret_cast<const unsigned short*>(x.data_ptr()),
+                       reinterpret_cast<const unsigned short*>(w.data_ptr()),
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       reinterpret_cast<unsigned short*>(dx.data_ptr()),
+                       partial.data_ptr<float>(), rows, C);
+    HIP_CHECK_LAST();
+    return nblocks;
 }
 
 }  // namespace
@@ -321,33 +375,30 @@ std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor x, torch::Tens
     auto dx = torch::empty_like(x);
     if (rows == 0) return {dx, torch::zeros_like(w), torch::zeros_like(w)};
     int chunks = (C + 511) / 512;
-    if (chunks == 1)      launch_ln_bwd_dx<1>(dy, x, wc, mean, rstd, dx, rows, C);
-    else if (chunks == 2) launch_ln_bwd_dx<2>(dy, x, wc, mean, rstd, dx, rows, C);
-    else if (chunks == 3) launch_ln_bwd_dx<3>(dy, x, wc, mean, rstd, dx, rows, C);
-    else                  launch_ln_bwd_dx<4>(dy, x, wc, mean, rstd, dx, rows, C);
-
-    torch::Tensor dw, db;
-    if (needs_dwdb) {
-        dim3 block(DW_LANES, DW_ROWS);
-        static const long kRedBlocks = [] {
-            const char* e = getenv("PERCEIVER_RED_BLOCKS");
-            return e ? atol(e) : 1024L;
-        }();
-        long nblocks = std::min((rows + DW_ROWS - 1) / DW_ROWS, kRedBlocks);
-        long cgroups = (2 * (long)C + DW_LANES - 1) / DW_LANES;
-        auto partial = torch::empty({cgroups * nblocks, (long)DW_LANES},
-                                    x.options().dtype(torch::kFloat32));
-        hipLaunchKernelGGL(ln_bwd_dwdb_fast_kernel, dim3(nblocks), block, 0,
-                           at::cuda::getCurrentCUDAStream(),
-                           reinterpret_cast<const unsigned short*>(dy.data_ptr()),
-                           reinterpret_cast<const unsigned short*>(x.data_ptr()),
-                           mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                           partial.data_ptr<float>(), nullptr, rows, C);
-        HIP_CHECK_LAST();
-        auto acc = torch::empty({2, (long)C}, x.options().dtype(torch::kFloat32));
-        colsum_reduce_partials(partial, acc, (int)nblocks, 2 * C);
-        dw = acc[0].to(x.scalar_type());
-        db = acc[1].to(x.scalar_type());
+    if (!needs_dwdb) {
+        if (chunks == 1)      launch_ln_bwd_dx<1>(dy, x, wc, mean, rstd, dx, rows, C);
+        else if (chunks == 2) launch_ln_bwd_dx<2>(dy, x, wc, mean, rstd, dx, rows, C);
+        else if (chunks == 3) launch_ln_bwd_dx<3>(dy, x, wc, mean, rstd, dx, rows, C);
+        else                  launch_ln_bwd_dx<4>(dy, x, wc, mean, rstd, dx, rows, C);
+        return {dx, torch::Tensor(), torch::Tensor()};
     }
+
+    // fused single pass: dx + per-block dw/db partials + shared finalize
+    static const long kRedBlocks = [] {
+        const char* e = getenv("PERCEIVER_RED_BLOCKS");
+        return e ? atol(e) : 1024L;
+    }();
+    long nblocks = std::min((rows + 3) / 4, kRedBlocks);
+    long cgroups = (2 * (long)C + DW_LANES - 1) / DW_LANES;
+    auto partial = torch::empty({cgroups * nblocks, (long)DW_LANES},
+                                x.options().dtype(torch::kFloat32));
+    if (chunks == 1)      launch_ln_bwd_fused<1>(dy, x, wc, mean, rstd, dx, partial, rows, C, nblocks);
+    else if (chunks == 2) launch_ln_bwd_fused<2>(dy, x, wc, mean, rstd, dx, partial, rows, C, nblocks);
+    else if (chunks == 3) launch_ln_bwd_fused<3>(dy, x, wc, mean, rstd, dx, partial, rows, C, nblocks);
+    else                  launch_ln_bwd_fused<4>(dy, x, wc, mean, rstd, dx, partial, rows, C, nblocks);
+    auto acc = torch::empty({2, (long)C}, x.options().dtype(torch::kFloat32));
+    colsum_reduce_partials(partial, acc, (int)nblocks, 2 * C);
+    auto dw = acc[0].to(x.scalar_type());
+    auto db = acc[1].to(x.scalar_type());
     return {dx, dw, db};
 }
