@@ -1,73 +1,71 @@
 // Column sums over tall bf16 matrices — bias gradients (db[c] = sum_r dy[r][c])
-// and the LN dw/db reductions share this machinery.
+// and the LN dw/db finalize share this machinery.
 //
 // torch's generic reduce runs ~18x off the bandwidth roofline on these
-// tall-skinny shapes; a first atomic-publishing version here was atomic-bound
-// (blocks x C fp32 atomicAdds measured slower than the data read). Final
-// shape: each block streams a strided row range at 16 B/lane into register
-// partials, combines its waves through LDS, writes ONE partial row per block
-// (plain stores, no zero-init needed), and a small finalize kernel reduces
-// the (nblocks, C) partial matrix column-wise.
+// tall-skinny shapes. Design history: an atomic-publishing version was
+// atomic-bound; a block-combined (LDS atomics + one partial row per block)
+// version measured ~1.3 TB/s — the per-block epilogue (zero/combine/write)
+// throttled how many blocks could be worth launching. Final shape: each WAVE
+// owns whole rows (grid-stride), accumulates per-lane fp32 partials in
+// registers, and writes its OWN partial row — no LDS, no atomics, no
+// barrier. The slab is tiled [c>>6][part][c&63] so the finalize kernel
+// streams 256-B bursts (the row-major slab cost it 16-32x DRAM
+// amplification reading 4-B columns at a 4*C stride).
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
 #include "common.h"
 
 namespace {
 
-constexpr int LANES = 64;   // x: one wave covers 64 granules = 512 columns
-constexpr int ROWS = 4;     // y: rows in flight per block
+constexpr int LANES = 64;
+constexpr int WPB = 4;       // waves per block
+constexpr int CS_MAXC = 2048;
 
+// CHUNKS = ceil(C / 512); lane covers 8 contiguous elems per chunk.
+template <int CHUNKS>
+__launch_bounds__(LANES* WPB, CHUNKS >= 3 ? 2 : 4)
 __global__ void colsum_partial_kernel(const unsigned short* __restrict__ x,
                                       float* __restrict__ partial,
                                       long rows, int C) {
-    // statically-indexed accumulators (runtime bounds would scratch them)
-    float acc[4][8];
-#pragma unroll
-    for (int j = 0; j < 4; ++j)
-#pragma unroll
-        for (int e = 0; e < 8; ++e) acc[j][e] = 0.f;
+    const int lane = threadIdx.x % 64;
+    const long wave0 = (long)blockIdx.x * WPB + threadIdx.x / 64;
+    const long wstride = (long)gridDim.x * WPB;
+    const long nparts = (long)gridDim.x * WPB;
 
-    const int gpr = (C + 7) / 8;
-#pragma unroll 4
-    for (long r = (long)blockIdx.x * ROWS + threadIdx.y; r < rows;
-         r += (long)gridDim.x * ROWS) {
+    float acc[CHUNKS][8];
+#pragma unroll
+    for (int i = 0; i < CHUNKS; ++i)
+#pragma unroll
+        for (int e = 0; e < 8; ++e) acc[i][e] = 0.f;
+
+    for (long r = wave0; r < rows; r += wstride) {
         const unsigned short* row = x + r * C;
 #pragma unroll
-        for (int j = 0; j < 4; ++j) {
-            int g = threadIdx.x + j * LANES;
-            int c0 = g * 8;
-            if (g < gpr) {
-                if (c0 + 8 <= C) {
-                    short8v v = *reinterpret_cast<const short8v*>(row + c0);
+        for (int i = 0; i < CHUNKS; ++i) {
+            int c0 = lane * 8 + i * 512;
+            if (c0 + 8 <= C) {
+                short8v v = *reinterpret_cast<const short8v*>(row + c0);
 #pragma unroll
-                    for (int e = 0; e < 8; ++e) acc[j][e] += bf2f((unsigned short)v[e]);
-                } else {
+                for (int e = 0; e < 8; ++e) acc[i][e] += bf2f((unsigned short)v[e]);
+            } else if (c0 < C) {
 #pragma unroll
-                    for (int e = 0; e < 8; ++e)
-                        if (c0 + e < C) acc[j][e] += bf2f(row[c0 + e]);
-                }
+                for (int e = 0; e < 8; ++e)
+                    if (c0 + e < C) acc[i][e] += bf2f(row[c0 + e]);
             }
         }
     }
 
-    __shared__ float red[2048];
-    for (int i = threadIdx.y * LANES + threadIdx.x; i < 2048; i += LANES * ROWS)
-        red[i] = 0.f;
-    __syncthreads();
+    // wave-private tiled partial row; whole-granule stores per chunk
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
-        int c0 = (threadIdx.x + j * LANES) * 8;
+    for (int i = 0; i < CHUNKS; ++i) {
+        int c0 = lane * 8 + i * 512;
 #pragma unroll
-        for (int e = 0; e < 8; ++e)
-            if (c0 + e < C) atomicAdd(&red[c0 + e], acc[j][e]);  // LDS only
+        for (int e = 0; e < 8; ++e) {
+            int c = c0 + e;
+            if (c < C)
+                partial[((long)(c >> 6) * nparts + wave0) * LANES + (c & 63)] = acc[i][e];
+        }
     }
-    __syncthreads();
-    // partial slab layout [c>>6][part][c&63]: the finalize then streams each
-    // 64-column group CONTIGUOUSLY (the row-major [part][C] layout made it
-    // read 4-B columns at a 4*C stride — 16-32x DRAM amplification, and the
-    // whole reason more partial blocks measured SLOWER)
-    for (int c = threadIdx.y * LANES + threadIdx.x; c < C; c += LANES * ROWS)
-        partial[((long)(c >> 6) * gridDim.x + blockIdx.x) * LANES + (c & 63)] = red[c];
 }
 
 // out[cb*64+lane] = sum_p partial[cb][p][lane]: one block per 64-column
@@ -111,23 +109,34 @@ torch::Tensor colsum_bf16(torch::Tensor x) {
     x = x.contiguous();
     long rows = x.size(0);
     int C = x.size(1);
-    TORCH_CHECK(C <= 2048, "colsum_bf16: C must be <= 2048");
+    TORCH_CHECK(C <= CS_MAXC, "colsum_bf16: C must be <= 2048");
     if (rows == 0) return torch::zeros({(long)C}, x.options().dtype(torch::kFloat32));
     static const long kRedBlocks = [] {
         const char* e = getenv("PERCEIVER_RED_BLOCKS");
         return e ? atol(e) : 1024L;
     }();
-    long nblocks = std::min((rows + ROWS - 1) / ROWS, kRedBlocks);
+    long nblocks = std::min((rows + WPB - 1) / WPB, kRedBlocks);
+    long nparts = nblocks * WPB;  // one partial row per WAVE
     long cgroups = (C + LANES - 1) / LANES;
-    auto partial = torch::empty({cgroups * nblocks, (long)LANES},
+    auto partial = torch::empty({cgroups * nparts, (long)LANES},
                                 x.options().dtype(torch::kFloat32));
     auto out = torch::empty({(long)C}, x.options().dtype(torch::kFloat32));
-    dim3 block(LANES, ROWS);
-    hipLaunchKernelGGL(colsum_partial_kernel, dim3(nblocks), block, 0,
-                       at::cuda::getCurrentCUDAStream(),
-                       reinterpret_cast<const unsigned short*>(x.data_ptr()),
-                       partial.data_ptr<float>(), rows, C);
+    const unsigned short* xp = reinterpret_cast<const unsigned short*>(x.data_ptr());
+    auto stream = at::cuda::getCurrentCUDAStream();
+    int chunks = (C + 511) / 512;
+    if (chunks == 1)
+        hipLaunchKernelGGL((colsum_partial_kernel<1>), dim3(nblocks), dim3(LANES * WPB), 0,
+                           stream, xp, partial.data_ptr<float>(), rows, C);
+    else if (chunks == 2)
+        hipLaunchKernelGGL((colsum_partial_kernel<2>), dim3(nblocks), dim3(LANES * WPB), 0,
+                           stream, xp, partial.data_ptr<float>(), rows, C);
+    else if (chunks == 3)
+        hipLaunchKernelGGL((colsum_partial_kernel<3>), dim3(nblocks), dim3(LANES * WPB), 0,
+                           stream, xp, partial.data_ptr<float>(), rows, C);
+    else
+        hipLaunchKernelGGL((colsum_partial_kernel<4>), dim3(nblocks), dim3(LANES * WPB), 0,
+                           stream, xp, partial.data_ptr<float>(), rows, C);
     HIP_CHECK_LAST();
-    colsum_reduce_partials(partial, out, (int)nblocks, C);
+    colsum_reduce_partials(partial, out, (int)nparts, C);
     return out;
 }
