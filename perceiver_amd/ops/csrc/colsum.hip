@@ -115,7 +115,10 @@ torch::Tensor colsum_bf16(torch::Tensor x) {
         const char* e = getenv("PERCEIVER_RED_BLOCKS");
         return e ? atol(e) : 1024L;
     }();
-    long nblocks = std::min((rows + WPB - 1) / WPB, kRedBlocks);
+    // row-adaptive block count (measured: 512 best for <=32k rows, 1024 for
+    // 64k+; more blocks shrink rows/wave until latency dominates)
+    long nblocks = std::min(std::max(rows / 32, (long)256), kRedBlocks);
+    nblocks = std::min(nblocks, (rows + WPB - 1) / WPB);
     long nparts = nblocks * WPB;  // one partial row per WAVE
     long cgroups = (C + LANES - 1) / LANES;
     auto partial = torch::empty({cgroups * nparts, (long)LANES},

@@ -388,7 +388,9 @@ std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor x, torch::Tens
         const char* e = getenv("PERCEIVER_RED_BLOCKS");
         return e ? atol(e) : 1024L;
     }();
-    long nblocks = std::min((rows + 3) / 4, kRedBlocks);
+    // row-adaptive block count (same sweep-derived heuristic as colsum)
+    long nblocks = std::min(std::max(rows / 32, (long)256), kRedBlocks);
+    nblocks = std::min(nblocks, (rows + 3) / 4);
     long cgroups = (2 * (long)C + DW_LANES - 1) / DW_LANES;
     auto partial = torch::empty({cgroups * nblocks, (long)DW_LANES},
                                 x.options().dtype(torch::kFloat32));
