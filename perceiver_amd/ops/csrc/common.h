@@ -111,6 +111,9 @@ DEVINL void write_rm(const short8v (&st)[NG], char* lds_rm, int ldst_bytes,
 #pragma unroll
     for (int i = 0; i < NG; ++i) {
         int g = tid + i * 256;
+        // NG rounds up for tiles that don't divide into whole 2048-elem
+        // sweeps (288/352-wide pad tiers); the surplus granules are no-ops
+        if ((ROWS_TILE * DD) % 2048 != 0 && g >= (ROWS_TILE * DD) / 8) continue;
         int row = g / GPR, c0 = (g % GPR) * 8;
         short8v val = st[i];
         if (tail && row >= rows_valid) val = short8v{};
@@ -126,6 +129,7 @@ DEVINL void write_sub16(const short8v (&st)[NG], char* lds16, int rows_valid, in
 #pragma unroll
     for (int i = 0; i < NG; ++i) {
         int g = tid + i * 256;
+        if ((ROWS_TILE * DD) % 2048 != 0 && g >= (ROWS_TILE * DD) / 8) continue;
         int row = g / GPR, c0 = (g % GPR) * 8;
         short8v val = st[i];
         if (tail && row >= rows_valid) val = short8v{};
@@ -143,6 +147,7 @@ DEVINL void write_rm_sub16_c(const short8v (&st)[NG], char* lds_rm, int ldst_byt
 #pragma unroll
     for (int i = 0; i < NG; ++i) {
         int g = tid + i * 256;
+        if ((ROWS_TILE * DD) % 2048 != 0 && g >= (ROWS_TILE * DD) / 8) continue;
         int row = g / GPR, c0 = (g % GPR) * 8;
         short8v val = st[i];
         if (tail && row >= rows_valid) val = short8v{};

@@ -261,11 +261,12 @@ __global__ void flash_dq_kernel(
 
     // T14 split staging on exact template matches: K/V loads for tile t+1 fly
     // during tile t's compute instead of serializing between the barriers
-    constexpr bool kFast = (DMAX % 32 == 0) && (DVMAX % 32 == 0) &&
-                           (DMAX <= 64 || DMAX == 288) &&
-                           ((TILE * DMAX) % 2048 == 0) && ((TILE * DVMAX) % 2048 == 0);
-    constexpr int NG_K = kFast ? (TILE * DMAX) / 2048 : 1;
-    constexpr int NG_V = kFast ? (TILE * DVMAX) / 2048 : 1;
+    // dkv stays on the slow stage for the wide pad tiers: fast staging at
+    // 288/352 spilled 37-298 VGPRs (this kernel's accumulators already fill
+    // the file at TILE=32)
+    constexpr bool kFast = (DMAX % 32 == 0) && (DVMAX % 32 == 0) && (DMAX <= 64);
+    constexpr int NG_K = kFast ? (TILE * DMAX + 2047) / 2048 : 1;
+    constexpr int NG_V = kFast ? (TILE * DVMAX + 2047) / 2048 : 1;
     short8v st_k[NG_K], st_v[NG_V];
     const bool fast = kFast && D == DMAX && Dv == DVMAX;
     const int n_tiles = (kv_end > kv_begin) ? (kv_end - kv_begin + TILE - 1) / TILE : 0;
@@ -539,10 +540,9 @@ __global__ void flash_dkv_kernel(
     // T14 split staging (see flash_dq_kernel): Q/dO tile t+1 loads fly under
     // tile t's MFMA work on exact template matches
     constexpr bool kFast = (DMAX % 32 == 0) && (DVMAX % 32 == 0) &&
-                           (DMAX <= 128 || DMAX == 288) &&
-                           ((TILE * DMAX) % 2048 == 0) && ((TILE * DVMAX) % 2048 == 0);
-    constexpr int NG_Q = kFast ? (TILE * DMAX) / 2048 : 1;
-    constexpr int NG_DO = kFast ? (TILE * DVMAX) / 2048 : 1;
+                           (DMAX <= 128 || DMAX == 288 || DMAX == 352);
+    constexpr int NG_Q = kFast ? (TILE * DMAX + 2047) / 2048 : 1;
+    constexpr int NG_DO = kFast ? (TILE * DVMAX + 2047) / 2048 : 1;
     short8v st_q[NG_Q], st_do[NG_DO];
     const bool fast = kFast && D == DMAX && Dv == DVMAX;
     const int nq_tiles = (q_end > q_start) ? (q_end - q_start + TILE - 1) / TILE : 0;
