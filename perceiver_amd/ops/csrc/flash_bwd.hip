@@ -539,8 +539,10 @@ __global__ void flash_dkv_kernel(
 
     // T14 split staging (see flash_dq_kernel): Q/dO tile t+1 loads fly under
     // tile t's MFMA work on exact template matches
+    // 352 measured SLOWER with fast staging (96 staged VGPRs at occupancy 1
+    // starve the S/dP pipeline); 288 gains — img encoder-CA backward
     constexpr bool kFast = (DMAX % 32 == 0) && (DVMAX % 32 == 0) &&
-                           (DMAX <= 128 || DMAX == 288 || DMAX == 352);
+                           (DMAX <= 128 || DMAX == 288);
     constexpr int NG_Q = kFast ? (TILE * DMAX + 2047) / 2048 : 1;
     constexpr int NG_DO = kFast ? (TILE * DVMAX + 2047) / 2048 : 1;
     short8v st_q[NG_Q], st_do[NG_DO];
