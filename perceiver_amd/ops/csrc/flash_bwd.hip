@@ -510,16 +510,15 @@ __global__ void flash_dkv_kernel(
         for (int r = 0; r < 4; ++r) any_keypad |= key_pad[h][r];
     any_keypad = __any(any_keypad);
 
-    // Column-split for the wide pad tiers: full-width dk/dv accumulators at
-    // DMAX >= 288 need 176+ VGPRs on top of the K/V fragments (the 352 tier
-    // measured 298 spilled VGPRs, 600 B/lane of scratch in the hot loop).
-    // Each half-pass recomputes S/dP (k/v fragments stay full-width) and
-    // accumulates + stores only its half of the columns.
-    constexpr int CSPLIT = (DMAX >= 288) ? 2 : 1;
-    constexpr int CBK = DMAX / 16 / CSPLIT;
-    constexpr int CBV = DVMAX / 16 / CSPLIT;
-    float4v dk_acc[QH][CBK];
-    float4v dv_acc[QH][CBV];
+    float4v dk_acc[QH][DMAX / 16];
+    float4v dv_acc[QH][DVMAX / 16];
+#pragma unroll
+    for (int h = 0; h < QH; ++h) {
+#pragma unroll
+        for (int cb = 0; cb < DMAX / 16; ++cb) dk_acc[h][cb] = float4v{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int cb = 0; cb < DVMAX / 16; ++cb) dv_acc[h][cb] = float4v{0.f, 0.f, 0.f, 0.f};
+    }
 
     int q_start = 0;
     if (causal) {
@@ -537,15 +536,6 @@ __global__ void flash_dkv_kernel(
         q_end = (int)min((long)Nq, (long)(blockIdx.z + 1) * q_chunk);
     }
     const unsigned int drop_thresh = (unsigned int)(drop_p * 4294967296.0);
-
-    for (int ch = 0; ch < CSPLIT; ++ch) {
-#pragma unroll
-    for (int h = 0; h < QH; ++h) {
-#pragma unroll
-        for (int cb = 0; cb < CBK; ++cb) dk_acc[h][cb] = float4v{0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-        for (int cb = 0; cb < CBV; ++cb) dv_acc[h][cb] = float4v{0.f, 0.f, 0.f, 0.f};
-    }
 
     // T14 split staging (see flash_dq_kernel): Q/dO tile t+1 loads fly under
     // tile t's MFMA work on exact template matches
@@ -680,19 +670,17 @@ __global__ void flash_dkv_kernel(
                 }
             }
 
-        // dV += P^T dO : B[k=qrow][j=col] via transpose reads of the subtiled
-        // image; only this pass's column half accumulates
+        // dV += P^T dO : B[k=qrow][j=ch] via transpose reads of the subtiled image
 #pragma unroll
-        for (int cb2 = 0; cb2 < CBV; ++cb2) {
-            int cb = ch * CBV + cb2;
+        for (int cb = 0; cb < DVMAX / 16; ++cb) {
             if (cb * 16 < dv_pad) {
 #pragma unroll
                 for (int t32 = 0; t32 < TBLKS / 2; ++t32) {
                     bf16x8 bfrag = read_bfrag_tr16<TILE>(do16_lds, cb, t32 * 32, hi4, lo16);
 #pragma unroll
                     for (int h = 0; h < QH; ++h)
-                        dv_acc[h][cb2] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                            pt_frag[h][t32], bfrag, dv_acc[h][cb2], 0, 0, 0);
+                        dv_acc[h][cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            pt_frag[h][t32], bfrag, dv_acc[h][cb], 0, 0, 0);
                 }
             }
         }
@@ -705,26 +693,24 @@ __global__ void flash_dkv_kernel(
                 dst_frag[h][t32] = read_bfrag_tr16<TILE>(
                     p_mine + h * SUBE_T * 2, 0, t32 * 32, hi4, lo16);
 
-        // dK += dS^T Q : B[k=qrow][j=col] via transpose reads of the subtiled
-        // image; only this pass's column half accumulates
+        // dK += dS^T Q : B[k=qrow][j=ch] via transpose reads of the subtiled image
 #pragma unroll
-        for (int cb2 = 0; cb2 < CBK; ++cb2) {
-            int cb = ch * CBK + cb2;
+        for (int cb = 0; cb < DMAX / 16; ++cb) {
             if (cb * 16 < d_pad) {
 #pragma unroll
                 for (int t32 = 0; t32 < TBLKS / 2; ++t32) {
                     bf16x8 bfrag = read_bfrag_tr16<TILE>(q16_lds, cb, t32 * 32, hi4, lo16);
 #pragma unroll
                     for (int h = 0; h < QH; ++h)
-                        dk_acc[h][cb2] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                            dst_frag[h][t32], bfrag, dk_acc[h][cb2], 0, 0, 0);
+                        dk_acc[h][cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            dst_frag[h][t32], bfrag, dk_acc[h][cb], 0, 0, 0);
                 }
             }
         }
     }
 
-    // store this half's dK/dV columns (C layout: key row h*16 + hi4*4+r,
-    // col lo16+16cb); fp32 partial slabs when Q-split (summed by the launcher)
+    // store dK/dV (C layout: key row h*16 + hi4*4+r, col lo16+16cb);
+    // fp32 partial slabs when Q-split (summed by the launcher)
 #pragma unroll
     for (int h = 0; h < QH; ++h)
 #pragma unroll
@@ -736,31 +722,30 @@ __global__ void flash_dkv_kernel(
                 float* dkrow = dk_part + base * D;
                 float* dvrow = dv_part + base * Dv;
 #pragma unroll
-                for (int cb2 = 0; cb2 < CBK; ++cb2) {
-                    int c = (ch * CBK + cb2) * 16 + lo16;
-                    if (c < D) dkrow[c] = dk_acc[h][cb2][r];
+                for (int cb = 0; cb < DMAX / 16; ++cb) {
+                    int c = cb * 16 + lo16;
+                    if (c < D) dkrow[c] = dk_acc[h][cb][r];
                 }
 #pragma unroll
-                for (int cb2 = 0; cb2 < CBV; ++cb2) {
-                    int c = (ch * CBV + cb2) * 16 + lo16;
-                    if (c < Dv) dvrow[c] = dv_acc[h][cb2][r];
+                for (int cb = 0; cb < DVMAX / 16; ++cb) {
+                    int c = cb * 16 + lo16;
+                    if (c < Dv) dvrow[c] = dv_acc[h][cb][r];
                 }
             } else {
                 unsigned short* dkrow = dkp + ((long)bh * Lk + ki) * D;
                 unsigned short* dvrow = dvp + ((long)bh * Lk + ki) * Dv;
 #pragma unroll
-                for (int cb2 = 0; cb2 < CBK; ++cb2) {
-                    int c = (ch * CBK + cb2) * 16 + lo16;
-                    if (c < D) dkrow[c] = f2bf(dk_acc[h][cb2][r]);
+                for (int cb = 0; cb < DMAX / 16; ++cb) {
+                    int c = cb * 16 + lo16;
+                    if (c < D) dkrow[c] = f2bf(dk_acc[h][cb][r]);
                 }
 #pragma unroll
-                for (int cb2 = 0; cb2 < CBV; ++cb2) {
-                    int c = (ch * CBV + cb2) * 16 + lo16;
-                    if (c < Dv) dvrow[c] = f2bf(dv_acc[h][cb2][r]);
+                for (int cb = 0; cb < DVMAX / 16; ++cb) {
+                    int c = cb * 16 + lo16;
+                    if (c < Dv) dvrow[c] = f2bf(dv_acc[h][cb][r]);
                 }
             }
         }
-    }  // ch (column-half) loop
 }
 
 template <int DMAX, int DVMAX, int DQ_TILE, int DQ_QH, int DKV_TILE, int DKV_QH>
