@@ -73,8 +73,11 @@ __global__ void colsum_partial_kernel(const unsigned short* __restrict__ x,
 // split the parts so the streams stay deep
 constexpr int FIN_ROWS = 16;
 
+// OUT_BF16: emit bf16 directly — the fp32->bf16 `.to()` after every finalize
+// was ~200 tiny launch-bound conversion kernels per MLM step (~1.6 ms)
+template <bool OUT_BF16>
 __global__ void colsum_finalize_kernel(const float* __restrict__ partial,
-                                       float* __restrict__ out,
+                                       void* __restrict__ out,
                                        int nparts, int C) {
     const float* grp = partial + (long)blockIdx.x * nparts * LANES;
     float acc = 0.f;
@@ -89,7 +92,10 @@ __global__ void colsum_finalize_kernel(const float* __restrict__ partial,
         float total = red[0][threadIdx.x];
 #pragma unroll
         for (int y = 1; y < FIN_ROWS; ++y) total += red[y][threadIdx.x];
-        out[c] = total;
+        if (OUT_BF16)
+            reinterpret_cast<unsigned short*>(out)[c] = f2bf(total);
+        else
+            reinterpret_cast<float*>(out)[c] = total;
     }
 }
 
@@ -98,9 +104,14 @@ __global__ void colsum_finalize_kernel(const float* __restrict__ partial,
 void colsum_reduce_partials(const torch::Tensor& partial, torch::Tensor& out,
                             int nparts, int C) {
     dim3 block(LANES, FIN_ROWS);
-    hipLaunchKernelGGL(colsum_finalize_kernel, dim3((C + LANES - 1) / LANES), block, 0,
-                       at::cuda::getCurrentCUDAStream(),
-                       partial.data_ptr<float>(), out.data_ptr<float>(), nparts, C);
+    if (out.scalar_type() == torch::kBFloat16)
+        hipLaunchKernelGGL((colsum_finalize_kernel<true>), dim3((C + LANES - 1) / LANES),
+                           block, 0, at::cuda::getCurrentCUDAStream(),
+                           partial.data_ptr<float>(), out.data_ptr(), nparts, C);
+    else
+        hipLaunchKernelGGL((colsum_finalize_kernel<false>), dim3((C + LANES - 1) / LANES),
+                           block, 0, at::cuda::getCurrentCUDAStream(),
+                           partial.data_ptr<float>(), out.data_ptr(), nparts, C);
     HIP_CHECK_LAST();
 }
 
@@ -110,7 +121,7 @@ torch::Tensor colsum_bf16(torch::Tensor x) {
     long rows = x.size(0);
     int C = x.size(1);
     TORCH_CHECK(C <= CS_MAXC, "colsum_bf16: C must be <= 2048");
-    if (rows == 0) return torch::zeros({(long)C}, x.options().dtype(torch::kFloat32));
+    if (rows == 0) return torch::zeros({(long)C}, x.options());
     static const long kRedBlocks = [] {
         const char* e = getenv("PERCEIVER_RED_BLOCKS");
         return e ? atol(e) : 1024L;
@@ -123,7 +134,7 @@ torch::Tensor colsum_bf16(torch::Tensor x) {
     long cgroups = (C + LANES - 1) / LANES;
     auto partial = torch::empty({cgroups * nparts, (long)LANES},
                                 x.options().dtype(torch::kFloat32));
-    auto out = torch::empty({(long)C}, x.options().dtype(torch::kFloat32));
+    auto out = torch::empty({(long)C}, x.options());  // bf16, rounded in-kernel
     const unsigned short* xp = reinterpret_cast<const unsigned short*>(x.data_ptr());
     auto stream = at::cuda::getCurrentCUDAStream();
     int chunks = (C + 511) / 512;

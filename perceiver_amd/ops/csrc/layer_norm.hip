@@ -398,9 +398,7 @@ std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor x, torch::Tens
     else if (chunks == 2) launch_ln_bwd_fused<2>(dy, x, wc, mean, rstd, dx, partial, rows, C, nblocks);
     else if (chunks == 3) launch_ln_bwd_fused<3>(dy, x, wc, mean, rstd, dx, partial, rows, C, nblocks);
     else                  launch_ln_bwd_fused<4>(dy, x, wc, mean, rstd, dx, partial, rows, C, nblocks);
-    auto acc = torch::empty({2, (long)C}, x.options().dtype(torch::kFloat32));
+    auto acc = torch::empty({2, (long)C}, x.options());  // bf16 straight out
     colsum_reduce_partials(partial, acc, (int)nblocks, 2 * C);
-    auto dw = acc[0].to(x.scalar_type());
-    auto db = acc[1].to(x.scalar_type());
-    return {dx, dw, db};
+    return {dx, acc[0], acc[1]};
 }

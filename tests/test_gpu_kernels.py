@@ -618,11 +618,15 @@ def test_residual_module_uses_fused_path_in_training():
 
 
 def test_colsum_matches_torch_sum():
+    # the kernel emits bf16 directly (single final rounding, same as the old
+    # fp32-out + .to(bf16) pair); compare against the identically-rounded
+    # torch sum with tolerance for fp32 accumulation-order ulps
     for rows, C in [(401408, 261), (16384, 1792), (1000, 7)]:
         x = (torch.randn(rows, C) * 0.5).bfloat16().cuda()
         got = _ext().colsum_bf16(x)
-        want = x.float().sum(0)
-        torch.testing.assert_close(got, want, atol=0.5, rtol=1e-3)
+        assert got.dtype == torch.bfloat16
+        want = x.float().sum(0).bfloat16().float()
+        torch.testing.assert_close(got.float(), want, atol=4.0, rtol=1e-2)
 
 
 def test_perceiver_linear_bias_grad_matches_torch():
