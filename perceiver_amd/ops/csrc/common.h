@@ -36,6 +36,13 @@ DEVINL unsigned short f2bf(float f) {
 // original splitmix64 chain lowered to ~3x the VALU ops, and the hash runs
 // once per attention element — it was a measurable share of the backward
 // kernels' 21:1 VALU:MFMA instruction ratio (profiles/ PMC).
+// Dropout draw mapping: one 32-bit hash yields TWO 16-bit draws for the
+// qi-pair (qi>>1, qi&1) at a given key j; every kernel (fwd, dq, dkv) uses
+// drop16(seed,bh,qi,j) so forward and backward regenerate identical masks.
+// The q-major pairing halves the hash VALU in the three kernels whose inner
+// r-loop walks consecutive qi (fwd pipe, fwd v3, dq — the compiler hoists
+// the shared hash); dkv (ki-major inner loop) pays the same cost as before.
+// Keep threshold is 16-bit: drop_p quantized to 1/65536.
 DEVINL unsigned int rng_hash(unsigned long long seed, int bh, int i, int j) {
     unsigned int x = (unsigned int)seed + (unsigned int)(seed >> 32) * 0x9E3779B9u;
     x += (unsigned int)bh * 0x85EBCA6Bu + (unsigned int)i * 0xC2B2AE35u +
@@ -46,6 +53,11 @@ DEVINL unsigned int rng_hash(unsigned long long seed, int bh, int i, int j) {
     x *= 0x846CA68Bu;
     x ^= x >> 16;
     return x;
+}
+
+DEVINL unsigned int drop16(unsigned long long seed, int bh, int qi, int j) {
+    unsigned int h = rng_hash(seed, bh, qi >> 1, j);
+    return (qi & 1) ? (h >> 16) : (h & 0xffffu);
 }
 
 // erf via the Abramowitz-Stegun 7.1.26 rational approximation (|err| <=

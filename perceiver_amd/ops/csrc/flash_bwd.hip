@@ -257,7 +257,7 @@ __global__ void flash_dq_kernel(
         kv_begin = (int)((long)blockIdx.z * kv_chunk);
         kv_end = min((long)kv_end, (long)(blockIdx.z + 1) * kv_chunk);
     }
-    const unsigned int drop_thresh = (unsigned int)(drop_p * 4294967296.0);
+    const unsigned int drop_thresh = (unsigned int)(drop_p * 65536.0);
 
     // T14 split staging on exact template matches: K/V loads for tile t+1 fly
     // during tile t's compute instead of serializing between the barriers
@@ -332,6 +332,14 @@ __global__ void flash_dq_kernel(
 #pragma unroll
             for (int h = 0; h < QH; ++h) {
                 float ds_pack[4];
+                unsigned int hh[2];
+                if (drop_p > 0.f) {
+                    // explicit qi-pair hashes (see common.h drop16 mapping)
+                    int j = kv0 + t * 16 + lo16;
+                    int qb2 = (q0 + h * 16 + hi4 * 4) >> 1;
+                    hh[0] = rng_hash(drop_seed, bh, qb2, j);
+                    hh[1] = rng_hash(drop_seed, bh, qb2 + 1, j);
+                }
 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
                     int qi = q0 + h * 16 + hi4 * 4 + r;
@@ -342,8 +350,8 @@ __global__ void flash_dq_kernel(
                     float p = masked ? 0.f : __expf(s_acc[h][r] - lse_r[h][r]);
                     float dprobs = dp_acc[h][r];
                     if (drop_p > 0.f) {
-                        bool kept = rng_hash(drop_seed, bh, qi, j) >= drop_thresh;
-                        dprobs = kept ? dprobs / (1.0f - drop_p) : 0.f;
+                        unsigned int d = (r & 1) ? (hh[r >> 1] >> 16) : (hh[r >> 1] & 0xffffu);
+                        dprobs = (d >= drop_thresh) ? dprobs / (1.0f - drop_p) : 0.f;
                     }
                     ds_pack[r] = p * (dprobs - delta_r[h][r]);
                 }
@@ -535,7 +543,7 @@ __global__ void flash_dkv_kernel(
         q_start = max(q_start, (int)((long)blockIdx.z * q_chunk));
         q_end = (int)min((long)Nq, (long)(blockIdx.z + 1) * q_chunk);
     }
-    const unsigned int drop_thresh = (unsigned int)(drop_p * 4294967296.0);
+    const unsigned int drop_thresh = (unsigned int)(drop_p * 65536.0);
 
     // T14 split staging (see flash_dq_kernel): Q/dO tile t+1 loads fly under
     // tile t's MFMA work on exact template matches
@@ -624,7 +632,7 @@ __global__ void flash_dkv_kernel(
                     float p_eff = p;
                     float dprobs = dpt_acc[h][r];
                     if (drop_p > 0.f) {
-                        bool kept = rng_hash(drop_seed, bh, qi, ki) >= drop_thresh;
+                        bool kept = drop16(drop_seed, bh, qi, ki) >= drop_thresh;
                         float inv_keep = 1.0f / (1.0f - drop_p);
                         p_eff = kept ? p * inv_keep : 0.f;
                         dprobs = kept ? dprobs * inv_keep : 0.f;

@@ -226,7 +226,7 @@ __global__ void flash_fwd_kernel(            // budget allows (not the 352 templ
         kv_end = min((long)kv_end, (long)(blockIdx.z + 1) * kv_chunk);
     }
 
-    const unsigned int drop_thresh = (unsigned int)(drop_p * 4294967296.0);
+    const unsigned int drop_thresh = (unsigned int)(drop_p * 65536.0);
 
     // T14 split staging on exact template matches (odd dims are padded to the
     // template by the host): K/V loads for tile t+1 fly under tile t's MFMAs
@@ -344,7 +344,7 @@ __global__ void flash_fwd_kernel(            // budget allows (not the 352 templ
                     if (drop_p > 0.f) {
                         int qi = q0 + h * 16 + hi4 * 4 + r;
                         int j = kv0 + kb * 16 + lo16;
-                        if (rng_hash(drop_seed, bh, qi, j) < drop_thresh) pv = 0.f;
+                        if (drop16(drop_seed, bh, qi, j) < drop_thresh) pv = 0.f;
                     }
                     *reinterpret_cast<unsigned short*>(
                         p_mine + (h * 16 + hi4 * 4 + r) * vt_stride + (kb * 16 + lo16) * 2) =

@@ -236,7 +236,7 @@ __global__ void flash_fwd_pipe_kernel(
     // prefetch clamp: issue addresses never run past the last real tile
     const int kv_last = kv_begin + (n_tiles > 0 ? (n_tiles - 1) * KVBLK : 0);
 
-    const unsigned int drop_thresh = (unsigned int)(drop_p * 4294967296.0);
+    const unsigned int drop_thresh = (unsigned int)(drop_p * 65536.0);
 
     Staged<KGR, VGR, DMAX, DVMAX> st;
     // ---- prologue: tile 0 staged synchronously, tile 1 issued ----
@@ -363,11 +363,16 @@ __global__ void flash_fwd_pipe_kernel(
                     psum_r[r] += pv[r];
                 }
                 if (drop_p > 0.f) {
+                    // explicit qi-pair hashes (qi base is a multiple of 4, so
+                    // rows r=0,1 and r=2,3 share one 32-bit hash each)
                     int j = kv0 + kb * 16 + lo16;
+                    int qb2 = (q0 + h * 16 + hi4 * 4) >> 1;
+                    unsigned int hh[2] = {rng_hash(drop_seed, bh, qb2, j),
+                                          rng_hash(drop_seed, bh, qb2 + 1, j)};
 #pragma unroll
                     for (int r = 0; r < 4; ++r) {
-                        int qi = q0 + h * 16 + hi4 * 4 + r;
-                        if (rng_hash(drop_seed, bh, qi, j) < drop_thresh) pv[r] = 0.f;
+                        unsigned int d = (r & 1) ? (hh[r >> 1] >> 16) : (hh[r >> 1] & 0xffffu);
+                        if (d < drop_thresh) pv[r] = 0.f;
                     }
                 }
                 unsigned short* dst = reinterpret_cast<unsigned short*>(
