@@ -23,11 +23,13 @@ DEVINL float bf2f(unsigned short b) {
 }
 
 DEVINL unsigned short f2bf(float f) {
-    unsigned int u;
-    __builtin_memcpy(&u, &f, 4);
-    unsigned int rounding = 0x7FFFu + ((u >> 16) & 1u);
-    u += rounding;
-    return (unsigned short)(u >> 16);
+    // native single-instruction convert (v_cvt_pk_bf16_f32, RTNE on gfx950);
+    // the manual bit-math round used to cost 3-4 VALU ops per convert and
+    // f2bf sits in every epilogue and the softmax P-packing hot paths
+    __bf16 b = (__bf16)f;
+    unsigned short u;
+    __builtin_memcpy(&u, &b, 2);
+    return u;
 }
 
 // counter-based RNG for attention dropout: the same (seed, bh, i, j) always
