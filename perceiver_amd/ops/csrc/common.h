@@ -22,6 +22,18 @@ DEVINL float bf2f(unsigned short b) {
     return f;
 }
 
+typedef __attribute__((ext_vector_type(2))) __bf16 bf16x2_t;
+
+// packed pair convert: ONE v_cvt_pk_bf16_f32 for two floats (the scalar
+// form needs 2 converts + shift + or); used by the P/dS image packing
+DEVINL unsigned int f2bf2(float a, float b) {
+    float2v f = {a, b};
+    bf16x2_t p = __builtin_convertvector(f, bf16x2_t);
+    unsigned int u;
+    __builtin_memcpy(&u, &p, 4);
+    return u;
+}
+
 DEVINL unsigned short f2bf(float f) {
     // native single-instruction convert (v_cvt_pk_bf16_f32, RTNE on gfx950);
     // the manual bit-math round used to cost 3-4 VALU ops per convert and

@@ -358,12 +358,9 @@ __global__ void flash_dq_kernel(
                 unsigned short* dst = reinterpret_cast<unsigned short*>(
                     p_mine + (h * SUBE_T + (t * 16 + lo16) * 16 + hi4 * 4) * 2);
 #pragma unroll
-                for (int rp = 0; rp < 2; ++rp) {
-                    ushort2 pk;
-                    pk.x = f2bf(ds_pack[2 * rp]);
-                    pk.y = f2bf(ds_pack[2 * rp + 1]);
-                    *reinterpret_cast<ushort2*>(dst + 2 * rp) = pk;
-                }
+                for (int rp = 0; rp < 2; ++rp)
+                    *reinterpret_cast<unsigned int*>(dst + 2 * rp) =
+                        f2bf2(ds_pack[2 * rp], ds_pack[2 * rp + 1]);
             }
         }
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
@@ -643,12 +640,9 @@ __global__ void flash_dkv_kernel(
                 unsigned short* dst = reinterpret_cast<unsigned short*>(
                     p_mine + (h * SUBE_T + (t * 16 + lo16) * 16 + hi4 * 4) * 2);
 #pragma unroll
-                for (int rp = 0; rp < 2; ++rp) {
-                    ushort2 pk;
-                    pk.x = f2bf(p_pack[2 * rp]);
-                    pk.y = f2bf(p_pack[2 * rp + 1]);
-                    *reinterpret_cast<ushort2*>(dst + 2 * rp) = pk;
-                }
+                for (int rp = 0; rp < 2; ++rp)
+                    *reinterpret_cast<unsigned int*>(dst + 2 * rp) =
+                        f2bf2(p_pack[2 * rp], p_pack[2 * rp + 1]);
             }
         }
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
@@ -670,12 +664,9 @@ __global__ void flash_dkv_kernel(
                 unsigned short* dst = reinterpret_cast<unsigned short*>(
                     p_mine + (h * SUBE_T + (t * 16 + lo16) * 16 + hi4 * 4) * 2);
 #pragma unroll
-                for (int rp = 0; rp < 2; ++rp) {
-                    ushort2 pk;
-                    pk.x = f2bf(ds_keep[h][t][2 * rp]);
-                    pk.y = f2bf(ds_keep[h][t][2 * rp + 1]);
-                    *reinterpret_cast<ushort2*>(dst + 2 * rp) = pk;
-                }
+                for (int rp = 0; rp < 2; ++rp)
+                    *reinterpret_cast<unsigned int*>(dst + 2 * rp) =
+                        f2bf2(ds_keep[h][t][2 * rp], ds_keep[h][t][2 * rp + 1]);
             }
 
         // dV += P^T dO : B[k=qrow][j=ch] via transpose reads of the subtiled image
