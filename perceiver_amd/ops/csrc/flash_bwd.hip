@@ -202,7 +202,7 @@ __global__ void flash_dq_kernel(
 
     short8v q_frag[QH][DMAX / 32];
     short8v do_frag[QH][DVMAX / 32];
-    float lse_r[QH][4], delta_r[QH][4];
+    float4v lse_r[QH], delta_r[QH];
 #pragma unroll
     for (int h = 0; h < QH; ++h) {
         int qi = q0 + h * 16 + lo16;
@@ -240,7 +240,7 @@ __global__ void flash_dq_kernel(
         for (int r = 0; r < 4; ++r) {
             int qi2 = q0 + h * 16 + hi4 * 4 + r;
             lse_r[h][r] = (qi2 < Nq) ? lse_row[qi2] : 0.f;
-            delta_r[h][r] = (qi2 < Nq) ? delta_row[qi2] : 0.f;
+            delta_r[h][r] = (qi2 < Nq) ? delta_row[qi2] : 0.f;  // float4v lanes
         }
     }
 
@@ -332,14 +332,10 @@ __global__ void flash_dq_kernel(
 #pragma unroll
             for (int h = 0; h < QH; ++h) {
                 float ds_pack[4];
-                unsigned int hh[2];
-                if (drop_p > 0.f) {
-                    // explicit qi-pair hashes (see common.h drop16 mapping)
-                    int j = kv0 + t * 16 + lo16;
-                    int qb2 = (q0 + h * 16 + hi4 * 4) >> 1;
-                    hh[0] = rng_hash(drop_seed, bh, qb2, j);
-                    hh[1] = rng_hash(drop_seed, bh, qb2 + 1, j);
-                }
+                // packed fp32 dS math (v_pk halves the VALU); the dropout
+                // scale folds into the per-element multiplier m4
+                float4v e4 = s_acc[h] - lse_r[h];
+                float4v p4;
 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
                     int qi = q0 + h * 16 + hi4 * 4 + r;
@@ -347,14 +343,25 @@ __global__ void flash_dq_kernel(
                     bool masked = tile_masked &&
                                   (j >= Lk || (padrow && j < Lk && padrow[j]) ||
                                    (causal && j > Lk - Nq + qi));
-                    float p = masked ? 0.f : __expf(s_acc[h][r] - lse_r[h][r]);
-                    float dprobs = dp_acc[h][r];
-                    if (drop_p > 0.f) {
-                        unsigned int d = (r & 1) ? (hh[r >> 1] >> 16) : (hh[r >> 1] & 0xffffu);
-                        dprobs = (d >= drop_thresh) ? dprobs / (1.0f - drop_p) : 0.f;
-                    }
-                    ds_pack[r] = p * (dprobs - delta_r[h][r]);
+                    p4[r] = masked ? 0.f : __expf(e4[r]);
                 }
+                float4v m4 = {1.f, 1.f, 1.f, 1.f};
+                if (drop_p > 0.f) {
+                    // explicit qi-pair hashes (see common.h drop16 mapping)
+                    int j = kv0 + t * 16 + lo16;
+                    int qb2 = (q0 + h * 16 + hi4 * 4) >> 1;
+                    unsigned int hh[2] = {rng_hash(drop_seed, bh, qb2, j),
+                                          rng_hash(drop_seed, bh, qb2 + 1, j)};
+                    const float inv_keep = 1.0f / (1.0f - drop_p);
+#pragma unroll
+                    for (int r = 0; r < 4; ++r) {
+                        unsigned int d = (r & 1) ? (hh[r >> 1] >> 16) : (hh[r >> 1] & 0xffffu);
+                        m4[r] = (d >= drop_thresh) ? inv_keep : 0.f;
+                    }
+                }
+                float4v ds4 = p4 * (dp_acc[h] * m4 - delta_r[h]);
+#pragma unroll
+                for (int r = 0; r < 4; ++r) ds_pack[r] = ds4[r];
                 unsigned short* dst = reinterpret_cast<unsigned short*>(
                     p_mine + (h * SUBE_T + (t * 16 + lo16) * 16 + hi4 * 4) * 2);
 #pragma unroll
@@ -616,27 +623,35 @@ __global__ void flash_dkv_kernel(
             }
 #pragma unroll
             for (int h = 0; h < QH; ++h) {
-                float p_pack[4];
+                // qi (and so lse/delta) is fixed across the 4 key rows, so the
+                // dS/P math runs on packed fp32 (v_pk_mul/add halve the VALU);
+                // the dropout scale folds into a per-element multiplier m4
+                const int qi = qt0 + t * 16 + lo16;
+                const float lse_i = (qi < Nq) ? lse_row[qi] : 0.f;
+                const float delta_i = (qi < Nq) ? delta_row[qi] : 0.f;
+                float4v e4 = st_acc[h] - lse_i;
+                float4v p4;
 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
                     int ki = k0 + h * 16 + hi4 * 4 + r;
-                    int qi = qt0 + t * 16 + lo16;
                     bool masked = tile_masked &&
                                   (key_pad[h][r] || qi >= Nq || (causal && ki > Lk - Nq + qi));
-                    float lse_i = (qi < Nq) ? lse_row[qi] : 0.f;
-                    float delta_i = (qi < Nq) ? delta_row[qi] : 0.f;
-                    float p = masked ? 0.f : __expf(st_acc[h][r] - lse_i);
-                    float p_eff = p;
-                    float dprobs = dpt_acc[h][r];
-                    if (drop_p > 0.f) {
-                        bool kept = drop16(drop_seed, bh, qi, ki) >= drop_thresh;
-                        float inv_keep = 1.0f / (1.0f - drop_p);
-                        p_eff = kept ? p * inv_keep : 0.f;
-                        dprobs = kept ? dprobs * inv_keep : 0.f;
-                    }
-                    ds_keep[h][t][r] = p * (dprobs - delta_i);
-                    p_pack[r] = p_eff;
+                    p4[r] = masked ? 0.f : __expf(e4[r]);
                 }
+                float4v m4 = {1.f, 1.f, 1.f, 1.f};
+                if (drop_p > 0.f) {
+                    const float inv_keep = 1.0f / (1.0f - drop_p);
+#pragma unroll
+                    for (int r = 0; r < 4; ++r) {
+                        int ki = k0 + h * 16 + hi4 * 4 + r;
+                        m4[r] = (drop16(drop_seed, bh, qi, ki) >= drop_thresh) ? inv_keep : 0.f;
+                    }
+                }
+                float4v ds4 = p4 * (dpt_acc[h] * m4 - delta_i);
+                float4v pe4 = p4 * m4;
+#pragma unroll
+                for (int r = 0; r < 4; ++r) ds_keep[h][t][r] = ds4[r];
+                float p_pack[4] = {pe4[0], pe4[1], pe4[2], pe4[3]};
                 unsigned short* dst = reinterpret_cast<unsigned short*>(
                     p_mine + (h * SUBE_T + (t * 16 + lo16) * 16 + hi4 * 4) * 2);
 #pragma unroll

@@ -318,50 +318,54 @@ __global__ void flash_fwd_pipe_kernel(
 #pragma unroll
         for (int h = 0; h < QH; ++h) {
             float rowmax[4];
+            // bias add + running max on packed fp32 (v_pk_add_f32 halves the
+            // VALU); the causal mask stays a per-element select on the (rare,
+            // wave-uniform-hoisted) cut tiles only
+            float4v mx4 = {-FLT_MAX, -FLT_MAX, -FLT_MAX, -FLT_MAX};
 #pragma unroll
-            for (int r = 0; r < 4; ++r) {
-                int qi = q0 + h * 16 + hi4 * 4 + r;
-                int jmax = Lk - Nq + qi;  // causal bound
-                float mx = -FLT_MAX;
+            for (int kb = 0; kb < KEYBLKS; ++kb) {
+                float4v s4 = s_acc[h][kb] + kbias[kb];
+                if (causal_tile) {
+                    int j = kv0 + kb * 16 + lo16;
 #pragma unroll
-                for (int kb = 0; kb < KEYBLKS; ++kb) {
-                    float sv = s_acc[h][kb][r] + kbias[kb];
-                    if (causal_tile) {
-                        int j = kv0 + kb * 16 + lo16;
-                        sv = (j > jmax) ? -FLT_MAX : sv;
+                    for (int r = 0; r < 4; ++r) {
+                        int jmax = Lk - Nq + (q0 + h * 16 + hi4 * 4 + r);
+                        if (j > jmax) s4[r] = -FLT_MAX;
                     }
-                    s_acc[h][kb][r] = sv;
-                    mx = fmaxf(mx, sv);
                 }
-                rowmax[r] = warp16_max(mx);
+                s_acc[h][kb] = s4;
+                mx4 = __builtin_elementwise_max(mx4, s4);
             }
+#pragma unroll
+            for (int r = 0; r < 4; ++r) rowmax[r] = warp16_max(mx4[r]);
             bool need = false;
 #pragma unroll
             for (int r = 0; r < 4; ++r) need |= rowmax[r] > m_run[h][r];
             if (__any(need)) {
+                float4v alpha4;
 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
                     float m_new = fmaxf(m_run[h][r], rowmax[r]);
-                    float alpha = __expf(m_run[h][r] - m_new);
+                    alpha4[r] = __expf(m_run[h][r] - m_new);
                     m_run[h][r] = m_new;
-                    l_run[h][r] *= alpha;
-#pragma unroll
-                    for (int cb = 0; cb < CBLOCKS; ++cb) o_acc[h][cb][r] *= alpha;
+                    l_run[h][r] *= alpha4[r];
                 }
+#pragma unroll
+                for (int cb = 0; cb < CBLOCKS; ++cb) o_acc[h][cb] *= alpha4;  // v_pk_mul
             }
             // exp + dropout + packed transposed write: rows (hi4*4+r, +r+1) are
             // adjacent elements of the [key][row] image -> one ushort2 store.
             // Row sums accumulate pre-dropout (the epilogue divides by
             // l * (1 - p), matching the backward's regenerated mask).
-            float psum_r[4] = {0.f, 0.f, 0.f, 0.f};
+            float4v psum4 = {0.f, 0.f, 0.f, 0.f};
+            const float4v mrun4 = {m_run[h][0], m_run[h][1], m_run[h][2], m_run[h][3]};
 #pragma unroll
             for (int kb = 0; kb < KEYBLKS; ++kb) {
+                float4v e4 = s_acc[h][kb] - mrun4;  // v_pk_add
                 float pv[4];
 #pragma unroll
-                for (int r = 0; r < 4; ++r) {
-                    pv[r] = __expf(s_acc[h][kb][r] - m_run[h][r]);
-                    psum_r[r] += pv[r];
-                }
+                for (int r = 0; r < 4; ++r) pv[r] = __expf(e4[r]);
+                psum4 += float4v{pv[0], pv[1], pv[2], pv[3]};
                 if (drop_p > 0.f) {
                     // explicit qi-pair hashes (qi base is a multiple of 4, so
                     // rows r=0,1 and r=2,3 share one 32-bit hash each)
@@ -383,7 +387,7 @@ __global__ void flash_fwd_pipe_kernel(
                         f2bf2(pv[2 * rp], pv[2 * rp + 1]);
             }
 #pragma unroll
-            for (int r = 0; r < 4; ++r) l_run[h][r] += warp16_sum(psum_r[r]);
+            for (int r = 0; r < 4; ++r) l_run[h][r] += warp16_sum(psum4[r]);
         }
 
         __syncthreads();  // B1: all waves done with tile ti's LDS buffers
