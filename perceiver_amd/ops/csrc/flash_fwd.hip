@@ -323,34 +323,48 @@ __global__ void flash_fwd_kernel(            // budget allows (not the 352 templ
 #pragma unroll
             for (int r = 0; r < 4; ++r) need |= rowmax[r] > m_run[h][r];
             if (__any(need)) {
+                float4v alpha4;
 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
                     float m_new = fmaxf(m_run[h][r], rowmax[r]);
-                    float alpha = __expf(m_run[h][r] - m_new);
+                    alpha4[r] = __expf(m_run[h][r] - m_new);
                     m_run[h][r] = m_new;
-                    l_run[h][r] *= alpha;
-#pragma unroll
-                    for (int cb = 0; cb < DVMAX / 16; ++cb) o_acc[h][cb][r] *= alpha;
+                    l_run[h][r] *= alpha4[r];
                 }
-            }
 #pragma unroll
-            for (int r = 0; r < 4; ++r) {
-                float m_new = m_run[h][r];
-                float psum = 0.f;
+                for (int cb = 0; cb < DVMAX / 16; ++cb) o_acc[h][cb] *= alpha4;  // v_pk_mul
+            }
+            {
+                // kb-outer so the exp input and row sums run on packed fp32;
+                // the P stores stay scalar (rows are vt_stride apart)
+                const float4v mrun4 = {m_run[h][0], m_run[h][1], m_run[h][2], m_run[h][3]};
+                float4v psum4 = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
                 for (int kb = 0; kb < KEYBLKS; ++kb) {
-                    float pv = __expf(s_acc[h][kb][r] - m_new);
-                    psum += pv;
+                    float4v e4 = s_acc[h][kb] - mrun4;  // v_pk_add
+                    float pv[4];
+#pragma unroll
+                    for (int r = 0; r < 4; ++r) pv[r] = __expf(e4[r]);
+                    psum4 += float4v{pv[0], pv[1], pv[2], pv[3]};
                     if (drop_p > 0.f) {
-                        int qi = q0 + h * 16 + hi4 * 4 + r;
                         int j = kv0 + kb * 16 + lo16;
-                        if (drop16(drop_seed, bh, qi, j) < drop_thresh) pv = 0.f;
+                        int qb2 = (q0 + h * 16 + hi4 * 4) >> 1;
+                        unsigned int hh[2] = {rng_hash(drop_seed, bh, qb2, j),
+                                              rng_hash(drop_seed, bh, qb2 + 1, j)};
+#pragma unroll
+                        for (int r = 0; r < 4; ++r) {
+                            unsigned int d = (r & 1) ? (hh[r >> 1] >> 16) : (hh[r >> 1] & 0xffffu);
+                            if (d < drop_thresh) pv[r] = 0.f;
+                        }
                     }
-                    *reinterpret_cast<unsigned short*>(
-                        p_mine + (h * 16 + hi4 * 4 + r) * vt_stride + (kb * 16 + lo16) * 2) =
-                        f2bf(pv);
+#pragma unroll
+                    for (int r = 0; r < 4; ++r)
+                        *reinterpret_cast<unsigned short*>(
+                            p_mine + (h * 16 + hi4 * 4 + r) * vt_stride + (kb * 16 + lo16) * 2) =
+                            f2bf(pv[r]);
                 }
-                l_run[h][r] += warp16_sum(psum);
+#pragma unroll
+                for (int r = 0; r < 4; ++r) l_run[h][r] += warp16_sum(psum4[r]);
             }
         }
         // wave-local LDS ordering for the P roundtrip (lgkmcnt only)
