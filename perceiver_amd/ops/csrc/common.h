@@ -34,6 +34,15 @@ DEVINL unsigned int f2bf2(float a, float b) {
     return u;
 }
 
+typedef __attribute__((ext_vector_type(2))) unsigned int uint2v;
+
+// four floats -> one 8-B store (two v_cvt_pk + one ds_write_b64): the two
+// 4-B stores per lane at 32-B stride were a 2-way LDS bank conflict on the
+// packed P/dS images (PMC: 10-12% of wave cycles)
+DEVINL uint2v f2bf4(float a, float b, float c, float d) {
+    return uint2v{f2bf2(a, b), f2bf2(c, d)};
+}
+
 DEVINL unsigned short f2bf(float f) {
     // native single-instruction convert (v_cvt_pk_bf16_f32, RTNE on gfx950);
     // the manual bit-math round used to cost 3-4 VALU ops per convert and

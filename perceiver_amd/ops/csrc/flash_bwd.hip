@@ -364,10 +364,8 @@ __global__ void flash_dq_kernel(
                 for (int r = 0; r < 4; ++r) ds_pack[r] = ds4[r];
                 unsigned short* dst = reinterpret_cast<unsigned short*>(
                     p_mine + (h * SUBE_T + (t * 16 + lo16) * 16 + hi4 * 4) * 2);
-#pragma unroll
-                for (int rp = 0; rp < 2; ++rp)
-                    *reinterpret_cast<unsigned int*>(dst + 2 * rp) =
-                        f2bf2(ds_pack[2 * rp], ds_pack[2 * rp + 1]);
+                *reinterpret_cast<uint2v*>(dst) =
+                    f2bf4(ds_pack[0], ds_pack[1], ds_pack[2], ds_pack[3]);
             }
         }
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
@@ -654,10 +652,8 @@ __global__ void flash_dkv_kernel(
                 float p_pack[4] = {pe4[0], pe4[1], pe4[2], pe4[3]};
                 unsigned short* dst = reinterpret_cast<unsigned short*>(
                     p_mine + (h * SUBE_T + (t * 16 + lo16) * 16 + hi4 * 4) * 2);
-#pragma unroll
-                for (int rp = 0; rp < 2; ++rp)
-                    *reinterpret_cast<unsigned int*>(dst + 2 * rp) =
-                        f2bf2(p_pack[2 * rp], p_pack[2 * rp + 1]);
+                *reinterpret_cast<uint2v*>(dst) =
+                    f2bf4(p_pack[0], p_pack[1], p_pack[2], p_pack[3]);
             }
         }
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
@@ -678,10 +674,9 @@ __global__ void flash_dkv_kernel(
             for (int t = 0; t < TBLKS; ++t) {
                 unsigned short* dst = reinterpret_cast<unsigned short*>(
                     p_mine + (h * SUBE_T + (t * 16 + lo16) * 16 + hi4 * 4) * 2);
-#pragma unroll
-                for (int rp = 0; rp < 2; ++rp)
-                    *reinterpret_cast<unsigned int*>(dst + 2 * rp) =
-                        f2bf2(ds_keep[h][t][2 * rp], ds_keep[h][t][2 * rp + 1]);
+                *reinterpret_cast<uint2v*>(dst) =
+                    f2bf4(ds_keep[h][t][0], ds_keep[h][t][1],
+                          ds_keep[h][t][2], ds_keep[h][t][3]);
             }
 
         // dV += P^T dO : B[k=qrow][j=ch] via transpose reads of the subtiled image

@@ -381,10 +381,7 @@ __global__ void flash_fwd_pipe_kernel(
                 }
                 unsigned short* dst = reinterpret_cast<unsigned short*>(
                     p_mine + (h * SUB_ELEMS + (kb * 16 + lo16) * 16 + hi4 * 4) * 2);
-#pragma unroll
-                for (int rp = 0; rp < 2; ++rp)
-                    *reinterpret_cast<unsigned int*>(dst + 2 * rp) =
-                        f2bf2(pv[2 * rp], pv[2 * rp + 1]);
+                *reinterpret_cast<uint2v*>(dst) = f2bf4(pv[0], pv[1], pv[2], pv[3]);
             }
 #pragma unroll
             for (int r = 0; r < 4; ++r) l_run[h][r] += warp16_sum(psum4[r]);
