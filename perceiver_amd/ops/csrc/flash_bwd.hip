@@ -468,8 +468,11 @@ __global__ void flash_dkv_kernel(
     // per-wave transposed P/dS image ([q][16 key-rows] subtile per h) —
     // packed ushort2 writes + tr16 A-fragment reads, see the dq kernel
     constexpr int SUBE_T = TILE * 16 + 8;
-    char* p_lds = do16_lds + (DVMAX / 16) * (TILE * 16 + 8) * 2;  // NWAVES*QH*SUBE_T*2
-    char* p_mine = p_lds + wave * QH * SUBE_T * 2;
+    // two per-wave image halves: P and dS written back-to-back, ONE lgkm
+    // wait covers both reads (the second roundtrip stall is gone)
+    char* p_lds = do16_lds + (DVMAX / 16) * (TILE * 16 + 8) * 2;  // NWAVES*2*QH*SUBE_T*2
+    char* p_mine = p_lds + wave * 2 * QH * SUBE_T * 2;
+    char* ds_mine = p_mine + QH * SUBE_T * 2;
 
     short8v k_frag[QH][DMAX / 32];
     short8v v_frag[QH][DVMAX / 32];
@@ -588,7 +591,6 @@ __global__ void flash_dkv_kernel(
 
         // t-outer: per 16-q-row block compute S^T and dP^T with short-lived
         // accumulators; P^T goes to LDS now, dS^T is kept in a small register array
-        float ds_keep[QH][TBLKS][4];
 #pragma unroll
         for (int t = 0; t < TBLKS; ++t) {
             float4v st_acc[QH], dpt_acc[QH];
@@ -647,36 +649,23 @@ __global__ void flash_dkv_kernel(
                 }
                 float4v ds4 = p4 * (dpt_acc[h] * m4 - delta_i);
                 float4v pe4 = p4 * m4;
-#pragma unroll
-                for (int r = 0; r < 4; ++r) ds_keep[h][t][r] = ds4[r];
-                float p_pack[4] = {pe4[0], pe4[1], pe4[2], pe4[3]};
-                unsigned short* dst = reinterpret_cast<unsigned short*>(
-                    p_mine + (h * SUBE_T + (t * 16 + lo16) * 16 + hi4 * 4) * 2);
-                *reinterpret_cast<uint2v*>(dst) =
-                    f2bf4(p_pack[0], p_pack[1], p_pack[2], p_pack[3]);
+                const long ioff = (h * SUBE_T + (t * 16 + lo16) * 16 + hi4 * 4) * 2;
+                *reinterpret_cast<uint2v*>(p_mine + ioff) =
+                    f2bf4(pe4[0], pe4[1], pe4[2], pe4[3]);
+                *reinterpret_cast<uint2v*>(ds_mine + ioff) =
+                    f2bf4(ds4[0], ds4[1], ds4[2], ds4[3]);
             }
         }
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-        bf16x8 pt_frag[QH][TBLKS / 2];
+        bf16x8 pt_frag[QH][TBLKS / 2], dst_frag[QH][TBLKS / 2];
 #pragma unroll
         for (int h = 0; h < QH; ++h)
 #pragma unroll
-            for (int t32 = 0; t32 < TBLKS / 2; ++t32)
+            for (int t32 = 0; t32 < TBLKS / 2; ++t32) {
                 pt_frag[h][t32] = read_bfrag_tr16<TILE>(
                     p_mine + h * SUBE_T * 2, 0, t32 * 32, hi4, lo16);
-
-        // second pass: dS^T through the same per-wave buffer — issued BEFORE the
-        // dV MFMAs (the P^T fragments are already in registers), so the whole
-        // write+read roundtrip latency hides under the dV matrix work
-#pragma unroll
-        for (int h = 0; h < QH; ++h)
-#pragma unroll
-            for (int t = 0; t < TBLKS; ++t) {
-                unsigned short* dst = reinterpret_cast<unsigned short*>(
-                    p_mine + (h * SUBE_T + (t * 16 + lo16) * 16 + hi4 * 4) * 2);
-                *reinterpret_cast<uint2v*>(dst) =
-                    f2bf4(ds_keep[h][t][0], ds_keep[h][t][1],
-                          ds_keep[h][t][2], ds_keep[h][t][3]);
+                dst_frag[h][t32] = read_bfrag_tr16<TILE>(
+                    ds_mine + h * SUBE_T * 2, 0, t32 * 32, hi4, lo16);
             }
 
         // dV += P^T dO : B[k=qrow][j=ch] via transpose reads of the subtiled image
@@ -693,15 +682,6 @@ __global__ void flash_dkv_kernel(
                 }
             }
         }
-        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-        bf16x8 dst_frag[QH][TBLKS / 2];
-#pragma unroll
-        for (int h = 0; h < QH; ++h)
-#pragma unroll
-            for (int t32 = 0; t32 < TBLKS / 2; ++t32)
-                dst_frag[h][t32] = read_bfrag_tr16<TILE>(
-                    p_mine + h * SUBE_T * 2, 0, t32 * 32, hi4, lo16);
-
         // dK += dS^T Q : B[k=qrow][j=ch] via transpose reads of the subtiled image
 #pragma unroll
         for (int cb = 0; cb < DMAX / 16; ++cb) {
@@ -820,7 +800,7 @@ void launch_flash_bwd(const torch::Tensor& dout, const torch::Tensor& q, const t
         const int kblk = 16 * DKV_QH * NWAVES;
         size_t smem = (size_t)DKV_TILE * q_stride + (size_t)(DMAX / 16) * (DKV_TILE * 16 + 8) * 2 +
                       (size_t)DKV_TILE * do_stride + (size_t)(DVMAX / 16) * (DKV_TILE * 16 + 8) * 2 +
-                      (size_t)NWAVES * DKV_QH * (DKV_TILE * 16 + 8) * 2;
+                      (size_t)NWAVES * 2 * DKV_QH * (DKV_TILE * 16 + 8) * 2;
         int gx = (Lk + kblk - 1) / kblk, gy = B * H;
         int nsplit = 1;
         long q_chunk = Nq;
