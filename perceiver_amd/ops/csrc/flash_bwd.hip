@@ -163,6 +163,7 @@ __global__ void flash_dq_kernel(
     long kv_chunk,
     long qsb, long qsh, long qsn, long ksb, long ksh, long ksn,
     long vsb, long vsh, long vsn, long dsb, long dsh, long dsn,
+    long oqsb, long oqsh, long oqsn,  // dq OUT strides (merged-heads memory)
     int B, int H, int Nq, int Lk, int D, int Dv, int causal,
     float drop_p, unsigned long long drop_seed) {
     constexpr int TBLKS = TILE / 16;
@@ -409,7 +410,7 @@ __global__ void flash_dq_kernel(
                     if (c < D) dqrow[c] = dq_acc[h][cb][r];
                 }
             } else {
-                unsigned short* dqrow = dqp + ((long)bh * Nq + qi) * D;
+                unsigned short* dqrow = dqp + (long)b * oqsb + (long)hh * oqsh + (long)qi * oqsn;
 #pragma unroll
                 for (int cb = 0; cb < DMAX / 16; ++cb) {
                     int c = cb * 16 + lo16;
@@ -434,6 +435,7 @@ __global__ void flash_dkv_kernel(
     long q_chunk,
     long qsb, long qsh, long qsn, long ksb, long ksh, long ksn,
     long vsb, long vsh, long vsn, long dsb, long dsh, long dsn,
+    long oksb, long oksh, long oksn, long ovsb, long ovsh, long ovsn,  // dk/dv OUT strides
     int B, int H, int Nq, int Lk, int D, int Dv, int causal,
     float drop_p, unsigned long long drop_seed) {
     constexpr int TBLKS = TILE / 16;
@@ -721,8 +723,8 @@ __global__ void flash_dkv_kernel(
                     if (c < Dv) dvrow[c] = dv_acc[h][cb][r];
                 }
             } else {
-                unsigned short* dkrow = dkp + ((long)bh * Lk + ki) * D;
-                unsigned short* dvrow = dvp + ((long)bh * Lk + ki) * Dv;
+                unsigned short* dkrow = dkp + (long)b * oksb + (long)hh * oksh + (long)ki * oksn;
+                unsigned short* dvrow = dvp + (long)b * ovsb + (long)hh * ovsh + (long)ki * ovsn;
 #pragma unroll
                 for (int cb = 0; cb < DMAX / 16; ++cb) {
                     int c = cb * 16 + lo16;
@@ -788,6 +790,7 @@ void launch_flash_bwd(const torch::Tensor& dout, const torch::Tensor& q, const t
                            k.stride(0), k.stride(1), k.stride(2),
                            v.stride(0), v.stride(1), v.stride(2),
                            dout.stride(0), dout.stride(1), dout.stride(2),
+                           dq.stride(0), dq.stride(1), dq.stride(2),
                            B, H, Nq, Lk, D, Dv, (int)causal, drop_p, drop_seed);
         HIP_CHECK_LAST();
         if (nsplit > 1) {
@@ -847,6 +850,8 @@ void launch_flash_bwd(const torch::Tensor& dout, const torch::Tensor& q, const t
                            k.stride(0), k.stride(1), k.stride(2),
                            v.stride(0), v.stride(1), v.stride(2),
                            dout.stride(0), dout.stride(1), dout.stride(2),
+                           dk.stride(0), dk.stride(1), dk.stride(2),
+                           dv.stride(0), dv.stride(1), dv.stride(2),
                            B, H, Nq, Lk, D, Dv, (int)causal, drop_p, drop_seed);
         HIP_CHECK_LAST();
         if (nsplit > 1) {
@@ -903,9 +908,16 @@ std::vector<torch::Tensor> flash_bwd(torch::Tensor dout, torch::Tensor q, torch:
         HIP_CHECK_LAST();
     }
 
-    auto dq = torch::empty(q.sizes(), q.options());
-    auto dk = torch::empty(k.sizes(), k.options());
-    auto dv = torch::empty(v.sizes(), v.options());
+    // grads live in merged-heads (B, N, H*D) memory and are returned as the
+    // (B, H, N, D) permuted views: the module's head-merge transpose+reshape
+    // on each grad becomes a free view instead of a 40 MB copy per tensor
+    // (the forward's out uses the same trick)
+    auto dq = torch::empty({q.size(0), q.size(2), q.size(1), q.size(3)}, q.options())
+                  .permute({0, 2, 1, 3});
+    auto dk = torch::empty({k.size(0), k.size(2), k.size(1), k.size(3)}, k.options())
+                  .permute({0, 2, 1, 3});
+    auto dv = torch::empty({v.size(0), v.size(2), v.size(1), v.size(3)}, v.options())
+                  .permute({0, 2, 1, 3});
 
     c10::optional<torch::Tensor> pm;
     if (pad_mask.has_value() && pad_mask->defined()) pm = pad_mask->contiguous();
