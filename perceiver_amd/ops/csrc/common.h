@@ -159,7 +159,7 @@ DEVINL void write_rm(const short8v (&st)[NG], char* lds_rm, int ldst_bytes,
 template <int ROWS_TILE, int DD, int NG>
 DEVINL void write_sub16(const short8v (&st)[NG], char* lds16, int rows_valid, int tid) {
     constexpr int GPR = DD / 8;
-    constexpr int SUBE = ROWS_TILE * 16 + 40;
+    constexpr int SUBE = ROWS_TILE * 16 + 8;
     const bool tail = rows_valid < ROWS_TILE;
 #pragma unroll
     for (int i = 0; i < NG; ++i) {
@@ -177,7 +177,7 @@ template <int ROWS_TILE, int DD, int NG>
 DEVINL void write_rm_sub16_c(const short8v (&st)[NG], char* lds_rm, int ldst_bytes,
                              char* lds16, int rows_valid, int tid) {
     constexpr int GPR = DD / 8;
-    constexpr int SUBE = ROWS_TILE * 16 + 40;
+    constexpr int SUBE = ROWS_TILE * 16 + 8;
     const bool tail = rows_valid < ROWS_TILE;
 #pragma unroll
     for (int i = 0; i < NG; ++i) {

@@ -56,7 +56,7 @@ typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
 template <int ROWS_TILE>
 DEVINL void stage_sub16(const unsigned short* __restrict__ src, long src_stride,
                         int rows_valid, int d, int d_pad, char* lds, int tid) {
-    constexpr int SUBE = ROWS_TILE * 16 + 40;
+    constexpr int SUBE = ROWS_TILE * 16 + 8;
     const int gpr = d_pad / 8;
     const int total = ROWS_TILE * gpr;
     for (int g = tid; g < total; g += 256) {
@@ -82,7 +82,7 @@ DEVINL void stage_sub16(const unsigned short* __restrict__ src, long src_stride,
 // delivers column lo16.
 template <int ROWS_TILE>
 DEVINL bf16x8 read_bfrag_tr16(const char* lds, int sub, int row0, int hi4, int lo16) {
-    constexpr int SUBE = ROWS_TILE * 16 + 40;
+    constexpr int SUBE = ROWS_TILE * 16 + 8;
     const __bf16* base = reinterpret_cast<const __bf16*>(lds) +
                          sub * SUBE + (row0 + hi4 * 8) * 16 + lo16 * 4;
     auto pp = (__attribute__((address_space(3))) bf16x4*)base;
@@ -102,7 +102,7 @@ template <int ROWS_TILE>
 DEVINL void stage_rm_sub16(const unsigned short* __restrict__ src, long src_stride,
                            int rows_valid, int d, int d_pad,
                            char* lds_rm, int ldst_bytes, char* lds16, int tid) {
-    constexpr int SUBE = ROWS_TILE * 16 + 40;
+    constexpr int SUBE = ROWS_TILE * 16 + 8;
     const int gpr = d_pad / 8;
     const int total = ROWS_TILE * gpr;
     for (int g = tid; g < total; g += 256) {
@@ -193,11 +193,11 @@ __global__ void flash_dq_kernel(
     const int v_stride = dv_pad * 2 + 16;
     char* k_lds = smem;                                  // TILE * k_stride
     char* kt16_lds = k_lds + TILE * k_stride;            // (DMAX/16) * (TILE*16+8) elems
-    char* v_lds = kt16_lds + (DMAX / 16) * (TILE * 16 + 40) * 2;   // TILE * v_stride
+    char* v_lds = kt16_lds + (DMAX / 16) * (TILE * 16 + 8) * 2;   // TILE * v_stride
     // per-wave transposed dS image ([key][16 q-rows] subtile per h): packed
     // ushort2 writes + ds_read_b64_tr_b16 A-fragment reads (the row-major
     // image needed 32 scalar b16 stores per tile per wave)
-    constexpr int SUBE_T = TILE * 16 + 40;
+    constexpr int SUBE_T = TILE * 16 + 8;
     char* p_lds = v_lds + TILE * v_stride;               // NWAVES * QH * SUBE_T * 2
     char* p_mine = p_lds + wave * QH * SUBE_T * 2;
 
@@ -465,14 +465,14 @@ __global__ void flash_dkv_kernel(
     const int do_stride = dv_pad * 2 + 16;    // dO row-major: TILE rows
     char* q_lds = smem;
     char* q16_lds = q_lds + TILE * q_stride;              // (DMAX/16)*(TILE*16+8) elems
-    char* do_lds = q16_lds + (DMAX / 16) * (TILE * 16 + 40) * 2;
+    char* do_lds = q16_lds + (DMAX / 16) * (TILE * 16 + 8) * 2;
     char* do16_lds = do_lds + TILE * do_stride;           // (DVMAX/16)*(TILE*16+8) elems
     // per-wave transposed P/dS image ([q][16 key-rows] subtile per h) —
     // packed ushort2 writes + tr16 A-fragment reads, see the dq kernel
-    constexpr int SUBE_T = TILE * 16 + 40;
+    constexpr int SUBE_T = TILE * 16 + 8;
     // two per-wave image halves: P and dS written back-to-back, ONE lgkm
     // wait covers both reads (the second roundtrip stall is gone)
-    char* p_lds = do16_lds + (DVMAX / 16) * (TILE * 16 + 40) * 2;  // NWAVES*2*QH*SUBE_T*2
+    char* p_lds = do16_lds + (DVMAX / 16) * (TILE * 16 + 8) * 2;  // NWAVES*2*QH*SUBE_T*2
     char* p_mine = p_lds + wave * 2 * QH * SUBE_T * 2;
     char* ds_mine = p_mine + QH * SUBE_T * 2;
 
@@ -756,8 +756,8 @@ void launch_flash_bwd(const torch::Tensor& dout, const torch::Tensor& q, const t
     {   // dQ (with KV-split when the grid would underfill the chip)
         const int k_stride = d_pad * 2 + 16, kt_stride = DQ_TILE * 2 + 16, v_stride = dv_pad * 2 + 16;
         const int qblk = 16 * DQ_QH * NWAVES;
-        size_t smem = (size_t)DQ_TILE * k_stride + (size_t)(DMAX / 16) * (DQ_TILE * 16 + 40) * 2 +
-                      (size_t)DQ_TILE * v_stride + (size_t)NWAVES * DQ_QH * (DQ_TILE * 16 + 40) * 2;
+        size_t smem = (size_t)DQ_TILE * k_stride + (size_t)(DMAX / 16) * (DQ_TILE * 16 + 8) * 2 +
+                      (size_t)DQ_TILE * v_stride + (size_t)NWAVES * DQ_QH * (DQ_TILE * 16 + 8) * 2;
         int gx = (Nq + qblk - 1) / qblk, gy = B * H;
         int nsplit = 1;
         long kv_chunk = Lk;
@@ -801,9 +801,9 @@ void launch_flash_bwd(const torch::Tensor& dout, const torch::Tensor& q, const t
         const int q_stride = d_pad * 2 + 16, qt_stride = DKV_TILE * 2 + 16;
         const int do_stride = dv_pad * 2 + 16;
         const int kblk = 16 * DKV_QH * NWAVES;
-        size_t smem = (size_t)DKV_TILE * q_stride + (size_t)(DMAX / 16) * (DKV_TILE * 16 + 40) * 2 +
-                      (size_t)DKV_TILE * do_stride + (size_t)(DVMAX / 16) * (DKV_TILE * 16 + 40) * 2 +
-                      (size_t)NWAVES * 2 * DKV_QH * (DKV_TILE * 16 + 40) * 2;
+        size_t smem = (size_t)DKV_TILE * q_stride + (size_t)(DMAX / 16) * (DKV_TILE * 16 + 8) * 2 +
+                      (size_t)DKV_TILE * do_stride + (size_t)(DVMAX / 16) * (DKV_TILE * 16 + 8) * 2 +
+                      (size_t)NWAVES * 2 * DKV_QH * (DKV_TILE * 16 + 8) * 2;
         int gx = (Lk + kblk - 1) / kblk, gy = B * H;
         int nsplit = 1;
         long q_chunk = Nq;

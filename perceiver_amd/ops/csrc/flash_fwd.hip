@@ -82,7 +82,7 @@ typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
 // 32x16 MFMA B-fragment — V is staged ROW-major (one 16-B write per granule,
 // no 8-row write lockstep: that pattern measured ~11% of wave cycles in bank
 // conflicts) yet consumed column-major for P@V.
-constexpr int SUB_ELEMS = 64 * 16 + 40;  // KVBLK rows x subtile plane pad (20-bank stagger)
+constexpr int SUB_ELEMS = 64 * 16 + 8;  // KVBLK rows x 16 cols + 16-B pad
 
 DEVINL void stage_tile_sub16(const unsigned short* __restrict__ src, long src_stride,
                              int rows_valid, int dv, int dv_pad,

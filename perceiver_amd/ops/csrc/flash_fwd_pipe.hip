@@ -42,7 +42,7 @@ constexpr int NWAVES = 4;
 constexpr int QH = 2;            // 16-row q fragments per wave
 constexpr int QROWS = 16 * QH;   // 32 q rows per wave
 constexpr int QBLK = QROWS * NWAVES;
-constexpr int SUB_ELEMS = KVBLK * 16 + 40;  // 16-col subtile + 80-B plane stagger
+constexpr int SUB_ELEMS = KVBLK * 16 + 8;  // 16-col subtile + 16-B bank shift
 
 DEVINL float warp16_max(float x) {
 #pragma unroll
